@@ -1,0 +1,128 @@
+"""Typed Kubernetes client (the client-go / generated-clientset analog).
+
+``KubeClient`` is the single seam reconcilers use; it is implemented by:
+
+* ``InMemoryClient`` — over :class:`~kuberay_amd.kube.store.InMemoryApiServer`
+  (tests, bench, local-node runtime),
+* ``RestClient`` (kuberay_amd/kube/rest.py) — over a real kube-apiserver via
+  httpx, same verb surface, for actual cluster deployments.
+
+Typed objects are pydantic models from ``kuberay_amd.kube.objects`` and
+``kuberay_amd.models``; the client converts at the boundary.
+"""
+from __future__ import annotations
+
+from typing import Any, Dict, List, Optional, Type, TypeVar
+
+from ..models import RayCluster, RayCronJob, RayJob, RayService
+from . import objects as k8s
+from .store import InMemoryApiServer, NotFoundError
+
+T = TypeVar("T")
+
+KIND_TO_MODEL: Dict[str, Any] = {
+    "RayCluster": RayCluster,
+    "RayJob": RayJob,
+    "RayService": RayService,
+    "RayCronJob": RayCronJob,
+    "Pod": k8s.Pod,
+    "Service": k8s.Service,
+    "Secret": k8s.Secret,
+    "ConfigMap": k8s.ConfigMap,
+    "PersistentVolumeClaim": k8s.PersistentVolumeClaim,
+    "ServiceAccount": k8s.ServiceAccount,
+    "Role": k8s.Role,
+    "RoleBinding": k8s.RoleBinding,
+    "Job": k8s.Job,
+    "NetworkPolicy": k8s.NetworkPolicy,
+    "Ingress": k8s.Ingress,
+    "EndpointSlice": k8s.EndpointSlice,
+    "Event": k8s.Event,
+}
+
+
+def model_for_kind(kind: str):
+    return KIND_TO_MODEL.get(kind)
+
+
+class KubeClient:
+    """Abstract verb surface. All methods accept/return typed models."""
+
+    def create(self, obj: T) -> T:
+        raise NotImplementedError
+
+    def get(self, model: Type[T], namespace: str, name: str) -> T:
+        raise NotImplementedError
+
+    def try_get(self, model: Type[T], namespace: str, name: str) -> Optional[T]:
+        try:
+            return self.get(model, namespace, name)
+        except NotFoundError:
+            return None
+
+    def list(
+        self,
+        model: Type[T],
+        namespace: Optional[str] = None,
+        label_selector: Optional[Dict[str, str]] = None,
+    ) -> List[T]:
+        raise NotImplementedError
+
+    def update(self, obj: T) -> T:
+        raise NotImplementedError
+
+    def update_status(self, obj: T) -> T:
+        raise NotImplementedError
+
+    def patch(self, model: Type[T], namespace: str, name: str,
+              patch: Dict[str, Any], subresource: Optional[str] = None) -> T:
+        raise NotImplementedError
+
+    def delete(self, model_or_obj, namespace: Optional[str] = None,
+               name: Optional[str] = None) -> None:
+        raise NotImplementedError
+
+
+def _kind_of(model_or_obj) -> str:
+    if isinstance(model_or_obj, type):
+        # model class: read the default of the `kind` field
+        return model_or_obj.model_fields["kind"].default
+    return model_or_obj.kind
+
+
+class InMemoryClient(KubeClient):
+    def __init__(self, server: Optional[InMemoryApiServer] = None):
+        self.server = server or InMemoryApiServer()
+
+    def create(self, obj):
+        data = obj.to_dict()
+        data["kind"] = obj.kind
+        out = self.server.create(data)
+        return type(obj).from_dict(out)
+
+    def get(self, model, namespace, name):
+        out = self.server.get(_kind_of(model), namespace, name)
+        return model.from_dict(out)
+
+    def list(self, model, namespace=None, label_selector=None):
+        kind = _kind_of(model)
+        return [model.from_dict(o) for o in self.server.list(kind, namespace, label_selector)]
+
+    def update(self, obj):
+        out = self.server.update(obj.to_dict())
+        return type(obj).from_dict(out)
+
+    def update_status(self, obj):
+        out = self.server.update(obj.to_dict(), subresource="status")
+        return type(obj).from_dict(out)
+
+    def patch(self, model, namespace, name, patch, subresource=None):
+        out = self.server.patch_merge(_kind_of(model), namespace, name, patch,
+                                      subresource=subresource)
+        return model.from_dict(out)
+
+    def delete(self, model_or_obj, namespace=None, name=None):
+        if namespace is None:
+            namespace = model_or_obj.metadata.namespace or "default"
+            name = model_or_obj.metadata.name
+        self.server.delete(_kind_of(model_or_obj), namespace, name)
