@@ -1,0 +1,283 @@
+"""Spec validation for all four CRDs
+(reference: ray-operator/controllers/ray/utils/validation.go).
+
+Pure functions returning a list of error strings (empty = valid). The same
+functions back the reconcilers' early validation, the validating webhooks,
+and the apiserver.
+"""
+from __future__ import annotations
+
+import re
+from typing import List
+
+from ..kube.objects import ObjectMeta
+from ..models.raycluster import RayCluster, RayClusterSpec
+from ..models.raycronjob import RayCronJob
+from ..models.rayjob import (
+    DeletionPolicyType,
+    JobDeploymentStatus,
+    JobStatus,
+    JobSubmissionMode,
+    RayJob,
+)
+from ..models.rayservice import RayService, RayServiceUpgradeType
+from . import constants as C
+
+_NAME_RE = re.compile(r"^[a-z0-9]([-a-z0-9]*[a-z0-9])?$")
+
+
+def validate_raycluster_metadata(meta: ObjectMeta) -> List[str]:
+    """validation.go ValidateRayClusterMetadata."""
+    errs = []
+    name = meta.name or ""
+    if len(name) > 63:
+        errs.append(f"RayCluster name '{name}' exceeds 63 characters")
+    if name and not _NAME_RE.match(name):
+        errs.append(f"RayCluster name '{name}' is not a valid DNS-1035 label")
+    return errs
+
+
+def validate_raycluster_spec(cluster: RayCluster) -> List[str]:
+    """validation.go:103 ValidateRayClusterSpec."""
+    errs = []
+    spec = cluster.spec
+    if not spec.head_group_spec.template.spec.containers:
+        errs.append("headGroupSpec should have at least one container")
+    group_names = set()
+    for group in spec.worker_group_specs:
+        if not group.group_name:
+            errs.append("workerGroupSpec groupName must not be empty")
+        if group.group_name in group_names:
+            errs.append(f"duplicate worker group name '{group.group_name}'")
+        group_names.add(group.group_name)
+        if not group.template.spec.containers:
+            errs.append(f"worker group '{group.group_name}' should have at least one container")
+        min_r = group.min_replicas or 0
+        max_r = group.max_replicas if group.max_replicas is not None else 2**31 - 1
+        if min_r > max_r:
+            errs.append(
+                f"worker group '{group.group_name}': minReplicas {min_r} > maxReplicas {max_r}")
+        if group.replicas is not None and group.replicas < 0:
+            errs.append(f"worker group '{group.group_name}': replicas must be >= 0")
+        if group.num_of_hosts < 0:
+            errs.append(f"worker group '{group.group_name}': numOfHosts must be >= 0")
+        if group.idle_timeout_seconds is not None and not _autoscaler_v2(spec):
+            errs.append(
+                f"worker group '{group.group_name}': idleTimeoutSeconds requires "
+                "autoscaler v2 (spec.autoscalerOptions.version: v2)")
+    errs += _validate_gcs_ft(cluster)
+    errs += _validate_auth(spec)
+    errs += _validate_network_policy(spec)
+    if spec.suspend and _autoscaler_enabled(spec) and False:
+        pass  # suspend+autoscaler is allowed; kept for structural parity
+    return errs
+
+
+def _autoscaler_enabled(spec: RayClusterSpec) -> bool:
+    return bool(spec.enable_in_tree_autoscaling)
+
+
+def _autoscaler_v2(spec: RayClusterSpec) -> bool:
+    return bool(spec.autoscaler_options and spec.autoscaler_options.version == "v2")
+
+
+def _validate_gcs_ft(cluster: RayCluster) -> List[str]:
+    """validation.go:333 (GCS FT backend rules)."""
+    errs = []
+    spec = cluster.spec
+    opts = spec.gcs_fault_tolerance_options
+    annotations = cluster.metadata.annotations or {}
+    ft_annotation = annotations.get(C.RAY_FT_ENABLED_ANNOTATION_KEY)
+    if opts is not None and ft_annotation is not None:
+        errs.append(
+            f"annotation {C.RAY_FT_ENABLED_ANNOTATION_KEY} and "
+            "gcsFaultToleranceOptions are mutually exclusive")
+    if opts is not None:
+        backend = opts.backend or ("redis" if opts.redis_address else None)
+        if backend == "embedded":
+            if opts.redis_address or opts.redis_username or opts.redis_password:
+                errs.append("embedded GCS storage backend cannot set redis fields")
+        elif backend == "redis":
+            if not opts.redis_address:
+                errs.append("redis GCS backend requires redisAddress")
+        if opts.storage is not None and backend != "embedded":
+            errs.append("gcsFaultToleranceOptions.storage requires backend: embedded")
+    head_container = (spec.head_group_spec.template.spec.containers[0]
+                      if spec.head_group_spec.template.spec.containers else None)
+    if head_container is not None and opts is not None:
+        for env in head_container.env or []:
+            if env.name == C.RAY_REDIS_ADDRESS and opts.backend == "embedded":
+                errs.append("RAY_REDIS_ADDRESS env is invalid with embedded GCS backend")
+    return errs
+
+
+def _validate_auth(spec: RayClusterSpec) -> List[str]:
+    errs = []
+    ao = spec.auth_options
+    if ao is None:
+        return errs
+    if ao.mode not in (None, "token", "disabled"):
+        errs.append(f"authOptions.mode must be 'token' or 'disabled', got '{ao.mode}'")
+    return errs
+
+
+def _validate_network_policy(spec: RayClusterSpec) -> List[str]:
+    """validation.go:380."""
+    errs = []
+    np = spec.network_policy
+    if np is None:
+        return errs
+    if np.mode not in (None, "DenyAll", "DenyAllIngress", "DenyAllEgress"):
+        errs.append(f"networkPolicy.mode invalid: '{np.mode}'")
+    group_names = {g.group_name for g in spec.worker_group_specs}
+    for wg in np.worker_groups or []:
+        if wg.group_name not in group_names:
+            errs.append(f"networkPolicy.workerGroups references unknown group '{wg.group_name}'")
+    return errs
+
+
+# ---------------------------------------------------------------------------
+# RayJob (validation.go:543-967)
+# ---------------------------------------------------------------------------
+
+def validate_rayjob_metadata(meta: ObjectMeta) -> List[str]:
+    errs = []
+    name = meta.name or ""
+    if len(name) > 63:
+        errs.append(f"RayJob name '{name}' exceeds 63 characters")
+    return errs
+
+
+def validate_rayjob_spec(rayjob: RayJob) -> List[str]:
+    errs = []
+    spec = rayjob.spec
+    if spec.submission_mode not in JobSubmissionMode.ALL:
+        errs.append(f"invalid submissionMode '{spec.submission_mode}'")
+    if spec.ray_cluster_spec is None and not spec.cluster_selector:
+        errs.append("one of rayClusterSpec or clusterSelector must be set")
+    if spec.ray_cluster_spec is not None and spec.cluster_selector:
+        errs.append("rayClusterSpec and clusterSelector are mutually exclusive")
+    if spec.submission_mode == JobSubmissionMode.INTERACTIVE and spec.entrypoint:
+        errs.append("entrypoint must be empty in InteractiveMode")
+    if spec.submission_mode != JobSubmissionMode.INTERACTIVE and not spec.entrypoint:
+        errs.append("entrypoint is required unless submissionMode is InteractiveMode")
+    if spec.submission_mode == JobSubmissionMode.SIDECAR and spec.cluster_selector:
+        errs.append("SidecarMode requires an operator-managed cluster (no clusterSelector)")
+    if spec.submitter_pod_template is not None and spec.submission_mode not in (
+            JobSubmissionMode.K8S_JOB,):
+        errs.append("submitterPodTemplate only applies to K8sJobMode")
+    if spec.backoff_limit is not None and spec.backoff_limit < 0:
+        errs.append("backoffLimit must be >= 0")
+    if spec.active_deadline_seconds is not None and spec.active_deadline_seconds <= 0:
+        errs.append("activeDeadlineSeconds must be > 0")
+    if spec.ttl_seconds_after_finished < 0:
+        errs.append("ttlSecondsAfterFinished must be >= 0")
+    if spec.ttl_seconds_after_finished > 0 and not spec.shutdown_after_job_finishes:
+        errs.append("ttlSecondsAfterFinished requires shutdownAfterJobFinishes: true")
+    if spec.shutdown_after_job_finishes and spec.cluster_selector:
+        errs.append("shutdownAfterJobFinishes is invalid with clusterSelector "
+                    "(the job does not own the cluster)")
+    errs += _validate_deletion_strategy(rayjob)
+    if spec.ray_cluster_spec is not None:
+        sub = RayCluster(metadata=rayjob.metadata, spec=spec.ray_cluster_spec)
+        errs += validate_raycluster_spec(sub)
+    return errs
+
+
+def _validate_deletion_strategy(rayjob: RayJob) -> List[str]:
+    errs = []
+    ds = rayjob.spec.deletion_strategy
+    if ds is None:
+        return errs
+    legacy = ds.on_success is not None or ds.on_failure is not None
+    rules = ds.deletion_rules is not None
+    if legacy and rules:
+        errs.append("deletionStrategy: onSuccess/onFailure and deletionRules are mutually exclusive")
+    if legacy and (ds.on_success is None or ds.on_failure is None):
+        errs.append("deletionStrategy: both onSuccess and onFailure must be set")
+    for block in (ds.on_success, ds.on_failure):
+        if block is not None and block.policy not in DeletionPolicyType.ALL:
+            errs.append(f"deletionStrategy: invalid policy '{block.policy}'")
+    for rule in ds.deletion_rules or []:
+        if rule.policy not in DeletionPolicyType.ALL:
+            errs.append(f"deletionRules: invalid policy '{rule.policy}'")
+        if rule.condition.ttl_seconds < 0:
+            errs.append("deletionRules: ttlSeconds must be >= 0")
+        js = rule.condition.job_status
+        if js is not None and js not in JobStatus.TERMINAL:
+            errs.append(f"deletionRules: jobStatus '{js}' is not terminal")
+        jds = rule.condition.job_deployment_status
+        if jds is not None and jds not in (JobDeploymentStatus.COMPLETE,
+                                           JobDeploymentStatus.FAILED):
+            errs.append(f"deletionRules: jobDeploymentStatus '{jds}' is not terminal")
+    if (rayjob.spec.shutdown_after_job_finishes and rules):
+        errs.append("deletionRules and shutdownAfterJobFinishes are mutually exclusive")
+    return errs
+
+
+# ---------------------------------------------------------------------------
+# RayService (validation.go:680)
+# ---------------------------------------------------------------------------
+
+def validate_rayservice_metadata(meta: ObjectMeta) -> List[str]:
+    errs = []
+    name = meta.name or ""
+    if len(name) > 63:
+        errs.append(f"RayService name '{name}' exceeds 63 characters")
+    return errs
+
+
+def validate_rayservice_spec(rayservice: RayService) -> List[str]:
+    errs = []
+    spec = rayservice.spec
+    sub = RayCluster(metadata=rayservice.metadata, spec=spec.ray_cluster_spec)
+    errs += validate_raycluster_spec(sub)
+    us = spec.upgrade_strategy
+    if us is not None and us.type not in (
+            None, RayServiceUpgradeType.NEW_CLUSTER,
+            RayServiceUpgradeType.NEW_CLUSTER_WITH_INCREMENTAL_UPGRADE,
+            RayServiceUpgradeType.NONE):
+        errs.append(f"invalid upgradeStrategy.type '{us.type}'")
+    if us is not None and us.type == RayServiceUpgradeType.NEW_CLUSTER_WITH_INCREMENTAL_UPGRADE:
+        opts = us.cluster_upgrade_options
+        if opts is None:
+            errs.append("NewClusterWithIncrementalUpgrade requires clusterUpgradeOptions")
+        else:
+            if opts.gateway_class_name in (None, ""):
+                errs.append("clusterUpgradeOptions.gatewayClassName is required")
+            for field, val in (("stepSizePercent", opts.step_size_percent),
+                               ("intervalSeconds", opts.interval_seconds)):
+                if val is None:
+                    errs.append(f"clusterUpgradeOptions.{field} is required")
+                elif val < 0 or (field == "stepSizePercent" and val > 100):
+                    errs.append(f"clusterUpgradeOptions.{field} out of range")
+            if opts.max_surge_percent is not None and not (0 <= opts.max_surge_percent <= 100):
+                errs.append("clusterUpgradeOptions.maxSurgePercent must be in [0,100]")
+    if spec.serve_config_v2:
+        import yaml
+        try:
+            data = yaml.safe_load(spec.serve_config_v2)
+            if not isinstance(data, dict) or "applications" not in data:
+                errs.append("serveConfigV2 must be a YAML map with an 'applications' list")
+        except yaml.YAMLError as e:
+            errs.append(f"serveConfigV2 is not valid YAML: {e}")
+    return errs
+
+
+# ---------------------------------------------------------------------------
+# RayCronJob (validation.go:969)
+# ---------------------------------------------------------------------------
+
+def validate_raycronjob_spec(cronjob: RayCronJob) -> List[str]:
+    errs = []
+    from .cron import parse_cron
+    if not cronjob.spec.schedule:
+        errs.append("schedule is required")
+    else:
+        try:
+            parse_cron(cronjob.spec.schedule)
+        except ValueError as e:
+            errs.append(f"invalid schedule: {e}")
+    job = RayJob(metadata=cronjob.metadata, spec=cronjob.spec.job_template)
+    errs += validate_rayjob_spec(job)
+    return errs
