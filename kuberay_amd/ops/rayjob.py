@@ -56,6 +56,7 @@ class RayJobReconciler(Reconciler):
         self.recorder = recorder or NullRecorder()
         self.dashboard_factory = dashboard_factory or self._default_dashboard
         self.metrics = metrics
+        self.requeue_seconds = REQUEUE_SECONDS
 
     @staticmethod
     def _default_dashboard(url: str):
@@ -153,12 +154,12 @@ class RayJobReconciler(Reconciler):
 
         cluster = self._get_or_create_cluster(rayjob)
         if cluster is None:
-            return Result(requeue_after=REQUEUE_SECONDS)
+            return Result(requeue_after=self.requeue_seconds)
         rayjob.status.ray_cluster_status = cluster.status
 
         if cluster.status.state != ClusterState.READY:
             self._set_status(rayjob, rayjob.status.job_deployment_status)
-            return Result(requeue_after=REQUEUE_SECONDS)
+            return Result(requeue_after=self.requeue_seconds)
 
         # cluster ready → resolve dashboard URL
         head_svc = names.head_service_name(C.KIND_RAYCLUSTER, cluster.spec,
@@ -178,13 +179,13 @@ class RayJobReconciler(Reconciler):
                     dashboard.submit_job(self._build_http_submission(rayjob))
             except DashboardClientError as e:
                 self.recorder.eventf(rayjob, "Warning", "FailedToSubmitJob", str(e))
-                return Result(requeue_after=REQUEUE_SECONDS)
+                return Result(requeue_after=self.requeue_seconds)
         elif mode == Mode.SIDECAR:
             pass  # sidecar container injected into the head pod via cluster spec
         # InteractiveMode: user submits; we just watch
 
         self._set_status(rayjob, JDS.RUNNING)
-        return Result(requeue_after=REQUEUE_SECONDS)
+        return Result(requeue_after=self.requeue_seconds)
 
     def _build_http_submission(self, rayjob: RayJob) -> dict:
         import yaml
@@ -208,7 +209,7 @@ class RayJobReconciler(Reconciler):
         if job_id:
             rayjob.status.job_id = job_id
             self._set_status(rayjob, JDS.RUNNING)
-        return Result(requeue_after=REQUEUE_SECONDS)
+        return Result(requeue_after=self.requeue_seconds)
 
     # ------------------------------------------------------------------
     def _handle_running(self, rayjob: RayJob) -> Result:
@@ -257,9 +258,9 @@ class RayJobReconciler(Reconciler):
 
         if mode == Mode.K8S_JOB and submitter_failed and not job_terminal:
             rayjob.status.end_time = now_iso()
-            self._set_status(rayjob, JDS.FAILED, JobFailedReason.SUBMISSION_FAILED,
-                             "Submitter K8s Job failed")
-            return self._maybe_retry(rayjob)
+            rayjob.status.failed = (rayjob.status.failed or 0) + 1
+            return self._fail_or_retry(rayjob, JobFailedReason.SUBMISSION_FAILED,
+                                       "Submitter K8s Job failed")
 
         if job_terminal:
             # For K8sJobMode wait for the submitter to finish too, with a grace
@@ -267,7 +268,7 @@ class RayJobReconciler(Reconciler):
             wait_submitter = mode == Mode.K8S_JOB and not submitter_finished
             if wait_submitter and not self._transition_grace_exceeded(rayjob):
                 self._set_status(rayjob, JDS.RUNNING)
-                return Result(requeue_after=REQUEUE_SECONDS)
+                return Result(requeue_after=self.requeue_seconds)
             rayjob.status.end_time = now_iso()
             if rayjob.status.job_status == JS.SUCCEEDED:
                 rayjob.status.succeeded = (rayjob.status.succeeded or 0) + 1
@@ -276,22 +277,26 @@ class RayJobReconciler(Reconciler):
                     self.metrics.observe_job_finished(rayjob, succeeded=True)
                 return Result(requeue_after=0.0)
             rayjob.status.failed = (rayjob.status.failed or 0) + 1
-            self._set_status(rayjob, JDS.FAILED, JobFailedReason.APP_FAILED,
-                             f"Ray job finished with status {rayjob.status.job_status}")
             if self.metrics is not None:
                 self.metrics.observe_job_finished(rayjob, succeeded=False)
-            return self._maybe_retry(rayjob)
+            return self._fail_or_retry(
+                rayjob, JobFailedReason.APP_FAILED,
+                f"Ray job finished with status {rayjob.status.job_status}")
 
         self._set_status(rayjob, JDS.RUNNING)
-        return Result(requeue_after=REQUEUE_SECONDS)
+        return Result(requeue_after=self.requeue_seconds)
 
-    def _maybe_retry(self, rayjob: RayJob) -> Result:
+    def _fail_or_retry(self, rayjob: RayJob, reason: str, message: str) -> Result:
+        """Decide Retrying vs Failed BEFORE writing status, so the job never
+        transiently reads as Failed while retries remain
+        (rayjob_controller.go:518-558)."""
         backoff = rayjob.spec.backoff_limit or 0
         attempts = (rayjob.status.succeeded or 0) + (rayjob.status.failed or 0)
         if backoff > 0 and attempts <= backoff and \
-                rayjob.status.reason != JobFailedReason.DEADLINE_EXCEEDED:
-            self._set_status(rayjob, JDS.RETRYING)
-            return Result(requeue_after=0.0)
+                reason != JobFailedReason.DEADLINE_EXCEEDED:
+            self._set_status(rayjob, JDS.RETRYING, reason, message)
+        else:
+            self._set_status(rayjob, JDS.FAILED, reason, message)
         return Result(requeue_after=0.0)
 
     # ------------------------------------------------------------------
@@ -303,7 +308,7 @@ class RayJobReconciler(Reconciler):
         rayjob.status.dashboard_url = None
         rayjob.status.ray_cluster_status = type(rayjob.status.ray_cluster_status)()
         self.client.update_status(rayjob)
-        return Result(requeue_after=REQUEUE_SECONDS)
+        return Result(requeue_after=self.requeue_seconds)
 
     def _handle_suspended(self, rayjob: RayJob) -> Result:
         if not rayjob.spec.suspend:

@@ -60,6 +60,7 @@ class RayServiceReconciler(Reconciler):
         self.http_proxy_client = http_proxy_client
         self.metrics = metrics
         self.cluster_deletion_delay_s = cluster_deletion_delay_s
+        self.requeue_seconds = REQUEUE_SECONDS
         # serve-config cache: (namespace, service, cluster) -> config hash
         self._serve_config_cache: Dict[Tuple[str, str, str], str] = {}
         # delayed old-cluster GC: (namespace, cluster) -> not-before time
@@ -114,7 +115,7 @@ class RayServiceReconciler(Reconciler):
 
         self._update_head_pod_serve_label(svc, active)
         self._calculate_status(svc, active, pending)
-        return Result(requeue_after=REQUEUE_SECONDS)
+        return Result(requeue_after=self.requeue_seconds)
 
     # ------------------------------------------------------------------
     # cluster lifecycle
@@ -405,7 +406,7 @@ class RayServiceReconciler(Reconciler):
             svc.status.conditions = set_condition(
                 svc.status.conditions, Cond.SUSPENDING, "True", "SuspendRequested")
             self.client.update_status(svc)
-            return Result(requeue_after=REQUEUE_SECONDS)
+            return Result(requeue_after=self.requeue_seconds)
         svc.status.conditions = set_condition(
             svc.status.conditions, Cond.SUSPENDING, "False", "Suspended")
         svc.status.conditions = set_condition(

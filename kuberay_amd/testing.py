@@ -26,7 +26,8 @@ class ControlPlane:
                  workers: int = 4, record_events: bool = True,
                  gpu_gate: Optional[Callable[[dict], bool]] = None,
                  dashboard_client=None, enable_kubelet: bool = True,
-                 requeue_seconds: Optional[int] = 300):
+                 requeue_seconds: Optional[int] = 300,
+                 poll_seconds: Optional[float] = None):
         from .ops.raycluster import RayClusterReconciler, RayClusterReconcilerOptions
         from .ops.rayjob import RayJobReconciler
         from .ops.rayservice import RayServiceReconciler
@@ -49,6 +50,9 @@ class ControlPlane:
             self.client, recorder=self.recorder, dashboard_factory=lambda url: self.dashboard)
         self.raycronjob_reconciler = RayCronJobReconciler(
             self.client, recorder=self.recorder)
+        if poll_seconds is not None:
+            self.rayjob_reconciler.requeue_seconds = poll_seconds
+            self.rayservice_reconciler.requeue_seconds = poll_seconds
 
         self.manager = Manager(self.server)
         self.manager.add_controller(Controller(

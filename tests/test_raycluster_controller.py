@@ -1,0 +1,248 @@
+"""RayCluster reconciler tests against the in-process control plane
+(reference analog: raycluster_controller_test.go / _unit_test.go under
+envtest; SURVEY.md §4 tier 2)."""
+import time
+
+import pytest
+
+from kuberay_amd.kube import objects as k8s
+from kuberay_amd.models import RayCluster
+from kuberay_amd.ops.raycluster import should_delete_pod
+from kuberay_amd.testing import simple_raycluster
+from kuberay_amd.utils import constants as C
+
+
+def get_cluster(cp, name="demo"):
+    return cp.client.get(RayCluster, "default", name)
+
+
+def pods_of(cp, name="demo"):
+    return [p for p in cp.server.list("Pod")
+            if (p["metadata"].get("labels") or {}).get(C.RAY_CLUSTER_LABEL_KEY) == name]
+
+
+class TestLifecycle:
+    def test_cluster_reaches_ready(self, control_plane):
+        control_plane.client.create(simple_raycluster("demo", workers=2))
+        assert control_plane.wait_cluster_state("default", "demo", "ready")
+        rc = get_cluster(control_plane)
+        assert rc.status.desired_worker_replicas == 2
+        assert rc.status.available_worker_replicas == 2
+        assert rc.status.ready_worker_replicas == 2
+        assert rc.status.state_transition_times.get("ready")
+        assert rc.status.observed_generation == rc.metadata.generation
+        assert rc.status.head.pod_ip
+        assert rc.status.endpoints["dashboard"] == "8265"
+
+    def test_conditions(self, control_plane):
+        control_plane.client.create(simple_raycluster("demo"))
+        assert control_plane.wait_cluster_condition("default", "demo", "HeadPodReady")
+        assert control_plane.wait_cluster_condition("default", "demo", "RayClusterProvisioned")
+
+    def test_head_service_created(self, control_plane):
+        control_plane.client.create(simple_raycluster("demo"))
+        assert control_plane.wait_for(
+            lambda: control_plane.server.try_get("Service", "default", "demo-head-svc"))
+        svc = control_plane.server.get("Service", "default", "demo-head-svc")
+        assert svc["metadata"]["ownerReferences"][0]["kind"] == "RayCluster"
+
+    def test_desired_gpu_status(self, control_plane):
+        control_plane.client.create(simple_raycluster("demo", workers=2, gpus_per_worker=4))
+        assert control_plane.wait_cluster_state("default", "demo", "ready")
+        assert get_cluster(control_plane).status.desired_gpu == "8"
+
+    def test_cluster_deletion_cascades_to_pods(self, control_plane):
+        control_plane.client.create(simple_raycluster("demo"))
+        assert control_plane.wait_cluster_state("default", "demo", "ready")
+        control_plane.server.delete("RayCluster", "default", "demo")
+        assert control_plane.wait_for(lambda: len(pods_of(control_plane)) == 0)
+
+
+class TestScaling:
+    def test_scale_up(self, control_plane):
+        control_plane.client.create(simple_raycluster("demo", workers=1))
+        assert control_plane.wait_cluster_state("default", "demo", "ready")
+        rc = get_cluster(control_plane)
+        rc.spec.worker_group_specs[0].replicas = 3
+        control_plane.client.update(rc)
+        assert control_plane.wait_for(
+            lambda: get_cluster(control_plane).status.ready_worker_replicas == 3)
+
+    def test_scale_down_without_autoscaler_deletes_random(self, control_plane):
+        control_plane.client.create(simple_raycluster("demo", workers=3))
+        assert control_plane.wait_cluster_state("default", "demo", "ready")
+        rc = get_cluster(control_plane)
+        rc.spec.worker_group_specs[0].replicas = 1
+        control_plane.client.update(rc)
+        assert control_plane.wait_for(
+            lambda: get_cluster(control_plane).status.available_worker_replicas == 1)
+
+    def test_workers_to_delete_honored(self, control_plane):
+        control_plane.client.create(simple_raycluster("demo", workers=2))
+        assert control_plane.wait_cluster_state("default", "demo", "ready")
+        victim = next(p["metadata"]["name"] for p in pods_of(control_plane)
+                      if p["metadata"]["labels"][C.RAY_NODE_TYPE_LABEL_KEY] == "worker")
+        rc = get_cluster(control_plane)
+        rc.spec.worker_group_specs[0].replicas = 1
+        rc.spec.worker_group_specs[0].scale_strategy.workers_to_delete = [victim]
+        control_plane.client.update(rc)
+        assert control_plane.wait_for(
+            lambda: victim not in [p["metadata"]["name"] for p in pods_of(control_plane)])
+        assert control_plane.wait_for(
+            lambda: get_cluster(control_plane).status.available_worker_replicas == 1)
+
+    def test_autoscaler_blocks_random_delete(self, control_plane):
+        cluster = simple_raycluster("demo", workers=3, enableInTreeAutoscaling=True)
+        control_plane.client.create(cluster)
+        assert control_plane.wait_for(
+            lambda: get_cluster(control_plane).status.available_worker_replicas == 3)
+        rc = get_cluster(control_plane)
+        rc.spec.worker_group_specs[0].replicas = 1
+        control_plane.client.update(rc)
+        time.sleep(0.5)
+        # without WorkersToDelete, operator must NOT delete pods when autoscaling
+        assert get_cluster(control_plane).status.available_worker_replicas == 3
+
+    def test_group_suspend(self, control_plane):
+        control_plane.client.create(simple_raycluster("demo", workers=2))
+        assert control_plane.wait_cluster_state("default", "demo", "ready")
+        rc = get_cluster(control_plane)
+        rc.spec.worker_group_specs[0].suspend = True
+        control_plane.client.update(rc)
+        assert control_plane.wait_for(
+            lambda: get_cluster(control_plane).status.available_worker_replicas == 0)
+        assert get_cluster(control_plane).status.desired_worker_replicas == 0
+
+    def test_multihost_group(self, control_plane):
+        control_plane.client.create(simple_raycluster("demo", workers=2, num_of_hosts=2))
+        assert control_plane.wait_for(
+            lambda: get_cluster(control_plane).status.available_worker_replicas == 4)
+        # headless service for multi-host groups
+        assert control_plane.server.try_get("Service", "default", "demo-headless")
+
+
+class TestSuspendResume:
+    def test_suspend_deletes_all_pods_and_sets_conditions(self, control_plane):
+        control_plane.client.create(simple_raycluster("demo"))
+        assert control_plane.wait_cluster_state("default", "demo", "ready")
+        rc = get_cluster(control_plane)
+        rc.spec.suspend = True
+        control_plane.client.update(rc)
+        assert control_plane.wait_cluster_state("default", "demo", "suspended")
+        assert len(pods_of(control_plane)) == 0
+        assert control_plane.wait_cluster_condition("default", "demo", "RayClusterSuspended")
+
+    def test_resume(self, control_plane):
+        control_plane.client.create(simple_raycluster("demo", suspend=True))
+        assert control_plane.wait_cluster_state("default", "demo", "suspended")
+        rc = get_cluster(control_plane)
+        rc.spec.suspend = False
+        control_plane.client.update(rc)
+        assert control_plane.wait_cluster_state("default", "demo", "ready")
+
+
+class TestFailureRecovery:
+    def test_failed_worker_pod_recreated(self, control_plane):
+        control_plane.client.create(simple_raycluster("demo", workers=1))
+        assert control_plane.wait_cluster_state("default", "demo", "ready")
+        worker = next(p for p in pods_of(control_plane)
+                      if p["metadata"]["labels"][C.RAY_NODE_TYPE_LABEL_KEY] == "worker")
+        control_plane.server.patch_merge(
+            "Pod", "default", worker["metadata"]["name"],
+            {"status": {"phase": "Failed"}}, subresource="status")
+        def recreated():
+            workers = [p for p in pods_of(control_plane)
+                       if p["metadata"]["labels"][C.RAY_NODE_TYPE_LABEL_KEY] == "worker"]
+            return (len(workers) == 1
+                    and workers[0]["metadata"]["name"] != worker["metadata"]["name"]
+                    and workers[0].get("status", {}).get("phase") == "Running")
+        assert control_plane.wait_for(recreated)
+
+    def test_failed_head_pod_recreated(self, control_plane):
+        control_plane.client.create(simple_raycluster("demo", workers=0))
+        assert control_plane.wait_cluster_state("default", "demo", "ready")
+        head = next(p for p in pods_of(control_plane)
+                    if p["metadata"]["labels"][C.RAY_NODE_TYPE_LABEL_KEY] == "head")
+        control_plane.server.patch_merge(
+            "Pod", "default", head["metadata"]["name"],
+            {"status": {"phase": "Failed"}}, subresource="status")
+        def recreated():
+            heads = [p for p in pods_of(control_plane)
+                     if p["metadata"]["labels"][C.RAY_NODE_TYPE_LABEL_KEY] == "head"]
+            return (len(heads) == 1
+                    and heads[0]["metadata"]["name"] != head["metadata"]["name"])
+        assert control_plane.wait_for(recreated)
+
+
+class TestShouldDeletePod:
+    def _pod(self, phase, restart_policy="Always", terminated=False):
+        pod = k8s.Pod.from_dict({
+            "metadata": {"name": "p"},
+            "spec": {"containers": [{"name": "ray"}], "restartPolicy": restart_policy},
+            "status": {"phase": phase},
+        })
+        if terminated:
+            pod.status.container_statuses = [k8s.ContainerStatus.from_dict(
+                {"name": "ray", "state": {"terminated": {"exitCode": 1}}})]
+        return pod
+
+    def test_terminal_phases(self):
+        assert should_delete_pod(self._pod("Failed"), "worker")[0]
+        assert should_delete_pod(self._pod("Succeeded"), "worker")[0]
+
+    def test_running_healthy(self):
+        assert not should_delete_pod(self._pod("Running"), "worker")[0]
+
+    def test_terminated_container_restart_never(self):
+        pod = self._pod("Running", restart_policy="Never", terminated=True)
+        assert should_delete_pod(pod, "worker")[0]
+
+    def test_terminated_container_restart_always(self):
+        pod = self._pod("Running", restart_policy="Always", terminated=True)
+        assert not should_delete_pod(pod, "worker")[0]
+
+
+class TestValidationRejection:
+    def test_invalid_spec_sets_failed_state(self, control_plane):
+        bad = simple_raycluster("demo")
+        bad.spec.worker_group_specs[0].min_replicas = 5
+        bad.spec.worker_group_specs[0].max_replicas = 1
+        control_plane.client.create(bad)
+        assert control_plane.wait_for(
+            lambda: get_cluster(control_plane).status.state == "failed")
+        assert "minReplicas" in get_cluster(control_plane).status.reason
+
+
+class TestGcsFaultTolerance:
+    def test_redis_cleanup_finalizer_machine(self, control_plane):
+        cluster = simple_raycluster("demo", gcsFaultToleranceOptions={
+            "backend": "redis", "redisAddress": "redis://r:6379"})
+        control_plane.client.create(cluster)
+        assert control_plane.wait_for(
+            lambda: C.GCS_FT_REDIS_CLEANUP_FINALIZER in
+            (get_cluster(control_plane).metadata.finalizers or []))
+        assert control_plane.wait_cluster_state("default", "demo", "ready")
+        control_plane.server.delete("RayCluster", "default", "demo")
+        # cleanup job created, completes (sim kubelet), finalizer removed, CR gone
+        assert control_plane.wait_for(
+            lambda: control_plane.server.try_get("RayCluster", "default", "demo") is None,
+            timeout=15)
+
+    def test_embedded_backend_creates_pvc(self, control_plane):
+        cluster = simple_raycluster("demo", gcsFaultToleranceOptions={
+            "backend": "embedded", "storage": {"size": "2Gi"}})
+        control_plane.client.create(cluster)
+        assert control_plane.wait_for(
+            lambda: control_plane.server.try_get(
+                "PersistentVolumeClaim", "default", "demo-gcs-pvc"))
+
+
+class TestAutoscalerRBAC:
+    def test_rbac_objects_created(self, control_plane):
+        control_plane.client.create(
+            simple_raycluster("demo", enableInTreeAutoscaling=True))
+        assert control_plane.wait_for(
+            lambda: control_plane.server.try_get("ServiceAccount", "default", "demo"))
+        role = control_plane.server.get("Role", "default", "demo")
+        resources = {r for rule in role["rules"] for r in rule["resources"]}
+        assert "rayclusters" in resources and "pods" in resources

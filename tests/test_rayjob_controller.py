@@ -1,0 +1,192 @@
+"""RayJob state-machine tests (reference analog: rayjob_controller_test.go +
+rayjob_controller_suspended_test.go)."""
+import time
+
+import pytest
+
+from kuberay_amd.models import RayCluster, RayJob
+from kuberay_amd.testing import simple_raycluster
+from kuberay_amd.utils import constants as C
+
+
+def make_rayjob(name="job1", **spec_overrides):
+    spec = {
+        "entrypoint": "python script.py",
+        "rayClusterSpec": simple_raycluster("x", workers=1).spec.to_dict(),
+    }
+    spec.update(spec_overrides)
+    return RayJob.from_dict({
+        "apiVersion": "ray.io/v1", "kind": "RayJob",
+        "metadata": {"name": name, "namespace": "default"},
+        "spec": spec,
+    })
+
+
+def job_of(cp, name="job1"):
+    return cp.client.get(RayJob, "default", name)
+
+
+def wait_deployment_status(cp, name, status, timeout=20):
+    return cp.wait_for(
+        lambda: job_of(cp, name).status.job_deployment_status == status, timeout)
+
+
+class TestHappyPath:
+    def test_k8s_job_mode_completes(self, control_plane):
+        control_plane.client.create(make_rayjob())
+        assert wait_deployment_status(control_plane, "job1", "Complete")
+        job = job_of(control_plane)
+        assert job.status.job_status == "SUCCEEDED"
+        assert job.status.job_id.startswith("job1-")
+        assert job.status.start_time and job.status.end_time
+        assert job.status.succeeded == 1
+        # submitter K8s Job was created
+        assert control_plane.server.try_get("Job", "default", "job1") is not None
+
+    def test_cluster_created_with_owner_labels(self, control_plane):
+        control_plane.client.create(make_rayjob())
+        assert control_plane.wait_for(
+            lambda: job_of(control_plane).status.ray_cluster_name)
+        name = job_of(control_plane).status.ray_cluster_name
+        assert control_plane.wait_for(
+            lambda: control_plane.server.try_get("RayCluster", "default", name))
+        rc = control_plane.server.get("RayCluster", "default", name)
+        labels = rc["metadata"]["labels"]
+        assert labels[C.RAY_ORIGINATED_FROM_CR_NAME_LABEL_KEY] == "job1"
+        assert labels[C.RAY_ORIGINATED_FROM_CRD_LABEL_KEY] == "RayJob"
+
+    def test_shutdown_after_job_finishes(self, control_plane):
+        control_plane.client.create(make_rayjob(shutdownAfterJobFinishes=True))
+        assert wait_deployment_status(control_plane, "job1", "Complete")
+        assert control_plane.wait_for(
+            lambda: control_plane.server.count("RayCluster") == 0)
+
+    def test_ttl_seconds_after_finished(self, control_plane):
+        control_plane.client.create(make_rayjob(
+            shutdownAfterJobFinishes=True, ttlSecondsAfterFinished=1))
+        assert wait_deployment_status(control_plane, "job1", "Complete")
+        # cluster still exists within TTL
+        assert control_plane.server.count("RayCluster") == 1
+        assert control_plane.wait_for(
+            lambda: control_plane.server.count("RayCluster") == 0, timeout=10)
+
+    def test_http_mode(self, control_plane):
+        control_plane.client.create(make_rayjob(submissionMode="HTTPMode"))
+        assert wait_deployment_status(control_plane, "job1", "Complete")
+        # no submitter K8s Job in HTTP mode
+        assert control_plane.server.try_get("Job", "default", "job1") is None
+
+    def test_sidecar_mode_injects_submitter_container(self, control_plane):
+        control_plane.client.create(make_rayjob(submissionMode="SidecarMode"))
+        assert control_plane.wait_for(
+            lambda: control_plane.server.count("RayCluster") == 1)
+        rc = control_plane.server.list("RayCluster")[0]
+        containers = rc["spec"]["headGroupSpec"]["template"]["spec"]["containers"]
+        assert any(c["name"] == "ray-job-submitter" for c in containers)
+
+
+class TestFailure:
+    def test_app_failure_marks_failed(self, control_plane):
+        control_plane.dashboard.get_job_info_mock = lambda job_id: {
+            "submission_id": job_id, "status": "FAILED", "message": "boom"}
+        control_plane.client.create(make_rayjob())
+        assert wait_deployment_status(control_plane, "job1", "Failed")
+        job = job_of(control_plane)
+        assert job.status.reason == "AppFailed"
+        assert job.status.failed == 1
+
+    def test_backoff_limit_retries(self, control_plane):
+        calls = {"n": 0}
+        def mock(job_id):
+            calls["n"] += 1
+            return {"submission_id": job_id, "status": "FAILED"}
+        control_plane.dashboard.get_job_info_mock = mock
+        control_plane.client.create(make_rayjob(backoffLimit=1))
+        assert wait_deployment_status(control_plane, "job1", "Failed", timeout=30)
+        job = job_of(control_plane)
+        assert job.status.failed == 2  # initial attempt + 1 retry
+
+    def test_validation_failure(self, control_plane):
+        bad = make_rayjob()
+        bad.spec.entrypoint = None
+        control_plane.client.create(bad)
+        assert wait_deployment_status(control_plane, "job1", "ValidationFailed")
+
+    def test_active_deadline(self, control_plane):
+        control_plane.dashboard.get_job_info_mock = lambda job_id: {
+            "submission_id": job_id, "status": "RUNNING"}
+        control_plane.client.create(make_rayjob(activeDeadlineSeconds=1))
+        assert wait_deployment_status(control_plane, "job1", "Failed", timeout=30)
+        assert job_of(control_plane).status.reason == "DeadlineExceeded"
+
+
+class TestSuspend:
+    def test_created_suspended(self, control_plane):
+        control_plane.client.create(make_rayjob(suspend=True))
+        assert wait_deployment_status(control_plane, "job1", "Suspended")
+        assert control_plane.server.count("RayCluster") == 0
+
+    def test_suspend_running_job_tears_down_cluster(self, control_plane):
+        control_plane.dashboard.get_job_info_mock = lambda job_id: {
+            "submission_id": job_id, "status": "RUNNING"}
+        control_plane.client.create(make_rayjob())
+        assert wait_deployment_status(control_plane, "job1", "Running")
+        job = job_of(control_plane)
+        job.spec.suspend = True
+        control_plane.client.update(job)
+        assert wait_deployment_status(control_plane, "job1", "Suspended")
+        assert control_plane.wait_for(
+            lambda: control_plane.server.count("RayCluster") == 0)
+
+    def test_resume_creates_fresh_cluster(self, control_plane):
+        control_plane.client.create(make_rayjob(suspend=True))
+        assert wait_deployment_status(control_plane, "job1", "Suspended")
+        job = job_of(control_plane)
+        job.spec.suspend = False
+        control_plane.client.update(job)
+        assert wait_deployment_status(control_plane, "job1", "Complete", timeout=30)
+
+
+class TestDeletionPolicies:
+    def test_deletion_rules_delete_self(self, control_plane):
+        control_plane.client.create(make_rayjob(deletionStrategy={
+            "deletionRules": [{"policy": "DeleteSelf",
+                               "condition": {"jobStatus": "SUCCEEDED"}}]}))
+        assert control_plane.wait_for(
+            lambda: control_plane.client.try_get(RayJob, "default", "job1") is None,
+            timeout=30)
+
+    def test_deletion_rules_delete_workers(self, control_plane):
+        control_plane.client.create(make_rayjob(deletionStrategy={
+            "deletionRules": [{"policy": "DeleteWorkers",
+                               "condition": {"jobStatus": "SUCCEEDED"}}]}))
+        assert wait_deployment_status(control_plane, "job1", "Complete")
+        def workers_suspended():
+            clusters = control_plane.server.list("RayCluster")
+            return clusters and all(
+                g.get("suspend") for c in clusters
+                for g in c["spec"].get("workerGroupSpecs", []))
+        assert control_plane.wait_for(workers_suspended)
+
+    def test_legacy_on_success_delete_cluster(self, control_plane):
+        control_plane.client.create(make_rayjob(deletionStrategy={
+            "onSuccess": {"policy": "DeleteCluster"},
+            "onFailure": {"policy": "DeleteNone"}}))
+        assert wait_deployment_status(control_plane, "job1", "Complete")
+        assert control_plane.wait_for(
+            lambda: control_plane.server.count("RayCluster") == 0)
+
+
+class TestClusterSelector:
+    def test_uses_existing_cluster(self, control_plane):
+        existing = simple_raycluster("shared", workers=1)
+        existing.metadata.labels = {"pool": "a"}
+        control_plane.client.create(existing)
+        assert control_plane.wait_cluster_state("default", "shared", "ready")
+        control_plane.client.create(make_rayjob(
+            clusterSelector={"pool": "a"}, rayClusterSpec=None))
+        assert wait_deployment_status(control_plane, "job1", "Complete")
+        job = job_of(control_plane)
+        assert job.status.ray_cluster_name == "shared"
+        # selected cluster must never be deleted by the job
+        assert control_plane.server.try_get("RayCluster", "default", "shared")
