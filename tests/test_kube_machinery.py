@@ -1,0 +1,222 @@
+"""In-memory apiserver / workqueue / expectations / cron unit tests."""
+import threading
+import time
+
+import pytest
+
+from kuberay_amd.kube.store import (
+    AlreadyExistsError,
+    ConflictError,
+    InMemoryApiServer,
+    NotFoundError,
+)
+from kuberay_amd.kube.workqueue import RateLimitingQueue
+from kuberay_amd.ops.expectations import FakeScaleExpectations, RayClusterScaleExpectations
+from kuberay_amd.utils.cron import parse_cron
+import datetime as dt
+
+
+def mk(kind, name, namespace="default", labels=None, **extra):
+    obj = {"apiVersion": "v1", "kind": kind,
+           "metadata": {"name": name, "namespace": namespace}}
+    if labels:
+        obj["metadata"]["labels"] = labels
+    obj.update(extra)
+    return obj
+
+
+class TestStore:
+    def test_create_get_uid_rv(self):
+        s = InMemoryApiServer()
+        out = s.create(mk("Pod", "p1"))
+        assert out["metadata"]["uid"]
+        assert out["metadata"]["resourceVersion"] == "1"
+        assert s.get("Pod", "default", "p1")["metadata"]["name"] == "p1"
+
+    def test_duplicate_create_conflicts(self):
+        s = InMemoryApiServer()
+        s.create(mk("Pod", "p1"))
+        with pytest.raises(AlreadyExistsError):
+            s.create(mk("Pod", "p1"))
+
+    def test_generate_name(self):
+        s = InMemoryApiServer()
+        out = s.create({"kind": "Pod", "metadata": {"generateName": "x-"}})
+        assert out["metadata"]["name"].startswith("x-")
+
+    def test_optimistic_concurrency(self):
+        s = InMemoryApiServer()
+        a = s.create(mk("Pod", "p1"))
+        b = s.get("Pod", "default", "p1")
+        a["spec"] = {"x": 1}
+        s.update(a)
+        b["spec"] = {"x": 2}
+        with pytest.raises(ConflictError):
+            s.update(b)
+
+    def test_generation_bumps_on_spec_change_only(self):
+        s = InMemoryApiServer()
+        obj = s.create(mk("RayCluster", "c1", spec={"a": 1}))
+        assert obj["metadata"]["generation"] == 1
+        obj["spec"] = {"a": 2}
+        obj = s.update(obj)
+        assert obj["metadata"]["generation"] == 2
+        obj["status"] = {"s": 1}
+        obj2 = s.update(obj, subresource="status")
+        assert obj2["metadata"]["generation"] == 2
+        assert obj2["status"] == {"s": 1}
+
+    def test_status_subresource_does_not_touch_spec(self):
+        s = InMemoryApiServer()
+        obj = s.create(mk("RayCluster", "c1", spec={"a": 1}))
+        stale = dict(obj)
+        stale["spec"] = {"a": 999}
+        stale["status"] = {"ok": True}
+        out = s.update(stale, subresource="status")
+        assert out["spec"] == {"a": 1}
+        assert out["status"] == {"ok": True}
+
+    def test_label_selector_list(self):
+        s = InMemoryApiServer()
+        s.create(mk("Pod", "a", labels={"g": "1", "t": "w"}))
+        s.create(mk("Pod", "b", labels={"g": "2", "t": "w"}))
+        s.create(mk("Pod", "c", labels={"g": "1", "t": "h"}))
+        assert len(s.list("Pod", label_selector={"g": "1"})) == 2
+        assert len(s.list("Pod", label_selector={"g": "1", "t": "w"})) == 1
+        assert len(s.list("Pod")) == 3
+
+    def test_label_index_updated_on_update(self):
+        s = InMemoryApiServer()
+        obj = s.create(mk("Pod", "a", labels={"g": "1"}))
+        obj["metadata"]["labels"] = {"g": "2"}
+        s.update(obj)
+        assert len(s.list("Pod", label_selector={"g": "1"})) == 0
+        assert len(s.list("Pod", label_selector={"g": "2"})) == 1
+
+    def test_finalizer_blocks_deletion(self):
+        s = InMemoryApiServer()
+        obj = mk("RayCluster", "c1")
+        obj["metadata"]["finalizers"] = ["f1"]
+        s.create(obj)
+        s.delete("RayCluster", "default", "c1")
+        cur = s.get("RayCluster", "default", "c1")
+        assert cur["metadata"]["deletionTimestamp"]
+        cur["metadata"]["finalizers"] = []
+        s.update(cur)
+        assert s.try_get("RayCluster", "default", "c1") is None
+
+    def test_owner_gc_cascade(self):
+        s = InMemoryApiServer()
+        owner = s.create(mk("RayCluster", "c1"))
+        child = mk("Pod", "p1")
+        child["metadata"]["ownerReferences"] = [{
+            "kind": "RayCluster", "name": "c1", "uid": owner["metadata"]["uid"]}]
+        s.create(child)
+        s.delete("RayCluster", "default", "c1")
+        assert s.try_get("Pod", "default", "p1") is None
+
+    def test_watch_events(self):
+        s = InMemoryApiServer()
+        w = s.watch({"Pod"})
+        s.create(mk("Pod", "p1"))
+        s.create(mk("Service", "s1"))
+        ev = w.next(timeout=1)
+        assert ev[0] == "ADDED" and ev[1]["metadata"]["name"] == "p1"
+        assert w.next(timeout=0.1) is None  # Service filtered out
+        w.stop()
+
+    def test_patch_merge(self):
+        s = InMemoryApiServer()
+        s.create(mk("Pod", "p1", labels={"a": "1"}))
+        s.patch_merge("Pod", "default", "p1", {"metadata": {"labels": {"b": "2"}}})
+        assert s.get("Pod", "default", "p1")["metadata"]["labels"] == {"a": "1", "b": "2"}
+
+
+class TestWorkqueue:
+    def test_dedup_while_queued(self):
+        q = RateLimitingQueue()
+        q.add("a"); q.add("a"); q.add("b")
+        assert len(q) == 2
+
+    def test_dirty_requeue_while_processing(self):
+        q = RateLimitingQueue()
+        q.add("a")
+        item = q.get()
+        q.add("a")  # re-added while processing -> dirty
+        q.done(item)
+        assert q.get(timeout=0.5) == "a"
+
+    def test_add_after(self):
+        q = RateLimitingQueue()
+        q.add_after("a", 0.1)
+        assert q.get(timeout=0.02) is None
+        assert q.get(timeout=1.0) == "a"
+
+    def test_rate_limited_backoff_grows(self):
+        q = RateLimitingQueue(base_delay=0.01, max_delay=1.0)
+        t0 = time.monotonic()
+        q.add_rate_limited("a")
+        assert q.get(timeout=2.0) == "a"
+        q.done("a")
+        q.add_rate_limited("a")
+        assert q.get(timeout=2.0) == "a"
+        assert time.monotonic() - t0 >= 0.02
+
+
+class TestExpectations:
+    def test_create_expectation_blocks_until_observed(self):
+        s = InMemoryApiServer()
+        e = RayClusterScaleExpectations()
+        e.expect_create_pod("default", "c1", "g1", "pod-x")
+        assert not e.is_satisfied(s, "default", "c1", "g1")
+        s.create(mk("Pod", "pod-x"))
+        assert e.is_satisfied(s, "default", "c1", "g1")
+
+    def test_delete_expectation(self):
+        s = InMemoryApiServer()
+        s.create(mk("Pod", "pod-x"))
+        e = RayClusterScaleExpectations()
+        e.expect_delete_pod("default", "c1", "g1", "pod-x")
+        assert not e.is_satisfied(s, "default", "c1", "g1")
+        s.delete("Pod", "default", "pod-x")
+        assert e.is_satisfied(s, "default", "c1", "g1")
+
+    def test_timeout_unblocks(self):
+        s = InMemoryApiServer()
+        e = RayClusterScaleExpectations(timeout_s=0.05)
+        e.expect_create_pod("default", "c1", "g1", "never-created")
+        time.sleep(0.1)
+        assert e.is_satisfied(s, "default", "c1", "g1")
+
+    def test_fake_always_satisfied(self):
+        e = FakeScaleExpectations()
+        e.expect_create_pod("d", "c", "g", "p")
+        assert e.is_satisfied(None, "d", "c", "g")
+
+
+class TestCron:
+    def test_parse_basic(self):
+        sched = parse_cron("*/5 * * * *")
+        nxt = sched.next_after(dt.datetime(2026, 1, 1, 0, 1))
+        assert nxt == dt.datetime(2026, 1, 1, 0, 5)
+
+    def test_daily(self):
+        sched = parse_cron("30 2 * * *")
+        nxt = sched.next_after(dt.datetime(2026, 1, 1, 3, 0))
+        assert nxt == dt.datetime(2026, 1, 2, 2, 30)
+
+    def test_macro(self):
+        sched = parse_cron("@hourly")
+        nxt = sched.next_after(dt.datetime(2026, 1, 1, 0, 30))
+        assert nxt == dt.datetime(2026, 1, 1, 1, 0)
+
+    def test_invalid(self):
+        with pytest.raises(ValueError):
+            parse_cron("61 * * * *")
+        with pytest.raises(ValueError):
+            parse_cron("* * *")
+
+    def test_dow(self):
+        sched = parse_cron("0 0 * * 1")  # Mondays
+        nxt = sched.next_after(dt.datetime(2026, 1, 1))  # Thursday
+        assert nxt == dt.datetime(2026, 1, 5)

@@ -1,0 +1,171 @@
+"""RayService controller tests (reference analog: rayservice_controller_test.go)."""
+import time
+
+import pytest
+
+from kuberay_amd.models import RayCluster, RayService
+from kuberay_amd.models.rayservice import ApplicationStatus
+from kuberay_amd.testing import simple_raycluster
+from kuberay_amd.utils import constants as C
+
+SERVE_CONFIG = """\
+applications:
+- name: app1
+  import_path: mod.graph
+  deployments:
+  - name: D1
+  - name: D2
+"""
+
+
+def make_rayservice(name="svc1", **spec_overrides):
+    spec = {
+        "serveConfigV2": SERVE_CONFIG,
+        "rayClusterConfig": simple_raycluster("x", workers=1).spec.to_dict(),
+    }
+    spec.update(spec_overrides)
+    return RayService.from_dict({
+        "apiVersion": "ray.io/v1", "kind": "RayService",
+        "metadata": {"name": name, "namespace": "default"},
+        "spec": spec,
+    })
+
+
+def svc_of(cp, name="svc1"):
+    return cp.client.get(RayService, "default", name)
+
+
+def wait_ready(cp, name="svc1", timeout=20):
+    return cp.wait_for(lambda: svc_of(cp, name).condition_true("Ready"), timeout)
+
+
+@pytest.fixture()
+def fast_gc(control_plane):
+    control_plane.rayservice_reconciler.cluster_deletion_delay_s = 0.2
+    return control_plane
+
+
+class TestDeploy:
+    def test_service_becomes_ready(self, control_plane):
+        control_plane.client.create(make_rayservice())
+        assert wait_ready(control_plane)
+        svc = svc_of(control_plane)
+        assert svc.status.service_status == "Running"
+        assert svc.status.num_serve_endpoints > 0
+        active = svc.status.active_service_status
+        assert active.ray_cluster_name
+        assert active.applications["app1"].status == ApplicationStatus.RUNNING
+        assert set(active.applications["app1"].deployments) == {"D1", "D2"}
+
+    def test_services_created_and_point_to_active(self, control_plane):
+        control_plane.client.create(make_rayservice())
+        assert wait_ready(control_plane)
+        active = svc_of(control_plane).status.active_service_status.ray_cluster_name
+        head = control_plane.server.get("Service", "default", "svc1-head-svc")
+        serve = control_plane.server.get("Service", "default", "svc1-serve-svc")
+        assert head["spec"]["selector"][C.RAY_CLUSTER_LABEL_KEY] == active
+        assert serve["spec"]["selector"][C.RAY_CLUSTER_LABEL_KEY] == active
+        assert serve["spec"]["selector"][C.RAY_CLUSTER_SERVING_SERVICE_LABEL_KEY] == "true"
+
+    def test_serve_config_submitted_once_per_config(self, control_plane):
+        control_plane.client.create(make_rayservice())
+        assert wait_ready(control_plane)
+        time.sleep(0.3)  # several reconcile cycles
+        assert len(control_plane.dashboard.update_serve_calls) == 1
+
+    def test_head_pod_serve_label_flipped(self, control_plane):
+        control_plane.client.create(make_rayservice())
+        assert wait_ready(control_plane)
+        active = svc_of(control_plane).status.active_service_status.ray_cluster_name
+        def head_labeled():
+            pods = control_plane.server.list(
+                "Pod", "default", {C.RAY_CLUSTER_LABEL_KEY: active,
+                                   C.RAY_NODE_TYPE_LABEL_KEY: "head"})
+            return pods and pods[0]["metadata"]["labels"].get(
+                C.RAY_CLUSTER_SERVING_SERVICE_LABEL_KEY) == "true"
+        assert control_plane.wait_for(head_labeled)
+
+
+class TestZeroDowntimeUpgrade:
+    def test_spec_change_promotes_new_cluster(self, fast_gc):
+        cp = fast_gc
+        cp.client.create(make_rayservice())
+        assert wait_ready(cp)
+        old_active = svc_of(cp).status.active_service_status.ray_cluster_name
+
+        svc = svc_of(cp)
+        svc.spec.ray_cluster_spec.worker_group_specs[0].template.spec \
+            .containers[0].image = "rayproject/ray:2.47.0-rocm"
+        cp.client.update(svc)
+
+        def promoted():
+            s = svc_of(cp)
+            return (s.status.active_service_status.ray_cluster_name
+                    not in (None, old_active) and s.condition_true("Ready"))
+        assert cp.wait_for(promoted, timeout=25)
+        # old cluster eventually GC'd after the deletion delay
+        assert cp.wait_for(
+            lambda: cp.server.try_get("RayCluster", "default", old_active) is None,
+            timeout=15)
+
+    def test_replicas_only_change_updates_in_place(self, control_plane):
+        cp = control_plane
+        cp.client.create(make_rayservice())
+        assert wait_ready(cp)
+        active = svc_of(cp).status.active_service_status.ray_cluster_name
+        svc = svc_of(cp)
+        svc.spec.ray_cluster_spec.worker_group_specs[0].replicas = 3
+        cp.client.update(svc)
+        def scaled():
+            rc = cp.client.try_get(RayCluster, "default", active)
+            return rc is not None and rc.status.available_worker_replicas == 3
+        assert cp.wait_for(scaled, timeout=20)
+        # no new cluster was prepared
+        assert svc_of(cp).status.active_service_status.ray_cluster_name == active
+
+    def test_upgrade_strategy_none_blocks_new_cluster(self, control_plane):
+        cp = control_plane
+        cp.client.create(make_rayservice(upgradeStrategy={"type": "None"}))
+        assert wait_ready(cp)
+        active = svc_of(cp).status.active_service_status.ray_cluster_name
+        svc = svc_of(cp)
+        svc.spec.ray_cluster_spec.worker_group_specs[0].template.spec \
+            .containers[0].image = "rayproject/ray:2.47.0-rocm"
+        cp.client.update(svc)
+        time.sleep(0.5)
+        s = svc_of(cp)
+        assert s.status.active_service_status.ray_cluster_name == active
+        assert not s.status.pending_service_status.ray_cluster_name
+
+
+class TestSuspend:
+    def test_suspend_and_resume(self, control_plane):
+        cp = control_plane
+        cp.client.create(make_rayservice())
+        assert wait_ready(cp)
+        svc = svc_of(cp)
+        svc.spec.suspend = True
+        cp.client.update(svc)
+        assert cp.wait_for(lambda: svc_of(cp).condition_true("Suspended"), timeout=15)
+        assert cp.wait_for(lambda: cp.server.count("RayCluster") == 0)
+        assert svc_of(cp).status.num_serve_endpoints == 0
+
+        svc = svc_of(cp)
+        svc.spec.suspend = False
+        cp.client.update(svc)
+        assert cp.wait_for(lambda: wait_ready(cp), timeout=25)
+
+
+class TestUnhealthy:
+    def test_deploy_failed_app_keeps_service_not_ready(self, control_plane):
+        cp = control_plane
+        cp.dashboard.serve_statuses_mock = {"applications": {
+            "app1": {"status": "DEPLOY_FAILED", "message": "import error",
+                     "deployments": {}}}}
+        cp.client.create(make_rayservice())
+        time.sleep(1.0)
+        s = svc_of(cp)
+        # first cluster stays pending; service never flips Ready
+        assert not s.condition_true("Ready")
+        apps = s.status.pending_service_status.applications or {}
+        assert apps.get("app1") and apps["app1"].status == "DEPLOY_FAILED"

@@ -1,0 +1,199 @@
+"""Utils unit tests: names, quantities, resources, hashing, validation
+(reference analogs: utils/util_test.go, utils/validation_test.go)."""
+import pytest
+
+from kuberay_amd.models import RayCluster, RayJob
+from kuberay_amd.testing import simple_raycluster
+from kuberay_amd.utils import constants as C
+from kuberay_amd.utils import names
+from kuberay_amd.utils.hashing import hash_without_replicas_and_workers_to_delete
+from kuberay_amd.utils.quantity import add_quantities, format_quantity, parse_quantity
+from kuberay_amd.utils.resources import (
+    calculate_desired_replicas,
+    calculate_desired_resources,
+    calculate_max_replicas,
+    calculate_min_replicas,
+    container_gpu_count,
+    is_amd_gpu_resource,
+    worker_group_desired_replicas,
+)
+from kuberay_amd.utils.validation import (
+    validate_raycluster_spec,
+    validate_rayjob_spec,
+    validate_rayservice_spec,
+)
+
+
+class TestNames:
+    def test_check_name_truncates_from_front(self):
+        long = "a" * 60
+        out = names.check_name(long)
+        assert len(out) == 50
+
+    def test_check_name_fixes_leading_digit(self):
+        assert names.check_name("9abc") == "rabc"
+
+    def test_pod_name(self):
+        assert names.pod_name("demo", "head", True) == "demo-head-"
+        assert names.pod_name("demo", "worker", False) == "demo-worker"
+        long = "x" * 60
+        assert len(names.pod_name(long, "worker", True)) == 50 + len("-worker-")
+
+    def test_head_service_name(self):
+        cluster = simple_raycluster("demo")
+        assert names.head_service_name("RayCluster", cluster.spec, "demo") == "demo-head-svc"
+        assert names.head_service_name("RayService", cluster.spec, "svc") == "svc-head-svc"
+
+    def test_fqdn(self):
+        cluster = simple_raycluster("demo")
+        assert names.fqdn_service_name(cluster, "ns1") == \
+            "demo-head-svc.ns1.svc.cluster.local"
+        assert names.extract_ray_ip_from_fqdn("a.b.svc.cluster.local") == "a"
+
+    def test_check_label(self):
+        assert len(names.check_label("a" * 80)) == 63
+
+
+class TestQuantity:
+    @pytest.mark.parametrize("value,expected", [
+        ("500m", 0.5), ("2", 2), ("2Gi", 2 * 1024**3), ("1k", 1000),
+        ("100Mi", 100 * 1024**2), (None, 0), ("0.5", 0.5), ("1e3", 1000),
+    ])
+    def test_parse(self, value, expected):
+        assert float(parse_quantity(value)) == expected
+
+    def test_format_roundtrip(self):
+        assert format_quantity(parse_quantity("2Gi")) == "2Gi"
+        assert format_quantity(parse_quantity("500m")) == "500m"
+        assert format_quantity(parse_quantity("3")) == "3"
+
+    def test_add(self):
+        assert add_quantities("500m", "1500m") == "2"
+        assert add_quantities("1Gi", "1Gi") == "2Gi"
+
+
+class TestResources:
+    def test_amd_gpu_only(self):
+        assert is_amd_gpu_resource("amd.com/gpu")
+        assert not is_amd_gpu_resource("nvidia.com/gpu")
+        assert not is_amd_gpu_resource("nvidia.com/mig-1g.5gb")
+        assert not is_amd_gpu_resource("google.com/tpu")
+
+    def test_worker_group_replicas_clamped(self):
+        cluster = simple_raycluster("demo", workers=5)
+        g = cluster.spec.worker_group_specs[0]
+        g.min_replicas, g.max_replicas = 2, 3
+        assert worker_group_desired_replicas(g) == 3
+        g.replicas = 1
+        assert worker_group_desired_replicas(g) == 2
+        g.suspend = True
+        assert worker_group_desired_replicas(g) == 0
+
+    def test_desired_counts_num_of_hosts(self):
+        cluster = simple_raycluster("demo", workers=2, num_of_hosts=4)
+        assert calculate_desired_replicas(cluster) == 8
+
+    def test_min_max(self):
+        cluster = simple_raycluster("demo", workers=2)
+        g = cluster.spec.worker_group_specs[0]
+        g.min_replicas, g.max_replicas = 1, 10
+        assert calculate_min_replicas(cluster) == 1
+        assert calculate_max_replicas(cluster) == 10
+
+    def test_desired_resources_sums_head_and_workers(self):
+        cluster = simple_raycluster("demo", workers=2, gpus_per_worker=2)
+        totals = calculate_desired_resources(cluster)
+        assert totals["desiredGPU"] == "4"
+        assert totals["desiredCPU"] == "3"  # head 1 + 2 workers * 1
+
+
+class TestHashing:
+    def test_stable_under_scale_fields(self):
+        cluster = simple_raycluster("demo")
+        h1 = hash_without_replicas_and_workers_to_delete(cluster.spec)
+        cluster.spec.worker_group_specs[0].replicas = 99
+        cluster.spec.worker_group_specs[0].min_replicas = 9
+        cluster.spec.worker_group_specs[0].scale_strategy.workers_to_delete = ["a"]
+        assert hash_without_replicas_and_workers_to_delete(cluster.spec) == h1
+
+    def test_changes_on_template_change(self):
+        cluster = simple_raycluster("demo")
+        h1 = hash_without_replicas_and_workers_to_delete(cluster.spec)
+        cluster.spec.worker_group_specs[0].template.spec.containers[0].image = "z"
+        assert hash_without_replicas_and_workers_to_delete(cluster.spec) != h1
+
+    def test_tolerations_muted(self):
+        cluster = simple_raycluster("demo")
+        h1 = hash_without_replicas_and_workers_to_delete(cluster.spec)
+        cluster.spec.worker_group_specs[0].template.spec.tolerations = [
+            {"key": "k", "operator": "Exists"}]
+        assert hash_without_replicas_and_workers_to_delete(cluster.spec) == h1
+
+
+class TestValidation:
+    def test_valid_cluster(self):
+        assert validate_raycluster_spec(simple_raycluster("demo")) == []
+
+    def test_min_gt_max(self):
+        c = simple_raycluster("demo")
+        c.spec.worker_group_specs[0].min_replicas = 5
+        c.spec.worker_group_specs[0].max_replicas = 2
+        assert any("minReplicas" in e for e in validate_raycluster_spec(c))
+
+    def test_duplicate_group_names(self):
+        c = simple_raycluster("demo")
+        c.spec.worker_group_specs.append(c.spec.worker_group_specs[0].clone())
+        assert any("duplicate" in e for e in validate_raycluster_spec(c))
+
+    def test_gcs_ft_embedded_rejects_redis_fields(self):
+        c = simple_raycluster("demo", gcsFaultToleranceOptions={
+            "backend": "embedded", "redisAddress": "redis://x"})
+        assert any("embedded" in e for e in validate_raycluster_spec(c))
+
+    def test_idle_timeout_requires_v2(self):
+        c = simple_raycluster("demo")
+        c.spec.worker_group_specs[0].idle_timeout_seconds = 60
+        assert any("v2" in e for e in validate_raycluster_spec(c))
+        c.spec.autoscaler_options = type(c.spec).model_fields["autoscaler_options"] \
+            .annotation.__args__[0](version="v2")
+        assert validate_raycluster_spec(c) == []
+
+    def test_rayjob_requires_cluster_source(self):
+        job = RayJob.from_dict({
+            "apiVersion": "ray.io/v1", "kind": "RayJob",
+            "metadata": {"name": "j"},
+            "spec": {"entrypoint": "x"}})
+        assert any("rayClusterSpec" in e for e in validate_rayjob_spec(job))
+
+    def test_rayjob_ttl_requires_shutdown(self):
+        job = RayJob.from_dict({
+            "apiVersion": "ray.io/v1", "kind": "RayJob",
+            "metadata": {"name": "j"},
+            "spec": {"entrypoint": "x", "ttlSecondsAfterFinished": 10,
+                     "rayClusterSpec": simple_raycluster("x").spec.to_dict()}})
+        assert any("shutdownAfterJobFinishes" in e for e in validate_rayjob_spec(job))
+
+    def test_rayjob_deletion_rules_exclusive(self):
+        job = RayJob.from_dict({
+            "apiVersion": "ray.io/v1", "kind": "RayJob",
+            "metadata": {"name": "j"},
+            "spec": {"entrypoint": "x",
+                     "rayClusterSpec": simple_raycluster("x").spec.to_dict(),
+                     "deletionStrategy": {
+                         "onSuccess": {"policy": "DeleteCluster"},
+                         "onFailure": {"policy": "DeleteNone"},
+                         "deletionRules": [{"policy": "DeleteSelf",
+                                            "condition": {"jobStatus": "FAILED"}}]}}})
+        assert any("mutually exclusive" in e for e in validate_rayjob_spec(job))
+
+    def test_rayservice_incremental_requires_gateway(self):
+        from kuberay_amd.models import RayService
+        svc = RayService.from_dict({
+            "apiVersion": "ray.io/v1", "kind": "RayService",
+            "metadata": {"name": "s"},
+            "spec": {
+                "serveConfigV2": "applications:\n- name: a\n",
+                "rayClusterConfig": simple_raycluster("x").spec.to_dict(),
+                "upgradeStrategy": {"type": "NewClusterWithIncrementalUpgrade"}}})
+        errs = validate_rayservice_spec(svc)
+        assert any("clusterUpgradeOptions" in e for e in errs)
