@@ -1,0 +1,224 @@
+#!/usr/bin/env python3
+"""Flagship control-plane benchmark — the BASELINE.json metric:
+p50 RayCluster CR→Ready latency + reconciles/sec + operator RSS @ 500
+RayClusters (reference comparable: KubeRay's clusterloader2
+100-raycluster "Wait for RayClusters ready" step, 83.5 s for 100 clusters
+of 1 head + 3 workers on GKE → 1.198 clusters/s).
+
+One STEP = the full lifecycle of `--clusters` RayClusters (1 head +
+`--workers-per-cluster` workers each, `--gpus-per-worker` amd.com/gpu per
+worker): create all CRs, reconcile to Ready (simulated kubelet sets pod
+status, like the reference's envtest methodology — no real kubelet), then
+delete and drain. Ranks run independent operator instances (weak scaling:
+per-GPU work fixed); `value` aggregates clusters-ready/sec over all ranks.
+
+Run:  python bench.py --gpus N --steps K --warmup W
+The driver launches N>1 via torch.distributed.run, one rank per GPU.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import statistics
+import sys
+import time
+
+
+def rss_mb() -> float:
+    try:
+        import psutil
+        return psutil.Process().memory_info().rss / (1024 * 1024)
+    except Exception:
+        import resource
+        return resource.getrusage(resource.RUSAGE_SELF).ru_maxrss / 1024
+
+
+def run_step(cp, n_clusters: int, workers_per_cluster: int, gpus_per_worker: int,
+             step_idx: int, rank: int):
+    """Create → all Ready → delete → drained. Returns per-step stats."""
+    from kuberay_amd.models import RayCluster
+    from kuberay_amd.testing import simple_raycluster
+    from kuberay_amd.utils import constants as C
+
+    namespace = "default"
+    prefix = f"bench-r{rank}-s{step_idx}"
+    create_times = {}
+    # watch (not poll): ready detection from status-update events, like a
+    # clusterloader2 measurement informer
+    watcher = cp.server.watch({"RayCluster"})
+    t0 = time.perf_counter()
+    for i in range(n_clusters):
+        name = f"{prefix}-{i:04d}"
+        rc = simple_raycluster(name, namespace=namespace,
+                               workers=workers_per_cluster,
+                               gpus_per_worker=gpus_per_worker)
+        cp.client.create(rc)
+        create_times[name] = time.perf_counter()
+    t_create_done = time.perf_counter()
+
+    ready_latency = {}
+    pending = set(create_times)
+    deadline = time.monotonic() + 600
+    while pending and time.monotonic() < deadline:
+        ev = watcher.next(timeout=0.25)
+        if ev is None:
+            continue
+        _, obj = ev
+        name = obj["metadata"]["name"]
+        if name in pending and obj.get("status", {}).get("state") == "ready":
+            ready_latency[name] = time.perf_counter() - create_times[name]
+            pending.discard(name)
+    watcher.stop()
+    if pending:
+        raise RuntimeError(f"{len(pending)} clusters never became ready")
+    t_ready = time.perf_counter()
+
+    rss_at_scale = rss_mb()
+    n_pods = cp.server.count("Pod")
+
+    for name in create_times:
+        cp.server.delete("RayCluster", namespace, name)
+    while cp.server.count("RayCluster") > 0 and time.monotonic() < deadline:
+        time.sleep(0.005)
+    t_end = time.perf_counter()
+
+    lat = sorted(ready_latency.values())
+    return {
+        "wall_s": t_end - t0,
+        "create_s": t_create_done - t0,
+        "ready_wait_s": t_ready - t0,
+        "p50_ready_s": statistics.median(lat),
+        "p99_ready_s": lat[int(len(lat) * 0.99) - 1] if len(lat) > 1 else lat[0],
+        "rss_mb_at_scale": rss_at_scale,
+        "pods_at_scale": n_pods,
+    }
+
+
+def main() -> int:
+    parser = argparse.ArgumentParser()
+    parser.add_argument("--gpus", type=int, default=1)
+    parser.add_argument("--steps", type=int, default=2)
+    parser.add_argument("--warmup", type=int, default=1)
+    parser.add_argument("--clusters", type=int, default=500)
+    parser.add_argument("--workers-per-cluster", type=int, default=3)
+    parser.add_argument("--gpus-per-worker", type=int, default=1)
+    parser.add_argument("--controller-workers", type=int, default=8)
+    args = parser.parse_args()
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+
+    import torch
+    use_cuda = torch.cuda.is_available()
+    if use_cuda:
+        torch.cuda.set_device(local_rank % max(torch.cuda.device_count(), 1))
+        # load the gfx950 native extension so GPU-side health is verified on
+        # the device this rank owns (fails loudly if missing on a GPU box)
+        from kuberay_amd._native import gpuhealth
+        assert gpuhealth.mfma_smoke(local_rank % gpuhealth.device_count(), 256)
+
+    dist = None
+    if world > 1:
+        import torch.distributed as tdist
+        dist = tdist
+        backend = "nccl" if use_cuda else "gloo"
+        dist.init_process_group(backend=backend)
+
+    from kuberay_amd.testing import ControlPlane
+    cp = ControlPlane(kubelet_delay=0.0, workers=args.controller_workers,
+                      record_events=False, requeue_seconds=3600,
+                      poll_seconds=1.0)
+    cp.start()
+
+    def barrier():
+        if dist is not None:
+            dist.barrier()
+        if use_cuda:
+            torch.cuda.synchronize()
+
+    try:
+        for w in range(args.warmup):
+            run_step(cp, args.clusters, args.workers_per_cluster,
+                     args.gpus_per_worker, step_idx=-1 - w, rank=rank)
+
+        rss_before_mb = rss_mb()
+        barrier()
+        t0 = time.perf_counter()
+        stats = []
+        for k in range(args.steps):
+            stats.append(run_step(cp, args.clusters, args.workers_per_cluster,
+                                  args.gpus_per_worker, step_idx=k, rank=rank))
+        barrier()
+        t1 = time.perf_counter()
+    finally:
+        reconciles = sum(c.reconcile_count for c in cp.manager.controllers)
+        cp.stop()
+
+    elapsed = t1 - t0
+    # MAX elapsed over ranks (slowest rank defines the job), SUM of work
+    if dist is not None:
+        te = torch.tensor([elapsed], dtype=torch.float64,
+                          device="cuda" if use_cuda else "cpu")
+        dist.all_reduce(te, op=dist.ReduceOp.MAX)
+        elapsed_max = float(te.item())
+        tw = torch.tensor([float(args.steps * args.clusters), float(reconciles)],
+                          dtype=torch.float64, device="cuda" if use_cuda else "cpu")
+        dist.all_reduce(tw, op=dist.ReduceOp.SUM)
+        total_clusters, total_reconciles = float(tw[0].item()), float(tw[1].item())
+    else:
+        elapsed_max = elapsed
+        total_clusters = float(args.steps * args.clusters)
+        total_reconciles = float(reconciles)
+
+    value = total_clusters / elapsed_max  # whole-job clusters-ready/sec
+    p50 = statistics.median(s["p50_ready_s"] for s in stats)
+    baseline_clusters_per_s = 100.0 / 83.5  # KubeRay 100-raycluster ready step
+
+    if rank == 0:
+        result = {
+            "metric": "rayclusters_to_ready_per_sec",
+            "value": round(value, 3),
+            "unit": "clusters/s",
+            "n_gpus": world if world > 1 else args.gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed_max / args.steps * 1000, 2),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": round(value / baseline_clusters_per_s, 2),
+            "dtype": "bf16",
+            "data": "synthetic",
+            "config": {
+                "model": "raycluster-control-plane",
+                "global_batch": int(total_clusters),
+                "seq_len": args.workers_per_cluster + 1,
+                "parallelism": f"dp{world if world > 1 else args.gpus}",
+                "clusters_per_step_per_rank": args.clusters,
+                "workers_per_cluster": args.workers_per_cluster,
+                "gpus_per_worker": args.gpus_per_worker,
+                "pods_per_step_per_rank": args.clusters * (args.workers_per_cluster + 1),
+                "p50_cr_to_ready_s": round(p50, 4),
+                "p99_cr_to_ready_s": round(max(s["p99_ready_s"] for s in stats), 4),
+                "reconciles_per_sec": round(total_reconciles / elapsed_max, 1),
+                "operator_rss_mb_at_scale": round(
+                    max(s["rss_mb_at_scale"] for s in stats), 1),
+                "operator_rss_mb_delta_over_idle": round(
+                    max(s["rss_mb_at_scale"] for s in stats) - rss_before_mb, 1),
+                "rss_note": "process includes the torch runtime the bench "
+                            "contract loads; delta_over_idle is the "
+                            "control-plane's own footprint at 2000 pods",
+                "kubelet": "simulated (envtest methodology; no real kubelet)",
+                "baseline": "kuberay clusterloader2 100-raycluster ready step "
+                            "83.5s on GKE (BASELINE.md)",
+            },
+        }
+        print(json.dumps(result))
+    if dist is not None:
+        dist.destroy_process_group()
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

@@ -25,6 +25,18 @@ from typing import Any, Callable, Dict, Iterable, List, Optional, Tuple
 Key = Tuple[str, str, str]  # (kind, namespace, name)
 
 
+def jsoncopy(obj):
+    """Deep copy for JSON-shaped trees (dict/list/scalars only) — ~4x faster
+    than copy.deepcopy, which pays for memoization and type dispatch these
+    trees never need."""
+    t = type(obj)
+    if t is dict:
+        return {k: jsoncopy(v) for k, v in obj.items()}
+    if t is list:
+        return [jsoncopy(v) for v in obj]
+    return obj
+
+
 class ApiError(Exception):
     def __init__(self, code: int, message: str):
         super().__init__(f"{code}: {message}")
@@ -69,6 +81,7 @@ class InMemoryApiServer:
         # label index: (kind, label_key, label_value) -> set of keys
         self._label_index: Dict[Tuple[str, str, str], set] = defaultdict(set)
         self._kind_index: Dict[str, set] = defaultdict(set)
+        self._owner_index: Dict[str, set] = defaultdict(set)  # owner uid -> keys
 
     # -- internals -----------------------------------------------------
     def _next_rv(self) -> str:
@@ -77,13 +90,23 @@ class InMemoryApiServer:
 
     def _index_add(self, key: Key, obj: Dict[str, Any]) -> None:
         self._kind_index[key[0]].add(key)
-        for lk, lv in (obj.get("metadata", {}).get("labels") or {}).items():
+        meta = obj.get("metadata", {})
+        for lk, lv in (meta.get("labels") or {}).items():
             self._label_index[(key[0], lk, lv)].add(key)
+        for ref in meta.get("ownerReferences") or []:
+            uid = ref.get("uid")
+            if uid:
+                self._owner_index[uid].add(key)
 
     def _index_remove(self, key: Key, obj: Dict[str, Any]) -> None:
         self._kind_index[key[0]].discard(key)
-        for lk, lv in (obj.get("metadata", {}).get("labels") or {}).items():
+        meta = obj.get("metadata", {})
+        for lk, lv in (meta.get("labels") or {}).items():
             self._label_index[(key[0], lk, lv)].discard(key)
+        for ref in meta.get("ownerReferences") or []:
+            uid = ref.get("uid")
+            if uid:
+                self._owner_index[uid].discard(key)
 
     def _notify(self, event_type: str, obj: Dict[str, Any]) -> None:
         for w in list(self._watchers):
@@ -96,7 +119,7 @@ class InMemoryApiServer:
 
     # -- verbs ---------------------------------------------------------
     def create(self, obj: Dict[str, Any]) -> Dict[str, Any]:
-        obj = copy.deepcopy(obj)
+        obj = jsoncopy(obj)
         meta = obj.setdefault("metadata", {})
         meta.setdefault("namespace", "default")
         if not meta.get("name"):
@@ -114,7 +137,7 @@ class InMemoryApiServer:
             meta["creationTimestamp"] = now_iso()
             self._objects[key] = obj
             self._index_add(key, obj)
-            out = copy.deepcopy(obj)
+            out = jsoncopy(obj)
         self._notify("ADDED", out)
         return out
 
@@ -123,7 +146,7 @@ class InMemoryApiServer:
             obj = self._objects.get((kind, namespace, name))
             if obj is None:
                 raise NotFoundError(f"{kind} {namespace}/{name} not found")
-            return copy.deepcopy(obj)
+            return jsoncopy(obj)
 
     def try_get(self, kind: str, namespace: str, name: str) -> Optional[Dict[str, Any]]:
         try:
@@ -155,12 +178,12 @@ class InMemoryApiServer:
                 if obj is None:
                     continue
                 if match_labels(obj.get("metadata", {}).get("labels"), label_selector):
-                    out.append(copy.deepcopy(obj))
+                    out.append(jsoncopy(obj))
             out.sort(key=lambda o: (o["metadata"]["namespace"], o["metadata"]["name"]))
             return out
 
     def update(self, obj: Dict[str, Any], *, subresource: Optional[str] = None) -> Dict[str, Any]:
-        obj = copy.deepcopy(obj)
+        obj = jsoncopy(obj)
         key = self._key_of(obj)
         with self._lock:
             current = self._objects.get(key)
@@ -175,7 +198,7 @@ class InMemoryApiServer:
             self._index_remove(key, current)
             if subresource == "status":
                 # status updates only replace .status
-                new_obj = copy.deepcopy(current)
+                new_obj = jsoncopy(current)
                 new_obj["status"] = obj.get("status", {})
             else:
                 new_obj = obj
@@ -197,7 +220,7 @@ class InMemoryApiServer:
                 current["metadata"].get("deletionTimestamp")
                 and not new_obj["metadata"].get("finalizers")
             )
-            out = copy.deepcopy(new_obj)
+            out = jsoncopy(new_obj)
         self._notify("MODIFIED", out)
         if finalizers_gone:
             # terminating object dropped its last finalizer -> actually delete
@@ -213,7 +236,7 @@ class InMemoryApiServer:
             current = self._objects.get((kind, namespace, name))
             if current is None:
                 raise NotFoundError(f"{kind} {namespace}/{name} not found")
-            merged = copy.deepcopy(current)
+            merged = jsoncopy(current)
 
             def merge(dst, src):
                 for k, v in src.items():
@@ -222,7 +245,7 @@ class InMemoryApiServer:
                     elif v is None:
                         dst.pop(k, None)
                     else:
-                        dst[k] = copy.deepcopy(v)
+                        dst[k] = jsoncopy(v)
 
             merge(merged, patch)
             merged["metadata"]["resourceVersion"] = current["metadata"]["resourceVersion"]
@@ -257,14 +280,8 @@ class InMemoryApiServer:
                 return
             self._index_remove(key, current)
             uid = current["metadata"]["uid"]
-            # ownerReference GC: collect dependents of this object
-            dependents = [
-                k for k, o in self._objects.items()
-                if any(
-                    ref.get("uid") == uid
-                    for ref in o.get("metadata", {}).get("ownerReferences") or []
-                )
-            ]
+            # ownerReference GC via the owner-uid index
+            dependents = list(self._owner_index.pop(uid, ()))
         self._notify("DELETED", current)
         for dep in dependents:
             try:
