@@ -1,0 +1,86 @@
+"""Operator configuration (reference: ray-operator/apis/config/v1alpha1/
+configuration_types.go:18-90 + defaults.go:8-13 + main.go:80-143).
+
+Flags mirror into a versioned Configuration object loadable from a YAML
+file (``--config``); explicit flags win over the file.
+"""
+from __future__ import annotations
+
+import argparse
+from typing import List, Optional
+
+import yaml
+from pydantic import Field
+
+from .kube.objects import K8sModel
+
+
+class Configuration(K8sModel):
+    api_version: str = "config.ray.io/v1alpha1"
+    kind: str = "Configuration"
+
+    metrics_addr: str = ":8080"
+    probe_addr: str = ":8082"
+    enable_leader_election: bool = True
+    leader_election_namespace: str = ""
+    reconcile_concurrency: int = 4          # reference default 1; Python
+                                            # reconcilers are pure cache fns
+    watch_namespaces: Optional[List[str]] = None
+    log_file: Optional[str] = None
+    log_file_encoder: str = "json"
+    log_stdout_encoder: str = "console"
+    batch_scheduler: Optional[str] = None   # volcano|yunikorn|scheduler-plugins|xgmi-gang
+    enable_batch_scheduler: bool = False
+    head_sidecar_containers: Optional[List[dict]] = None
+    worker_sidecar_containers: Optional[List[dict]] = None
+    feature_gates: str = ""
+    qps: float = 100.0
+    burst: int = 200
+    enable_metrics: bool = True
+    # MI355X-native knobs
+    enable_mi355x_autoscaler: bool = True
+    mi355x_autoscaler_interval_s: float = 5.0
+    # backend: "memory" (self-contained control plane, tests/bench/local) or
+    # "kubernetes" (real cluster via REST)
+    backend: str = "memory"
+    kubeconfig: Optional[str] = None
+
+
+def load_config(argv: Optional[List[str]] = None) -> Configuration:
+    parser = argparse.ArgumentParser(prog="kuberay-amd-operator")
+    parser.add_argument("--config", help="YAML Configuration file")
+    parser.add_argument("--metrics-addr")
+    parser.add_argument("--probe-addr")
+    parser.add_argument("--reconcile-concurrency", type=int)
+    parser.add_argument("--watch-namespace", action="append", dest="watch_namespaces")
+    parser.add_argument("--batch-scheduler")
+    parser.add_argument("--enable-batch-scheduler", action="store_true", default=None)
+    parser.add_argument("--feature-gates", default=None)
+    parser.add_argument("--backend", choices=["memory", "kubernetes"])
+    parser.add_argument("--kubeconfig")
+    parser.add_argument("--log-file")
+    parser.add_argument("--no-metrics", action="store_true", default=None)
+    args = parser.parse_args(argv)
+
+    data = {}
+    if args.config:
+        with open(args.config) as f:
+            data = yaml.safe_load(f) or {}
+    cfg = Configuration.from_dict(data)
+
+    for flag, attr in [
+        ("metrics_addr", "metrics_addr"), ("probe_addr", "probe_addr"),
+        ("reconcile_concurrency", "reconcile_concurrency"),
+        ("watch_namespaces", "watch_namespaces"),
+        ("batch_scheduler", "batch_scheduler"),
+        ("enable_batch_scheduler", "enable_batch_scheduler"),
+        ("feature_gates", "feature_gates"),
+        ("backend", "backend"), ("kubeconfig", "kubeconfig"),
+        ("log_file", "log_file"),
+    ]:
+        val = getattr(args, flag, None)
+        if val is not None:
+            setattr(cfg, attr, val)
+    if args.no_metrics:
+        cfg.enable_metrics = False
+    return cfg

@@ -1,0 +1,227 @@
+"""Operator wiring, feature gates, metrics, networkpolicy, gang scheduling."""
+import time
+
+import pytest
+
+import kuberay_amd.features as features
+from kuberay_amd.config import Configuration, load_config
+from kuberay_amd.kube.client import InMemoryClient
+from kuberay_amd.metrics import OperatorMetrics
+from kuberay_amd.models import RayCluster
+from kuberay_amd.operator import build_manager
+from kuberay_amd.ops.networkpolicy import (
+    NetworkPolicyReconciler,
+    build_head_network_policy,
+    build_worker_network_policies,
+)
+from kuberay_amd.parallel import (
+    VolcanoBatchScheduler,
+    XgmiGangScheduler,
+    YunikornBatchScheduler,
+    scheduler_for,
+)
+from kuberay_amd.testing import simple_raycluster
+from kuberay_amd.utils import constants as C
+
+
+@pytest.fixture(autouse=True)
+def reset_gates():
+    yield
+    features.reset()
+
+
+class TestFeatureGates:
+    def test_defaults(self):
+        assert features.enabled("RayClusterStatusConditions")
+        assert not features.enabled("RayClusterMTLS")
+        assert features.enabled("MI355XGpuHealthProbes")
+
+    def test_parse(self):
+        features.parse_feature_gates("RayClusterMTLS=true,RayCronJob=false")
+        assert features.enabled("RayClusterMTLS")
+        assert not features.enabled("RayCronJob")
+
+    def test_unknown_gate_rejected(self):
+        with pytest.raises(KeyError):
+            features.set_gate("NoSuchGate", True)
+
+
+class TestConfig:
+    def test_defaults(self):
+        cfg = load_config([])
+        assert cfg.backend == "memory"
+        assert cfg.reconcile_concurrency == 4
+
+    def test_flags_override_file(self, tmp_path):
+        f = tmp_path / "cfg.yaml"
+        f.write_text("reconcileConcurrency: 2\nbatchScheduler: volcano\n")
+        cfg = load_config(["--config", str(f), "--reconcile-concurrency", "8"])
+        assert cfg.reconcile_concurrency == 8
+        assert cfg.batch_scheduler == "volcano"
+
+
+class TestOperatorWiring:
+    def test_build_manager_controllers(self):
+        cfg = Configuration()
+        m, client, metrics, autoscaler = build_manager(cfg)
+        names_ = [c.name for c in m.controllers]
+        assert names_ == ["raycluster", "rayjob", "rayservice", "raycronjob"]
+        assert metrics is not None and autoscaler is not None
+
+    def test_networkpolicy_controller_gated(self):
+        features.set_gate("RayClusterNetworkPolicy", True)
+        m, *_ = build_manager(Configuration())
+        assert "networkpolicy" in [c.name for c in m.controllers]
+
+    def test_end_to_end_via_operator_manager(self):
+        cfg = Configuration(enable_mi355x_autoscaler=False)
+        m, client, metrics, _ = build_manager(cfg)
+        from kuberay_amd.kube.kubelet import SimKubelet
+        kubelet = SimKubelet(m.server)
+        m.start(); kubelet.start()
+        try:
+            client.create(simple_raycluster("op-e2e"))
+            deadline = time.monotonic() + 15
+            state = None
+            while time.monotonic() < deadline:
+                rc = client.try_get(RayCluster, "default", "op-e2e")
+                state = rc.status.state if rc else None
+                if state == "ready":
+                    break
+                time.sleep(0.05)
+            assert state == "ready"
+            # metrics recorded the provisioned cluster
+            expo = metrics.exposition().decode()
+            assert "kuberay_cluster_provisioned_duration_seconds" in expo
+        finally:
+            kubelet.stop(); m.stop()
+
+
+class TestMetrics:
+    def test_exposition_names(self):
+        m = OperatorMetrics()
+        expo = m.exposition().decode()
+        for name in ("kuberay_cluster_provisioned_duration_seconds",
+                     "kuberay_job_execution_duration_seconds",
+                     "kuberay_service_info",
+                     "kuberay_mi355x_gpu_utilization_pct"):
+            assert name in expo, name
+
+    def test_gpu_stats_observation(self):
+        from kuberay_amd.gpu.rocm_smi import GpuStats
+        m = OperatorMetrics()
+        m.observe_gpu_stats([GpuStats(index=0, utilization_pct=50,
+                                      vram_used_bytes=144 * 1024**3)])
+        expo = m.exposition().decode()
+        assert 'kuberay_mi355x_gpu_utilization_pct{gpu="0"} 50.0' in expo
+
+
+class TestNetworkPolicy:
+    def _cluster(self, mode="DenyAll", **np_extra):
+        return simple_raycluster("demo", networkPolicy={"mode": mode, **np_extra})
+
+    def test_head_policy_deny_all(self):
+        p = build_head_network_policy(self._cluster())
+        assert set(p.spec["policyTypes"]) == {"Ingress", "Egress"}
+        # intra-cluster always allowed
+        assert p.spec["ingress"][0]["from"][0]["podSelector"]["matchLabels"] == \
+            {C.RAY_CLUSTER_LABEL_KEY: "demo"}
+        # DNS egress open
+        assert any(any(pp.get("port") == 53 for pp in r.get("ports", []))
+                   for r in p.spec["egress"])
+
+    def test_ingress_only_mode(self):
+        p = build_head_network_policy(self._cluster(mode="DenyAllIngress"))
+        assert p.spec["policyTypes"] == ["Ingress"]
+        assert "egress" not in p.spec
+
+    def test_worker_group_override(self):
+        c = self._cluster(workerGroups=[{
+            "groupName": "default-group",
+            "ingressRules": [{"from": [{"ipBlock": {"cidr": "10.0.0.0/8"}}]}]}])
+        policies = build_worker_network_policies(c)
+        assert len(policies) == 2
+        override = policies[1]
+        assert override.spec["podSelector"]["matchLabels"][
+            C.RAY_NODE_GROUP_LABEL_KEY] == "default-group"
+
+    def test_reconciler_creates_and_gcs(self):
+        client = InMemoryClient()
+        cluster = self._cluster()
+        client.create(cluster)
+        r = NetworkPolicyReconciler(client)
+        r.reconcile(("default", "demo"))
+        assert client.server.count("NetworkPolicy") == 2
+        # turn off policy -> stale GC
+        rc = client.get(RayCluster, "default", "demo")
+        rc.spec.network_policy = None
+        client.update(rc)
+        r.reconcile(("default", "demo"))
+        assert client.server.count("NetworkPolicy") == 0
+
+
+class TestGangScheduling:
+    def _pod(self, cluster, gpus=1):
+        from kuberay_amd.common import pod as podlib
+        group = cluster.spec.worker_group_specs[0]
+        fqdn = "x.default.svc.cluster.local"
+        t = podlib.default_worker_pod_template(cluster, group, "p-", fqdn, "6379")
+        return podlib.build_pod(t, "worker", group.ray_start_params, "6379",
+                                False, None, fqdn)
+
+    def test_selector_registry(self):
+        assert isinstance(scheduler_for("volcano"), VolcanoBatchScheduler)
+        assert isinstance(scheduler_for("xgmi-gang"), XgmiGangScheduler)
+        assert scheduler_for(None) is None
+        with pytest.raises(ValueError):
+            scheduler_for("nope")
+
+    def test_volcano_podgroup_and_annotations(self):
+        client = InMemoryClient()
+        cluster = simple_raycluster("demo", workers=3, gpus_per_worker=1)
+        client.create(cluster)
+        sched = VolcanoBatchScheduler()
+        sched.do_batch_scheduling_on_submission(client, cluster)
+        pgs = client.server.list("PodGroup")
+        assert len(pgs) == 1
+        assert pgs[0]["spec"]["minMember"] == 4  # head + 3 workers
+        assert pgs[0]["spec"]["minResources"]["amd.com/gpu"] == "3"
+        pod = self._pod(cluster)
+        sched.add_metadata_to_pod(client, cluster, "default-group", pod)
+        assert pod.spec.scheduler_name == "volcano"
+        assert pod.metadata.annotations["scheduling.k8s.io/group-name"] == \
+            "ray-demo-pg"
+
+    def test_xgmi_gang_adds_island_affinity_for_gpu_pods(self):
+        client = InMemoryClient()
+        cluster = simple_raycluster("demo", workers=2, gpus_per_worker=4)
+        client.create(cluster)
+        sched = XgmiGangScheduler()
+        sched.do_batch_scheduling_on_submission(client, cluster)
+        assert client.server.count("PodGroup") == 1
+        pod = self._pod(cluster)
+        sched.add_metadata_to_pod(client, cluster, "default-group", pod)
+        terms = pod.spec.affinity["podAffinity"][
+            "preferredDuringSchedulingIgnoredDuringExecution"]
+        assert terms[0]["podAffinityTerm"]["topologyKey"] == "amd.com/xgmi-island"
+        assert pod.metadata.labels["ray.io/xgmi-gang"] == "demo-default-group"
+
+    def test_xgmi_gang_skips_cpu_pods(self):
+        client = InMemoryClient()
+        cluster = simple_raycluster("demo", workers=2, gpus_per_worker=0)
+        client.create(cluster)
+        sched = XgmiGangScheduler()
+        sched.do_batch_scheduling_on_submission(client, cluster)
+        assert client.server.count("PodGroup") == 0
+        pod = self._pod(cluster, gpus=0)
+        sched.add_metadata_to_pod(client, cluster, "default-group", pod)
+        assert pod.spec.affinity is None
+
+    def test_yunikorn_labels(self):
+        client = InMemoryClient()
+        cluster = simple_raycluster("demo")
+        pod = self._pod(cluster)
+        YunikornBatchScheduler().add_metadata_to_pod(client, cluster,
+                                                     "default-group", pod)
+        assert pod.metadata.labels["applicationId"] == "default-demo"
+        assert pod.spec.scheduler_name == "yunikorn"
