@@ -1,0 +1,349 @@
+"""kray — the kubectl-plugin analog CLI
+(reference: kubectl-plugin/pkg/cmd/ray.go:27-57: get / create / delete /
+scale / session / log / job submit / version).
+
+Targets a kuberay-amd apiserver (``--server``) or a real K8s apiserver via
+its ray.io/v1 surface. The ``--gpu`` flag requests ``amd.com/gpu`` — there
+is no NVIDIA resource path (reference: kubectl-plugin generation.go:175-214
+defaulted to nvidia.com/gpu; dropped).
+"""
+from __future__ import annotations
+
+import json
+import os
+import sys
+from typing import Optional
+
+import click
+import yaml
+
+import kuberay_amd
+from ..kube.httpclient import HttpKubeClient
+from ..models import RayCluster, RayCronJob, RayJob, RayService
+from ..utils import constants as C
+
+
+def make_client(server: Optional[str]):
+    server = server or os.environ.get("KURAY_SERVER", "http://127.0.0.1:8888")
+    return HttpKubeClient(server)
+
+
+@click.group()
+@click.option("--server", envvar="KURAY_SERVER", default=None,
+              help="kuberay-amd apiserver URL (default http://127.0.0.1:8888)")
+@click.option("--namespace", "-n", default="default")
+@click.pass_context
+def cli(ctx, server, namespace):
+    """kray — manage Ray clusters on MI355X Kubernetes."""
+    ctx.ensure_object(dict)
+    ctx.obj["server"] = server
+    ctx.obj["namespace"] = namespace
+    ctx.obj["client"] = None
+
+
+def client_of(ctx):
+    if ctx.obj["client"] is None:
+        ctx.obj["client"] = make_client(ctx.obj["server"])
+    return ctx.obj["client"]
+
+
+@cli.command()
+def version():
+    """Print the kuberay-amd version."""
+    click.echo(f"kuberay-amd {kuberay_amd.__version__} (ray.io/v1)")
+
+
+# ---------------------------------------------------------------------------
+# get
+# ---------------------------------------------------------------------------
+@cli.group()
+def get():
+    """Get Ray resources."""
+
+
+@get.command("cluster")
+@click.argument("name", required=False)
+@click.pass_context
+def get_cluster(ctx, name):
+    client = client_of(ctx)
+    ns = ctx.obj["namespace"]
+    rows = []
+    clusters = ([client.get(RayCluster, ns, name)] if name
+                else client.list(RayCluster, ns))
+    for rc in clusters:
+        rows.append((rc.metadata.name, rc.status.state or "-",
+                     rc.status.desired_worker_replicas,
+                     rc.status.ready_worker_replicas,
+                     rc.status.desired_cpu or "-", rc.status.desired_gpu or "-"))
+    _table(["NAME", "STATE", "DESIRED", "READY", "CPUS", "GPUS"], rows)
+
+
+@get.command("workergroup")
+@click.argument("cluster_name")
+@click.pass_context
+def get_workergroup(ctx, cluster_name):
+    client = client_of(ctx)
+    rc = client.get(RayCluster, ctx.obj["namespace"], cluster_name)
+    rows = [(g.group_name, g.replicas, g.min_replicas, g.max_replicas,
+             g.num_of_hosts, "yes" if g.suspend else "no")
+            for g in rc.spec.worker_group_specs]
+    _table(["GROUP", "REPLICAS", "MIN", "MAX", "HOSTS", "SUSPENDED"], rows)
+
+
+@get.command("job")
+@click.argument("name", required=False)
+@click.pass_context
+def get_job(ctx, name):
+    client = client_of(ctx)
+    ns = ctx.obj["namespace"]
+    jobs = [client.get(RayJob, ns, name)] if name else client.list(RayJob, ns)
+    rows = [(j.metadata.name, j.status.job_deployment_status or "-",
+             j.status.job_status or "-", j.status.ray_cluster_name or "-")
+            for j in jobs]
+    _table(["NAME", "DEPLOYMENT STATUS", "JOB STATUS", "CLUSTER"], rows)
+
+
+@get.command("service")
+@click.argument("name", required=False)
+@click.pass_context
+def get_service(ctx, name):
+    client = client_of(ctx)
+    ns = ctx.obj["namespace"]
+    svcs = ([client.get(RayService, ns, name)] if name
+            else client.list(RayService, ns))
+    rows = [(s.metadata.name, s.status.service_status or "-",
+             s.status.num_serve_endpoints,
+             s.status.active_service_status.ray_cluster_name or "-")
+            for s in svcs]
+    _table(["NAME", "STATUS", "ENDPOINTS", "ACTIVE CLUSTER"], rows)
+
+
+# ---------------------------------------------------------------------------
+# create
+# ---------------------------------------------------------------------------
+@cli.group()
+def create():
+    """Create Ray resources."""
+
+
+def _cluster_spec(image, head_cpu, head_memory, worker_replicas, worker_cpu,
+                  worker_memory, worker_gpu, autoscaler):
+    worker_limits = {"cpu": worker_cpu, "memory": worker_memory}
+    if worker_gpu:
+        worker_limits[C.AMD_GPU_RESOURCE_NAME] = str(worker_gpu)
+    spec = {
+        "rayVersion": "2.46.0",
+        "headGroupSpec": {
+            "rayStartParams": {},
+            "template": {"spec": {"containers": [{
+                "name": "ray-head", "image": image,
+                "resources": {"limits": {"cpu": head_cpu, "memory": head_memory},
+                              "requests": {"cpu": head_cpu, "memory": head_memory}},
+            }]}},
+        },
+        "workerGroupSpecs": [{
+            "groupName": "default-group",
+            "replicas": worker_replicas,
+            "minReplicas": 0,
+            "maxReplicas": max(worker_replicas, 8),
+            "rayStartParams": {},
+            "template": {"spec": {"containers": [{
+                "name": "ray-worker", "image": image,
+                "resources": {"limits": dict(worker_limits),
+                              "requests": dict(worker_limits)},
+            }]}},
+        }],
+    }
+    if autoscaler:
+        spec["enableInTreeAutoscaling"] = True
+    return spec
+
+
+@create.command("cluster")
+@click.argument("name")
+@click.option("--image", default=C.DEFAULT_RAY_ROCM_IMAGE, show_default=True)
+@click.option("--head-cpu", default="2")
+@click.option("--head-memory", default="4Gi")
+@click.option("--worker-replicas", default=1, type=int)
+@click.option("--worker-cpu", default="4")
+@click.option("--worker-memory", default="8Gi")
+@click.option("--worker-gpu", default=0, type=int,
+              help="amd.com/gpu per worker (MI355X)")
+@click.option("--autoscaler", is_flag=True)
+@click.option("--dry-run", is_flag=True, help="print YAML instead of creating")
+@click.pass_context
+def create_cluster(ctx, name, image, head_cpu, head_memory, worker_replicas,
+                   worker_cpu, worker_memory, worker_gpu, autoscaler, dry_run):
+    spec = _cluster_spec(image, head_cpu, head_memory, worker_replicas,
+                         worker_cpu, worker_memory, worker_gpu, autoscaler)
+    obj = {"apiVersion": "ray.io/v1", "kind": "RayCluster",
+           "metadata": {"name": name, "namespace": ctx.obj["namespace"]},
+           "spec": spec}
+    if dry_run:
+        click.echo(yaml.safe_dump(obj, sort_keys=False))
+        return
+    client_of(ctx).create(RayCluster.from_dict(obj))
+    click.echo(f"raycluster.ray.io/{name} created")
+
+
+@create.command("workergroup")
+@click.argument("cluster_name")
+@click.option("--group-name", required=True)
+@click.option("--image", default=None)
+@click.option("--worker-replicas", default=1, type=int)
+@click.option("--worker-cpu", default="4")
+@click.option("--worker-memory", default="8Gi")
+@click.option("--worker-gpu", default=0, type=int)
+@click.pass_context
+def create_workergroup(ctx, cluster_name, group_name, image, worker_replicas,
+                       worker_cpu, worker_memory, worker_gpu):
+    client = client_of(ctx)
+    rc = client.get(RayCluster, ctx.obj["namespace"], cluster_name)
+    limits = {"cpu": worker_cpu, "memory": worker_memory}
+    if worker_gpu:
+        limits[C.AMD_GPU_RESOURCE_NAME] = str(worker_gpu)
+    head_image = rc.spec.head_group_spec.template.spec.containers[0].image
+    from ..models.raycluster import WorkerGroupSpec
+    rc.spec.worker_group_specs.append(WorkerGroupSpec.from_dict({
+        "groupName": group_name, "replicas": worker_replicas,
+        "minReplicas": 0, "maxReplicas": max(worker_replicas, 8),
+        "rayStartParams": {},
+        "template": {"spec": {"containers": [{
+            "name": "ray-worker", "image": image or head_image,
+            "resources": {"limits": limits, "requests": dict(limits)}}]}},
+    }))
+    client.update(rc)
+    click.echo(f"worker group {group_name} added to {cluster_name}")
+
+
+# ---------------------------------------------------------------------------
+# delete / scale / suspend
+# ---------------------------------------------------------------------------
+@cli.command()
+@click.argument("kind", type=click.Choice(["cluster", "job", "service", "cronjob"]))
+@click.argument("name")
+@click.pass_context
+def delete(ctx, kind, name):
+    """Delete a Ray resource."""
+    model = {"cluster": RayCluster, "job": RayJob,
+             "service": RayService, "cronjob": RayCronJob}[kind]
+    client_of(ctx).delete(model, ctx.obj["namespace"], name)
+    click.echo(f"{kind} {name} deleted")
+
+
+@cli.group()
+def scale():
+    """Scale Ray resources."""
+
+
+@scale.command("cluster")
+@click.argument("name")
+@click.option("--worker-group", default=None)
+@click.option("--replicas", required=True, type=int)
+@click.pass_context
+def scale_cluster(ctx, name, worker_group, replicas):
+    client = client_of(ctx)
+    rc = client.get(RayCluster, ctx.obj["namespace"], name)
+    groups = rc.spec.worker_group_specs
+    target = None
+    if worker_group:
+        target = next((g for g in groups if g.group_name == worker_group), None)
+        if target is None:
+            raise click.ClickException(
+                f"worker group '{worker_group}' not found "
+                f"(have: {[g.group_name for g in groups]})")
+    elif len(groups) == 1:
+        target = groups[0]
+    else:
+        raise click.ClickException("--worker-group required (multiple groups)")
+    target.replicas = replicas
+    client.update(rc)
+    click.echo(f"cluster {name}/{target.group_name} scaled to {replicas}")
+
+
+# ---------------------------------------------------------------------------
+# job submit
+# ---------------------------------------------------------------------------
+@cli.group()
+def job():
+    """Ray job operations."""
+
+
+@job.command("submit")
+@click.option("--name", required=True)
+@click.option("--entrypoint", required=True)
+@click.option("--image", default=C.DEFAULT_RAY_ROCM_IMAGE)
+@click.option("--worker-replicas", default=1, type=int)
+@click.option("--worker-gpu", default=0, type=int)
+@click.option("--runtime-env", default=None, help="runtime env YAML file")
+@click.option("--shutdown-after-finish/--keep-cluster", default=True)
+@click.option("--dry-run", is_flag=True)
+@click.pass_context
+def job_submit(ctx, name, entrypoint, image, worker_replicas, worker_gpu,
+               runtime_env, shutdown_after_finish, dry_run):
+    spec = {
+        "entrypoint": entrypoint,
+        "shutdownAfterJobFinishes": shutdown_after_finish,
+        "rayClusterSpec": _cluster_spec(image, "2", "4Gi", worker_replicas,
+                                        "4", "8Gi", worker_gpu, False),
+    }
+    if runtime_env:
+        with open(runtime_env) as f:
+            spec["runtimeEnvYAML"] = f.read()
+    obj = {"apiVersion": "ray.io/v1", "kind": "RayJob",
+           "metadata": {"name": name, "namespace": ctx.obj["namespace"]},
+           "spec": spec}
+    if dry_run:
+        click.echo(yaml.safe_dump(obj, sort_keys=False))
+        return
+    client_of(ctx).create(RayJob.from_dict(obj))
+    click.echo(f"rayjob.ray.io/{name} submitted")
+
+
+# ---------------------------------------------------------------------------
+# session / log
+# ---------------------------------------------------------------------------
+@cli.command()
+@click.argument("cluster_name")
+@click.pass_context
+def session(ctx, cluster_name):
+    """Print connection endpoints for a cluster (port-forward analog)."""
+    client = client_of(ctx)
+    rc = client.get(RayCluster, ctx.obj["namespace"], cluster_name)
+    head_svc = f"{cluster_name}-head-svc.{ctx.obj['namespace']}.svc"
+    click.echo(f"cluster:   {cluster_name} ({rc.status.state})")
+    click.echo(f"dashboard: http://{head_svc}:8265")
+    click.echo(f"client:    ray://{head_svc}:10001")
+    click.echo(f"serve:     http://{head_svc}:8000")
+    click.echo("kubectl port-forward "
+               f"svc/{cluster_name}-head-svc 8265:8265 10001:10001 8000:8000")
+
+
+@cli.command()
+@click.argument("cluster_name")
+@click.pass_context
+def log(ctx, cluster_name):
+    """Show recent events for a cluster (log-download analog)."""
+    client = client_of(ctx)
+    rc = client.get(RayCluster, ctx.obj["namespace"], cluster_name)
+    click.echo(f"state={rc.status.state} head={rc.status.head.pod_name} "
+               f"ready={rc.status.ready_worker_replicas}/"
+               f"{rc.status.desired_worker_replicas}")
+    for cond in rc.status.conditions or []:
+        click.echo(f"  {cond.type}={cond.status} ({cond.reason})")
+
+
+def _table(headers, rows):
+    widths = [max(len(str(h)), *(len(str(r[i])) for r in rows)) if rows else len(h)
+              for i, h in enumerate(headers)]
+    click.echo("  ".join(str(h).ljust(w) for h, w in zip(headers, widths)))
+    for r in rows:
+        click.echo("  ".join(str(c).ljust(w) for c, w in zip(r, widths)))
+
+
+def main(argv=None):
+    return cli(args=argv, standalone_mode=True)
+
+
+if __name__ == "__main__":
+    main()

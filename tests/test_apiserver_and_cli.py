@@ -1,0 +1,208 @@
+"""APIServer v1/v2 + HttpKubeClient + kray CLI tests
+(reference analogs: apiserver unit/e2e tests, kubectl-plugin e2e suites)."""
+import pytest
+from click.testing import CliRunner
+from fastapi.testclient import TestClient
+
+from kuberay_amd.apiserver import create_app
+from kuberay_amd.cli.main import cli
+from kuberay_amd.kube.client import InMemoryClient
+from kuberay_amd.kube.httpclient import HttpKubeClient
+from kuberay_amd.models import RayCluster, RayJob
+from kuberay_amd.utils.fake_dashboard import FakeRayDashboardClient
+
+CLUSTER_BODY = {
+    "name": "c1", "version": "2.46.0", "user": "alice@example.com",
+    "clusterSpec": {
+        "headGroupSpec": {"computeTemplate": "tpl", "rayStartParams": {}},
+        "workerGroupSpec": [{
+            "groupName": "g", "computeTemplate": "tpl",
+            "replicas": 2, "minReplicas": 0, "maxReplicas": 4}],
+    },
+}
+
+
+@pytest.fixture()
+def api():
+    client = InMemoryClient()
+    fake = FakeRayDashboardClient()
+    app = create_app(client, dashboard_factory=lambda name: fake)
+    t = TestClient(app)
+    t.post("/apis/v1/namespaces/ns1/compute_templates",
+           json={"name": "tpl", "cpu": 4, "memory": 8, "gpu": 2})
+    return t, client, fake
+
+
+class TestComputeTemplates:
+    def test_crud(self, api):
+        t, _, _ = api
+        r = t.get("/apis/v1/namespaces/ns1/compute_templates")
+        assert r.status_code == 200
+        tpl = r.json()["computeTemplates"][0]
+        assert tpl["gpuAccelerator"] == "amd.com/gpu"
+        assert t.get("/apis/v1/namespaces/ns1/compute_templates/tpl").status_code == 200
+        assert t.delete("/apis/v1/namespaces/ns1/compute_templates/tpl").status_code == 200
+        assert t.get("/apis/v1/namespaces/ns1/compute_templates/tpl").status_code == 404
+
+
+class TestClusterService:
+    def test_create_expands_template_to_amd_gpu(self, api):
+        t, client, _ = api
+        r = t.post("/apis/v1/namespaces/ns1/clusters", json=CLUSTER_BODY)
+        assert r.status_code == 200, r.text
+        rc = client.get(RayCluster, "ns1", "c1")
+        limits = rc.spec.worker_group_specs[0].template.spec.containers[0] \
+            .resources.limits
+        assert limits["amd.com/gpu"] == "2"
+        assert limits["cpu"] == "4"
+        # user label sanitized
+        assert rc.metadata.labels["ray.io/user"] == "alice-example.com"
+
+    def test_get_list_delete(self, api):
+        t, _, _ = api
+        t.post("/apis/v1/namespaces/ns1/clusters", json=CLUSTER_BODY)
+        assert len(t.get("/apis/v1/namespaces/ns1/clusters").json()["clusters"]) == 1
+        assert t.get("/apis/v1/namespaces/ns1/clusters/c1").status_code == 200
+        assert t.delete("/apis/v1/namespaces/ns1/clusters/c1").status_code == 200
+        assert t.get("/apis/v1/namespaces/ns1/clusters/c1").status_code == 404
+
+    def test_invalid_rejected(self, api):
+        t, _, _ = api
+        bad = dict(CLUSTER_BODY)
+        bad["clusterSpec"] = {
+            "headGroupSpec": {"computeTemplate": "tpl"},
+            "workerGroupSpec": [{"groupName": "g", "computeTemplate": "tpl",
+                                 "replicas": 2, "minReplicas": 5, "maxReplicas": 1}]}
+        r = t.post("/apis/v1/namespaces/ns1/clusters", json=bad)
+        assert r.status_code == 400
+
+
+class TestJobService:
+    def test_create_and_get(self, api):
+        t, client, _ = api
+        r = t.post("/apis/v1/namespaces/ns1/jobs", json={
+            "name": "j1", "entrypoint": "python x.py",
+            "clusterSpec": CLUSTER_BODY["clusterSpec"]})
+        assert r.status_code == 200, r.text
+        job = client.get(RayJob, "ns1", "j1")
+        assert job.spec.entrypoint == "python x.py"
+        assert job.spec.ray_cluster_spec is not None
+        assert t.get("/apis/v1/namespaces/ns1/jobs/j1").json()["name"] == "j1"
+
+
+class TestJobSubmissionProxy:
+    def test_submit_and_get(self, api):
+        t, _, fake = api
+        t.post("/apis/v1/namespaces/ns1/clusters", json=CLUSTER_BODY)
+        r = t.post("/apis/v1/namespaces/ns1/jobsubmissions/c1",
+                   json={"entrypoint": "python x.py", "submission_id": "sub-1"})
+        assert r.status_code == 200
+        assert r.json()["submissionId"] == "sub-1"
+        r = t.get("/apis/v1/namespaces/ns1/jobsubmissions/c1/sub-1")
+        assert r.status_code == 200
+
+
+class TestV2Proxy:
+    def test_restricted_to_ray_resources(self, api):
+        t, _, _ = api
+        assert t.get("/apis/ray.io/v1/namespaces/ns1/pods").status_code == 404
+
+    def test_full_cr_roundtrip(self, api):
+        t, _, _ = api
+        from kuberay_amd.testing import simple_raycluster
+        body = simple_raycluster("v2c").to_dict()
+        r = t.post("/apis/ray.io/v1/namespaces/ns1/rayclusters", json=body)
+        assert r.status_code == 200
+        got = t.get("/apis/ray.io/v1/namespaces/ns1/rayclusters/v2c").json()
+        assert got["spec"]["workerGroupSpecs"][0]["groupName"] == "default-group"
+
+
+@pytest.fixture()
+def http_stack(api):
+    """HttpKubeClient wired to the FastAPI app via in-process transport."""
+    t, client, fake = api
+    hc = HttpKubeClient("http://testserver", http_client=t)
+    return hc, client
+
+
+class TestHttpKubeClient:
+    def test_crud(self, http_stack):
+        hc, backing = http_stack
+        from kuberay_amd.testing import simple_raycluster
+        created = hc.create(simple_raycluster("hc1", namespace="ns1"))
+        assert created.metadata.uid
+        got = hc.get(RayCluster, "ns1", "hc1")
+        got.spec.worker_group_specs[0].replicas = 5
+        hc.update(got)
+        assert backing.get(RayCluster, "ns1", "hc1") \
+            .spec.worker_group_specs[0].replicas == 5
+        assert len(hc.list(RayCluster, "ns1")) == 1
+        hc.delete(RayCluster, "ns1", "hc1")
+        assert hc.try_get(RayCluster, "ns1", "hc1") is None
+
+
+@pytest.fixture()
+def kray(api, monkeypatch):
+    t, client, _ = api
+    hc = HttpKubeClient("http://testserver", http_client=t)
+    import importlib
+    climod = importlib.import_module("kuberay_amd.cli.main")
+    monkeypatch.setattr(climod, "make_client", lambda server: hc)
+    return CliRunner(), client
+
+
+class TestKrayCli:
+    def test_version(self, kray):
+        runner, _ = kray
+        r = runner.invoke(cli, ["version"])
+        assert r.exit_code == 0 and "kuberay-amd" in r.output
+
+    def test_create_get_scale_delete_cluster(self, kray):
+        runner, backing = kray
+        r = runner.invoke(cli, ["-n", "ns1", "create", "cluster", "k1",
+                                "--worker-replicas", "2", "--worker-gpu", "1"])
+        assert r.exit_code == 0, r.output
+        rc = backing.get(RayCluster, "ns1", "k1")
+        limits = rc.spec.worker_group_specs[0].template.spec.containers[0] \
+            .resources.limits
+        assert limits["amd.com/gpu"] == "1"
+
+        r = runner.invoke(cli, ["-n", "ns1", "get", "cluster"])
+        assert r.exit_code == 0 and "k1" in r.output
+
+        r = runner.invoke(cli, ["-n", "ns1", "scale", "cluster", "k1",
+                                "--replicas", "4"])
+        assert r.exit_code == 0, r.output
+        assert backing.get(RayCluster, "ns1", "k1") \
+            .spec.worker_group_specs[0].replicas == 4
+
+        r = runner.invoke(cli, ["-n", "ns1", "delete", "cluster", "k1"])
+        assert r.exit_code == 0
+        assert backing.try_get(RayCluster, "ns1", "k1") is None
+
+    def test_create_workergroup(self, kray):
+        runner, backing = kray
+        runner.invoke(cli, ["-n", "ns1", "create", "cluster", "k1"])
+        r = runner.invoke(cli, ["-n", "ns1", "create", "workergroup", "k1",
+                                "--group-name", "gpu-group", "--worker-gpu", "8"])
+        assert r.exit_code == 0, r.output
+        rc = backing.get(RayCluster, "ns1", "k1")
+        assert [g.group_name for g in rc.spec.worker_group_specs] == \
+            ["default-group", "gpu-group"]
+
+    def test_job_submit(self, kray):
+        runner, backing = kray
+        r = runner.invoke(cli, ["-n", "ns1", "job", "submit", "--name", "j1",
+                                "--entrypoint", "python t.py",
+                                "--worker-gpu", "2"])
+        assert r.exit_code == 0, r.output
+        job = backing.get(RayJob, "ns1", "j1")
+        assert job.spec.entrypoint == "python t.py"
+
+    def test_dry_run_yaml(self, kray):
+        runner, backing = kray
+        r = runner.invoke(cli, ["-n", "ns1", "create", "cluster", "k1",
+                                "--worker-gpu", "4", "--dry-run"])
+        assert r.exit_code == 0
+        assert "amd.com/gpu: '4'" in r.output
+        assert backing.try_get(RayCluster, "ns1", "k1") is None
