@@ -121,6 +121,10 @@ class InMemoryClient(KubeClient):
                                       subresource=subresource)
         return model.from_dict(out)
 
+    def list_pod_views(self, namespace=None, label_selector=None):
+        """Reconcile hot-loop projection (no pod materialization)."""
+        return self.server.list_pod_views(namespace, label_selector)
+
     def delete(self, model_or_obj, namespace=None, name=None):
         if namespace is None:
             namespace = model_or_obj.metadata.namespace or "default"

@@ -7,20 +7,26 @@ seam natively: a thread-safe object store with resourceVersion semantics,
 optimistic-concurrency conflicts, finalizer-aware deletion, ownerReference
 garbage collection and label-selector lists + watches.
 
-It is also the substrate of the scale benchmark (500-cluster
-soak): all verbs are O(1) dict ops except list, which uses per-(kind,label)
-indices. A C++ native backend with the same verb surface lives in
-``kuberay_amd._native`` and is used when built.
+Storage is pluggable (the semantic layer is backend-agnostic):
+
+* ``PyBackend`` — plain python dict trees (always available),
+* ``NativeBackend`` (kuberay_amd/kube/native.py) — the C++ engine: objects
+  live as compact JSON blobs in C++ with label/owner indexes and
+  precomputed pod *views*, so the 500-cluster / 2000-pod cache holds no
+  Python object trees at all (RSS) and the reconcile hot loop reads tiny
+  projections instead of materializing pods.
+
+``InMemoryApiServer()`` picks the native backend automatically when the
+extension is built.
 """
 from __future__ import annotations
 
-import copy
-import fnmatch
+import dataclasses
 import threading
 import time
 import uuid
 from collections import defaultdict
-from typing import Any, Callable, Dict, Iterable, List, Optional, Tuple
+from typing import Any, Dict, Iterable, List, Optional, Tuple
 
 Key = Tuple[str, str, str]  # (kind, namespace, name)
 
@@ -70,24 +76,133 @@ def match_labels(labels: Optional[Dict[str, str]], selector: Optional[Dict[str, 
     return all(labels.get(k) == v for k, v in selector.items())
 
 
-class InMemoryApiServer:
-    """Thread-safe object store with Kubernetes verb semantics."""
+# ---------------------------------------------------------------------------
+# pod views — the reconcile hot-loop projection, precomputed at write time
+# ---------------------------------------------------------------------------
+
+@dataclasses.dataclass
+class PodView:
+    name: str
+    namespace: str
+    labels: Dict[str, str]
+    phase: str = ""
+    ready: bool = False
+    deletion_timestamp: str = ""
+    pod_ip: str = ""
+    restart_policy: str = ""
+    ray_container_terminated: bool = False
+    creation_timestamp: str = ""
+
+
+def compute_pod_view(obj: Dict[str, Any]) -> PodView:
+    meta = obj.get("metadata", {})
+    status = obj.get("status") or {}
+    spec = obj.get("spec") or {}
+    ready = any(c.get("type") == "Ready" and c.get("status") == "True"
+                for c in status.get("conditions") or [])
+    terminated = False
+    containers = spec.get("containers") or []
+    if containers:
+        ray_name = containers[0].get("name")
+        for cs in status.get("containerStatuses") or []:
+            if cs.get("name") == ray_name and (cs.get("state") or {}).get("terminated"):
+                terminated = True
+    return PodView(
+        name=meta.get("name", ""),
+        namespace=meta.get("namespace", "default"),
+        labels=dict(meta.get("labels") or {}),
+        phase=status.get("phase") or "",
+        ready=ready,
+        deletion_timestamp=meta.get("deletionTimestamp") or "",
+        pod_ip=status.get("podIP") or "",
+        restart_policy=spec.get("restartPolicy") or "",
+        ray_container_terminated=terminated,
+        creation_timestamp=meta.get("creationTimestamp") or "",
+    )
+
+
+# ---------------------------------------------------------------------------
+# storage backends
+# ---------------------------------------------------------------------------
+
+class PyBackend:
+    """Plain python dict storage. Copy-on-put + copy-on-fetch keeps stored
+    trees private."""
+
+    name = "python"
 
     def __init__(self) -> None:
-        self._lock = threading.RLock()
         self._objects: Dict[Key, Dict[str, Any]] = {}
-        self._rv = 0
-        self._watchers: List["Watcher"] = []
-        # label index: (kind, label_key, label_value) -> set of keys
+        self._views: Dict[Key, PodView] = {}
         self._label_index: Dict[Tuple[str, str, str], set] = defaultdict(set)
         self._kind_index: Dict[str, set] = defaultdict(set)
-        self._owner_index: Dict[str, set] = defaultdict(set)  # owner uid -> keys
+        self._owner_index: Dict[str, set] = defaultdict(set)
 
-    # -- internals -----------------------------------------------------
-    def _next_rv(self) -> str:
-        self._rv += 1
-        return str(self._rv)
+    def put(self, key: Key, obj: Dict[str, Any]) -> None:
+        old = self._objects.get(key)
+        if old is not None:
+            self._index_remove(key, old)
+        stored = jsoncopy(obj)
+        self._objects[key] = stored
+        self._index_add(key, stored)
+        if key[0] == "Pod":
+            self._views[key] = compute_pod_view(stored)
 
+    def fetch(self, key: Key) -> Optional[Dict[str, Any]]:
+        obj = self._objects.get(key)
+        return jsoncopy(obj) if obj is not None else None
+
+    def rv(self, key: Key) -> Optional[str]:
+        obj = self._objects.get(key)
+        return obj["metadata"].get("resourceVersion") if obj is not None else None
+
+    def contains(self, key: Key) -> bool:
+        return key in self._objects
+
+    def remove(self, key: Key) -> Optional[Dict[str, Any]]:
+        obj = self._objects.pop(key, None)
+        if obj is not None:
+            self._index_remove(key, obj)
+            self._views.pop(key, None)
+        return obj
+
+    def _select_keys(self, kind: str, namespace: Optional[str],
+                     selector: Optional[Dict[str, str]]) -> List[Key]:
+        if selector:
+            candidate_sets = [self._label_index.get((kind, k, v), set())
+                              for k, v in selector.items()]
+            keys: Iterable[Key] = (set.intersection(*candidate_sets)
+                                   if candidate_sets else set())
+        else:
+            keys = self._kind_index.get(kind, set())
+        return [k for k in keys if namespace is None or k[1] == namespace]
+
+    def list(self, kind: str, namespace: Optional[str],
+             selector: Optional[Dict[str, str]]) -> List[Dict[str, Any]]:
+        out = [jsoncopy(self._objects[k])
+               for k in self._select_keys(kind, namespace, selector)
+               if k in self._objects]
+        out.sort(key=lambda o: (o["metadata"]["namespace"], o["metadata"]["name"]))
+        return out
+
+    def list_views(self, namespace: Optional[str],
+                   selector: Optional[Dict[str, str]]) -> List[PodView]:
+        out = [self._views[k]
+               for k in self._select_keys("Pod", namespace, selector)
+               if k in self._views]
+        out.sort(key=lambda v: (v.namespace, v.name))
+        return out
+
+    def dependents(self, uid: str) -> List[Key]:
+        return list(self._owner_index.get(uid, ()))
+
+    def drop_owner(self, uid: str) -> None:
+        self._owner_index.pop(uid, None)
+
+    def count(self, kind: str) -> int:
+        return len(self._kind_index.get(kind, ()))
+
+    # -- index internals ----------------------------------------------
     def _index_add(self, key: Key, obj: Dict[str, Any]) -> None:
         self._kind_index[key[0]].add(key)
         meta = obj.get("metadata", {})
@@ -108,6 +223,38 @@ class InMemoryApiServer:
             if uid:
                 self._owner_index[uid].discard(key)
 
+
+def default_backend():
+    """Native C++ backend when built, python otherwise."""
+    try:
+        from .native import NativeBackend
+        return NativeBackend()
+    except ImportError:
+        return PyBackend()
+
+
+# ---------------------------------------------------------------------------
+# semantic layer
+# ---------------------------------------------------------------------------
+
+class InMemoryApiServer:
+    """Thread-safe object store with Kubernetes verb semantics."""
+
+    def __init__(self, backend=None) -> None:
+        self._lock = threading.RLock()
+        self._backend = backend if backend is not None else default_backend()
+        self._rv = 0
+        self._watchers: List["Watcher"] = []
+
+    @property
+    def backend_name(self) -> str:
+        return self._backend.name
+
+    # -- internals -----------------------------------------------------
+    def _next_rv(self) -> str:
+        self._rv += 1
+        return str(self._rv)
+
     def _notify(self, event_type: str, obj: Dict[str, Any]) -> None:
         for w in list(self._watchers):
             w.push(event_type, obj)
@@ -115,7 +262,8 @@ class InMemoryApiServer:
     @staticmethod
     def _key_of(obj: Dict[str, Any]) -> Key:
         meta = obj.get("metadata", {})
-        return (obj.get("kind", ""), meta.get("namespace", "default"), meta.get("name", ""))
+        return (obj.get("kind", ""), meta.get("namespace", "default"),
+                meta.get("name", ""))
 
     # -- verbs ---------------------------------------------------------
     def create(self, obj: Dict[str, Any]) -> Dict[str, Any]:
@@ -129,114 +277,83 @@ class InMemoryApiServer:
             meta["name"] = gen + uuid.uuid4().hex[:5]
         with self._lock:
             key = self._key_of(obj)
-            if key in self._objects:
+            if self._backend.contains(key):
                 raise AlreadyExistsError(f"{key} already exists")
             meta["uid"] = str(uuid.uuid4())
             meta["resourceVersion"] = self._next_rv()
             meta["generation"] = 1
             meta["creationTimestamp"] = now_iso()
-            self._objects[key] = obj
-            self._index_add(key, obj)
-            out = jsoncopy(obj)
-        self._notify("ADDED", out)
-        return out
+            self._backend.put(key, obj)
+        self._notify("ADDED", obj)
+        return obj
 
     def get(self, kind: str, namespace: str, name: str) -> Dict[str, Any]:
         with self._lock:
-            obj = self._objects.get((kind, namespace, name))
-            if obj is None:
-                raise NotFoundError(f"{kind} {namespace}/{name} not found")
-            return jsoncopy(obj)
+            obj = self._backend.fetch((kind, namespace, name))
+        if obj is None:
+            raise NotFoundError(f"{kind} {namespace}/{name} not found")
+        return obj
 
     def try_get(self, kind: str, namespace: str, name: str) -> Optional[Dict[str, Any]]:
-        try:
-            return self.get(kind, namespace, name)
-        except NotFoundError:
-            return None
-
-    def list(
-        self,
-        kind: str,
-        namespace: Optional[str] = None,
-        label_selector: Optional[Dict[str, str]] = None,
-    ) -> List[Dict[str, Any]]:
         with self._lock:
-            if label_selector:
-                # use the most selective label index entry
-                candidate_sets = [
-                    self._label_index.get((kind, k, v), set())
-                    for k, v in label_selector.items()
-                ]
-                keys = set.intersection(*candidate_sets) if candidate_sets else set()
-            else:
-                keys = set(self._kind_index.get(kind, set()))
-            out = []
-            for key in keys:
-                if namespace is not None and key[1] != namespace:
-                    continue
-                obj = self._objects.get(key)
-                if obj is None:
-                    continue
-                if match_labels(obj.get("metadata", {}).get("labels"), label_selector):
-                    out.append(jsoncopy(obj))
-            out.sort(key=lambda o: (o["metadata"]["namespace"], o["metadata"]["name"]))
-            return out
+            return self._backend.fetch((kind, namespace, name))
+
+    def list(self, kind: str, namespace: Optional[str] = None,
+             label_selector: Optional[Dict[str, str]] = None) -> List[Dict[str, Any]]:
+        with self._lock:
+            return self._backend.list(kind, namespace, label_selector)
+
+    def list_pod_views(self, namespace: Optional[str] = None,
+                       label_selector: Optional[Dict[str, str]] = None) -> List[PodView]:
+        """Reconcile-hot-loop projection (no pod materialization)."""
+        with self._lock:
+            return self._backend.list_views(namespace, label_selector)
 
     def update(self, obj: Dict[str, Any], *, subresource: Optional[str] = None) -> Dict[str, Any]:
         obj = jsoncopy(obj)
         key = self._key_of(obj)
         with self._lock:
-            current = self._objects.get(key)
+            current = self._backend.fetch(key)
             if current is None:
                 raise NotFoundError(f"{key} not found")
             meta = obj.setdefault("metadata", {})
             rv = meta.get("resourceVersion")
-            if rv is not None and rv != current["metadata"]["resourceVersion"]:
+            cur_meta = current["metadata"]
+            if rv is not None and rv != cur_meta["resourceVersion"]:
                 raise ConflictError(
-                    f"{key}: resourceVersion mismatch {rv} != {current['metadata']['resourceVersion']}"
-                )
-            self._index_remove(key, current)
+                    f"{key}: resourceVersion mismatch {rv} != {cur_meta['resourceVersion']}")
             if subresource == "status":
-                # status updates only replace .status
-                new_obj = jsoncopy(current)
+                new_obj = current
                 new_obj["status"] = obj.get("status", {})
             else:
                 new_obj = obj
-                # spec changes bump generation
                 if new_obj.get("spec") != current.get("spec"):
-                    new_obj["metadata"]["generation"] = current["metadata"].get("generation", 1) + 1
+                    new_obj["metadata"]["generation"] = cur_meta.get("generation", 1) + 1
                 else:
-                    new_obj["metadata"]["generation"] = current["metadata"].get("generation", 1)
-                # status only changes through the status subresource
+                    new_obj["metadata"]["generation"] = cur_meta.get("generation", 1)
                 new_obj["status"] = current.get("status", {})
-            new_obj["metadata"]["uid"] = current["metadata"]["uid"]
-            new_obj["metadata"]["creationTimestamp"] = current["metadata"]["creationTimestamp"]
-            if current["metadata"].get("deletionTimestamp"):
-                new_obj["metadata"]["deletionTimestamp"] = current["metadata"]["deletionTimestamp"]
+            new_obj["metadata"]["uid"] = cur_meta["uid"]
+            new_obj["metadata"]["creationTimestamp"] = cur_meta["creationTimestamp"]
+            if cur_meta.get("deletionTimestamp"):
+                new_obj["metadata"]["deletionTimestamp"] = cur_meta["deletionTimestamp"]
             new_obj["metadata"]["resourceVersion"] = self._next_rv()
-            self._objects[key] = new_obj
-            self._index_add(key, new_obj)
+            self._backend.put(key, new_obj)
             finalizers_gone = (
-                current["metadata"].get("deletionTimestamp")
-                and not new_obj["metadata"].get("finalizers")
-            )
-            out = jsoncopy(new_obj)
-        self._notify("MODIFIED", out)
+                cur_meta.get("deletionTimestamp")
+                and not new_obj["metadata"].get("finalizers"))
+        self._notify("MODIFIED", new_obj)
         if finalizers_gone:
-            # terminating object dropped its last finalizer -> actually delete
             self._finalize_delete(key)
-        return out
+        return new_obj
 
-    def patch_merge(
-        self, kind: str, namespace: str, name: str, patch: Dict[str, Any],
-        *, subresource: Optional[str] = None,
-    ) -> Dict[str, Any]:
+    def patch_merge(self, kind: str, namespace: str, name: str,
+                    patch: Dict[str, Any], *, subresource: Optional[str] = None,
+                    ) -> Dict[str, Any]:
         """Strategic-merge-ish patch (recursive dict merge; lists replaced)."""
         with self._lock:
-            current = self._objects.get((kind, namespace, name))
-            if current is None:
+            merged = self._backend.fetch((kind, namespace, name))
+            if merged is None:
                 raise NotFoundError(f"{kind} {namespace}/{name} not found")
-            merged = jsoncopy(current)
 
             def merge(dst, src):
                 for k, v in src.items():
@@ -247,25 +364,27 @@ class InMemoryApiServer:
                     else:
                         dst[k] = jsoncopy(v)
 
-            merge(merged, patch)
-            merged["metadata"]["resourceVersion"] = current["metadata"]["resourceVersion"]
+            target = merged if subresource is None else merged.setdefault(
+                "status", {})
+            if subresource == "status":
+                merge(target, patch.get("status", patch))
+            else:
+                merge(merged, patch)
         return self.update(merged, subresource=subresource)
 
     def delete(self, kind: str, namespace: str, name: str) -> None:
+        key = (kind, namespace, name)
         with self._lock:
-            key = (kind, namespace, name)
-            current = self._objects.get(key)
+            current = self._backend.fetch(key)
             if current is None:
                 raise NotFoundError(f"{kind} {namespace}/{name} not found")
             if current["metadata"].get("finalizers"):
-                if not current["metadata"].get("deletionTimestamp"):
-                    self._index_remove(key, current)
-                    current["metadata"]["deletionTimestamp"] = now_iso()
-                    current["metadata"]["resourceVersion"] = self._next_rv()
-                    self._index_add(key, current)
-                    out = copy.deepcopy(current)
-                else:
+                if current["metadata"].get("deletionTimestamp"):
                     return
+                current["metadata"]["deletionTimestamp"] = now_iso()
+                current["metadata"]["resourceVersion"] = self._next_rv()
+                self._backend.put(key, current)
+                out = current
             else:
                 out = None
         if out is not None:
@@ -275,13 +394,12 @@ class InMemoryApiServer:
 
     def _finalize_delete(self, key: Key) -> None:
         with self._lock:
-            current = self._objects.pop(key, None)
+            current = self._backend.remove(key)
             if current is None:
                 return
-            self._index_remove(key, current)
             uid = current["metadata"]["uid"]
-            # ownerReference GC via the owner-uid index
-            dependents = list(self._owner_index.pop(uid, ()))
+            dependents = self._backend.dependents(uid)
+            self._backend.drop_owner(uid)
         self._notify("DELETED", current)
         for dep in dependents:
             try:
@@ -304,7 +422,7 @@ class InMemoryApiServer:
     # -- introspection -------------------------------------------------
     def count(self, kind: str) -> int:
         with self._lock:
-            return len(self._kind_index.get(kind, set()))
+            return self._backend.count(kind)
 
 
 class Watcher:

@@ -43,6 +43,13 @@ from ..utils.validation import validate_raycluster_metadata, validate_raycluster
 logger = logging.getLogger("kuberay.raycluster")
 
 
+def _deletion_ts(p):
+    ts = getattr(p, "deletion_timestamp", None)
+    if ts is None and hasattr(p, "metadata"):
+        ts = p.metadata.deletion_timestamp
+    return ts or None
+
+
 def set_condition(conditions: Optional[List[k8s.Condition]], cond_type: str,
                   status: str, reason: str, message: str = "") -> List[k8s.Condition]:
     """meta.SetStatusCondition analog: update-in-place, keep transition time."""
@@ -80,6 +87,20 @@ def find_suspend_status(cluster: RayCluster) -> Optional[str]:
     if condition_true(cluster.status.conditions, Cond.SUSPENDED):
         return Cond.SUSPENDED
     return None
+
+
+def should_delete_pod_view(view, node_type: str) -> (bool, str):
+    """View-projection variant of shouldDeletePod (same decision table)."""
+    if view.phase in ("Failed", "Succeeded"):
+        return True, (f"The {node_type} Pod {view.name} status is {view.phase} "
+                      "which is a terminal state.")
+    if view.phase == "Running" and view.ray_container_terminated:
+        if view.restart_policy == "Never":
+            return True, (f"Pod {view.name} Ray container terminated and "
+                          "restartPolicy=Never.")
+        return False, (f"Pod {view.name} Ray container terminated but will "
+                       f"restart (restartPolicy={view.restart_policy}).")
+    return False, f"Pod {view.name} is healthy ({view.phase})."
 
 
 def should_delete_pod(pod: k8s.Pod, node_type: str) -> (bool, str):
@@ -352,8 +373,26 @@ class RayClusterReconciler(Reconciler):
             k8s.Pod, cluster.metadata.namespace or "default",
             association.cluster_all_pods_selector(cluster.metadata.name))
 
-    def _active(self, pods: List[k8s.Pod]) -> List[k8s.Pod]:
-        return [p for p in pods if not p.metadata.deletion_timestamp]
+    def _list_cluster_pod_views(self, cluster: RayCluster):
+        """Hot-loop projection; falls back to typed pods for clients without
+        native views."""
+        namespace = cluster.metadata.namespace or "default"
+        selector = association.cluster_all_pods_selector(cluster.metadata.name)
+        list_views = getattr(self.client, "list_pod_views", None)
+        if list_views is not None:
+            return list_views(namespace, selector)
+        from ..kube.store import compute_pod_view
+        return [compute_pod_view(p.to_dict())
+                for p in self.client.list(k8s.Pod, namespace, selector)]
+
+    def _delete_pod_by_name(self, namespace: str, name: str) -> None:
+        try:
+            self.client.delete(k8s.Pod, namespace, name)
+        except NotFoundError:
+            pass
+
+    def _active(self, pods):
+        return [p for p in pods if not _deletion_ts(p)]
 
     def _reconcile_pods(self, cluster: RayCluster) -> None:
         namespace = cluster.metadata.namespace or "default"
@@ -363,8 +402,8 @@ class RayClusterReconciler(Reconciler):
         suspend_status = find_suspend_status(cluster)
         if cluster.spec.suspend or suspend_status == Cond.SUSPENDING:
             if suspend_status != Cond.SUSPENDED:
-                for pod in self._active(self._list_cluster_pods(cluster)):
-                    self.client.delete(pod)
+                for view in self._active(self._list_cluster_pod_views(cluster)):
+                    self._delete_pod_by_name(namespace, view.name)
                 return
             if cluster.spec.suspend:
                 return  # stays suspended, no pods
@@ -374,9 +413,9 @@ class RayClusterReconciler(Reconciler):
         if self.batch_scheduler is not None:
             self.batch_scheduler.do_batch_scheduling_on_submission(self.client, cluster)
 
-        pods = self._list_cluster_pods(cluster)
+        pods = self._list_cluster_pod_views(cluster)
         head_pods = [p for p in pods
-                     if (p.metadata.labels or {}).get(C.RAY_NODE_TYPE_LABEL_KEY) == RayNodeType.HEAD]
+                     if p.labels.get(C.RAY_NODE_TYPE_LABEL_KEY) == RayNodeType.HEAD]
 
         # ---- head pod singleton
         if not self.expectations.is_satisfied(
@@ -385,7 +424,7 @@ class RayClusterReconciler(Reconciler):
         active_heads = self._active(head_pods)
         if len(active_heads) == 1:
             head = active_heads[0]
-            delete, reason = should_delete_pod(head, RayNodeType.HEAD)
+            delete, reason = should_delete_pod_view(head, RayNodeType.HEAD)
             if delete and (cluster.metadata.annotations or {}).get(
                     C.DISABLE_PROVISIONED_HEAD_RESTART_ANNOTATION_KEY) == "true" \
                     and condition_true(cluster.status.conditions, Cond.PROVISIONED):
@@ -393,19 +432,19 @@ class RayClusterReconciler(Reconciler):
             if delete:
                 self.recorder.eventf(cluster, "Normal", "DeletedHeadPod", reason)
                 self.expectations.expect_delete_pod(namespace, name, "__head__",
-                                                    head.metadata.name)
-                self.client.delete(head)
+                                                    head.name)
+                self._delete_pod_by_name(namespace, head.name)
         elif len(active_heads) == 0:
             self._create_head_pod(cluster)
         else:
             # too many heads: keep oldest, delete the rest
             self.recorder.eventf(cluster, "Warning", "TooManyHeadPods",
                                  "Found %d head pods; deleting extras", len(active_heads))
-            for pod in sorted(active_heads,
-                              key=lambda p: p.metadata.creation_timestamp or "")[1:]:
+            for view in sorted(active_heads,
+                               key=lambda p: p.creation_timestamp or "")[1:]:
                 self.expectations.expect_delete_pod(namespace, name, "__head__",
-                                                    pod.metadata.name)
-                self.client.delete(pod)
+                                                    view.name)
+                self._delete_pod_by_name(namespace, view.name)
 
         # ---- worker groups
         for group in cluster.spec.worker_group_specs:
@@ -416,44 +455,41 @@ class RayClusterReconciler(Reconciler):
         return getattr(self.client, "server", None) or self.client
 
     def _reconcile_worker_group(self, cluster: RayCluster, group: WorkerGroupSpec,
-                                all_pods: List[k8s.Pod]) -> None:
+                                all_pods) -> None:
         namespace = cluster.metadata.namespace or "default"
         name = cluster.metadata.name
         group_pods = [p for p in all_pods
-                      if (p.metadata.labels or {}).get(C.RAY_NODE_TYPE_LABEL_KEY) == RayNodeType.WORKER
-                      and (p.metadata.labels or {}).get(C.RAY_NODE_GROUP_LABEL_KEY) == group.group_name]
+                      if p.labels.get(C.RAY_NODE_TYPE_LABEL_KEY) == RayNodeType.WORKER
+                      and p.labels.get(C.RAY_NODE_GROUP_LABEL_KEY) == group.group_name]
 
         # per-group suspend (controller.go:1080-1106)
         if group.suspend:
-            for pod in self._active(group_pods):
-                self.client.delete(pod)
+            for view in self._active(group_pods):
+                self._delete_pod_by_name(namespace, view.name)
             return
 
         if not self.expectations.is_satisfied(self._cache(), namespace, name, group.group_name):
             return
 
         # unhealthy deletion
-        for pod in self._active(group_pods):
-            delete, reason = should_delete_pod(pod, RayNodeType.WORKER)
+        for view in self._active(group_pods):
+            delete, reason = should_delete_pod_view(view, RayNodeType.WORKER)
             if delete:
                 self.recorder.eventf(cluster, "Normal", "DeletedWorkerPod", reason)
                 self.expectations.expect_delete_pod(namespace, name, group.group_name,
-                                                    pod.metadata.name)
-                self.client.delete(pod)
-                group_pods = [p for p in group_pods if p.metadata.name != pod.metadata.name]
+                                                    view.name)
+                self._delete_pod_by_name(namespace, view.name)
+                group_pods = [p for p in group_pods if p.name != view.name]
 
         # honor ScaleStrategy.WorkersToDelete (controller.go:1135-1153)
         to_delete = set(group.scale_strategy.workers_to_delete or [])
         if to_delete:
-            for pod in self._active(group_pods):
-                if pod.metadata.name in to_delete:
+            for view in self._active(group_pods):
+                if view.name in to_delete:
                     self.expectations.expect_delete_pod(namespace, name, group.group_name,
-                                                        pod.metadata.name)
-                    try:
-                        self.client.delete(pod)
-                    except NotFoundError:
-                        pass
-            group_pods = [p for p in group_pods if p.metadata.name not in to_delete]
+                                                        view.name)
+                    self._delete_pod_by_name(namespace, view.name)
+            group_pods = [p for p in group_pods if p.name not in to_delete]
 
         running = self._active(group_pods)
         desired = res.worker_group_desired_replicas(group) * max(group.num_of_hosts, 1)
@@ -467,10 +503,10 @@ class RayClusterReconciler(Reconciler):
             random_delete_enabled = os.environ.get(
                 C.ENABLE_RANDOM_POD_DELETE, "").lower() == "true"
             if not podlib.is_autoscaling_enabled(cluster.spec) or random_delete_enabled:
-                for pod in running[:(-diff)]:
+                for view in running[:(-diff)]:
                     self.expectations.expect_delete_pod(namespace, name, group.group_name,
-                                                        pod.metadata.name)
-                    self.client.delete(pod)
+                                                        view.name)
+                    self._delete_pod_by_name(namespace, view.name)
             # else: wait for the autoscaler to name victims via WorkersToDelete
 
     def _owner_crd_type(self, cluster: RayCluster) -> Optional[str]:
@@ -550,10 +586,12 @@ class RayClusterReconciler(Reconciler):
 
         status.observed_generation = cluster.metadata.generation
 
-        pods = self._list_cluster_pods(cluster)
-        pods = self._active(pods)
-        status.ready_worker_replicas = res.calculate_ready_replicas(pods)
-        status.available_worker_replicas = res.calculate_available_replicas(pods)
+        pods = self._active(self._list_cluster_pod_views(cluster))
+        workers = [p for p in pods
+                   if p.labels.get(C.RAY_NODE_TYPE_LABEL_KEY) == RayNodeType.WORKER]
+        status.ready_worker_replicas = sum(1 for p in workers if p.ready)
+        status.available_worker_replicas = sum(
+            1 for p in workers if p.phase == "Running")
         status.desired_worker_replicas = res.calculate_desired_replicas(cluster)
         status.min_worker_replicas = res.calculate_min_replicas(cluster)
         status.max_worker_replicas = res.calculate_max_replicas(cluster)
@@ -562,7 +600,8 @@ class RayClusterReconciler(Reconciler):
         status.desired_memory = totals["desiredMemory"]
         status.desired_gpu = totals["desiredGPU"]
 
-        all_running = bool(pods) and all(res.is_pod_running_and_ready(p) for p in pods)
+        all_running = bool(pods) and all(
+            p.phase == "Running" and p.ready for p in pods)
         if (reconcile_err is None
                 and len(pods) == status.desired_worker_replicas + 1
                 and all_running):
@@ -570,13 +609,13 @@ class RayClusterReconciler(Reconciler):
             status.reason = None
 
         # HeadPodReady condition
-        head = next((p for p in pods if (p.metadata.labels or {}).get(
+        head = next((p for p in pods if p.labels.get(
             C.RAY_NODE_TYPE_LABEL_KEY) == RayNodeType.HEAD), None)
         if head is None:
             status.conditions = set_condition(
                 status.conditions, Cond.HEAD_POD_READY, "False",
                 Reason.HEAD_POD_NOT_FOUND, "Head Pod not found")
-        elif res.is_pod_running_and_ready(head):
+        elif head.phase == "Running" and head.ready:
             status.conditions = set_condition(
                 status.conditions, Cond.HEAD_POD_READY, "True",
                 Reason.HEAD_POD_RUNNING_AND_READY, "Head Pod is running and ready")
@@ -656,11 +695,11 @@ class RayClusterReconciler(Reconciler):
                 endpoints[port.name] = str(port.node_port or port.port)
         cluster.status.endpoints = endpoints
 
-    def _update_head_info(self, cluster: RayCluster, head: Optional[k8s.Pod]) -> None:
+    def _update_head_info(self, cluster: RayCluster, head) -> None:
         info = cluster.status.head
         if head is not None:
-            info.pod_name = head.metadata.name
-            info.pod_ip = head.status.pod_ip
+            info.pod_name = head.name
+            info.pod_ip = head.pod_ip or None
         else:
             info.pod_name = None
             info.pod_ip = None
