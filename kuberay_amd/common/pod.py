@@ -545,6 +545,26 @@ def default_head_pod_template(cluster: RayCluster, head_spec, pod_name: str,
             template.spec.restart_policy = "Never"
 
     head_spec.ray_start_params = params  # defaulting is visible to callers (reference semantics)
+
+    # history-server collector sidecar (pod.go:304-315; gated)
+    from .. import features
+    if (features.enabled("RayClusterHistoryServer")
+            and cluster.spec.history_server_options is not None
+            and cluster.spec.history_server_options.collector_options is not None):
+        from ..historyserver.collector import build_collector_container
+        collector = build_collector_container(
+            cluster.spec.history_server_options.collector_options,
+            RayNodeType.HEAD, cluster.metadata.name,
+            cluster.metadata.namespace or "default",
+            names.fqdn_service_name(cluster, cluster.metadata.namespace or "default"))
+        if collector.image is None:
+            collector.image = template.spec.containers[C.RAY_CONTAINER_INDEX].image
+        template.spec.containers.append(collector)
+        ray_c = template.spec.containers[C.RAY_CONTAINER_INDEX]
+        ray_c.set_env_if_absent("RAY_enable_ray_event", "true")
+        ray_c.set_env_if_absent(
+            "RAY_enable_core_worker_ray_event_to_aggregator", "true")
+
     configure_gcs_fault_tolerance(template, cluster, RayNodeType.HEAD)
     _ensure_metrics_port(template)
     if is_auth_enabled(cluster.spec):

@@ -1,0 +1,51 @@
+"""Multi-process bench-path test: world_size=2 over gloo on CPU (the same
+aggregation code the driver exercises at N=1..8 over RCCL on the GPU node).
+"""
+import json
+import os
+import subprocess
+import sys
+
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+class TestBenchSingleProcess:
+    def test_bench_contract_json(self):
+        out = subprocess.run(
+            [sys.executable, "bench.py", "--clusters", "20", "--steps", "1",
+             "--warmup", "0", "--workers-per-cluster", "1"],
+            capture_output=True, text=True, cwd=REPO, timeout=300)
+        assert out.returncode == 0, out.stderr[-2000:]
+        line = out.stdout.strip().splitlines()[-1]
+        d = json.loads(line)
+        assert d["metric"] == "rayclusters_to_ready_per_sec"
+        assert d["value"] > 0
+        assert d["higher_is_better"] is True
+        assert d["scaling"] == "weak"
+        assert d["n_gpus"] == 1
+        assert d["config"]["p50_cr_to_ready_s"] > 0
+        assert d["vs_baseline"] > 0
+
+
+@pytest.mark.timeout(600)
+class TestBenchDistributed:
+    def test_world2_gloo(self):
+        """Two ranks, gloo backend, 127.0.0.1 rendezvous — mirrors the
+        driver's torch.distributed.run launch."""
+        out = subprocess.run(
+            [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+             "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+             "--master-port", "29517", "bench.py", "--clusters", "15",
+             "--steps", "1", "--warmup", "0", "--workers-per-cluster", "1"],
+            capture_output=True, text=True, cwd=REPO, timeout=540,
+            env={**os.environ, "MASTER_ADDR": "127.0.0.1"})
+        assert out.returncode == 0, (out.stdout[-1500:], out.stderr[-1500:])
+        lines = [l for l in out.stdout.splitlines() if l.startswith("{")]
+        assert len(lines) == 1, "exactly one JSON line from rank 0"
+        d = json.loads(lines[0])
+        assert d["n_gpus"] == 2
+        # whole-job aggregate: 2 ranks x 15 clusters
+        assert d["config"]["global_batch"] == 30
+        assert d["config"]["parallelism"] == "dp2"
