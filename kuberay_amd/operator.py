@@ -90,6 +90,11 @@ def build_manager(cfg: Configuration, server=None, client=None):
             "networkpolicy", "RayCluster",
             NetworkPolicyReconciler(client, recorder=recorder),
             owned_kinds=["NetworkPolicy"], workers=1))
+    if features.enabled("RayClusterMTLS"):
+        from .ops.mtls import MTLSReconciler
+        manager.add_controller(Controller(
+            "mtls", "RayCluster", MTLSReconciler(client, recorder=recorder),
+            owned_kinds=["Secret", "Pod"], workers=1))
 
     autoscaler = None
     if cfg.enable_mi355x_autoscaler and features.enabled("MI355XAutoscaler"):
