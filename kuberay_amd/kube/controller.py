@@ -64,6 +64,7 @@ class Controller:
         owned_kinds: Optional[List[str]] = None,
         workers: int = 4,
         use_predicates: bool = True,
+        watch_namespaces: Optional[List[str]] = None,
     ):
         self.name = name
         self.primary_kind = primary_kind
@@ -71,6 +72,8 @@ class Controller:
         self.owned_kinds = set(owned_kinds or [])
         self.workers = workers
         self.use_predicates = use_predicates
+        # informer scoping (reference: internal/managercache/cache.go:19-39)
+        self.watch_namespaces = set(watch_namespaces) if watch_namespaces else None
         self.queue = RateLimitingQueue()
         self._fingerprints: Dict[Request, Tuple] = {}
         self._threads: List[threading.Thread] = []
@@ -82,6 +85,9 @@ class Controller:
     def observe(self, event_type: str, obj: Dict[str, Any]) -> None:
         kind = obj.get("kind")
         meta = obj.get("metadata", {})
+        if self.watch_namespaces is not None and \
+                meta.get("namespace", "default") not in self.watch_namespaces:
+            return
         if kind == self.primary_kind:
             key: Request = (meta.get("namespace", "default"), meta.get("name", ""))
             if event_type == "MODIFIED" and self.use_predicates:

@@ -36,6 +36,27 @@ class HttpKubeClient(KubeClient):
         p = f"/apis/ray.io/v1/namespaces/{namespace}/{plural}"
         return f"{p}/{name}" if name else p
 
+    def _request_with_retry(self, method: str, path: str, **kwargs):
+        """Retry with backoff on transient transport errors / 5xx
+        (apiserversdk proxy.go:106-208 retry round-tripper analog)."""
+        import time
+        last = None
+        for attempt in range(3):
+            try:
+                resp = self._http.request(method, path, **kwargs)
+            except httpx.TransportError as e:
+                last = e
+                time.sleep(0.1 * (2 ** attempt))
+                continue
+            if resp.status_code >= 500:
+                last = ApiError(resp.status_code, resp.text[:200])
+                time.sleep(0.1 * (2 ** attempt))
+                continue
+            return resp
+        if isinstance(last, ApiError):
+            raise last
+        raise ApiError(503, f"transport error after retries: {last}")
+
     @staticmethod
     def _check(resp: httpx.Response):
         if resp.status_code == 404:
@@ -49,18 +70,18 @@ class HttpKubeClient(KubeClient):
     def create(self, obj):
         kind = obj.kind
         ns = obj.metadata.namespace or "default"
-        resp = self._check(self._http.post(self._path(kind, ns), json=obj.to_dict()))
+        resp = self._check(self._request_with_retry("POST", self._path(kind, ns), json=obj.to_dict()))
         return type(obj).from_dict(resp.json())
 
     def get(self, model, namespace, name):
         kind = _kind_of(model)
-        resp = self._check(self._http.get(self._path(kind, namespace, name)))
+        resp = self._check(self._request_with_retry("GET", self._path(kind, namespace, name)))
         return model.from_dict(resp.json())
 
     def list(self, model, namespace=None, label_selector=None):
         kind = _kind_of(model)
         ns = namespace or "default"
-        resp = self._check(self._http.get(self._path(kind, ns)))
+        resp = self._check(self._request_with_retry("GET", self._path(kind, ns)))
         items = [model.from_dict(o) for o in resp.json().get("items", [])]
         if label_selector:
             items = [o for o in items
@@ -72,14 +93,14 @@ class HttpKubeClient(KubeClient):
         # v2 proxy: update via create-path PUT is not exposed; use full PUT
         kind = obj.kind
         ns = obj.metadata.namespace or "default"
-        resp = self._check(self._http.put(
+        resp = self._check(self._request_with_retry("PUT", 
             self._path(kind, ns, obj.metadata.name), json=obj.to_dict()))
         return type(obj).from_dict(resp.json())
 
     def update_status(self, obj):
         kind = obj.kind
         ns = obj.metadata.namespace or "default"
-        resp = self._check(self._http.put(
+        resp = self._check(self._request_with_retry("PUT", 
             self._path(kind, ns, obj.metadata.name) + "/status",
             json=obj.to_dict()))
         return type(obj).from_dict(resp.json())
@@ -89,7 +110,7 @@ class HttpKubeClient(KubeClient):
         path = self._path(kind, namespace, name)
         if subresource:
             path += f"/{subresource}"
-        resp = self._check(self._http.patch(path, json=patch))
+        resp = self._check(self._request_with_retry("PATCH", path, json=patch))
         return model.from_dict(resp.json())
 
     def delete(self, model_or_obj, namespace=None, name=None):
@@ -97,4 +118,4 @@ class HttpKubeClient(KubeClient):
             namespace = model_or_obj.metadata.namespace or "default"
             name = model_or_obj.metadata.name
         kind = _kind_of(model_or_obj)
-        self._check(self._http.delete(self._path(kind, namespace, name)))
+        self._check(self._request_with_retry("DELETE", self._path(kind, namespace, name)))

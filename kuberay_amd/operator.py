@@ -66,20 +66,23 @@ def build_manager(cfg: Configuration, server=None, client=None):
 
     manager = Manager(server)
     workers = cfg.reconcile_concurrency
+    ns_scope = cfg.watch_namespaces or None
     manager.add_controller(Controller(
         "raycluster", "RayCluster",
         RayClusterReconciler(client, recorder=recorder, batch_scheduler=scheduler,
                              options=options, metrics=metrics),
         owned_kinds=["Pod", "Service", "Secret", "PersistentVolumeClaim", "Job"],
-        workers=workers))
+        workers=workers, watch_namespaces=ns_scope))
     manager.add_controller(Controller(
         "rayjob", "RayJob",
         RayJobReconciler(client, recorder=recorder, metrics=metrics),
-        owned_kinds=["RayCluster", "Job"], workers=workers))
+        owned_kinds=["RayCluster", "Job"], workers=workers,
+        watch_namespaces=ns_scope))
     manager.add_controller(Controller(
         "rayservice", "RayService",
         RayServiceReconciler(client, recorder=recorder, metrics=metrics),
-        owned_kinds=["RayCluster", "Service"], workers=workers))
+        owned_kinds=["RayCluster", "Service"], workers=workers,
+        watch_namespaces=ns_scope))
     if features.enabled("RayCronJob"):
         manager.add_controller(Controller(
             "raycronjob", "RayCronJob",
