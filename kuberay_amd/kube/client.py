@@ -110,10 +110,14 @@ class InMemoryClient(KubeClient):
 
     def update(self, obj):
         out = self.server.update(obj.to_dict())
+        # client-go semantics: refresh the passed object's resourceVersion
+        obj.metadata.resource_version = out["metadata"]["resourceVersion"]
+        obj.metadata.generation = out["metadata"].get("generation")
         return type(obj).from_dict(out)
 
     def update_status(self, obj):
         out = self.server.update(obj.to_dict(), subresource="status")
+        obj.metadata.resource_version = out["metadata"]["resourceVersion"]
         return type(obj).from_dict(out)
 
     def patch(self, model, namespace, name, patch, subresource=None):

@@ -90,12 +90,13 @@ class HttpKubeClient(KubeClient):
         return items
 
     def update(self, obj):
-        # v2 proxy: update via create-path PUT is not exposed; use full PUT
         kind = obj.kind
         ns = obj.metadata.namespace or "default"
-        resp = self._check(self._request_with_retry("PUT", 
+        resp = self._check(self._request_with_retry("PUT",
             self._path(kind, ns, obj.metadata.name), json=obj.to_dict()))
-        return type(obj).from_dict(resp.json())
+        out = resp.json()
+        obj.metadata.resource_version = out.get("metadata", {}).get("resourceVersion")
+        return type(obj).from_dict(out)
 
     def update_status(self, obj):
         kind = obj.kind

@@ -184,7 +184,9 @@ class RestClient(KubeClient):
         ns = obj.metadata.namespace or "default"
         resp = self._check(self._http.put(
             self._path(kind, ns, obj.metadata.name), json=obj.to_dict()))
-        return type(obj).from_dict(resp.json())
+        out = resp.json()
+        obj.metadata.resource_version = out.get("metadata", {}).get("resourceVersion")
+        return type(obj).from_dict(out)
 
     def update_status(self, obj):
         kind = obj.kind
@@ -192,7 +194,9 @@ class RestClient(KubeClient):
         resp = self._check(self._http.put(
             self._path(kind, ns, obj.metadata.name, "status"),
             json=obj.to_dict()))
-        return type(obj).from_dict(resp.json())
+        out = resp.json()
+        obj.metadata.resource_version = out.get("metadata", {}).get("resourceVersion")
+        return type(obj).from_dict(out)
 
     def patch(self, model, namespace, name, patch, subresource=None):
         kind = _kind_of(model)

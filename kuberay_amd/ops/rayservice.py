@@ -61,6 +61,8 @@ class RayServiceReconciler(Reconciler):
         self.metrics = metrics
         self.cluster_deletion_delay_s = cluster_deletion_delay_s
         self.requeue_seconds = REQUEUE_SECONDS
+        from .incremental import IncrementalUpgrader
+        self.upgrader = IncrementalUpgrader(client, self.recorder)
         # serve-config cache: (namespace, service, cluster) -> config hash
         self._serve_config_cache: Dict[Tuple[str, str, str], str] = {}
         # delayed old-cluster GC: (namespace, cluster) -> not-before time
@@ -108,14 +110,33 @@ class RayServiceReconciler(Reconciler):
             ready = self._reconcile_serve(svc, target, is_pending)
 
         if is_pending and ready:
-            self._promote(svc, target)
-            active, pending = target, None
+            if self._incremental_enabled(svc) and active is not None:
+                # Gateway-API weighted migration instead of instant promote
+                self.upgrader.ensure_gateway_infra(svc, active, target)
+                done = self.upgrader.step_traffic(svc, active, target)
+                self.client.update_status(svc)
+                if done:
+                    self._promote(svc, target)
+                    self.upgrader.cleanup(svc)
+                    active, pending = target, None
+                else:
+                    self._reconcile_services(svc, active)
+            else:
+                self._promote(svc, target)
+                active, pending = target, None
         elif active is not None:
             self._reconcile_services(svc, active)
 
         self._update_head_pod_serve_label(svc, active)
         self._calculate_status(svc, active, pending)
         return Result(requeue_after=self.requeue_seconds)
+
+    def _incremental_enabled(self, svc: RayService) -> bool:
+        from .. import features
+        us = svc.spec.upgrade_strategy
+        return (features.enabled("RayServiceIncrementalUpgrade")
+                and us is not None
+                and us.type == RayServiceUpgradeType.NEW_CLUSTER_WITH_INCREMENTAL_UPGRADE)
 
     # ------------------------------------------------------------------
     # cluster lifecycle
@@ -143,6 +164,14 @@ class RayServiceReconciler(Reconciler):
             return
         if pending is not None:
             if self._cluster_hash(pending) != goal:
+                if active is not None and self._cluster_hash(active) == goal:
+                    # spec reverted to the live cluster mid-upgrade → rollback
+                    svc.status.conditions = set_condition(
+                        svc.status.conditions, "RollbackInProgress", "True",
+                        "SpecRevertedToActive")
+                    self.upgrader.rollback(svc, pending)
+                    self.client.update_status(svc)
+                    return
                 # spec changed while upgrading: replace the pending cluster
                 self._delete_cluster_later(pending, delay=0)
                 svc.status.pending_service_status.ray_cluster_name = \

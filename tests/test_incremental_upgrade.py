@@ -1,0 +1,139 @@
+"""Incremental upgrade tests (reference analog: test/e2eincrementalupgrade +
+rayservice_controller.go Gateway/HTTPRoute/TargetCapacity logic)."""
+import time
+
+import pytest
+
+import kuberay_amd.features as features
+from kuberay_amd.models import RayCluster, RayService
+from kuberay_amd.testing import ControlPlane, simple_raycluster
+
+SERVE_CONFIG = "applications:\n- name: app1\n  import_path: m.g\n"
+
+
+def make_service(name="svc1", interval=0, step=50):
+    return RayService.from_dict({
+        "apiVersion": "ray.io/v1", "kind": "RayService",
+        "metadata": {"name": name, "namespace": "default"},
+        "spec": {
+            "serveConfigV2": SERVE_CONFIG,
+            "rayClusterConfig": simple_raycluster("x", workers=1).spec.to_dict(),
+            "upgradeStrategy": {
+                "type": "NewClusterWithIncrementalUpgrade",
+                "clusterUpgradeOptions": {
+                    "gatewayClassName": "istio",
+                    "stepSizePercent": step,
+                    "intervalSeconds": interval,
+                    "maxSurgePercent": 100,
+                }},
+        },
+    })
+
+
+@pytest.fixture()
+def cp():
+    features.set_gate("RayServiceIncrementalUpgrade", True)
+    plane = ControlPlane(kubelet_delay=0.01, poll_seconds=0.05)
+    plane.rayservice_reconciler.cluster_deletion_delay_s = 0.2
+    plane.start()
+    yield plane
+    plane.stop()
+    features.reset()
+
+
+def svc_of(cp, name="svc1"):
+    return cp.client.get(RayService, "default", name)
+
+
+def wait_ready(cp, name="svc1", timeout=25):
+    return cp.wait_for(lambda: svc_of(cp, name).condition_true("Ready"), timeout)
+
+
+def trigger_upgrade(cp):
+    svc = svc_of(cp)
+    svc.spec.ray_cluster_spec.worker_group_specs[0].template.spec \
+        .containers[0].image = "rayproject/ray:2.47.0-rocm"
+    cp.client.update(svc)
+
+
+class TestIncrementalUpgrade:
+    def test_gateway_and_route_created_with_weighted_migration(self, cp):
+        cp.client.create(make_service())
+        assert wait_ready(cp)
+        old_active = svc_of(cp).status.active_service_status.ray_cluster_name
+        trigger_upgrade(cp)
+
+        # migration completes (2 steps of 50%) and promotes
+        def promoted():
+            s = svc_of(cp)
+            return (s.status.active_service_status.ray_cluster_name
+                    not in (None, old_active) and s.condition_true("Ready"))
+        assert cp.wait_for(promoted, timeout=30)
+
+        # gateway infra existed during migration and is cleaned after
+        assert cp.wait_for(lambda: cp.server.count("HTTPRoute") == 0, timeout=10)
+        assert cp.server.count("Gateway") == 0
+
+    def test_traffic_percent_tracked_in_status(self, cp):
+        cp.client.create(make_service(interval=3600, step=40))
+        assert wait_ready(cp)
+        trigger_upgrade(cp)
+
+        def first_step():
+            s = svc_of(cp)
+            return (s.status.pending_service_status.traffic_routed_percent or 0) >= 40
+        assert cp.wait_for(first_step, timeout=25)
+        s = svc_of(cp)
+        # interval=1h: held at the first step, not promoted
+        assert s.status.pending_service_status.traffic_routed_percent == 40
+        assert s.status.pending_service_status.last_traffic_migrated_time
+        assert s.status.active_service_status.ray_cluster_name  # still active
+        route = cp.server.list("HTTPRoute")[0]
+        weights = {b["name"]: b["weight"]
+                   for b in route["spec"]["rules"][0]["backendRefs"]}
+        assert sorted(weights.values()) == [40, 60]
+
+    def test_rollback_on_spec_revert(self, cp):
+        cp.client.create(make_service(interval=3600, step=10))
+        assert wait_ready(cp)
+        original_image = svc_of(cp).spec.ray_cluster_spec \
+            .worker_group_specs[0].template.spec.containers[0].image
+        active = svc_of(cp).status.active_service_status.ray_cluster_name
+        trigger_upgrade(cp)
+        assert cp.wait_for(
+            lambda: svc_of(cp).status.pending_service_status.ray_cluster_name,
+            timeout=20)
+        # revert the spec to the running cluster's shape mid-upgrade
+        svc = svc_of(cp)
+        svc.spec.ray_cluster_spec.worker_group_specs[0].template.spec \
+            .containers[0].image = original_image
+        cp.client.update(svc)
+
+        def rolled_back():
+            s = svc_of(cp)
+            return (not s.status.pending_service_status.ray_cluster_name
+                    and s.status.active_service_status.ray_cluster_name == active)
+        assert cp.wait_for(rolled_back, timeout=25)
+
+    def test_gate_off_promotes_instantly(self):
+        plane = ControlPlane(kubelet_delay=0.01, poll_seconds=0.05)
+        plane.rayservice_reconciler.cluster_deletion_delay_s = 0.2
+        plane.start()
+        try:
+            plane.client.create(make_service())
+            assert plane.wait_for(
+                lambda: plane.client.get(RayService, "default", "svc1")
+                .condition_true("Ready"), timeout=25)
+            old = plane.client.get(RayService, "default", "svc1") \
+                .status.active_service_status.ray_cluster_name
+            svc = plane.client.get(RayService, "default", "svc1")
+            svc.spec.ray_cluster_spec.worker_group_specs[0].template.spec \
+                .containers[0].image = "rayproject/ray:2.47.0-rocm"
+            plane.client.update(svc)
+            assert plane.wait_for(
+                lambda: plane.client.get(RayService, "default", "svc1")
+                .status.active_service_status.ray_cluster_name not in (None, old),
+                timeout=25)
+            assert plane.server.count("Gateway") == 0  # no gateway infra
+        finally:
+            plane.stop()
