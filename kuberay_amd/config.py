@@ -44,6 +44,9 @@ class Configuration(K8sModel):
     # "kubernetes" (real cluster via REST)
     backend: str = "memory"
     kubeconfig: Optional[str] = None
+    # memory backend durability: JSONL snapshot path ("" = off)
+    state_file: str = ""
+    state_snapshot_interval_s: float = 30.0
 
 
 def load_config(argv: Optional[List[str]] = None) -> Configuration:
@@ -59,6 +62,7 @@ def load_config(argv: Optional[List[str]] = None) -> Configuration:
     parser.add_argument("--backend", choices=["memory", "kubernetes"])
     parser.add_argument("--kubeconfig")
     parser.add_argument("--log-file")
+    parser.add_argument("--state-file")
     parser.add_argument("--no-metrics", action="store_true", default=None)
     args = parser.parse_args(argv)
 
@@ -76,7 +80,7 @@ def load_config(argv: Optional[List[str]] = None) -> Configuration:
         ("enable_batch_scheduler", "enable_batch_scheduler"),
         ("feature_gates", "feature_gates"),
         ("backend", "backend"), ("kubeconfig", "kubeconfig"),
-        ("log_file", "log_file"),
+        ("log_file", "log_file"), ("state_file", "state_file"),
     ]:
         val = getattr(args, flag, None)
         if val is not None:

@@ -65,6 +65,7 @@ class Controller:
         workers: int = 4,
         use_predicates: bool = True,
         watch_namespaces: Optional[List[str]] = None,
+        metrics=None,
     ):
         self.name = name
         self.primary_kind = primary_kind
@@ -80,6 +81,7 @@ class Controller:
         self._stopped = threading.Event()
         self.reconcile_count = 0
         self.error_count = 0
+        self.metrics = metrics
 
     # -- event routing -------------------------------------------------
     def observe(self, event_type: str, obj: Dict[str, Any]) -> None:
@@ -107,10 +109,13 @@ class Controller:
 
     # -- workers -------------------------------------------------------
     def _worker(self) -> None:
+        import time as _time
         while not self._stopped.is_set():
             item = self.queue.get(timeout=0.5)
             if item is None:
                 continue
+            outcome = "success"
+            t0 = _time.perf_counter()
             try:
                 result = self.reconciler.reconcile(item)
                 self.reconcile_count += 1
@@ -121,15 +126,23 @@ class Controller:
                     self.queue.add_rate_limited(item)
             except ConflictError:
                 # optimistic-concurrency loss: immediate-ish retry
+                outcome = "conflict"
                 self.reconcile_count += 1
                 self.queue.add_rate_limited(item)
             except Exception:
+                outcome = "error"
                 self.error_count += 1
                 logger.error("reconcile %s %s failed:\n%s", self.name, item,
                              traceback.format_exc())
                 self.queue.add_rate_limited(item)
             finally:
                 self.queue.done(item)
+                if self.metrics is not None:
+                    self.metrics.reconcile_total.labels(self.name, outcome).inc()
+                    self.metrics.reconcile_duration.labels(self.name).observe(
+                        _time.perf_counter() - t0)
+                    self.metrics.workqueue_depth.labels(self.name).set(
+                        len(self.queue))
 
     def start(self) -> None:
         for i in range(self.workers):

@@ -1,0 +1,96 @@
+"""Control-plane state checkpoint/resume for the in-memory backend.
+
+The reference's "resume" story is CR status as persisted state machine (the
+real apiserver/etcd persists everything; SURVEY.md §5 checkpoint/resume).
+The memory backend gets the same durability here: periodic JSONL snapshots
+of the whole object store, atomically swapped, restored on operator start.
+Restores preserve uid/resourceVersion/generation so controllers resume the
+exact state machines (suspend cycles, job retries, serve-config caches are
+recomputed)."""
+from __future__ import annotations
+
+import json
+import os
+import threading
+from typing import Optional
+
+from .store import InMemoryApiServer
+
+
+def save_snapshot(server: InMemoryApiServer, path: str) -> int:
+    """Write every stored object as one JSON line. Returns object count."""
+    objects = []
+    with server._lock:
+        backend = server._backend
+        kinds = set()
+        # enumerate via kind index (both backends expose count/list)
+        for kind in ("RayCluster", "RayJob", "RayService", "RayCronJob", "Pod",
+                     "Service", "Secret", "ConfigMap", "PersistentVolumeClaim",
+                     "ServiceAccount", "Role", "RoleBinding", "Job",
+                     "NetworkPolicy", "Ingress", "Event"):
+            kinds.add(kind)
+        for kind in kinds:
+            objects.extend(backend.list(kind, None, None))
+        rv = server._rv
+    tmp = path + ".tmp"
+    os.makedirs(os.path.dirname(os.path.abspath(path)), exist_ok=True)
+    with open(tmp, "w") as f:
+        f.write(json.dumps({"__meta__": {"resourceVersion": rv}}) + "\n")
+        for obj in objects:
+            f.write(json.dumps(obj, separators=(",", ":")) + "\n")
+    os.replace(tmp, path)
+    return len(objects)
+
+
+def load_snapshot(server: InMemoryApiServer, path: str) -> int:
+    """Restore a snapshot into an empty server. Returns object count."""
+    if not os.path.exists(path):
+        return 0
+    count = 0
+    with open(path) as f:
+        first = f.readline()
+        meta = json.loads(first).get("__meta__", {}) if first.strip() else {}
+        with server._lock:
+            server._rv = int(meta.get("resourceVersion", 0))
+            for line in f:
+                line = line.strip()
+                if not line:
+                    continue
+                obj = json.loads(line)
+                key = server._key_of(obj)
+                server._backend.put(key, obj)
+                count += 1
+    return count
+
+
+class SnapshotLoop:
+    """Periodic snapshotter (wire into the operator's memory backend)."""
+
+    def __init__(self, server: InMemoryApiServer, path: str,
+                 interval_s: float = 30.0):
+        self.server = server
+        self.path = path
+        self.interval_s = interval_s
+        self._stop = threading.Event()
+        self._thread: Optional[threading.Thread] = None
+
+    def start(self) -> None:
+        def loop():
+            while not self._stop.is_set():
+                self._stop.wait(self.interval_s)
+                try:
+                    save_snapshot(self.server, self.path)
+                except Exception:
+                    pass
+        self._thread = threading.Thread(target=loop, name="state-snapshot",
+                                        daemon=True)
+        self._thread.start()
+
+    def stop(self) -> None:
+        self._stop.set()
+        if self._thread:
+            self._thread.join(timeout=2)
+        try:
+            save_snapshot(self.server, self.path)
+        except Exception:
+            pass

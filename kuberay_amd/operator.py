@@ -72,17 +72,17 @@ def build_manager(cfg: Configuration, server=None, client=None):
         RayClusterReconciler(client, recorder=recorder, batch_scheduler=scheduler,
                              options=options, metrics=metrics),
         owned_kinds=["Pod", "Service", "Secret", "PersistentVolumeClaim", "Job"],
-        workers=workers, watch_namespaces=ns_scope))
+        workers=workers, watch_namespaces=ns_scope, metrics=metrics))
     manager.add_controller(Controller(
         "rayjob", "RayJob",
         RayJobReconciler(client, recorder=recorder, metrics=metrics),
         owned_kinds=["RayCluster", "Job"], workers=workers,
-        watch_namespaces=ns_scope))
+        watch_namespaces=ns_scope, metrics=metrics))
     manager.add_controller(Controller(
         "rayservice", "RayService",
         RayServiceReconciler(client, recorder=recorder, metrics=metrics),
         owned_kinds=["RayCluster", "Service"], workers=workers,
-        watch_namespaces=ns_scope))
+        watch_namespaces=ns_scope, metrics=metrics))
     if features.enabled("RayCronJob"):
         manager.add_controller(Controller(
             "raycronjob", "RayCronJob",
@@ -166,7 +166,17 @@ def main(argv=None) -> int:
     health.start()
 
     kubelet = None
+    snapshotter = None
     if cfg.backend == "memory":
+        if cfg.state_file:
+            from .kube.snapshot import SnapshotLoop, load_snapshot
+            restored = load_snapshot(manager.server, cfg.state_file)
+            if restored:
+                logger.info("restored %d objects from %s", restored,
+                            cfg.state_file)
+            snapshotter = SnapshotLoop(manager.server, cfg.state_file,
+                                       cfg.state_snapshot_interval_s)
+            snapshotter.start()
         from .kube.kubelet import SimKubelet
         gate = None
         if features.enabled("MI355XGpuHealthProbes"):
@@ -193,6 +203,8 @@ def main(argv=None) -> int:
             autoscaler.stop()
         if kubelet is not None:
             kubelet.stop()
+        if snapshotter is not None:
+            snapshotter.stop()
         manager.stop()
         health.stop()
     return 0
