@@ -285,6 +285,15 @@ def create_app(client: Optional[KubeClient] = None,
     def healthz():
         return {"status": "ok"}
 
+    # web dashboard (reference: dashboard/ Next.js UI)
+    from fastapi.responses import HTMLResponse
+
+    from .dashboard import DASHBOARD_HTML
+
+    @app.get("/", response_class=HTMLResponse)
+    def dashboard():
+        return DASHBOARD_HTML
+
     return app
 
 
