@@ -97,7 +97,7 @@ class InMemoryClient(KubeClient):
     def create(self, obj):
         data = obj.to_dict()
         data["kind"] = obj.kind
-        out = self.server.create(data)
+        out = self.server.create(data, assume_owned=True)
         return type(obj).from_dict(out)
 
     def get(self, model, namespace, name):
@@ -109,14 +109,15 @@ class InMemoryClient(KubeClient):
         return [model.from_dict(o) for o in self.server.list(kind, namespace, label_selector)]
 
     def update(self, obj):
-        out = self.server.update(obj.to_dict())
+        out = self.server.update(obj.to_dict(), assume_owned=True)
         # client-go semantics: refresh the passed object's resourceVersion
         obj.metadata.resource_version = out["metadata"]["resourceVersion"]
         obj.metadata.generation = out["metadata"].get("generation")
         return type(obj).from_dict(out)
 
     def update_status(self, obj):
-        out = self.server.update(obj.to_dict(), subresource="status")
+        out = self.server.update(obj.to_dict(), subresource="status",
+                                 assume_owned=True)
         obj.metadata.resource_version = out["metadata"]["resourceVersion"]
         return type(obj).from_dict(out)
 

@@ -25,7 +25,7 @@ import dataclasses
 import threading
 import time
 import uuid
-from collections import defaultdict
+from collections import defaultdict, deque
 from typing import Any, Dict, Iterable, List, Optional, Tuple
 
 Key = Tuple[str, str, str]  # (kind, namespace, name)
@@ -266,8 +266,9 @@ class InMemoryApiServer:
                 meta.get("name", ""))
 
     # -- verbs ---------------------------------------------------------
-    def create(self, obj: Dict[str, Any]) -> Dict[str, Any]:
-        obj = jsoncopy(obj)
+    def create(self, obj: Dict[str, Any], *, assume_owned: bool = False) -> Dict[str, Any]:
+        if not assume_owned:
+            obj = jsoncopy(obj)
         meta = obj.setdefault("metadata", {})
         meta.setdefault("namespace", "default")
         if not meta.get("name"):
@@ -309,8 +310,10 @@ class InMemoryApiServer:
         with self._lock:
             return self._backend.list_views(namespace, label_selector)
 
-    def update(self, obj: Dict[str, Any], *, subresource: Optional[str] = None) -> Dict[str, Any]:
-        obj = jsoncopy(obj)
+    def update(self, obj: Dict[str, Any], *, subresource: Optional[str] = None,
+               assume_owned: bool = False) -> Dict[str, Any]:
+        if not assume_owned:
+            obj = jsoncopy(obj)
         key = self._key_of(obj)
         with self._lock:
             current = self._backend.fetch(key)
@@ -424,6 +427,10 @@ class InMemoryApiServer:
         with self._lock:
             return self._backend.count(kind)
 
+    def contains(self, kind: str, namespace: str, name: str) -> bool:
+        with self._lock:
+            return self._backend.contains((kind, namespace, name))
+
 
 class Watcher:
     """A watch stream: buffered (event_type, object) pairs."""
@@ -432,7 +439,7 @@ class Watcher:
         self._server = server
         self._kinds = kinds
         self._cond = threading.Condition()
-        self._events: List[Tuple[str, Dict[str, Any]]] = []
+        self._events: deque = deque()
         self._stopped = False
 
     def push(self, event_type: str, obj: Dict[str, Any]) -> None:
@@ -449,12 +456,12 @@ class Watcher:
             if not self._events:
                 self._cond.wait(timeout)
             if self._events:
-                return self._events.pop(0)
+                return self._events.popleft()
             return None
 
     def drain(self) -> List[Tuple[str, Dict[str, Any]]]:
         with self._cond:
-            events, self._events = self._events, []
+            events, self._events = list(self._events), deque()
             return events
 
     def stop(self) -> None:

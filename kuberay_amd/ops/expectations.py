@@ -54,21 +54,28 @@ class RayClusterScaleExpectations(ScaleExpectations):
         self._expect((namespace, cluster, group), pod_name, "delete")
 
     def is_satisfied(self, cache, namespace, cluster, group) -> bool:
-        """cache: object with try_get(kind, namespace, name) -> Optional[dict]."""
+        """cache: InMemoryApiServer-shaped (contains()/try_get())."""
         key = (namespace, cluster, group)
         now = time.monotonic()
+        contains = getattr(cache, "contains", None)
         with self._lock:
             pending = self._pending.get(key)
             if not pending:
                 return True
             satisfied = []
             for pod_name, (op, deadline) in pending.items():
-                observed = cache.try_get("Pod", namespace, pod_name)
                 if op == "create":
-                    ok = observed is not None
+                    if contains is not None:
+                        ok = contains("Pod", namespace, pod_name)
+                    else:
+                        ok = cache.try_get("Pod", namespace, pod_name) is not None
                 else:
-                    ok = observed is None or bool(
-                        observed.get("metadata", {}).get("deletionTimestamp"))
+                    if contains is not None and not contains("Pod", namespace, pod_name):
+                        ok = True
+                    else:
+                        observed = cache.try_get("Pod", namespace, pod_name)
+                        ok = observed is None or bool(
+                            observed.get("metadata", {}).get("deletionTimestamp"))
                 if ok or now > deadline:
                     satisfied.append(pod_name)
             for pod_name in satisfied:
