@@ -302,8 +302,22 @@ def main(argv=None) -> int:
     import uvicorn
 
     parser = argparse.ArgumentParser(prog="kuberay-amd-apiserver")
-    parser.add_argument("--port", type=int, default=8888)
+    parser.add_argument("--port", type=int, default=8888,
+                        help="HTTP gateway port (reference :8888)")
+    parser.add_argument("--grpc-port", type=int, default=8887,
+                        help="gRPC port (reference :8887); 0 disables")
     parser.add_argument("--host", default="0.0.0.0")
     args = parser.parse_args(argv)
-    uvicorn.run(create_app(), host=args.host, port=args.port)
+    from ..kube.client import InMemoryClient
+    client = InMemoryClient()
+    grpc_server = None
+    if args.grpc_port:
+        from .grpc_api import create_grpc_server
+        grpc_server = create_grpc_server(client, port=args.grpc_port)
+        grpc_server.start()
+    try:
+        uvicorn.run(create_app(client), host=args.host, port=args.port)
+    finally:
+        if grpc_server is not None:
+            grpc_server.stop(2)
     return 0

@@ -206,3 +206,64 @@ class TestKrayCli:
         assert r.exit_code == 0
         assert "amd.com/gpu: '4'" in r.output
         assert backing.try_get(RayCluster, "ns1", "k1") is None
+
+
+class TestGrpcApi:
+    @pytest.fixture()
+    def grpc_stack(self):
+        import grpc
+        from kuberay_amd.apiserver.grpc_api import (
+            Cluster, ComputeTemplate, DeleteRequest, GetRequest, ListRequest,
+            ListClusterResponse, create_grpc_server)
+        from kuberay_amd.kube.client import InMemoryClient
+        client = InMemoryClient()
+        server = create_grpc_server(client, port=0)
+        port = server.add_insecure_port("127.0.0.1:0")
+        server.start()
+        channel = grpc.insecure_channel(f"127.0.0.1:{port}")
+        yield channel, client
+        server.stop(0)
+
+    def _call(self, channel, service, method, request, resp_cls):
+        import grpc
+        fn = channel.unary_unary(
+            f"/kuberayamd.v1.{service}/{method}",
+            request_serializer=lambda m: m.SerializeToString(),
+            response_deserializer=resp_cls.FromString)
+        return fn(request, timeout=5)
+
+    def test_cluster_crud_over_grpc(self, grpc_stack):
+        import json
+        import grpc
+        from kuberay_amd.apiserver.grpc_api import (
+            Cluster, ComputeTemplate, DeleteRequest, Empty, GetRequest,
+            ListRequest, ListClusterResponse)
+        channel, backing = grpc_stack
+        self._call(channel, "ComputeTemplateService", "CreateComputeTemplate",
+                   ComputeTemplate(name="tpl", namespace="ns1", cpu=4,
+                                   memory=8, gpu=2), ComputeTemplate)
+        spec = {"headGroupSpec": {"computeTemplate": "tpl"},
+                "workerGroupSpec": [{"groupName": "g", "computeTemplate": "tpl",
+                                     "replicas": 1, "maxReplicas": 2}]}
+        created = self._call(channel, "ClusterService", "CreateCluster",
+                             Cluster(name="gc1", namespace="ns1",
+                                     version="2.46.0",
+                                     spec_json=json.dumps(spec)), Cluster)
+        assert created.name == "gc1"
+        # expanded to amd.com/gpu in the backing store
+        rc = backing.server.get("RayCluster", "ns1", "gc1")
+        limits = rc["spec"]["workerGroupSpecs"][0]["template"]["spec"][
+            "containers"][0]["resources"]["limits"]
+        assert limits["amd.com/gpu"] == "2"
+
+        got = self._call(channel, "ClusterService", "GetCluster",
+                         GetRequest(name="gc1", namespace="ns1"), Cluster)
+        assert got.version == "2.46.0"
+        listed = self._call(channel, "ClusterService", "ListCluster",
+                            ListRequest(namespace="ns1"), ListClusterResponse)
+        assert len(listed.clusters) == 1
+        self._call(channel, "ClusterService", "DeleteCluster",
+                   DeleteRequest(name="gc1", namespace="ns1"), Empty)
+        with pytest.raises(grpc.RpcError):
+            self._call(channel, "ClusterService", "GetCluster",
+                       GetRequest(name="gc1", namespace="ns1"), Cluster)
