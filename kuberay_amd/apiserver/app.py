@@ -21,7 +21,7 @@ from fastapi.responses import JSONResponse
 
 from ..kube import objects as k8s
 from ..kube.client import InMemoryClient, KubeClient, model_for_kind
-from ..kube.store import AlreadyExistsError, ApiError, NotFoundError
+from ..kube.store import ApiError
 from ..models import RayCluster, RayJob, RayService
 from ..utils import constants as C
 from ..utils.validation import (

@@ -18,7 +18,7 @@ from __future__ import annotations
 
 import json
 from concurrent import futures
-from typing import Any, Dict, Optional
+from typing import Any, Dict
 
 import grpc
 from google.protobuf import descriptor_pb2, descriptor_pool, message_factory

@@ -9,9 +9,7 @@ defaulted to nvidia.com/gpu; dropped).
 """
 from __future__ import annotations
 
-import json
 import os
-import sys
 from typing import Optional
 
 import click

@@ -15,7 +15,6 @@ import yaml
 
 from ..kube.objects import (
     Container,
-    EnvVar,
     Job,
     JobSpec,
     ObjectMeta,

@@ -10,7 +10,7 @@
 from __future__ import annotations
 
 import os
-from typing import Dict, List, Optional
+from typing import Dict, Optional
 
 from ..kube.objects import Service, ServicePort, ServiceSpec, ObjectMeta
 from ..models.raycluster import RayCluster, RayNodeType

@@ -10,8 +10,6 @@ import argparse
 from typing import List, Optional
 
 import yaml
-from pydantic import Field
-
 from .kube.objects import K8sModel
 
 

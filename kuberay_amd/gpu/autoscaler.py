@@ -31,7 +31,6 @@ from ..common import association
 from ..kube import objects as k8s
 from ..kube.client import KubeClient
 from ..models import RayCluster
-from ..utils import constants as C
 from ..utils.resources import pod_gpu_count
 
 AMD_AUTOSCALER_ANNOTATION = "ray.io/amd-gpu-autoscaler"

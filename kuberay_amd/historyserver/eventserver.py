@@ -7,7 +7,7 @@ historyserver/pkg/historyserver/timeline.go).
 from __future__ import annotations
 
 import json
-from typing import Any, Dict, List, Optional
+from typing import Any, Dict, List
 
 from .storage import StorageReader, decompress
 

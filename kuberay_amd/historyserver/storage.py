@@ -10,10 +10,9 @@ from __future__ import annotations
 
 import gzip
 import io
-import json
 import os
 import threading
-from typing import Dict, Iterable, List, Optional
+from typing import Dict, List
 
 
 class StorageWriter:

@@ -21,7 +21,7 @@ import dataclasses
 import logging
 import threading
 import traceback
-from typing import Any, Callable, Dict, List, Optional, Tuple
+from typing import Any, Dict, List, Optional, Tuple
 
 from .store import ConflictError, InMemoryApiServer
 from .workqueue import RateLimitingQueue

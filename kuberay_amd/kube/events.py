@@ -8,7 +8,7 @@ scale.
 from __future__ import annotations
 
 import threading
-from typing import Any, Dict, Optional
+from typing import Any
 
 from .store import InMemoryApiServer, NotFoundError, now_iso
 

@@ -5,7 +5,7 @@ in-memory client, through kuberay_amd.apiserver's v2 routes (or any real
 K8s apiserver exposing /apis/ray.io/v1)."""
 from __future__ import annotations
 
-from typing import Any, Dict, Optional, Type
+from typing import Optional
 
 import httpx
 

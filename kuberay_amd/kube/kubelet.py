@@ -19,7 +19,7 @@ import heapq
 import itertools
 import threading
 import time
-from typing import Callable, Dict, List, Optional, Tuple
+from typing import Callable, List, Optional, Tuple
 
 from ..utils import constants as C
 from .store import InMemoryApiServer, NotFoundError, now_iso

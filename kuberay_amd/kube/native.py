@@ -8,7 +8,7 @@ reads precomputed pod views without touching the blob at all.
 from __future__ import annotations
 
 import json
-from typing import Any, Dict, List, Optional, Tuple
+from typing import Any, Dict, List, Optional
 
 from .._native import engine  # raises ImportError if not built
 from .store import Key, PodView

@@ -14,7 +14,6 @@ import base64
 import json
 import logging
 import os
-import ssl
 import tempfile
 import threading
 from typing import Any, Dict, Iterable, List, Optional, Tuple
@@ -22,12 +21,11 @@ from typing import Any, Dict, Iterable, List, Optional, Tuple
 import httpx
 import yaml
 
-from .client import KubeClient, _kind_of, model_for_kind
+from .client import KubeClient, _kind_of
 from .store import (
     AlreadyExistsError,
     ApiError,
     ConflictError,
-    Key,
     NotFoundError,
     PodView,
     compute_pod_view,

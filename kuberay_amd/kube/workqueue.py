@@ -10,7 +10,7 @@ from __future__ import annotations
 import heapq
 import threading
 import time
-from typing import Any, Dict, Hashable, List, Optional, Set, Tuple
+from typing import Dict, Hashable, List, Optional, Set, Tuple
 
 
 class RateLimitingQueue:

@@ -1,7 +1,7 @@
 """ray.io/v1 RayJob types (reference: ray-operator/apis/ray/v1/rayjob_types.go)."""
 from __future__ import annotations
 
-from typing import Any, Dict, List, Optional
+from typing import Dict, List, Optional
 
 from pydantic import Field
 

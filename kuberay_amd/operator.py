@@ -16,7 +16,6 @@ import logging
 import signal
 import sys
 import threading
-import time
 
 from . import features
 from .config import Configuration, load_config

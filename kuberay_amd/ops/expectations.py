@@ -11,7 +11,7 @@ from __future__ import annotations
 
 import threading
 import time
-from typing import Dict, Optional, Set, Tuple
+from typing import Dict, Tuple
 
 GroupKey = Tuple[str, str, str]  # (namespace, cluster, group)
 

@@ -28,7 +28,7 @@ from ..kube import objects as k8s
 from ..kube.client import KubeClient
 from ..kube.controller import Reconciler, Request, Result
 from ..kube.events import EventRecorder, NullRecorder
-from ..kube.store import AlreadyExistsError, NotFoundError
+from ..kube.store import AlreadyExistsError
 from ..models import RayCluster
 from ..utils import constants as C
 from ..utils import names

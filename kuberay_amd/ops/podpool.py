@@ -13,8 +13,7 @@ from typing import Dict, List, Optional
 
 from ..kube import objects as k8s
 from ..kube.client import KubeClient
-from ..kube.store import AlreadyExistsError, NotFoundError
-from ..utils import constants as C
+from ..kube.store import NotFoundError
 
 WARM_POD_LABEL = "ray.io/warm-pod"
 WARM_POOL_LABEL = "ray.io/warm-pool"

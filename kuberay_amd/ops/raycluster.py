@@ -37,7 +37,7 @@ from ..models.raycluster import (
 from ..utils import constants as C
 from ..utils import names
 from ..utils import resources as res
-from ..utils.hashing import hash_without_replicas_and_workers_to_delete, json_hash
+from ..utils.hashing import hash_without_replicas_and_workers_to_delete
 from ..utils.validation import validate_raycluster_metadata, validate_raycluster_spec
 
 logger = logging.getLogger("kuberay.raycluster")

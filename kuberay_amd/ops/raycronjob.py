@@ -14,7 +14,7 @@ from ..kube import objects as k8s
 from ..kube.client import KubeClient
 from ..kube.controller import Reconciler, Request, Result
 from ..kube.events import EventRecorder, NullRecorder
-from ..kube.store import AlreadyExistsError, now_iso
+from ..kube.store import AlreadyExistsError
 from ..models import RayCronJob, RayJob
 from ..utils import constants as C
 from ..utils.cron import parse_cron

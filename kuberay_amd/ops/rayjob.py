@@ -14,7 +14,7 @@ import os
 import time
 from typing import Callable, Optional
 
-from ..common import association, job as joblib
+from ..common import job as joblib
 from ..kube import objects as k8s
 from ..kube.client import KubeClient
 from ..kube.controller import Reconciler, Request, Result
@@ -29,7 +29,7 @@ from ..models import (
     RayCluster,
     RayJob,
 )
-from ..models.raycluster import ClusterState, RayClusterConditionType
+from ..models.raycluster import ClusterState
 from ..utils import constants as C
 from ..utils import names
 from ..utils.dashboard_client import DashboardClientError

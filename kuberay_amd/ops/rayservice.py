@@ -12,7 +12,6 @@ in this build (see kuberay_amd.features.RAYSERVICE_INCREMENTAL_UPGRADE).
 """
 from __future__ import annotations
 
-import calendar
 import logging
 import threading
 import time

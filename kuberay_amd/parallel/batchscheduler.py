@@ -15,7 +15,6 @@ calls ``do_batch_scheduling_on_submission`` before creating pods and
 """
 from __future__ import annotations
 
-import math
 from typing import Dict, Optional
 
 from ..kube import objects as k8s

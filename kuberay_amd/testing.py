@@ -8,7 +8,7 @@ FakeRayDashboardClient (ray-operator/controllers/ray/suite_test.go:57-120).
 from __future__ import annotations
 
 import time
-from typing import Callable, Dict, List, Optional
+from typing import Callable, Dict, Optional
 
 from .kube.client import InMemoryClient
 from .kube.controller import Controller, Manager

@@ -2,9 +2,8 @@
 used by the RayCronJob reconciler; reference: raycronjob_controller.go)."""
 from __future__ import annotations
 
-import calendar
 import datetime as dt
-from typing import List, Optional, Set, Tuple
+from typing import Optional, Set
 
 _FIELDS = [
     ("minute", 0, 59),

@@ -10,7 +10,6 @@ ClientProvider seam of the reference (suite_test.go:57-69).
 """
 from __future__ import annotations
 
-import json
 from typing import Any, Dict, List, Optional
 
 import httpx

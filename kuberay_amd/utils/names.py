@@ -9,7 +9,6 @@ from __future__ import annotations
 
 import os
 import random
-import string
 
 from . import constants as C
 

@@ -8,10 +8,10 @@ reference's generic ``*gpu`` suffix matching plus the NVIDIA MIG regex
 from __future__ import annotations
 
 from decimal import Decimal
-from typing import Dict, List, Optional
+from typing import Dict, List
 
 from ..kube.objects import Pod
-from ..models.raycluster import RayCluster, RayClusterSpec, WorkerGroupSpec
+from ..models.raycluster import RayCluster, WorkerGroupSpec
 from . import constants as C
 from .quantity import format_quantity, parse_quantity
 
