@@ -66,6 +66,7 @@ class Controller:
         use_predicates: bool = True,
         watch_namespaces: Optional[List[str]] = None,
         metrics=None,
+        owned_event_coalesce_s: float = 0.0,
     ):
         self.name = name
         self.primary_kind = primary_kind
@@ -82,6 +83,10 @@ class Controller:
         self.reconcile_count = 0
         self.error_count = 0
         self.metrics = metrics
+        # optional delay on owned-object events to fold bursts into one
+        # reconcile; 0 (default) measured fastest — the dedup queue already
+        # coalesces while workers are busy
+        self.owned_event_coalesce_s = owned_event_coalesce_s
 
     # -- event routing -------------------------------------------------
     def observe(self, event_type: str, obj: Dict[str, Any]) -> None:
@@ -105,7 +110,11 @@ class Controller:
         elif kind in self.owned_kinds:
             for ref in meta.get("ownerReferences") or []:
                 if ref.get("kind") == self.primary_kind:
-                    self.queue.add((meta.get("namespace", "default"), ref.get("name", "")))
+                    key = (meta.get("namespace", "default"), ref.get("name", ""))
+                    if self.owned_event_coalesce_s > 0:
+                        self.queue.add_after(key, self.owned_event_coalesce_s)
+                    else:
+                        self.queue.add(key)
 
     # -- workers -------------------------------------------------------
     def _worker(self) -> None:
