@@ -40,13 +40,22 @@ logger = logging.getLogger("kuberay.rayjob")
 REQUEUE_SECONDS = 3
 
 
-def _parse_ts(ts: Optional[str]) -> Optional[float]:
-    if not ts:
+def _parse_ts(ts) -> Optional[float]:
+    """ISO-8601 (operator-written) or numeric epoch (Ray dashboard job info
+    reports start/end times as epoch milliseconds)."""
+    if ts in (None, "", 0):
         return None
+    if isinstance(ts, (int, float)):
+        value = float(ts)
+        return value / 1000.0 if value > 1e12 else value
     try:
         return calendar.timegm(time.strptime(ts, "%Y-%m-%dT%H:%M:%SZ"))
     except ValueError:
-        return None
+        try:
+            value = float(ts)
+            return value / 1000.0 if value > 1e12 else value
+        except ValueError:
+            return None
 
 
 class RayJobReconciler(Reconciler):

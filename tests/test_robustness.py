@@ -1,0 +1,143 @@
+"""Robustness tests: operator restart convergence (e2eupgrade analog),
+submitter idempotency, submitter-finished vs job-status races
+(SURVEY.md §7 hard parts (e) and (f))."""
+import os
+import time
+
+import pytest
+
+from kuberay_amd.common.job import build_job_submit_command
+from kuberay_amd.kube.client import InMemoryClient
+from kuberay_amd.kube.kubelet import SimKubelet
+from kuberay_amd.models import RayCluster, RayJob
+from kuberay_amd.testing import ControlPlane, simple_raycluster
+from kuberay_amd.utils import constants as C
+
+
+class TestOperatorRestart:
+    def test_new_operator_converges_on_existing_state(self):
+        """Kill the controllers mid-provisioning; a fresh operator instance
+        over the same apiserver state must converge the cluster to ready
+        (reference analog: test/e2eupgrade operator-version upgrade)."""
+        cp = ControlPlane(kubelet_delay=0.05, poll_seconds=0.05)
+        cp.start()
+        server = cp.server
+        cp.client.create(simple_raycluster("survivor", workers=3))
+        # let it half-start: wait until at least one pod exists, then kill
+        deadline = time.monotonic() + 10
+        while server.count("Pod") == 0 and time.monotonic() < deadline:
+            time.sleep(0.01)
+        cp.stop()  # old operator + kubelet die (pods stuck mid-lifecycle)
+
+        cp2 = ControlPlane(kubelet_delay=0.0, poll_seconds=0.05)
+        cp2.server = server
+        cp2.client = InMemoryClient(server)
+        # rebuild a control plane bound to the surviving server
+        from kuberay_amd.testing import ControlPlane as CP
+        cp2 = CP.__new__(CP)
+        cp2.__init__(kubelet_delay=0.0, poll_seconds=0.05)
+        # swap in the surviving state before start
+        cp2.server = server
+        cp2.client = InMemoryClient(server)
+        cp2.raycluster_reconciler.client = cp2.client
+        cp2.rayjob_reconciler.client = cp2.client
+        cp2.rayservice_reconciler.client = cp2.client
+        cp2.raycronjob_reconciler.client = cp2.client
+        cp2.manager.server = server
+        cp2.kubelet.server = server
+        cp2.start()
+        try:
+            ok = cp2.wait_cluster_state("default", "survivor", "ready",
+                                        timeout=20)
+            assert ok
+            rc = cp2.client.get(RayCluster, "default", "survivor")
+            assert rc.status.available_worker_replicas == 3
+        finally:
+            cp2.stop()
+
+
+class TestSubmitterCommand:
+    def _job(self, **spec):
+        base = {"entrypoint": "python t.py",
+                "rayClusterSpec": simple_raycluster("x").spec.to_dict()}
+        base.update(spec)
+        job = RayJob.from_dict({
+            "apiVersion": "ray.io/v1", "kind": "RayJob",
+            "metadata": {"name": "j1"}, "spec": base})
+        job.status.job_id = "j1-abcde"
+        job.status.dashboard_url = "head-svc.default.svc.cluster.local:8265"
+        return job
+
+    def test_k8s_mode_is_idempotent(self):
+        """job.go:119-130 — status-check before submit, logs --follow after,
+        so submitter pod retries never double-submit."""
+        cmd = " ".join(build_job_submit_command(self._job(), "K8sJobMode"))
+        assert "if ! ray job status" in cmd
+        assert "--no-wait" in cmd
+        assert cmd.index("ray job submit") > cmd.index("ray job status")
+        assert "ray job logs" in cmd and "--follow" in cmd
+        assert "--submission-id j1-abcde" in cmd
+
+    def test_gcs_health_wait_precedes_submit(self):
+        cmd = " ".join(build_job_submit_command(self._job(), "K8sJobMode"))
+        assert cmd.index("until") < cmd.index("ray job submit")
+        assert "api/gcs_healthz" in cmd
+
+    def test_sidecar_mode_uses_localhost(self):
+        cmd = " ".join(build_job_submit_command(self._job(), "SidecarMode"))
+        assert "http://127.0.0.1:8265" in cmd
+        assert "if ! ray job status" not in cmd  # restartPolicy=Never, no retry
+
+    def test_runtime_env_and_resources_flags(self):
+        job = self._job(runtimeEnvYAML="pip:\n- requests\n",
+                        entrypointNumCpus=2.0, entrypointNumGpus=1.0)
+        cmd = " ".join(build_job_submit_command(job, "K8sJobMode"))
+        assert "--runtime-env-json" in cmd
+        assert "--entrypoint-num-cpus" in cmd
+        assert "--entrypoint-num-gpus" in cmd
+
+
+class TestSubmitterJobStatusRace:
+    def test_job_succeeded_but_submitter_hangs_grace_period(self, control_plane,
+                                                            monkeypatch):
+        """rayjob_controller.go:334-356: JobStatus terminal but submitter K8s
+        Job never finishes → transition after the grace period."""
+        monkeypatch.setenv(
+            C.RAYJOB_DEPLOYMENT_STATUS_TRANSITION_GRACE_PERIOD_SECONDS, "1")
+        # stop the sim kubelet from completing Jobs (hang the submitter)
+        control_plane.kubelet.job_runtime = 10_000
+        control_plane.client.create(RayJob.from_dict({
+            "apiVersion": "ray.io/v1", "kind": "RayJob",
+            "metadata": {"name": "race1"},
+            "spec": {"entrypoint": "python t.py",
+                     "rayClusterSpec": simple_raycluster("x").spec.to_dict()}}))
+
+        def complete():
+            j = control_plane.client.try_get(RayJob, "default", "race1")
+            return j is not None and j.status.job_deployment_status == "Complete"
+        assert control_plane.wait_for(complete, timeout=30)
+
+    def test_submitter_failure_before_app_terminal_fails_job(self, control_plane):
+        """Submitter K8s Job fails while the app never reports terminal →
+        SubmissionFailed."""
+        control_plane.dashboard.get_job_info_mock = lambda jid: {
+            "submission_id": jid, "status": "PENDING"}
+        control_plane.client.create(RayJob.from_dict({
+            "apiVersion": "ray.io/v1", "kind": "RayJob",
+            "metadata": {"name": "race2"},
+            "spec": {"entrypoint": "python t.py",
+                     "rayClusterSpec": simple_raycluster("x").spec.to_dict()}}))
+        # wait for the submitter job to exist, then fail it
+        assert control_plane.wait_for(
+            lambda: control_plane.server.try_get("Job", "default", "race2"),
+            timeout=20)
+        control_plane.server.patch_merge("Job", "default", "race2", {
+            "status": {"failed": 1,
+                       "conditions": [{"type": "Failed", "status": "True"}]}},
+            subresource="status")
+
+        def failed():
+            j = control_plane.client.try_get(RayJob, "default", "race2")
+            return (j is not None and j.status.job_deployment_status == "Failed"
+                    and j.status.reason == "SubmissionFailed")
+        assert control_plane.wait_for(failed, timeout=20)
