@@ -191,7 +191,11 @@ class RayJobReconciler(Reconciler):
                 return Result(requeue_after=self.requeue_seconds)
         elif mode == Mode.SIDECAR:
             pass  # sidecar container injected into the head pod via cluster spec
-        # InteractiveMode: user submits; we just watch
+        if mode == Mode.INTERACTIVE:
+            # user submits out-of-band; wait for the submission-id annotation
+            # (rayjob_controller.go Waiting state)
+            self._set_status(rayjob, JDS.WAITING)
+            return Result(requeue_after=self.requeue_seconds)
 
         self._set_status(rayjob, JDS.RUNNING)
         return Result(requeue_after=self.requeue_seconds)

@@ -190,3 +190,18 @@ class TestClusterSelector:
         assert job.status.ray_cluster_name == "shared"
         # selected cluster must never be deleted by the job
         assert control_plane.server.try_get("RayCluster", "default", "shared")
+
+
+class TestInteractiveMode:
+    def test_waits_for_submission_id_annotation(self, control_plane):
+        job = make_rayjob(submissionMode="InteractiveMode", entrypoint=None)
+        control_plane.client.create(job)
+        assert wait_deployment_status(control_plane, "job1", "Waiting")
+        # user submits out-of-band and annotates with the submission id
+        j = job_of(control_plane)
+        control_plane.dashboard.set_job_status("user-sub-1", "_pinned")
+        control_plane.dashboard.jobs["user-sub-1"]["status"] = "SUCCEEDED"
+        j.metadata.annotations = {"ray.io/ray-job-submission-id": "user-sub-1"}
+        control_plane.client.update(j)
+        assert wait_deployment_status(control_plane, "job1", "Complete")
+        assert job_of(control_plane).status.job_id == "user-sub-1"
