@@ -1,0 +1,121 @@
+"""Deployable-path e2e: the operator running through RestApiServerAdapter
+against a real HTTP kube-apiserver surface (KubeApiFacade) — REST verbs,
+labelSelector lists, status subresource, chunked watch streams, informer
+pod-view cache. This is the '--backend kubernetes' stack end to end."""
+import time
+
+import pytest
+
+from kuberay_amd.kube.httpserver import KubeApiFacade
+from kuberay_amd.kube.kubelet import SimKubelet
+from kuberay_amd.kube.rest import RestApiServerAdapter, RestClient
+from kuberay_amd.kube.store import InMemoryApiServer, NotFoundError
+from kuberay_amd.models import RayCluster
+from kuberay_amd.testing import simple_raycluster
+
+
+@pytest.fixture()
+def facade():
+    f = KubeApiFacade().start()
+    yield f
+    f.stop()
+
+
+class TestRestClientAgainstFacade:
+    def test_crud_and_selectors(self, facade):
+        client = RestClient(base_url=facade.url)
+        a = simple_raycluster("a", namespace="ns1")
+        a.metadata.labels = {"team": "x"}
+        client.create(a)
+        client.create(simple_raycluster("b", namespace="ns1"))
+        assert client.get(RayCluster, "ns1", "a").metadata.uid
+        assert [o.metadata.name
+                for o in client.list(RayCluster, "ns1", {"team": "x"})] == ["a"]
+        got = client.get(RayCluster, "ns1", "a")
+        got.spec.worker_group_specs[0].replicas = 4
+        client.update(got)
+        assert facade.store.get("RayCluster", "ns1", "a")["spec"][
+            "workerGroupSpecs"][0]["replicas"] == 4
+        # status subresource does not clobber spec
+        got.status.state = "ready"
+        client.update_status(got)
+        fresh = facade.store.get("RayCluster", "ns1", "a")
+        assert fresh["status"]["state"] == "ready"
+        assert fresh["spec"]["workerGroupSpecs"][0]["replicas"] == 4
+        client.delete(RayCluster, "ns1", "a")
+        with pytest.raises(NotFoundError):
+            client.get(RayCluster, "ns1", "a")
+
+    def test_watch_stream(self, facade):
+        client = RestClient(base_url=facade.url)
+        events = []
+        import threading
+        done = threading.Event()
+
+        def consume():
+            for ev in client.raw_watch_stream("RayCluster"):
+                events.append(ev)
+                if len(events) >= 2:
+                    break
+            done.set()
+
+        t = threading.Thread(target=consume, daemon=True)
+        t.start()
+        time.sleep(0.2)
+        client.create(simple_raycluster("w1", namespace="ns1"))
+        client.delete(RayCluster, "ns1", "w1")
+        assert done.wait(5), "watch events not delivered"
+        types = [e[0] for e in events]
+        assert "ADDED" in types
+
+
+class TestOperatorOverRest:
+    def test_full_reconcile_through_http(self, facade):
+        """Manager + reconcilers on the REST adapter; kubelet sim runs on the
+        backing store (the 'real cluster' side)."""
+        from kuberay_amd.kube.controller import Controller, Manager
+        from kuberay_amd.ops.raycluster import (
+            RayClusterReconciler,
+            RayClusterReconcilerOptions,
+        )
+
+        adapter = RestApiServerAdapter(
+            rest_client=RestClient(base_url=facade.url))
+        client = adapter.client()
+        options = RayClusterReconcilerOptions()
+        options.requeue_after_seconds = 300
+        reconciler = RayClusterReconciler(client, options=options)
+        manager = Manager(adapter)
+        manager.add_controller(Controller(
+            "raycluster", "RayCluster", reconciler,
+            owned_kinds=["Pod", "Service", "Secret",
+                         "PersistentVolumeClaim", "Job"], workers=2))
+        kubelet = SimKubelet(facade.store, startup_delay=0.01)
+
+        manager.start()
+        kubelet.start()
+        try:
+            client.create(simple_raycluster("rest-e2e", workers=2))
+            deadline = time.monotonic() + 30
+            state = None
+            while time.monotonic() < deadline:
+                rc = client.try_get(RayCluster, "default", "rest-e2e")
+                state = rc.status.state if rc else None
+                if state == "ready":
+                    break
+                time.sleep(0.05)
+            assert state == "ready"
+            rc = client.try_get(RayCluster, "default", "rest-e2e")
+            assert rc.status.available_worker_replicas == 2
+            # informer-backed pod views flowed over HTTP watch
+            views = client.list_pod_views(
+                "default", {"ray.io/cluster": "rest-e2e"})
+            assert len(views) == 3
+            assert all(v.ready for v in views)
+            # head service was created over REST
+            assert facade.store.try_get("Service", "default",
+                                        "rest-e2e-head-svc") is not None
+        finally:
+            kubelet.stop()
+            adapter.stop()
+            manager.stop()
