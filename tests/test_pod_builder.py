@@ -261,3 +261,34 @@ class TestServices:
         assert svc.spec.cluster_ip == "None"
         assert svc.spec.publish_not_ready_addresses is True
         assert svc.metadata.name == "demo-headless"
+
+
+class TestEnvFlagToggles:
+    def test_deterministic_head_pod_name(self, monkeypatch):
+        monkeypatch.setenv(C.ENABLE_DETERMINISTIC_HEAD_POD_NAME, "true")
+        cluster = simple_raycluster("demo")
+        t = podlib.default_head_pod_template(cluster, cluster.spec.head_group_spec,
+                                             "demo-head-", "6379")
+        assert t.metadata.name == "demo-head-"
+        assert t.metadata.generate_name is None
+
+    def test_login_shell_flag(self, monkeypatch):
+        monkeypatch.setenv(C.ENABLE_LOGIN_SHELL, "true")
+        cluster = simple_raycluster("demo")
+        pod = build_worker_pod(cluster)
+        assert pod.spec.containers[0].command == ["/bin/bash", "-lc", "--"]
+
+    def test_probes_injection_disable(self, monkeypatch):
+        monkeypatch.setenv(C.ENABLE_PROBES_INJECTION, "false")
+        cluster = simple_raycluster("demo")
+        pod = build_worker_pod(cluster)
+        assert pod.spec.containers[0].readiness_probe is None
+        assert pod.spec.containers[0].liveness_probe is None
+
+    def test_init_container_injection_disable(self, monkeypatch):
+        monkeypatch.setenv(C.ENABLE_INIT_CONTAINER_INJECTION, "false")
+        cluster = simple_raycluster("demo")
+        group = cluster.spec.worker_group_specs[0]
+        t = podlib.default_worker_pod_template(
+            cluster, group, "w-", "x.default.svc.cluster.local", "6379")
+        assert not t.spec.init_containers

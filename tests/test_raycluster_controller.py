@@ -246,3 +246,20 @@ class TestAutoscalerRBAC:
         role = control_plane.server.get("Role", "default", "demo")
         resources = {r for rule in role["rules"] for r in rule["resources"]}
         assert "rayclusters" in resources and "pods" in resources
+
+
+class TestRandomPodDeleteFlag:
+    def test_enable_random_pod_delete_overrides_autoscaler_gate(
+            self, control_plane, monkeypatch):
+        monkeypatch.setenv(C.ENABLE_RANDOM_POD_DELETE, "true")
+        control_plane.client.create(
+            simple_raycluster("demo", workers=3, enableInTreeAutoscaling=True))
+        assert control_plane.wait_for(
+            lambda: get_cluster(control_plane).status.available_worker_replicas == 3)
+        rc = get_cluster(control_plane)
+        rc.spec.worker_group_specs[0].replicas = 1
+        control_plane.client.update(rc)
+        # with the flag on, the operator random-deletes down to 1 even
+        # though autoscaling is enabled (reference ENABLE_RANDOM_POD_DELETE)
+        assert control_plane.wait_for(
+            lambda: get_cluster(control_plane).status.available_worker_replicas == 1)
