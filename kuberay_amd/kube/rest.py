@@ -262,6 +262,10 @@ class _PodViewCache:
         out.sort(key=lambda v: (v.namespace, v.name))
         return out
 
+    def contains(self, namespace: str, name: str) -> bool:
+        with self._lock:
+            return (namespace, name) in self._views
+
 
 class RestApiServerAdapter:
     """Manager-facing surface over a real cluster: watch streams + list
@@ -272,6 +276,7 @@ class RestApiServerAdapter:
         self._client = rest_client or RestClient(kubeconfig=kubeconfig)
         self._view_cache = _PodViewCache()
         self._client.list_pod_views = self._view_cache.list  # type: ignore[attr-defined]
+        self._client.pod_cache_contains = self._view_cache.contains  # type: ignore[attr-defined]
         self._watchers: List["_AdapterWatcher"] = []
         self._threads: List[threading.Thread] = []
         self._stopped = threading.Event()

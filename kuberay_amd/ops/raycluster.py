@@ -129,6 +129,27 @@ def _ray_container_terminated(pod: k8s.Pod):
     return None
 
 
+class _TypedCacheShim:
+    """Expectations-cache adapter for typed clients (REST backend): contains()
+    via the informer pod-view cache when present, try_get via typed verbs."""
+
+    def __init__(self, client):
+        self.client = client
+
+    def contains(self, kind: str, namespace: str, name: str) -> bool:
+        if kind == "Pod":
+            cache_contains = getattr(self.client, "pod_cache_contains", None)
+            if cache_contains is not None:
+                return cache_contains(namespace, name)
+        return self.try_get(kind, namespace, name) is not None
+
+    def try_get(self, kind: str, namespace: str, name: str):
+        from ..kube.client import model_for_kind
+        model = model_for_kind(kind)
+        obj = self.client.try_get(model, namespace, name)
+        return obj.to_dict() if obj is not None else None
+
+
 class RayClusterReconcilerOptions:
     def __init__(self):
         self.requeue_after_seconds = int(os.environ.get(
@@ -451,8 +472,12 @@ class RayClusterReconciler(Reconciler):
             self._reconcile_worker_group(cluster, group, pods)
 
     def _cache(self):
-        # the in-memory client's server doubles as the informer cache
-        return getattr(self.client, "server", None) or self.client
+        # the in-memory client's server doubles as the informer cache; REST
+        # clients get a shim over the informer pod-view cache / typed verbs
+        server = getattr(self.client, "server", None)
+        if server is not None:
+            return server
+        return _TypedCacheShim(self.client)
 
     def _reconcile_worker_group(self, cluster: RayCluster, group: WorkerGroupSpec,
                                 all_pods) -> None:

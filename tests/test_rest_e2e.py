@@ -115,6 +115,8 @@ class TestOperatorOverRest:
             # head service was created over REST
             assert facade.store.try_get("Service", "default",
                                         "rest-e2e-head-svc") is not None
+            # the whole path must run clean — no reconcile errors retried away
+            assert manager.controllers[0].error_count == 0
         finally:
             kubelet.stop()
             adapter.stop()
