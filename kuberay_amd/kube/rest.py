@@ -49,6 +49,8 @@ RESOURCES: Dict[str, Tuple[str, str]] = {
     "NetworkPolicy": ("/apis/networking.k8s.io/v1", "networkpolicies"),
     "Ingress": ("/apis/networking.k8s.io/v1", "ingresses"),
     "EndpointSlice": ("/apis/discovery.k8s.io/v1", "endpointslices"),
+    "Gateway": ("/apis/gateway.networking.k8s.io/v1", "gateways"),
+    "HTTPRoute": ("/apis/gateway.networking.k8s.io/v1", "httproutes"),
     "RayCluster": ("/apis/ray.io/v1", "rayclusters"),
     "RayJob": ("/apis/ray.io/v1", "rayjobs"),
     "RayService": ("/apis/ray.io/v1", "rayservices"),
@@ -210,7 +212,30 @@ class RestClient(KubeClient):
         kind = _kind_of(model_or_obj)
         self._check(self._http.delete(self._path(kind, namespace, name)))
 
-    # -- raw dict surface (adapter/informer use) -----------------------
+    # -- raw dict surface (adapter/informer/CRD-less kinds) -------------
+    def raw_create(self, obj: Dict[str, Any]) -> Dict[str, Any]:
+        kind = obj.get("kind", "")
+        ns = obj.get("metadata", {}).get("namespace", "default")
+        resp = self._check(self._http.post(self._path(kind, ns), json=obj))
+        return resp.json()
+
+    def raw_patch(self, kind: str, namespace: str, name: str,
+                  patch: Dict[str, Any]) -> Dict[str, Any]:
+        resp = self._check(self._http.patch(
+            self._path(kind, namespace, name), json=patch,
+            headers={"Content-Type": "application/merge-patch+json"}))
+        return resp.json()
+
+    def raw_delete(self, kind: str, namespace: str, name: str) -> None:
+        self._check(self._http.delete(self._path(kind, namespace, name)))
+
+    def raw_try_get(self, kind: str, namespace: str, name: str) -> Optional[Dict[str, Any]]:
+        try:
+            return self._check(self._http.get(
+                self._path(kind, namespace, name))).json()
+        except NotFoundError:
+            return None
+
     def raw_list(self, kind: str, namespace: Optional[str] = None) -> List[Dict[str, Any]]:
         resp = self._check(self._http.get(self._path(kind, namespace)))
         items = resp.json().get("items", [])

@@ -50,6 +50,43 @@ def upgrade_options(svc: RayService):
     return us.cluster_upgrade_options if us else None
 
 
+class _RawObjects:
+    """Dict-object verbs over either backend: the in-memory server directly,
+    or a RestClient's raw_* methods (Gateway/HTTPRoute have no typed model)."""
+
+    def __init__(self, client):
+        self.server = getattr(client, "server", None)
+        self.client = client
+
+    def try_get(self, kind, namespace, name):
+        if self.server is not None:
+            return self.server.try_get(kind, namespace, name)
+        fn = getattr(self.client, "raw_try_get", None)
+        return fn(kind, namespace, name) if fn else None
+
+    def create(self, obj):
+        if self.server is not None:
+            return self.server.create(obj)
+        fn = getattr(self.client, "raw_create", None)
+        if fn is not None:
+            return fn(obj)
+        return None
+
+    def patch(self, kind, namespace, name, patch):
+        if self.server is not None:
+            return self.server.patch_merge(kind, namespace, name, patch)
+        fn = getattr(self.client, "raw_patch", None)
+        return fn(kind, namespace, name, patch) if fn else None
+
+    def delete(self, kind, namespace, name):
+        if self.server is not None:
+            self.server.delete(kind, namespace, name)
+            return
+        fn = getattr(self.client, "raw_delete", None)
+        if fn is not None:
+            fn(kind, namespace, name)
+
+
 class IncrementalUpgrader:
     """Composable helper driven by RayServiceReconciler."""
 
@@ -57,6 +94,7 @@ class IncrementalUpgrader:
         self.client = client
         self.recorder = recorder
         self.clock = clock
+        self.raw = _RawObjects(client)
 
     # ------------------------------------------------------------------
     def ensure_gateway_infra(self, svc: RayService, active: RayCluster,
@@ -65,7 +103,6 @@ class IncrementalUpgrader:
         from ..common import service as servicelib
         namespace = svc.metadata.namespace or "default"
         opts = upgrade_options(svc)
-        server = getattr(self.client, "server", None)
 
         for cluster in (active, pending):
             serve = servicelib.build_serve_service(cluster, cluster,
@@ -78,8 +115,6 @@ class IncrementalUpgrader:
                 except AlreadyExistsError:
                     pass
 
-        if server is None:
-            return
         gw = {
             "apiVersion": "gateway.networking.k8s.io/v1",
             "kind": "Gateway",
@@ -91,15 +126,15 @@ class IncrementalUpgrader:
                                "allowedRoutes": {"namespaces": {"from": "Same"}}}],
             },
         }
-        if server.try_get("Gateway", namespace, gateway_name(svc)) is None:
+        if self.raw.try_get("Gateway", namespace, gateway_name(svc)) is None:
             try:
-                server.create(gw)
+                self.raw.create(gw)
             except AlreadyExistsError:
                 pass
-        if server.try_get("HTTPRoute", namespace, route_name(svc)) is None:
+        if self.raw.try_get("HTTPRoute", namespace, route_name(svc)) is None:
             try:
-                server.create(self._route(svc, active.metadata.name,
-                                          pending.metadata.name, 0))
+                self.raw.create(self._route(svc, active.metadata.name,
+                                            pending.metadata.name, 0))
             except AlreadyExistsError:
                 pass
 
@@ -161,17 +196,15 @@ class IncrementalUpgrader:
         svc.status.active_service_status.target_capacity = 100 - new_weight \
             if new_weight == 100 else 100
 
-        server = getattr(self.client, "server", None)
-        if server is not None:
-            try:
-                server.patch_merge(
-                    "HTTPRoute", svc.metadata.namespace or "default",
-                    route_name(svc),
-                    {"spec": self._route(svc, active.metadata.name,
-                                         pending.metadata.name,
-                                         new_weight)["spec"]})
-            except NotFoundError:
-                pass
+        try:
+            self.raw.patch(
+                "HTTPRoute", svc.metadata.namespace or "default",
+                route_name(svc),
+                {"spec": self._route(svc, active.metadata.name,
+                                     pending.metadata.name,
+                                     new_weight)["spec"]})
+        except NotFoundError:
+            pass
         self.recorder.eventf(svc, "Normal", "TrafficMigrated",
                              "Routed %d%% of traffic to RayCluster %s",
                              new_weight, pending.metadata.name)
@@ -187,11 +220,10 @@ class IncrementalUpgrader:
                 self.client.delete(pending)
             except NotFoundError:
                 pass
-        server = getattr(self.client, "server", None)
         active_name = svc.status.active_service_status.ray_cluster_name
-        if server is not None and active_name:
+        if active_name:
             try:
-                server.patch_merge(
+                self.raw.patch(
                     "HTTPRoute", namespace, route_name(svc),
                     {"spec": self._route(svc, active_name, active_name, 0)["spec"]})
             except NotFoundError:
@@ -212,13 +244,10 @@ class IncrementalUpgrader:
     def cleanup(self, svc: RayService) -> None:
         """Remove Gateway/HTTPRoute after promotion completes."""
         namespace = svc.metadata.namespace or "default"
-        server = getattr(self.client, "server", None)
-        if server is None:
-            return
         for kind, name in (("HTTPRoute", route_name(svc)),
                            ("Gateway", gateway_name(svc))):
             try:
-                server.delete(kind, namespace, name)
+                self.raw.delete(kind, namespace, name)
             except NotFoundError:
                 pass
 
