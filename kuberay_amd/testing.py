@@ -27,14 +27,17 @@ class ControlPlane:
                  gpu_gate: Optional[Callable[[dict], bool]] = None,
                  dashboard_client=None, enable_kubelet: bool = True,
                  requeue_seconds: Optional[int] = 300,
-                 poll_seconds: Optional[float] = None):
+                 poll_seconds: Optional[float] = None,
+                 server: Optional[InMemoryApiServer] = None):
         from .ops.raycluster import RayClusterReconciler, RayClusterReconcilerOptions
         from .ops.rayjob import RayJobReconciler
         from .ops.rayservice import RayServiceReconciler
         from .ops.raycronjob import RayCronJobReconciler
         from .utils.fake_dashboard import FakeRayDashboardClient
 
-        self.server = InMemoryApiServer()
+        # an existing server can be adopted (operator-restart tests resume
+        # over surviving apiserver state)
+        self.server = server or InMemoryApiServer()
         self.client = InMemoryClient(self.server)
         self.recorder = StoreRecorder(self.server) if record_events else NullRecorder()
         self.dashboard = dashboard_client or FakeRayDashboardClient()

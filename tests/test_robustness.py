@@ -29,22 +29,8 @@ class TestOperatorRestart:
             time.sleep(0.01)
         cp.stop()  # old operator + kubelet die (pods stuck mid-lifecycle)
 
-        cp2 = ControlPlane(kubelet_delay=0.0, poll_seconds=0.05)
-        cp2.server = server
-        cp2.client = InMemoryClient(server)
-        # rebuild a control plane bound to the surviving server
-        from kuberay_amd.testing import ControlPlane as CP
-        cp2 = CP.__new__(CP)
-        cp2.__init__(kubelet_delay=0.0, poll_seconds=0.05)
-        # swap in the surviving state before start
-        cp2.server = server
-        cp2.client = InMemoryClient(server)
-        cp2.raycluster_reconciler.client = cp2.client
-        cp2.rayjob_reconciler.client = cp2.client
-        cp2.rayservice_reconciler.client = cp2.client
-        cp2.raycronjob_reconciler.client = cp2.client
-        cp2.manager.server = server
-        cp2.kubelet.server = server
+        # a fresh operator instance adopts the surviving apiserver state
+        cp2 = ControlPlane(kubelet_delay=0.0, poll_seconds=0.05, server=server)
         cp2.start()
         try:
             ok = cp2.wait_cluster_state("default", "survivor", "ready",
