@@ -36,6 +36,7 @@ logger = logging.getLogger("kuberay.rest")
 
 # kind -> (api prefix, plural, namespaced)
 RESOURCES: Dict[str, Tuple[str, str]] = {
+    "Node": ("/api/v1", "nodes"),
     "Pod": ("/api/v1", "pods"),
     "Service": ("/api/v1", "services"),
     "Secret": ("/api/v1", "secrets"),
@@ -44,6 +45,7 @@ RESOURCES: Dict[str, Tuple[str, str]] = {
     "ServiceAccount": ("/api/v1", "serviceaccounts"),
     "Event": ("/api/v1", "events"),
     "Job": ("/apis/batch/v1", "jobs"),
+    "Lease": ("/apis/coordination.k8s.io/v1", "leases"),
     "Role": ("/apis/rbac.authorization.k8s.io/v1", "roles"),
     "RoleBinding": ("/apis/rbac.authorization.k8s.io/v1", "rolebindings"),
     "NetworkPolicy": ("/apis/networking.k8s.io/v1", "networkpolicies"),

@@ -13,6 +13,7 @@ Backends:
 from __future__ import annotations
 
 import logging
+import os
 import signal
 import sys
 import threading
@@ -185,6 +186,20 @@ def main(argv=None) -> int:
         kubelet = SimKubelet(manager.server, gpu_gate=gate)
         kubelet.start()
 
+    elector = None
+    if cfg.enable_leader_election and cfg.backend == "kubernetes":
+        from .kube.leaderelection import LeaderElector
+        import socket
+        started = threading.Event()
+        elector = LeaderElector(
+            client,
+            namespace=cfg.leader_election_namespace or "ray-system",
+            identity=f"{socket.gethostname()}-{os.getpid()}",
+            on_started_leading=started.set)
+        elector.start()
+        logger.info("waiting for leader election (identity=%s)",
+                    elector.identity)
+        started.wait()
     manager.start()
     if autoscaler is not None:
         autoscaler.start(cfg.mi355x_autoscaler_interval_s)
@@ -204,6 +219,8 @@ def main(argv=None) -> int:
             kubelet.stop()
         if snapshotter is not None:
             snapshotter.stop()
+        if elector is not None:
+            elector.stop()
         manager.stop()
         health.stop()
     return 0

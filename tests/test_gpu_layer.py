@@ -274,3 +274,28 @@ class TestAutoscalerOnDevice:
         m.observe_gpu_stats(get_gpu_stats())
         expo = m.exposition().decode()
         assert 'kuberay_mi355x_hbm_used_fraction{gpu="0"}' in expo
+
+
+class TestNodeLabeller:
+    def test_compute_labels_full_island(self):
+        from kuberay_amd.gpu.labeller import compute_node_labels
+        topo = topology.parse_showtopo_text(TOPO_8GPU)
+        labels = compute_node_labels(topo, "mi355x-node-1")
+        assert labels["amd.com/xgmi-island"] == "mi355x-node-1-island0"
+        assert labels["amd.com/gpu.count"] == "8"
+        assert labels["amd.com/xgmi-fully-connected"] == "true"
+
+    def test_label_node_in_memory(self):
+        from kuberay_amd.gpu.labeller import label_node
+        client = InMemoryClient()
+        client.server.create({"kind": "Node",
+                              "metadata": {"name": "n1", "namespace": "default"}})
+        topo = topology.parse_showtopo_text(TOPO_8GPU)
+        labels = label_node(client, "n1", topo)
+        assert labels
+        node = client.server.get("Node", "default", "n1")
+        assert node["metadata"]["labels"]["amd.com/xgmi-island"] == "n1-island0"
+
+    def test_no_labels_without_gpus(self):
+        from kuberay_amd.gpu.labeller import compute_node_labels
+        assert compute_node_labels(None, "n1") == {}

@@ -269,3 +269,46 @@ class TestSnapshot:
         time.sleep(0.15)
         loop.stop()
         assert os.path.exists(tmp_path / "s.jsonl")
+
+
+class TestLeaderElection:
+    def test_single_candidate_acquires(self):
+        from kuberay_amd.kube.client import InMemoryClient
+        from kuberay_amd.kube.leaderelection import LeaderElector
+        client = InMemoryClient()
+        started = []
+        e = LeaderElector(client, identity="a",
+                          on_started_leading=lambda: started.append("a"))
+        assert e.try_acquire_or_renew() is True
+        assert started == ["a"]
+        # renewal keeps leadership
+        assert e.try_acquire_or_renew() is True
+
+    def test_second_candidate_waits_then_takes_over(self):
+        from kuberay_amd.kube.client import InMemoryClient
+        from kuberay_amd.kube.leaderelection import LeaderElector
+        client = InMemoryClient()
+        a = LeaderElector(client, identity="a", lease_duration_s=0.2)
+        b = LeaderElector(client, identity="b", lease_duration_s=0.2)
+        assert a.try_acquire_or_renew() is True
+        assert b.try_acquire_or_renew() is False
+        time.sleep(1.1)  # let the lease expire (timestamps have 1s granularity)
+        assert b.try_acquire_or_renew() is True
+        # a notices it lost on its next round (lease now held by b)
+        assert a.try_acquire_or_renew() is False
+        assert a.is_leader is False
+
+    def test_graceful_release_on_stop(self):
+        from kuberay_amd.kube.client import InMemoryClient
+        from kuberay_amd.kube.leaderelection import LeaderElector
+        client = InMemoryClient()
+        stopped = []
+        a = LeaderElector(client, identity="a", renew_period_s=0.05,
+                          on_stopped_leading=lambda: stopped.append(1))
+        a.start()
+        time.sleep(0.2)
+        assert a.is_leader
+        a.stop()
+        assert stopped == [1]
+        b = LeaderElector(client, identity="b")
+        assert b.try_acquire_or_renew() is True  # released lease is free
