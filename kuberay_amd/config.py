@@ -45,6 +45,9 @@ class Configuration(K8sModel):
     # memory backend durability: JSONL snapshot path ("" = off)
     state_file: str = ""
     state_snapshot_interval_s: float = 30.0
+    # memory backend: serve the kube-API facade on this port (0 = off) so
+    # kubectl-style tooling / kray --server can target the control plane
+    api_port: int = 0
 
 
 def load_config(argv: Optional[List[str]] = None) -> Configuration:
@@ -61,6 +64,7 @@ def load_config(argv: Optional[List[str]] = None) -> Configuration:
     parser.add_argument("--kubeconfig")
     parser.add_argument("--log-file")
     parser.add_argument("--state-file")
+    parser.add_argument("--api-port", type=int, default=None)
     parser.add_argument("--no-metrics", action="store_true", default=None)
     args = parser.parse_args(argv)
 
@@ -79,6 +83,7 @@ def load_config(argv: Optional[List[str]] = None) -> Configuration:
         ("feature_gates", "feature_gates"),
         ("backend", "backend"), ("kubeconfig", "kubeconfig"),
         ("log_file", "log_file"), ("state_file", "state_file"),
+        ("api_port", "api_port"),
     ]:
         val = getattr(args, flag, None)
         if val is not None:

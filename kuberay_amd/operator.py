@@ -167,7 +167,12 @@ def main(argv=None) -> int:
 
     kubelet = None
     snapshotter = None
+    facade = None
     if cfg.backend == "memory":
+        if cfg.api_port:
+            from .kube.httpserver import KubeApiFacade
+            facade = KubeApiFacade(manager.server, port=cfg.api_port).start()
+            logger.info("kube-API facade serving at %s", facade.url)
         if cfg.state_file:
             from .kube.snapshot import SnapshotLoop, load_snapshot
             restored = load_snapshot(manager.server, cfg.state_file)
@@ -219,6 +224,8 @@ def main(argv=None) -> int:
             kubelet.stop()
         if snapshotter is not None:
             snapshotter.stop()
+        if facade is not None:
+            facade.stop()
         if elector is not None:
             elector.stop()
         manager.stop()
