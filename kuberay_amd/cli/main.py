@@ -318,6 +318,22 @@ def session(ctx, cluster_name):
 
 
 @cli.command()
+@click.argument("kind", type=click.Choice(["cluster", "job", "service"]))
+@click.argument("name")
+@click.option("--resume/--suspend", "resume", default=False)
+@click.pass_context
+def suspend(ctx, kind, name, resume):
+    """Suspend (or --resume) a Ray resource."""
+    client = client_of(ctx)
+    ns = ctx.obj["namespace"]
+    model = {"cluster": RayCluster, "job": RayJob, "service": RayService}[kind]
+    obj = client.get(model, ns, name)
+    obj.spec.suspend = not resume
+    client.update(obj)
+    click.echo(f"{kind} {name} {'resumed' if resume else 'suspended'}")
+
+
+@cli.command()
 @click.argument("cluster_name")
 @click.pass_context
 def log(ctx, cluster_name):

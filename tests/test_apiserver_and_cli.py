@@ -267,3 +267,16 @@ class TestGrpcApi:
         with pytest.raises(grpc.RpcError):
             self._call(channel, "ClusterService", "GetCluster",
                        GetRequest(name="gc1", namespace="ns1"), Cluster)
+
+
+class TestKraySuspend:
+    def test_suspend_and_resume(self, kray):
+        runner, backing = kray
+        runner.invoke(cli, ["-n", "ns1", "create", "cluster", "k1"])
+        r = runner.invoke(cli, ["-n", "ns1", "suspend", "cluster", "k1"])
+        assert r.exit_code == 0, r.output
+        assert backing.get(RayCluster, "ns1", "k1").spec.suspend is True
+        r = runner.invoke(cli, ["-n", "ns1", "suspend", "cluster", "k1",
+                                "--resume"])
+        assert r.exit_code == 0
+        assert backing.get(RayCluster, "ns1", "k1").spec.suspend is False
