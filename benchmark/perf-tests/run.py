@@ -24,10 +24,11 @@ SCALES = {
     "500": {"clusters": 500, "namespaces": 10},
     "1000": {"clusters": 1000, "namespaces": 20},
     "5000": {"clusters": 5000, "namespaces": 50},
+    "10000": {"clusters": 10000, "namespaces": 100},
 }
 
 REFERENCE = {  # junit totals from the reference repo (GKE, v1.1.1)
-    "100": 135.2, "1000": 647.5, "5000": 2700.1,
+    "100": 135.2, "1000": 647.5, "5000": 2700.1, "10000": 2720.1,
 }
 
 
