@@ -78,9 +78,15 @@ def build_manager(cfg: Configuration, server=None, client=None):
         RayJobReconciler(client, recorder=recorder, metrics=metrics),
         owned_kinds=["RayCluster", "Job"], workers=workers,
         watch_namespaces=ns_scope, metrics=metrics))
+    # real clusters: serve-proxy healthz drives the head pod's serve label
+    http_proxy = None
+    if cfg.backend == "kubernetes":
+        from .utils.dashboard_client import RayHttpProxyClient
+        http_proxy = RayHttpProxyClient()
     manager.add_controller(Controller(
         "rayservice", "RayService",
-        RayServiceReconciler(client, recorder=recorder, metrics=metrics),
+        RayServiceReconciler(client, recorder=recorder, metrics=metrics,
+                             http_proxy_client=http_proxy),
         owned_kinds=["RayCluster", "Service"], workers=workers,
         watch_namespaces=ns_scope, metrics=metrics))
     if features.enabled("RayCronJob"):

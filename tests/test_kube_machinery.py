@@ -312,3 +312,29 @@ class TestLeaderElection:
         assert stopped == [1]
         b = LeaderElector(client, identity="b")
         assert b.try_acquire_or_renew() is True  # released lease is free
+
+
+class TestEventRecorder:
+    def test_events_aggregate_by_reason(self):
+        from kuberay_amd.kube.events import StoreRecorder
+        from kuberay_amd.testing import simple_raycluster
+        server = InMemoryApiServer()
+        rec = StoreRecorder(server)
+        obj = simple_raycluster("evt")
+        obj.metadata.uid = "u1"
+        for _ in range(3):
+            rec.eventf(obj, "Normal", "CreatedWorkerPod", "Created worker Pod")
+        events = server.list("Event")
+        assert len(events) == 1
+        assert events[0]["count"] == 3
+        assert events[0]["involvedObject"]["name"] == "evt"
+
+    def test_distinct_messages_create_distinct_events(self):
+        from kuberay_amd.kube.events import StoreRecorder
+        from kuberay_amd.testing import simple_raycluster
+        server = InMemoryApiServer()
+        rec = StoreRecorder(server)
+        obj = simple_raycluster("evt")
+        rec.eventf(obj, "Normal", "CreatedWorkerPod", "Created worker Pod %s", "a")
+        rec.eventf(obj, "Normal", "CreatedWorkerPod", "Created worker Pod %s", "b")
+        assert len(server.list("Event")) == 2
