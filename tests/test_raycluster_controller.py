@@ -263,3 +263,58 @@ class TestRandomPodDeleteFlag:
         # though autoscaling is enabled (reference ENABLE_RANDOM_POD_DELETE)
         assert control_plane.wait_for(
             lambda: get_cluster(control_plane).status.available_worker_replicas == 1)
+
+
+class TestWorkerGroupAddition:
+    def test_new_group_added_to_live_cluster(self, control_plane):
+        control_plane.client.create(simple_raycluster("demo", workers=1))
+        assert control_plane.wait_cluster_state("default", "demo", "ready")
+        rc = get_cluster(control_plane)
+        from kuberay_amd.models.raycluster import WorkerGroupSpec
+        new_group = {
+            "groupName": "gpu-group", "replicas": 2, "minReplicas": 0,
+            "maxReplicas": 4, "rayStartParams": {},
+            "template": {"spec": {"containers": [{
+                "name": "ray-worker", "image": "rayproject/ray:2.46.0-rocm",
+                "resources": {"limits": {"cpu": "1", "memory": "1Gi",
+                                         "amd.com/gpu": "1"}}}]}},
+        }
+        rc.spec.worker_group_specs.append(WorkerGroupSpec.from_dict(new_group))
+        control_plane.client.update(rc)
+        assert control_plane.wait_for(
+            lambda: get_cluster(control_plane).status.available_worker_replicas == 3)
+        group_pods = [p for p in pods_of(control_plane)
+                      if (p["metadata"]["labels"] or {}).get(
+                          C.RAY_NODE_GROUP_LABEL_KEY) == "gpu-group"]
+        assert len(group_pods) == 2
+        assert get_cluster(control_plane).status.desired_gpu == "2"
+
+
+class TestGcsPvcRetain:
+    def test_retain_policy_pvc_survives_cluster_deletion(self, control_plane):
+        cluster = simple_raycluster("demo", gcsFaultToleranceOptions={
+            "backend": "embedded",
+            "storage": {"size": "1Gi", "deletionPolicy": "Retain"}})
+        control_plane.client.create(cluster)
+        assert control_plane.wait_for(
+            lambda: control_plane.server.try_get(
+                "PersistentVolumeClaim", "default", "demo-gcs-pvc"))
+        control_plane.server.delete("RayCluster", "default", "demo")
+        assert control_plane.wait_for(
+            lambda: control_plane.server.try_get("RayCluster", "default", "demo") is None)
+        # Retain: the PVC is NOT owner-referenced, so it survives GC
+        assert control_plane.server.try_get(
+            "PersistentVolumeClaim", "default", "demo-gcs-pvc") is not None
+
+    def test_delete_policy_pvc_garbage_collected(self, control_plane):
+        cluster = simple_raycluster("demo", gcsFaultToleranceOptions={
+            "backend": "embedded",
+            "storage": {"size": "1Gi", "deletionPolicy": "Delete"}})
+        control_plane.client.create(cluster)
+        assert control_plane.wait_for(
+            lambda: control_plane.server.try_get(
+                "PersistentVolumeClaim", "default", "demo-gcs-pvc"))
+        control_plane.server.delete("RayCluster", "default", "demo")
+        assert control_plane.wait_for(
+            lambda: control_plane.server.try_get(
+                "PersistentVolumeClaim", "default", "demo-gcs-pvc") is None)
