@@ -281,6 +281,18 @@ def create_app(client: Optional[KubeClient] = None,
         client.delete(model_for_kind(kind), ns, name)
         return {"status": "Success"}
 
+    # ------------------------------------------------------------------
+    # v1: serve applications proxy (RayServeService analog)
+    # ------------------------------------------------------------------
+    @app.get("/apis/v1/namespaces/{ns}/serveapplications/{cluster}")
+    def get_serve_applications(ns: str, cluster: str):
+        return _dashboard_for_cluster(ns, cluster).get_serve_applications()
+
+    @app.put("/apis/v1/namespaces/{ns}/serveapplications/{cluster}")
+    def update_serve_applications(ns: str, cluster: str, body: Dict[str, Any]):
+        _dashboard_for_cluster(ns, cluster).update_serve_applications(body)
+        return {}
+
     @app.get("/healthz")
     def healthz():
         return {"status": "ok"}

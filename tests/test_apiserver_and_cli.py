@@ -280,3 +280,16 @@ class TestKraySuspend:
                                 "--resume"])
         assert r.exit_code == 0
         assert backing.get(RayCluster, "ns1", "k1").spec.suspend is False
+
+
+class TestServeApplicationsProxy:
+    def test_put_and_get(self, api):
+        t, _, fake = api
+        t.post("/apis/v1/namespaces/ns1/clusters", json=CLUSTER_BODY)
+        cfg = {"applications": [{"name": "app1", "import_path": "m.g"}]}
+        r = t.put("/apis/v1/namespaces/ns1/serveapplications/c1", json=cfg)
+        assert r.status_code == 200
+        assert fake.serve_config == cfg
+        r = t.get("/apis/v1/namespaces/ns1/serveapplications/c1")
+        assert r.status_code == 200
+        assert "applications" in r.json()

@@ -205,3 +205,38 @@ class TestInteractiveMode:
         control_plane.client.update(j)
         assert wait_deployment_status(control_plane, "job1", "Complete")
         assert job_of(control_plane).status.job_id == "user-sub-1"
+
+
+class TestCronEndToEnd:
+    def test_cron_fires_job_through_full_stack(self, control_plane):
+        import datetime as dt
+        from kuberay_amd.models import RayCronJob
+        # controllable clock anchored at real now (the CR's creationTimestamp
+        # is real wall time and is the first schedule basis)
+        real_now = dt.datetime.utcnow()
+        fake_now = {"t": real_now}
+        control_plane.raycronjob_reconciler.now_fn = lambda: fake_now["t"]
+        control_plane.client.create(RayCronJob.from_dict({
+            "apiVersion": "ray.io/v1", "kind": "RayCronJob",
+            "metadata": {"name": "nightly", "namespace": "default"},
+            "spec": {"schedule": "* * * * *",
+                     "jobTemplate": make_rayjob().spec.to_dict()}}))
+        import time as _t
+        _t.sleep(0.3)
+        assert control_plane.server.count("RayJob") == 0  # not due yet
+        fake_now["t"] = real_now + dt.timedelta(minutes=2)
+        # force a reconcile via annotation touch
+        control_plane.client.patch(RayCronJob, "default", "nightly",
+                                   {"metadata": {"annotations": {"touch": "1"}}})
+        def fired():
+            jobs = control_plane.server.list("RayJob")
+            return len(jobs) == 1 and jobs[0]["metadata"]["labels"][
+                "ray.io/cronjob-name"] == "nightly"
+        assert control_plane.wait_for(fired, timeout=10)
+        # the spawned RayJob runs to completion through the whole stack
+        job_name = control_plane.server.list("RayJob")[0]["metadata"]["name"]
+        assert wait_deployment_status(control_plane, job_name, "Complete",
+                                      timeout=30)
+        from kuberay_amd.models import RayCronJob as RCJ
+        cron = control_plane.client.get(RCJ, "default", "nightly")
+        assert cron.status.last_schedule_time  # recorded the fire time
