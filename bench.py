@@ -103,7 +103,8 @@ def main() -> int:
     parser.add_argument("--clusters", type=int, default=500)
     parser.add_argument("--workers-per-cluster", type=int, default=3)
     parser.add_argument("--gpus-per-worker", type=int, default=1)
-    parser.add_argument("--controller-workers", type=int, default=8)
+    parser.add_argument("--controller-workers", type=int, default=0,
+                        help="0 = auto: min(8, cores/world), >=2")
     args = parser.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -126,8 +127,13 @@ def main() -> int:
         backend = "nccl" if use_cuda else "gloo"
         dist.init_process_group(backend=backend)
 
+    workers = args.controller_workers
+    if workers <= 0:
+        cores = os.cpu_count() or 8
+        workers = max(2, min(8, cores // max(world, 1)))
+
     from kuberay_amd.testing import ControlPlane
-    cp = ControlPlane(kubelet_delay=0.0, workers=args.controller_workers,
+    cp = ControlPlane(kubelet_delay=0.0, workers=workers,
                       record_events=False, requeue_seconds=3600,
                       poll_seconds=1.0)
     cp.start()
