@@ -200,6 +200,7 @@ class RayClusterReconciler(Reconciler):
         self._ensure_finalizers(cluster)
 
         reconcile_err: Optional[Exception] = None
+        pods_complete = True
         try:
             if podlib.is_autoscaling_enabled(cluster.spec):
                 self._reconcile_autoscaler_rbac(cluster)
@@ -214,13 +215,22 @@ class RayClusterReconciler(Reconciler):
                     C.ENABLE_SERVE_SERVICE_KEY) == C.ENABLE_SERVE_SERVICE_TRUE:
                 self._reconcile_serve_service(cluster)
             self._reconcile_gcs_storage_pvc(cluster)
-            self._reconcile_pods(cluster)
+            pods_complete = self._reconcile_pods(cluster)
         except Exception as e:  # surface via ReplicaFailure condition + requeue
             reconcile_err = e
 
         self._calculate_and_update_status(cluster, reconcile_err)
         if reconcile_err is not None:
             raise reconcile_err
+        # suspend state machine advances through status-only writes, which
+        # the event predicates deliberately ignore — requeue hot while the
+        # machine is mid-flight (reference: 2s inconsistency requeue,
+        # raycluster_controller.go:400-419)
+        suspended_now = condition_true(cluster.status.conditions, Cond.SUSPENDED)
+        suspending_now = condition_true(cluster.status.conditions, Cond.SUSPENDING)
+        if (bool(cluster.spec.suspend) != suspended_now) or suspending_now \
+                or not pods_complete:
+            return Result(requeue_after=2)
         return Result(requeue_after=self.options.requeue_after_seconds)
 
     # ------------------------------------------------------------------
@@ -415,7 +425,10 @@ class RayClusterReconciler(Reconciler):
     def _active(self, pods):
         return [p for p in pods if not _deletion_ts(p)]
 
-    def _reconcile_pods(self, cluster: RayCluster) -> None:
+    def _reconcile_pods(self, cluster: RayCluster) -> bool:
+        """Returns False when work was deferred (expectations unsatisfied) —
+        the caller requeues hot, mirroring the reference's 2s inconsistency
+        requeue."""
         namespace = cluster.metadata.namespace or "default"
         name = cluster.metadata.name
 
@@ -425,9 +438,9 @@ class RayClusterReconciler(Reconciler):
             if suspend_status != Cond.SUSPENDED:
                 for view in self._active(self._list_cluster_pod_views(cluster)):
                     self._delete_pod_by_name(namespace, view.name)
-                return
+                return True
             if cluster.spec.suspend:
-                return  # stays suspended, no pods
+                return True  # stays suspended, no pods
         if suspend_status == Cond.SUSPENDED and not cluster.spec.suspend:
             pass  # resuming: fall through to create pods
 
@@ -438,10 +451,11 @@ class RayClusterReconciler(Reconciler):
         head_pods = [p for p in pods
                      if p.labels.get(C.RAY_NODE_TYPE_LABEL_KEY) == RayNodeType.HEAD]
 
+        complete = True
         # ---- head pod singleton
         if not self.expectations.is_satisfied(
                 self._cache(), namespace, name, "__head__"):
-            return
+            return False
         active_heads = self._active(head_pods)
         if len(active_heads) == 1:
             head = active_heads[0]
@@ -469,7 +483,9 @@ class RayClusterReconciler(Reconciler):
 
         # ---- worker groups
         for group in cluster.spec.worker_group_specs:
-            self._reconcile_worker_group(cluster, group, pods)
+            if not self._reconcile_worker_group(cluster, group, pods):
+                complete = False
+        return complete
 
     def _cache(self):
         # the in-memory client's server doubles as the informer cache; REST
@@ -480,7 +496,7 @@ class RayClusterReconciler(Reconciler):
         return _TypedCacheShim(self.client)
 
     def _reconcile_worker_group(self, cluster: RayCluster, group: WorkerGroupSpec,
-                                all_pods) -> None:
+                                all_pods) -> bool:
         namespace = cluster.metadata.namespace or "default"
         name = cluster.metadata.name
         group_pods = [p for p in all_pods
@@ -491,10 +507,10 @@ class RayClusterReconciler(Reconciler):
         if group.suspend:
             for view in self._active(group_pods):
                 self._delete_pod_by_name(namespace, view.name)
-            return
+            return True
 
         if not self.expectations.is_satisfied(self._cache(), namespace, name, group.group_name):
-            return
+            return False
 
         # unhealthy deletion
         for view in self._active(group_pods):
@@ -533,6 +549,7 @@ class RayClusterReconciler(Reconciler):
                                                         view.name)
                     self._delete_pod_by_name(namespace, view.name)
             # else: wait for the autoscaler to name victims via WorkersToDelete
+        return True
 
     def _owner_crd_type(self, cluster: RayCluster) -> Optional[str]:
         return (cluster.metadata.labels or {}).get(C.RAY_ORIGINATED_FROM_CRD_LABEL_KEY)
