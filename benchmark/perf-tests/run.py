@@ -30,6 +30,8 @@ SCALES = {
 REFERENCE = {  # junit totals from the reference repo (GKE, v1.1.1)
     "100": 135.2, "1000": 647.5, "5000": 2700.1, "10000": 2720.1,
 }
+REFERENCE_RAYJOB = {"100": 712.8, "1000": 997.2, "5000": 3059.0,
+                    "10000": 3459.0}
 
 
 def run_raycluster_tier(scale_cfg, workers=4):
@@ -70,30 +72,85 @@ def run_raycluster_tier(scale_cfg, workers=4):
             "overall_s": round(t_total, 2)}
 
 
+def run_rayjob_tier(scale_cfg, workers=4):
+    """N RayJobs (ephemeral clusters, shutdown on finish) to Complete —
+    the reference's N-rayjob clusterloader2 suite."""
+    from kuberay_amd.models import RayJob
+    from kuberay_amd.testing import ControlPlane, simple_raycluster
+    n = scale_cfg["clusters"]
+    cp = ControlPlane(kubelet_delay=0.0, record_events=False,
+                      requeue_seconds=3600, poll_seconds=0.2, workers=workers,
+                      job_runtime=0.01)
+    cp.dashboard.job_polls_to_succeeded = 1
+    cp.start()
+    try:
+        watcher = cp.server.watch({"RayJob"})
+        t0 = time.perf_counter()
+        for i in range(n):
+            cp.client.create(RayJob.from_dict({
+                "apiVersion": "ray.io/v1", "kind": "RayJob",
+                "metadata": {"name": f"perfjob-{i:05d}", "namespace": "default"},
+                "spec": {"entrypoint": "python train.py",
+                         "shutdownAfterJobFinishes": True,
+                         "rayClusterSpec": simple_raycluster(
+                             "x", workers=1).spec.to_dict()}}))
+        t_create = time.perf_counter() - t0
+        pending = {f"perfjob-{i:05d}" for i in range(n)}
+        while pending:
+            ev = watcher.next(timeout=1.0)
+            if ev is None:
+                continue
+            _, obj = ev
+            if obj.get("status", {}).get("jobDeploymentStatus") == "Complete":
+                pending.discard(obj["metadata"]["name"])
+        t_complete = time.perf_counter() - t0
+        watcher.stop()
+    finally:
+        cp.stop()
+    return {"create_s": round(t_create, 2),
+            "jobs_complete_s": round(t_complete, 2),
+            "overall_s": round(t_complete, 2)}
+
+
 def main():
     parser = argparse.ArgumentParser()
     parser.add_argument("--scale", default="100", choices=sorted(SCALES))
+    parser.add_argument("--suite", default="raycluster",
+                        choices=["raycluster", "rayjob"])
     args = parser.parse_args()
-    result = run_raycluster_tier(SCALES[args.scale])
-    ref = REFERENCE.get(args.scale)
+    if args.suite == "rayjob":
+        result = run_rayjob_tier(SCALES[args.scale])
+        ref = REFERENCE_RAYJOB.get(args.scale)
+    else:
+        result = run_raycluster_tier(SCALES[args.scale])
+        ref = REFERENCE.get(args.scale)
     out = {
-        "suite": f"{args.scale}-raycluster",
+        "suite": f"{args.scale}-{args.suite}",
         "kubelet": "simulated",
+        "note": ("rayjob suite uses synthetic instant applications: this "
+                 "measures operator lifecycle overhead; the reference's "
+                 "numbers include real MNIST workload runtime"
+                 if args.suite == "rayjob" else
+                 "measures operator-side steps; reference numbers include "
+                 "real GKE pod cold-starts"),
         **result,
         "reference_overall_s_gke": ref,
         "speedup_vs_reference": round(ref / result["overall_s"], 1) if ref else None,
     }
     print(json.dumps(out, indent=2))
     here = os.path.dirname(os.path.abspath(__file__))
-    with open(os.path.join(here, f"results-{args.scale}-raycluster.json"), "w") as f:
+    with open(os.path.join(here, f"results-{args.scale}-{args.suite}.json"), "w") as f:
         json.dump(out, f, indent=2)
     # junit for parity with the reference artifact format
-    xml = (f'<testsuite name="{args.scale}-raycluster" tests="3">'
+    wait_name = ("Wait for RayClusters ready" if args.suite == "raycluster"
+                 else "Wait for RayJobs complete")
+    wait_time = result.get("ready_wait_s", result.get("jobs_complete_s"))
+    xml = (f'<testsuite name="{args.scale}-{args.suite}" tests="3">'
            f'<testcase name="overall" time="{result["overall_s"]}"/>'
-           f'<testcase name="create RayClusters" time="{result["create_s"]}"/>'
-           f'<testcase name="Wait for RayClusters ready" time="{result["ready_wait_s"]}"/>'
+           f'<testcase name="create" time="{result["create_s"]}"/>'
+           f'<testcase name="{wait_name}" time="{wait_time}"/>'
            f'</testsuite>')
-    with open(os.path.join(here, f"junit-{args.scale}-raycluster.xml"), "w") as f:
+    with open(os.path.join(here, f"junit-{args.scale}-{args.suite}.xml"), "w") as f:
         f.write(xml)
 
 
