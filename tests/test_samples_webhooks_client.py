@@ -41,7 +41,7 @@ class TestSampleYamls:
             def done():
                 j = control_plane.client.try_get(RayJob, "default", name)
                 return j is not None and j.status.job_deployment_status in (
-                    "Running", "Complete")
+                    "Running", "Complete", "Waiting")
             assert control_plane.wait_for(done, timeout=25), fname
         elif kind == "RayService":
             def ready():
