@@ -2,6 +2,7 @@
 scheduler-plugins) and the MI355X-native xGMI gang scheduler."""
 from .batchscheduler import (  # noqa: F401
     BatchScheduler,
+    KaiBatchScheduler,
     SchedulerPluginsBatchScheduler,
     VolcanoBatchScheduler,
     XgmiGangScheduler,

@@ -165,6 +165,21 @@ class SchedulerPluginsBatchScheduler(BatchScheduler):
         pod.metadata.ensure_labels()[POD_GROUP_ANNOTATION] = pod_group_name(cluster)
 
 
+class KaiBatchScheduler(BatchScheduler):
+    """batchscheduler/kai analog: KAI's pod-grouper gang-groups pods by
+    their top owner (the RayCluster), so the operator only routes pods to
+    the kai-scheduler and propagates the queue label."""
+
+    name = "kai-scheduler"
+    QUEUE_LABEL = "kai.scheduler/queue"
+
+    def add_metadata_to_pod(self, client, cluster, group_name, pod) -> None:
+        pod.spec.scheduler_name = "kai-scheduler"
+        queue = (cluster.metadata.labels or {}).get(self.QUEUE_LABEL)
+        if queue:
+            pod.metadata.ensure_labels()[self.QUEUE_LABEL] = queue
+
+
 class XgmiGangScheduler(BatchScheduler):
     """MI355X-native gang scheduling with xGMI-topology affinity.
 
@@ -220,6 +235,7 @@ SCHEDULERS = {
     VolcanoBatchScheduler.name: VolcanoBatchScheduler,
     YunikornBatchScheduler.name: YunikornBatchScheduler,
     SchedulerPluginsBatchScheduler.name: SchedulerPluginsBatchScheduler,
+    KaiBatchScheduler.name: KaiBatchScheduler,
     XgmiGangScheduler.name: XgmiGangScheduler,
 }
 

@@ -225,3 +225,22 @@ class TestGangScheduling:
                                                      "default-group", pod)
         assert pod.metadata.labels["applicationId"] == "default-demo"
         assert pod.spec.scheduler_name == "yunikorn"
+
+
+class TestKaiScheduler:
+    def test_kai_routing_and_queue(self):
+        from kuberay_amd.parallel import KaiBatchScheduler
+        client = InMemoryClient()
+        cluster = simple_raycluster("demo", gpus_per_worker=1)
+        cluster.metadata.labels = {"kai.scheduler/queue": "team-a"}
+        client.create(cluster)
+        from kuberay_amd.common import pod as podlib
+        group = cluster.spec.worker_group_specs[0]
+        t = podlib.default_worker_pod_template(
+            cluster, group, "p-", "x.default.svc.cluster.local", "6379")
+        pod = podlib.build_pod(t, "worker", group.ray_start_params, "6379",
+                               False, None, "x.default.svc.cluster.local")
+        KaiBatchScheduler().add_metadata_to_pod(client, cluster,
+                                                "default-group", pod)
+        assert pod.spec.scheduler_name == "kai-scheduler"
+        assert pod.metadata.labels["kai.scheduler/queue"] == "team-a"
