@@ -169,3 +169,45 @@ class TestUnhealthy:
         assert not s.condition_true("Ready")
         apps = s.status.pending_service_status.applications or {}
         assert apps.get("app1") and apps["app1"].status == "DEPLOY_FAILED"
+
+
+class TestServeStatusEdgeCases:
+    def test_app_vanishes_from_dashboard_flips_ready_off(self, control_plane):
+        cp = control_plane
+        cp.client.create(make_rayservice())
+        assert wait_ready(cp)
+        # dashboard suddenly reports no applications (e.g. serve controller died)
+        cp.dashboard.serve_statuses_mock = {"applications": {}}
+        def not_ready():
+            s = svc_of(cp)
+            return (not s.condition_true("Ready")
+                    or s.status.num_serve_endpoints == 0)
+        # Ready condition tracks endpoints; pods are still serving so the
+        # endpoints stay — but application status must reflect the outage
+        def apps_gone():
+            s = svc_of(cp)
+            apps = s.status.active_service_status.applications
+            return apps == {} or apps is None or all(
+                a.status != "RUNNING" for a in apps.values())
+        assert cp.wait_for(apps_gone, timeout=15)
+
+    def test_dashboard_flap_resubmits_config(self, control_plane):
+        cp = control_plane
+        cp.client.create(make_rayservice())
+        assert wait_ready(cp)
+        calls_before = len(cp.dashboard.update_serve_calls)
+        # dashboard restart: next status poll raises, cache must invalidate
+        from kuberay_amd.utils.dashboard_client import DashboardClientError
+        original = cp.dashboard.get_serve_applications
+        state = {"fail": 2}
+        def flaky():
+            if state["fail"] > 0:
+                state["fail"] -= 1
+                raise DashboardClientError("dashboard restarting")
+            return original()
+        cp.dashboard.get_serve_applications = flaky
+        assert cp.wait_for(
+            lambda: len(cp.dashboard.update_serve_calls) > calls_before,
+            timeout=15), "config not resubmitted after dashboard flap"
+        assert cp.wait_for(lambda: svc_of(cp).condition_true("Ready"),
+                           timeout=15)
