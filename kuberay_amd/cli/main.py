@@ -101,6 +101,36 @@ def get_job(ctx, name):
     _table(["NAME", "DEPLOYMENT STATUS", "JOB STATUS", "CLUSTER"], rows)
 
 
+@get.command("events")
+@click.argument("name", required=False)
+@click.pass_context
+def get_events(ctx, name):
+    """List operator events (optionally for one object)."""
+    import httpx
+    client = client_of(ctx)
+    ns = ctx.obj["namespace"]
+    # events are core objects — fetch through the raw surface when available
+    raw = getattr(client, "raw_list", None)
+    server = getattr(client, "server", None)
+    if server is not None:
+        events = server.list("Event", ns)
+    elif raw is not None:
+        events = raw("Event", ns)
+    else:
+        raise click.ClickException("events not available over this server")
+    rows = []
+    for e in events:
+        involved = e.get("involvedObject") or {}
+        if name and involved.get("name") != name:
+            continue
+        rows.append((e.get("lastTimestamp", ""), e.get("type", ""),
+                     e.get("reason", ""),
+                     f"{involved.get('kind','')}/{involved.get('name','')}",
+                     e.get("count", 1), (e.get("message") or "")[:60]))
+    rows.sort()
+    _table(["LAST SEEN", "TYPE", "REASON", "OBJECT", "COUNT", "MESSAGE"], rows)
+
+
 @get.command("service")
 @click.argument("name", required=False)
 @click.pass_context

@@ -293,3 +293,20 @@ class TestServeApplicationsProxy:
         r = t.get("/apis/v1/namespaces/ns1/serveapplications/c1")
         assert r.status_code == 200
         assert "applications" in r.json()
+
+
+class TestKrayEvents:
+    def test_get_events_lists_operator_events(self):
+        import importlib
+        from kuberay_amd.testing import ControlPlane, simple_raycluster
+        climod = importlib.import_module("kuberay_amd.cli.main")
+        cp = ControlPlane(kubelet_delay=0.01, poll_seconds=0.05).start()
+        try:
+            cp.client.create(simple_raycluster("demo"))
+            assert cp.wait_cluster_state("default", "demo", "ready")
+            climod.make_client = lambda server: cp.client
+            r = CliRunner().invoke(cli, ["get", "events", "demo"])
+            assert r.exit_code == 0, r.output
+            assert "CreatedHeadPod" in r.output
+        finally:
+            cp.stop()
