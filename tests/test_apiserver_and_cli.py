@@ -301,6 +301,7 @@ class TestKrayEvents:
         from kuberay_amd.testing import ControlPlane, simple_raycluster
         climod = importlib.import_module("kuberay_amd.cli.main")
         cp = ControlPlane(kubelet_delay=0.01, poll_seconds=0.05).start()
+        original = climod.make_client
         try:
             cp.client.create(simple_raycluster("demo"))
             assert cp.wait_cluster_state("default", "demo", "ready")
@@ -309,4 +310,5 @@ class TestKrayEvents:
             assert r.exit_code == 0, r.output
             assert "CreatedHeadPod" in r.output
         finally:
+            climod.make_client = original
             cp.stop()
