@@ -1,9 +1,10 @@
 """CRD YAML generation (the kubebuilder codegen analog; reference CRDs at
 ray-operator/config/crd/bases/ray.io_*.yaml).
 
-Schemas are generated from the pydantic models: typed top-level fields with
-``x-kubernetes-preserve-unknown-fields`` on the pod-template subtrees (which
-embed arbitrary core/v1 content), same printer columns as the reference.
+Full structural openAPIV3Schemas are generated from the pydantic models
+(kuberay_amd.schema): every CR field is typed; the embedded pod-template /
+core-v1 subtrees carry ``x-kubernetes-preserve-unknown-fields`` (they hold
+arbitrary core/v1 content). Printer columns match the reference.
 """
 from __future__ import annotations
 
@@ -15,9 +16,22 @@ from .utils import constants as C
 
 _PRESERVE = {"type": "object", "x-kubernetes-preserve-unknown-fields": True}
 
+_MODELS = None
+
+
+def _model_for(kind: str):
+    global _MODELS
+    if _MODELS is None:
+        from .models import RayCluster, RayCronJob, RayJob, RayService
+        _MODELS = {"RayCluster": RayCluster, "RayJob": RayJob,
+                   "RayService": RayService, "RayCronJob": RayCronJob}
+    return _MODELS[kind]
+
 
 def _crd(kind: str, plural: str, printer_columns: List[Dict[str, Any]],
          extra_names: Dict[str, Any] = None) -> Dict[str, Any]:
+    from .schema import structural_schema
+    schema = structural_schema(_model_for(kind))
     singular = kind.lower()
     return {
         "apiVersion": "apiextensions.k8s.io/v1",
@@ -39,16 +53,7 @@ def _crd(kind: str, plural: str, printer_columns: List[Dict[str, Any]],
                 "storage": True,
                 "subresources": {"status": {}},
                 "additionalPrinterColumns": printer_columns,
-                "schema": {"openAPIV3Schema": {
-                    "type": "object",
-                    "properties": {
-                        "apiVersion": {"type": "string"},
-                        "kind": {"type": "string"},
-                        "metadata": {"type": "object"},
-                        "spec": _PRESERVE,
-                        "status": _PRESERVE,
-                    },
-                }},
+                "schema": {"openAPIV3Schema": schema},
             }],
         },
     }

@@ -197,3 +197,53 @@ class TestValidation:
                 "upgradeStrategy": {"type": "NewClusterWithIncrementalUpgrade"}}})
         errs = validate_rayservice_spec(svc)
         assert any("clusterUpgradeOptions" in e for e in errs)
+
+
+class TestStructuralSchemas:
+    def test_all_crds_structural_and_clean(self):
+        import json
+        from kuberay_amd.crds import all_crds
+        for crd in all_crds():
+            schema = crd["spec"]["versions"][0]["schema"]["openAPIV3Schema"]
+            text = json.dumps(schema)
+            for banned in ("$ref", "$defs", "anyOf", "allOf", "oneOf"):
+                assert banned not in text, (crd["metadata"]["name"], banned)
+            assert schema["type"] == "object"
+            assert "spec" in schema["properties"]
+            assert "status" in schema["properties"]
+
+    def test_samples_validate_against_schema(self):
+        """Every shipped sample must pass its CRD's structural schema
+        (jsonschema-lite check on typed fields)."""
+        import glob
+        import os
+        import yaml as _yaml
+        from kuberay_amd.crds import all_crds
+        schemas = {c["spec"]["names"]["kind"]:
+                   c["spec"]["versions"][0]["schema"]["openAPIV3Schema"]
+                   for c in all_crds()}
+
+        def check(node, schema, path):
+            if schema.get("x-kubernetes-preserve-unknown-fields"):
+                return
+            t = schema.get("type")
+            if t == "object" and isinstance(node, dict):
+                props = schema.get("properties", {})
+                for k, v in node.items():
+                    if k in props:
+                        check(v, props[k], f"{path}.{k}")
+            elif t == "array" and isinstance(node, list):
+                for i, v in enumerate(node):
+                    check(v, schema.get("items", {}), f"{path}[{i}]")
+            elif t == "integer":
+                assert isinstance(node, int), (path, node)
+            elif t == "string":
+                assert isinstance(node, str), (path, node)
+            elif t == "boolean":
+                assert isinstance(node, bool), (path, node)
+
+        samples = os.path.join(os.path.dirname(__file__), "..", "deploy", "samples")
+        for f in sorted(glob.glob(os.path.join(samples, "*.yaml"))):
+            for doc in _yaml.safe_load_all(open(f)):
+                if doc and doc.get("kind") in schemas:
+                    check(doc, schemas[doc["kind"]], os.path.basename(f))
