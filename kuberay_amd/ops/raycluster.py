@@ -533,7 +533,11 @@ class RayClusterReconciler(Reconciler):
             group_pods = [p for p in group_pods if p.name not in to_delete]
 
         running = self._active(group_pods)
-        desired = res.worker_group_desired_replicas(group) * max(group.num_of_hosts, 1)
+        n_hosts = max(group.num_of_hosts, 1)
+        if n_hosts > 1:
+            return self._reconcile_multihost(cluster, group, running)
+
+        desired = res.worker_group_desired_replicas(group)
         diff = desired - len(running)
         if diff > 0:
             for i in range(diff):
@@ -549,6 +553,61 @@ class RayClusterReconciler(Reconciler):
                                                         view.name)
                     self._delete_pod_by_name(namespace, view.name)
             # else: wait for the autoscaler to name victims via WorkersToDelete
+        return True
+
+    def _reconcile_multihost(self, cluster: RayCluster, group: WorkerGroupSpec,
+                             running) -> bool:
+        """Multi-host groups create/delete whole NumOfHosts replica units
+        (RayMultiHostIndexing; controller.go:1287, pod.go:673-682): a replica
+        that lost any host is torn down entirely and rebuilt."""
+        namespace = cluster.metadata.namespace or "default"
+        name = cluster.metadata.name
+        n_hosts = max(group.num_of_hosts, 1)
+        desired_replicas = res.worker_group_desired_replicas(group)
+
+        by_replica = {}
+        strays = []
+        for view in running:
+            rep = view.labels.get(C.RAY_WORKER_REPLICA_NAME_KEY)
+            if rep:
+                by_replica.setdefault(rep, []).append(view)
+            else:
+                strays.append(view)
+        # pods without a replica label cannot belong to a multi-host unit
+        for view in strays:
+            self.expectations.expect_delete_pod(namespace, name,
+                                                group.group_name, view.name)
+            self._delete_pod_by_name(namespace, view.name)
+
+        complete_replicas = []
+        for rep, members in sorted(by_replica.items()):
+            if len(members) == n_hosts:
+                complete_replicas.append(rep)
+            else:
+                # partial unit (a host died or creation raced): tear it down
+                for view in members:
+                    self.expectations.expect_delete_pod(
+                        namespace, name, group.group_name, view.name)
+                    self._delete_pod_by_name(namespace, view.name)
+
+        diff = desired_replicas - len(complete_replicas)
+        if diff > 0:
+            for r in range(diff):
+                replica_grp = names.worker_replica_group_name(group.group_name)
+                for host_idx in range(n_hosts):
+                    self._create_worker_pod(
+                        cluster, group,
+                        replica_index=len(complete_replicas) + r,
+                        host_index=host_idx, replica_grp_name=replica_grp)
+        elif diff < 0:
+            random_delete_enabled = os.environ.get(
+                C.ENABLE_RANDOM_POD_DELETE, "").lower() == "true"
+            if not podlib.is_autoscaling_enabled(cluster.spec) or random_delete_enabled:
+                for rep in complete_replicas[:(-diff)]:
+                    for view in by_replica[rep]:
+                        self.expectations.expect_delete_pod(
+                            namespace, name, group.group_name, view.name)
+                        self._delete_pod_by_name(namespace, view.name)
         return True
 
     def _owner_crd_type(self, cluster: RayCluster) -> Optional[str]:
@@ -581,19 +640,21 @@ class RayClusterReconciler(Reconciler):
                              "Created head Pod %s", created.metadata.name)
 
     def _create_worker_pod(self, cluster: RayCluster, group: WorkerGroupSpec,
-                           replica_index: int = 0) -> None:
+                           replica_index: int = 0, host_index: int = 0,
+                           replica_grp_name: Optional[str] = None) -> None:
         namespace = cluster.metadata.namespace or "default"
         pod_name = names.pod_name(
             names.check_name(f"{cluster.metadata.name}-{group.group_name}"),
             RayNodeType.WORKER, True)
         head_port = podlib.get_head_port(cluster.spec.head_group_spec.ray_start_params)
         fqdn = names.fqdn_service_name(cluster, namespace)
-        replica_grp = names.worker_replica_group_name(group.group_name)
+        replica_grp = replica_grp_name or names.worker_replica_group_name(
+            group.group_name)
         template = podlib.default_worker_pod_template(
             cluster, group, pod_name, fqdn, head_port,
             replica_grp_name=replica_grp,
-            replica_index=replica_index // max(group.num_of_hosts, 1),
-            num_host_index=replica_index % max(group.num_of_hosts, 1))
+            replica_index=replica_index,
+            num_host_index=host_index)
         for sidecar in self.options.worker_sidecar_containers:
             template.spec.containers.append(sidecar.clone())
         pod = podlib.build_pod(

@@ -318,3 +318,50 @@ class TestGcsPvcRetain:
         assert control_plane.wait_for(
             lambda: control_plane.server.try_get(
                 "PersistentVolumeClaim", "default", "demo-gcs-pvc") is None)
+
+
+class TestMultihostAtomicity:
+    def test_replica_units_share_group_label(self, control_plane):
+        control_plane.client.create(simple_raycluster("demo", workers=2,
+                                                      num_of_hosts=3))
+        assert control_plane.wait_for(
+            lambda: get_cluster(control_plane).status.available_worker_replicas == 6)
+        views = control_plane.client.list_pod_views(
+            "default", {C.RAY_CLUSTER_LABEL_KEY: "demo",
+                        C.RAY_NODE_TYPE_LABEL_KEY: "worker"})
+        by_rep = {}
+        for v in views:
+            rep = v.labels[C.RAY_WORKER_REPLICA_NAME_KEY]
+            by_rep.setdefault(rep, []).append(v)
+        assert len(by_rep) == 2
+        for rep, members in by_rep.items():
+            assert len(members) == 3
+            hosts = sorted(v.labels[C.RAY_HOST_INDEX_KEY] for v in members)
+            assert hosts == ["0", "1", "2"]
+
+    def test_partial_replica_rebuilt_whole(self, control_plane):
+        control_plane.client.create(simple_raycluster("demo", workers=1,
+                                                      num_of_hosts=3))
+        assert control_plane.wait_for(
+            lambda: get_cluster(control_plane).status.available_worker_replicas == 3)
+        views = control_plane.client.list_pod_views(
+            "default", {C.RAY_CLUSTER_LABEL_KEY: "demo",
+                        C.RAY_NODE_TYPE_LABEL_KEY: "worker"})
+        old_rep = views[0].labels[C.RAY_WORKER_REPLICA_NAME_KEY]
+        victim = views[0].name
+        # kill one host of the replica
+        control_plane.server.patch_merge("Pod", "default", victim,
+                                         {"status": {"phase": "Failed"}},
+                                         subresource="status")
+        def rebuilt():
+            vs = control_plane.client.list_pod_views(
+                "default", {C.RAY_CLUSTER_LABEL_KEY: "demo",
+                            C.RAY_NODE_TYPE_LABEL_KEY: "worker"})
+            live = [v for v in vs if not v.deletion_timestamp
+                    and v.phase == "Running"]
+            if len(live) != 3:
+                return False
+            reps = {v.labels[C.RAY_WORKER_REPLICA_NAME_KEY] for v in live}
+            # the whole unit was replaced under a fresh replica name
+            return len(reps) == 1 and old_rep not in reps
+        assert control_plane.wait_for(rebuilt, timeout=20)
