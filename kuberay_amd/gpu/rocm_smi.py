@@ -51,7 +51,13 @@ def query_rocm_smi_json(timeout: float = 10.0) -> Dict:
     out = subprocess.run(cmd, capture_output=True, text=True, timeout=timeout)
     if out.returncode != 0:
         raise RuntimeError(f"rocm-smi failed rc={out.returncode}: {out.stderr[:400]}")
-    return json.loads(out.stdout)
+    # rocm-smi sometimes prefixes warnings (e.g. low-power state) — take the
+    # JSON from the first brace
+    text = out.stdout
+    start = text.find("{")
+    if start < 0:
+        raise RuntimeError(f"rocm-smi produced no JSON: {text[:200]}")
+    return json.loads(text[start:])
 
 
 def parse_rocm_smi_json(data: Dict) -> List[GpuStats]:
@@ -69,7 +75,9 @@ def parse_rocm_smi_json(data: Dict) -> List[GpuStats]:
         for k, v in card.items():
             lk = k.lower()
             try:
-                if "gpu use" in lk or "gfx activity" in lk or lk == "gpu use (%)":
+                if "gpu use" in lk:
+                    # NOTE: "GFX Activity" is a monotonically increasing
+                    # counter, NOT a percentage — never treat it as one.
                     s.utilization_pct = float(v)
                 elif "vram" in lk and "used" in lk:
                     s.vram_used_bytes = int(v)
