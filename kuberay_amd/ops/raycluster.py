@@ -448,6 +448,30 @@ class RayClusterReconciler(Reconciler):
             self.batch_scheduler.do_batch_scheduling_on_submission(self.client, cluster)
 
         pods = self._list_cluster_pod_views(cluster)
+
+        # UpgradeStrategy Recreate: rebuild pods whose spec-hash label is
+        # stale (reference: UpgradeStrategyRecreateHashKey)
+        if (cluster.spec.upgrade_strategy is not None
+                and cluster.spec.upgrade_strategy.type == "Recreate"):
+            current_hash = hash_without_replicas_and_workers_to_delete(
+                cluster.spec)[:63]
+            stale = [p for p in self._active(pods)
+                     if p.labels.get(
+                         C.HASH_WITHOUT_REPLICAS_AND_WORKERS_TO_DELETE_KEY)
+                     not in (None, current_hash)]
+            for view in stale:
+                self.recorder.eventf(cluster, "Normal", "RecreatePodForUpgrade",
+                                     "Deleting Pod %s (stale spec hash, "
+                                     "upgradeStrategy: Recreate)", view.name)
+                group = view.labels.get(C.RAY_NODE_GROUP_LABEL_KEY, "__head__")
+                if view.labels.get(C.RAY_NODE_TYPE_LABEL_KEY) == RayNodeType.HEAD:
+                    group = "__head__"
+                self.expectations.expect_delete_pod(namespace, name, group,
+                                                    view.name)
+                self._delete_pod_by_name(namespace, view.name)
+            if stale:
+                pods = self._list_cluster_pod_views(cluster)
+
         head_pods = [p for p in pods
                      if p.labels.get(C.RAY_NODE_TYPE_LABEL_KEY) == RayNodeType.HEAD]
 
@@ -662,6 +686,8 @@ class RayClusterReconciler(Reconciler):
             podlib.is_autoscaling_enabled(cluster.spec), self._owner_crd_type(cluster),
             fqdn, self.options.default_container_envs, cluster.spec.ray_version)
         pod.metadata.owner_references = [k8s.owner_reference_for(cluster)]
+        pod.metadata.labels[C.HASH_WITHOUT_REPLICAS_AND_WORKERS_TO_DELETE_KEY] = \
+            hash_without_replicas_and_workers_to_delete(cluster.spec)[:63]
         if self.batch_scheduler is not None:
             self.batch_scheduler.add_metadata_to_pod(self.client, cluster,
                                                      group.group_name, pod)

@@ -365,3 +365,51 @@ class TestMultihostAtomicity:
             # the whole unit was replaced under a fresh replica name
             return len(reps) == 1 and old_rep not in reps
         assert control_plane.wait_for(rebuilt, timeout=20)
+
+
+class TestUpgradeStrategyRecreate:
+    def test_pods_rebuilt_on_spec_change(self, control_plane):
+        cluster = simple_raycluster("demo", workers=2,
+                                    upgradeStrategy={"type": "Recreate"})
+        control_plane.client.create(cluster)
+        assert control_plane.wait_cluster_state("default", "demo", "ready")
+        old_names = {p["metadata"]["name"] for p in pods_of(control_plane)}
+
+        rc = get_cluster(control_plane)
+        rc.spec.worker_group_specs[0].template.spec.containers[0].image = \
+            "rayproject/ray:2.47.0-rocm"
+        control_plane.client.update(rc)
+
+        def rebuilt():
+            current = {p["metadata"]["name"] for p in pods_of(control_plane)}
+            rc2 = get_cluster(control_plane)
+            return (current.isdisjoint(old_names) and len(current) == 3
+                    and rc2.status.state == "ready"
+                    and rc2.status.ready_worker_replicas == 2)
+        assert control_plane.wait_for(rebuilt, timeout=20)
+
+    def test_no_recreate_without_strategy(self, control_plane):
+        control_plane.client.create(simple_raycluster("demo", workers=1))
+        assert control_plane.wait_cluster_state("default", "demo", "ready")
+        old_names = {p["metadata"]["name"] for p in pods_of(control_plane)}
+        rc = get_cluster(control_plane)
+        rc.spec.worker_group_specs[0].template.spec.containers[0].image = \
+            "rayproject/ray:2.47.0-rocm"
+        control_plane.client.update(rc)
+        time.sleep(0.6)
+        current = {p["metadata"]["name"] for p in pods_of(control_plane)}
+        assert current == old_names  # default: in-place, no recreation
+
+    def test_replicas_change_does_not_recreate(self, control_plane):
+        cluster = simple_raycluster("demo", workers=1,
+                                    upgradeStrategy={"type": "Recreate"})
+        control_plane.client.create(cluster)
+        assert control_plane.wait_cluster_state("default", "demo", "ready")
+        old_names = {p["metadata"]["name"] for p in pods_of(control_plane)}
+        rc = get_cluster(control_plane)
+        rc.spec.worker_group_specs[0].replicas = 2
+        control_plane.client.update(rc)
+        assert control_plane.wait_for(
+            lambda: get_cluster(control_plane).status.available_worker_replicas == 2)
+        current = {p["metadata"]["name"] for p in pods_of(control_plane)}
+        assert old_names.issubset(current)  # hash mutes replica changes
