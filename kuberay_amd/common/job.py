@@ -52,7 +52,10 @@ def build_job_submit_command(rayjob: RayJob, submission_mode: str) -> List[str]:
                                    C.DEFAULT_DASHBOARD_PORT)
         address = f"http://127.0.0.1:{port}"
         health_url = f"http://localhost:{port}/{C.RAY_DASHBOARD_GCS_HEALTH_PATH}"
-        needs_status_check = False
+        # sidecar containers never retry unless the SidecarSubmitterRestart
+        # gate makes them restartable (job.go:160-165)
+        from .. import features
+        needs_status_check = features.enabled("SidecarSubmitterRestart")
     elif submission_mode == JobSubmissionMode.K8S_JOB:
         address = rayjob.status.dashboard_url or ""
         if not address.startswith("http://"):

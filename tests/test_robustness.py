@@ -127,3 +127,24 @@ class TestSubmitterJobStatusRace:
             return (j is not None and j.status.job_deployment_status == "Failed"
                     and j.status.reason == "SubmissionFailed")
         assert control_plane.wait_for(failed, timeout=20)
+
+
+class TestSidecarSubmitterRestartGate:
+    def test_gate_enables_idempotent_sidecar_command(self):
+        import kuberay_amd.features as features
+        from kuberay_amd.testing import simple_raycluster
+        job = RayJob.from_dict({
+            "apiVersion": "ray.io/v1", "kind": "RayJob",
+            "metadata": {"name": "j1"},
+            "spec": {"entrypoint": "python t.py",
+                     "rayClusterSpec": simple_raycluster("x").spec.to_dict()}})
+        job.status.job_id = "j1-x"
+        cmd_off = " ".join(build_job_submit_command(job, "SidecarMode"))
+        assert "if ! ray job status" not in cmd_off
+        features.set_gate("SidecarSubmitterRestart", True)
+        try:
+            cmd_on = " ".join(build_job_submit_command(job, "SidecarMode"))
+            assert "if ! ray job status" in cmd_on
+            assert "--no-wait" in cmd_on
+        finally:
+            features.reset()
