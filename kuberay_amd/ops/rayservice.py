@@ -64,6 +64,8 @@ class RayServiceReconciler(Reconciler):
         self.upgrader = IncrementalUpgrader(client, self.recorder)
         # serve-config cache: (namespace, service, cluster) -> config hash
         self._serve_config_cache: Dict[Tuple[str, str, str], str] = {}
+        # first-unhealthy timestamps: (namespace, service, cluster) -> monotonic
+        self._unhealthy_since: Dict[Tuple[str, str, str], float] = {}
         # delayed old-cluster GC: (namespace, cluster) -> not-before time
         self._pending_deletions: Dict[Tuple[str, str], float] = {}
         self._lock = threading.Lock()
@@ -107,6 +109,14 @@ class RayServiceReconciler(Reconciler):
         ready = False
         if target is not None and target.status.state == ClusterState.READY:
             ready = self._reconcile_serve(svc, target, is_pending)
+            self._track_health(svc, target, ready)
+            if (not is_pending and not ready
+                    and self._unhealthy_beyond_threshold(svc, target)):
+                # reference rayservice_controller.go:1409-1440: persistently
+                # unhealthy apps trigger a replacement cluster
+                self._prepare_replacement_cluster(svc, target)
+                pending = self._get_cluster(
+                    svc, svc.status.pending_service_status.ray_cluster_name)
 
         if is_pending and ready:
             if self._incremental_enabled(svc) and active is not None:
@@ -331,6 +341,46 @@ class RayServiceReconciler(Reconciler):
         target_status.applications = apps
         target_status.ray_cluster_status = cluster.status
         return all_running and bool(apps)
+
+    # ------------------------------------------------------------------
+    # unhealthy-cluster replacement
+    # ------------------------------------------------------------------
+    def _track_health(self, svc: RayService, cluster: RayCluster,
+                      healthy: bool) -> None:
+        key = (svc.metadata.namespace or "default", svc.metadata.name,
+               cluster.metadata.name)
+        if healthy:
+            self._unhealthy_since.pop(key, None)
+        else:
+            self._unhealthy_since.setdefault(key, time.monotonic())
+
+    def _unhealthy_beyond_threshold(self, svc: RayService,
+                                    cluster: RayCluster) -> bool:
+        key = (svc.metadata.namespace or "default", svc.metadata.name,
+               cluster.metadata.name)
+        since = self._unhealthy_since.get(key)
+        if since is None:
+            return False
+        threshold = svc.spec.service_unhealthy_second_threshold
+        if threshold is None:
+            threshold = 900  # reference default
+        return (time.monotonic() - since) > threshold
+
+    def _prepare_replacement_cluster(self, svc: RayService,
+                                     unhealthy: RayCluster) -> None:
+        if svc.status.pending_service_status.ray_cluster_name:
+            return  # replacement already underway
+        name = names.ray_cluster_name_for(svc.metadata.name)
+        svc.status.pending_service_status.ray_cluster_name = name
+        svc.status.conditions = set_condition(
+            svc.status.conditions, Cond.UPGRADE_IN_PROGRESS, "True",
+            "UnhealthyClusterReplacement",
+            f"Serve apps unhealthy on {unhealthy.metadata.name}; preparing a "
+            "replacement RayCluster")
+        self.client.update_status(svc)
+        self.recorder.eventf(svc, "Warning", "UnhealthyServeApps",
+                             "Preparing replacement RayCluster %s", name)
+        self._create_cluster(svc, name, self._goal_hash(svc))
 
     # ------------------------------------------------------------------
     # services & promotion

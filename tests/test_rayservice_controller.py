@@ -211,3 +211,42 @@ class TestServeStatusEdgeCases:
             timeout=15), "config not resubmitted after dashboard flap"
         assert cp.wait_for(lambda: svc_of(cp).condition_true("Ready"),
                            timeout=15)
+
+
+class TestUnhealthyReplacement:
+    def test_persistently_unhealthy_apps_trigger_replacement(self, control_plane):
+        cp = control_plane
+        cp.rayservice_reconciler.cluster_deletion_delay_s = 0.2
+        cp.client.create(make_rayservice(serviceUnhealthySecondThreshold=1))
+        assert wait_ready(cp)
+        first_active = svc_of(cp).status.active_service_status.ray_cluster_name
+        # the app goes permanently unhealthy on the active cluster
+        cp.dashboard.serve_statuses_mock = {"applications": {
+            "app1": {"status": "UNHEALTHY", "message": "actor died",
+                     "deployments": {}}}}
+
+        def replacement_started():
+            s = svc_of(cp)
+            return bool(s.status.pending_service_status.ray_cluster_name)
+        assert cp.wait_for(replacement_started, timeout=20)
+        # once the replacement cluster's apps are healthy, it promotes
+        cp.dashboard.serve_statuses_mock = None
+
+        def promoted():
+            s = svc_of(cp)
+            return (s.status.active_service_status.ray_cluster_name
+                    not in (None, first_active) and s.condition_true("Ready"))
+        assert cp.wait_for(promoted, timeout=25)
+
+    def test_transient_unhealth_does_not_replace(self, control_plane):
+        cp = control_plane
+        cp.client.create(make_rayservice(serviceUnhealthySecondThreshold=3600))
+        assert wait_ready(cp)
+        cp.dashboard.serve_statuses_mock = {"applications": {
+            "app1": {"status": "UNHEALTHY", "message": "blip",
+                     "deployments": {}}}}
+        time.sleep(0.8)
+        cp.dashboard.serve_statuses_mock = None
+        assert cp.wait_for(lambda: svc_of(cp).condition_true("Ready"),
+                           timeout=15)
+        assert not svc_of(cp).status.pending_service_status.ray_cluster_name
