@@ -103,7 +103,17 @@ def create_app(client: Optional[KubeClient] = None,
         rc = client.try_get(RayCluster, ns, name)
         if rc is None:
             raise HTTPException(404, f"cluster {name} not found")
-        return conv.raycluster_to_api_cluster(rc)
+        out = conv.raycluster_to_api_cluster(rc)
+        # operator events for this cluster (reference returns cluster events)
+        server = getattr(client, "server", None)
+        if server is not None:
+            out["events"] = [
+                {"reason": e.get("reason"), "type": e.get("type"),
+                 "message": e.get("message"), "count": e.get("count", 1),
+                 "lastTimestamp": e.get("lastTimestamp")}
+                for e in server.list("Event", ns)
+                if (e.get("involvedObject") or {}).get("name") == name]
+        return out
 
     @app.delete("/apis/v1/namespaces/{ns}/clusters/{name}")
     def delete_cluster(ns: str, name: str):
