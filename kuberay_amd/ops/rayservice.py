@@ -7,8 +7,8 @@ serve config submission with cache :1551-1642 + :1896-1925, serve status
 :1846-1894, head serve-label flipping :2074-2128, promotion by service
 repointing :568-641 + :1927-1985, suspend :383-470, endpoints :2130-2171.
 
-The Gateway-API incremental upgrade path (:985-1199) is feature-gated off
-in this build (see kuberay_amd.features.RAYSERVICE_INCREMENTAL_UPGRADE).
+The Gateway-API incremental upgrade path (:985-1199) lives in
+kuberay_amd.ops.incremental behind the RayServiceIncrementalUpgrade gate.
 """
 from __future__ import annotations
 
@@ -84,6 +84,13 @@ class RayServiceReconciler(Reconciler):
         if svc.spec.managed_by not in (None, "ray.io/kuberay-operator"):
             return Result()
         if svc.metadata.deletion_timestamp:
+            # cleanUpServeConfigCache analog (:1896-1925)
+            for key in [k for k in self._serve_config_cache
+                        if k[0] == namespace and k[1] == name]:
+                self._serve_config_cache.pop(key, None)
+            for key in [k for k in self._unhealthy_since
+                        if k[0] == namespace and k[1] == name]:
+                self._unhealthy_since.pop(key, None)
             return Result()
 
         self._gc_old_clusters(namespace)
