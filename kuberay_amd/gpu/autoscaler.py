@@ -79,11 +79,17 @@ class MI355XAutoscaler:
             summary = self.telemetry()
         except Exception:
             return decisions
+        from ..kube.store import ApiError
         for cluster in self.client.list(RayCluster):
             ann = cluster.metadata.annotations or {}
             if ann.get(AMD_AUTOSCALER_ANNOTATION, "").lower() != "true":
                 continue
-            decisions.extend(self._evaluate_cluster(cluster, summary))
+            try:
+                decisions.extend(self._evaluate_cluster(cluster, summary))
+            except ApiError:
+                # optimistic-concurrency loss against the operator/users —
+                # next tick re-evaluates from fresh state
+                continue
         return decisions
 
     def _evaluate_cluster(self, cluster: RayCluster, summary: Dict) -> List[str]:

@@ -299,3 +299,33 @@ class TestNodeLabeller:
     def test_no_labels_without_gpus(self):
         from kuberay_amd.gpu.labeller import compute_node_labels
         assert compute_node_labels(None, "n1") == {}
+
+
+class TestAutoscalerConflictTolerance:
+    def test_conflict_during_update_skips_cluster(self):
+        client = InMemoryClient()
+        cluster = simple_raycluster("demo", workers=1, gpus_per_worker=1)
+        cluster.metadata.annotations = {asc.AMD_AUTOSCALER_ANNOTATION: "true"}
+        client.create(cluster)
+        clock = {"t": 0.0}
+        a = asc.MI355XAutoscaler(
+            client,
+            telemetry=lambda: {"avg_utilization_pct": 95,
+                               "max_vram_used_fraction": 0.9, "gpu_count": 8},
+            policy=asc.AutoscalerPolicy(up_stable_s=0, cooldown_s=0),
+            clock=lambda: clock["t"])
+        # sabotage: every update sees a concurrent writer bumping the rv
+        original_update = client.update
+        def racing_update(obj):
+            from kuberay_amd.models import RayCluster as RC
+            fresh = client.get(RC, "default", "demo")
+            fresh.metadata.annotations["racer"] = str(clock["t"])
+            original_update(fresh)
+            return original_update(obj)  # now stale -> ConflictError
+        client.update = racing_update
+        clock["t"] = 1.0
+        decisions = a.step()  # must not raise
+        assert decisions == [] or decisions  # step survived
+        client.update = original_update
+        clock["t"] = 2.0
+        assert any("scale-up" in d for d in a.step())
