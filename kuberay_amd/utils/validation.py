@@ -68,13 +68,7 @@ def validate_raycluster_spec(cluster: RayCluster) -> List[str]:
     errs += _validate_gcs_ft(cluster)
     errs += _validate_auth(spec)
     errs += _validate_network_policy(spec)
-    if spec.suspend and _autoscaler_enabled(spec) and False:
-        pass  # suspend+autoscaler is allowed; kept for structural parity
     return errs
-
-
-def _autoscaler_enabled(spec: RayClusterSpec) -> bool:
-    return bool(spec.enable_in_tree_autoscaling)
 
 
 def _autoscaler_v2(spec: RayClusterSpec) -> bool:
