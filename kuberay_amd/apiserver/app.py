@@ -246,6 +246,34 @@ def create_app(client: Optional[KubeClient] = None,
             raise HTTPException(404, f"{kind} {name} not found")
         return obj.to_dict()
 
+    def _expand_compute_templates(ns: str, body: Dict[str, Any]) -> None:
+        """apiserversdk compute-template middleware analog
+        (apiserversdk/util/template.go): group specs naming a
+        ``computeTemplate`` get its resources injected."""
+        templates = _templates(ns)
+        spec = body.get("spec") or {}
+        groups = [spec.get("headGroupSpec") or {}]
+        groups += list(spec.get("workerGroupSpecs") or [])
+        for g in groups:
+            tpl_name = g.pop("computeTemplate", None)
+            if not tpl_name:
+                continue
+            tpl = templates.get(tpl_name)
+            if tpl is None:
+                raise HTTPException(400, f"compute template '{tpl_name}' not found")
+            containers = (((g.get("template") or {}).get("spec") or {})
+                          .get("containers") or [])
+            if not containers:
+                g.setdefault("template", {}).setdefault("spec", {})[
+                    "containers"] = [conv._container_from_template(
+                        "ray", C.DEFAULT_RAY_ROCM_IMAGE, tpl)]
+            elif not containers[0].get("resources"):
+                limits = {"cpu": str(tpl["cpu"]), "memory": f"{tpl['memory']}Gi"}
+                if tpl.get("gpu"):
+                    limits[tpl.get("gpuAccelerator", C.AMD_GPU_RESOURCE_NAME)] =                         str(tpl["gpu"])
+                containers[0]["resources"] = {"limits": limits,
+                                              "requests": dict(limits)}
+
     @app.post("/apis/ray.io/v1/namespaces/{ns}/{plural}")
     def v2_create(ns: str, plural: str, body: Dict[str, Any]):
         kind = KIND_BY_PLURAL.get(plural)
@@ -253,6 +281,8 @@ def create_app(client: Optional[KubeClient] = None,
             raise HTTPException(404, f"resource {plural} is not proxied")
         model = model_for_kind(kind)
         body.setdefault("metadata", {})["namespace"] = ns
+        if kind == "RayCluster":
+            _expand_compute_templates(ns, body)
         obj = model.from_dict(body)
         return client.create(obj).to_dict()
 

@@ -312,3 +312,33 @@ class TestKrayEvents:
         finally:
             climod.make_client = original
             cp.stop()
+
+
+class TestV2ComputeTemplateMiddleware:
+    def test_template_expanded_on_v2_create(self, api):
+        t, client, _ = api
+        body = {
+            "apiVersion": "ray.io/v1", "kind": "RayCluster",
+            "metadata": {"name": "mw1"},
+            "spec": {
+                "headGroupSpec": {"computeTemplate": "tpl",
+                                  "rayStartParams": {}},
+                "workerGroupSpecs": [{
+                    "groupName": "g", "replicas": 1, "maxReplicas": 2,
+                    "computeTemplate": "tpl", "rayStartParams": {}}],
+            },
+        }
+        r = t.post("/apis/ray.io/v1/namespaces/ns1/rayclusters", json=body)
+        assert r.status_code == 200, r.text
+        rc = client.server.get("RayCluster", "ns1", "mw1")
+        limits = rc["spec"]["workerGroupSpecs"][0]["template"]["spec"][
+            "containers"][0]["resources"]["limits"]
+        assert limits["amd.com/gpu"] == "2"
+
+    def test_unknown_template_rejected(self, api):
+        t, _, _ = api
+        body = {"apiVersion": "ray.io/v1", "kind": "RayCluster",
+                "metadata": {"name": "mw2"},
+                "spec": {"headGroupSpec": {"computeTemplate": "missing"}}}
+        r = t.post("/apis/ray.io/v1/namespaces/ns1/rayclusters", json=body)
+        assert r.status_code == 400
