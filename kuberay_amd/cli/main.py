@@ -101,6 +101,60 @@ def get_job(ctx, name):
     _table(["NAME", "DEPLOYMENT STATUS", "JOB STATUS", "CLUSTER"], rows)
 
 
+@get.command("node")
+@click.pass_context
+def get_node(ctx, ):
+    """List nodes with their MI355X capacity and xGMI labels."""
+    client = client_of(ctx)
+    server = getattr(client, "server", None)
+    raw = getattr(client, "raw_list", None)
+    if server is not None:
+        nodes = server.list("Node")
+    elif raw is not None:
+        nodes = raw("Node")
+    else:
+        raise click.ClickException("nodes not available over this server")
+    rows = []
+    for n in nodes:
+        labels = (n.get("metadata") or {}).get("labels") or {}
+        cap = ((n.get("status") or {}).get("capacity") or {})
+        rows.append((n["metadata"]["name"],
+                     cap.get("amd.com/gpu", labels.get("amd.com/gpu.count", "-")),
+                     labels.get("amd.com/xgmi-island", "-"),
+                     labels.get("amd.com/xgmi-fully-connected", "-")))
+    _table(["NODE", "AMD GPUS", "XGMI ISLAND", "FULLY CONNECTED"], rows)
+
+
+@get.command("token")
+@click.argument("cluster_name")
+@click.pass_context
+def get_token(ctx, cluster_name):
+    """Print the auth token of a token-auth RayCluster."""
+    import base64
+    client = client_of(ctx)
+    ns = ctx.obj["namespace"]
+    rc = client.get(RayCluster, ns, cluster_name)
+    from ..common.pod import is_auth_enabled
+    if not is_auth_enabled(rc.spec):
+        raise click.ClickException(f"cluster {cluster_name} does not use token auth")
+    from ..utils import names as _names
+    secret_name = (rc.spec.auth_options.secret_name
+                   if rc.spec.auth_options and rc.spec.auth_options.secret_name
+                   else _names.auth_secret_name(cluster_name))
+    server = getattr(client, "server", None)
+    raw = getattr(client, "raw_try_get", None)
+    if server is not None:
+        secret = server.try_get("Secret", ns, secret_name)
+    elif raw is not None:
+        secret = raw("Secret", ns, secret_name)
+    else:
+        raise click.ClickException("secrets not available over this server")
+    if secret is None:
+        raise click.ClickException(f"auth secret {secret_name} not found")
+    token = base64.b64decode((secret.get("data") or {}).get("auth_token", "")).decode()
+    click.echo(token)
+
+
 @get.command("events")
 @click.argument("name", required=False)
 @click.pass_context
