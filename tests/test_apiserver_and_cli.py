@@ -342,3 +342,28 @@ class TestV2ComputeTemplateMiddleware:
                 "spec": {"headGroupSpec": {"computeTemplate": "missing"}}}
         r = t.post("/apis/ray.io/v1/namespaces/ns1/rayclusters", json=body)
         assert r.status_code == 400
+
+
+class TestKrayNodeToken:
+    def test_get_node_and_token(self):
+        import importlib
+        from kuberay_amd.testing import ControlPlane, simple_raycluster
+        climod = importlib.import_module("kuberay_amd.cli.main")
+        cp = ControlPlane(kubelet_delay=0.01, poll_seconds=0.05).start()
+        original = climod.make_client
+        try:
+            climod.make_client = lambda server: cp.client
+            cp.server.create({"kind": "Node", "metadata": {
+                "name": "n1", "labels": {"amd.com/gpu.count": "8",
+                                         "amd.com/xgmi-island": "n1-island0"}}})
+            r = CliRunner().invoke(cli, ["get", "node"])
+            assert r.exit_code == 0 and "n1-island0" in r.output
+            cp.client.create(simple_raycluster("authy",
+                                               authOptions={"mode": "token"}))
+            assert cp.wait_for(lambda: cp.server.try_get(
+                "Secret", "default", "authy-auth-token"))
+            r = CliRunner().invoke(cli, ["get", "token", "authy"])
+            assert r.exit_code == 0 and len(r.output.strip()) == 64
+        finally:
+            climod.make_client = original
+            cp.stop()
