@@ -39,6 +39,13 @@ DASHBOARD_HTML = """<!doctype html>
 </tr></thead><tbody></tbody></table>
 
 <h2>RayJobs</h2>
+<p>
+  <input id="jname" placeholder="job name">
+  <input id="entrypoint" placeholder="entrypoint (python train.py)" size="32">
+  <input id="jgpus" type="number" value="1" min="0" max="8" style="width:4rem"
+         title="amd.com/gpu per worker">
+  <button onclick="submitJob()">submit job</button>
+</p>
 <table id="jobs"><thead><tr>
   <th>name</th><th>deployment status</th><th>job status</th><th>cluster</th><th></th>
 </tr></thead><tbody></tbody></table>
@@ -109,6 +116,24 @@ async function createCluster() {
         minReplicas: 0, maxReplicas: Math.max(workers, 8)}]}})});
   refresh();
 }
+async function submitJob() {
+  const name = document.getElementById('jname').value;
+  const entrypoint = document.getElementById('entrypoint').value;
+  if (!name || !entrypoint) return alert('name and entrypoint required');
+  const gpus = +document.getElementById('jgpus').value;
+  await fetch(`/apis/v1/namespaces/${ns()}/compute_templates`, {
+    method: 'POST', headers: {'Content-Type': 'application/json'},
+    body: JSON.stringify({name: `${name}-tpl`, cpu: 4, memory: 8, gpu: gpus})});
+  await fetch(`/apis/v1/namespaces/${ns()}/jobs`, {
+    method: 'POST', headers: {'Content-Type': 'application/json'},
+    body: JSON.stringify({name, entrypoint, clusterSpec: {
+      headGroupSpec: {computeTemplate: `${name}-tpl`},
+      workerGroupSpec: [{groupName: 'default-group',
+        computeTemplate: `${name}-tpl`, replicas: 1, minReplicas: 0,
+        maxReplicas: 4}]}})});
+  refresh();
+}
+
 refresh();
 setInterval(refresh, 5000);
 </script>
