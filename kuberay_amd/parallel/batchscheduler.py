@@ -19,7 +19,7 @@ from typing import Dict, Optional
 
 from ..kube import objects as k8s
 from ..kube.client import KubeClient
-from ..kube.store import AlreadyExistsError
+from ..kube.store import AlreadyExistsError, ApiError
 from ..models.raycluster import RayCluster
 from ..utils import constants as C
 from ..utils import names
@@ -231,11 +231,199 @@ class XgmiGangScheduler(BatchScheduler):
         pod.spec.affinity = affinity
 
 
+class SchedulingRetry(RuntimeError):
+    """Raised when immutable scheduling resources were deleted this pass and
+    must be recreated on a later reconcile (the workqueue's backoff retries)."""
+
+
+class KubernetesWASBatchScheduler(BatchScheduler):
+    """Kubernetes workload-aware scheduling (scheduling.k8s.io/v1alpha2).
+
+    Behavioral analog of batchscheduler/kubernetes-was (scheduler.go:1-81,
+    v1alpha2/kubernetes_was_v1alpha2.go): the whole RayCluster (head + every
+    worker group) is gang scheduled as ONE PodGroup driven by a Workload
+    with a single PodGroupTemplate named ``cluster``. Opt-in per cluster via
+    the ``ray.io/gang-scheduling-enabled: "true"`` label; skipped (and any
+    stale resources cleaned up) when the label is absent or in-tree
+    autoscaling is on. v1alpha2 scheduling resources are immutable, so a
+    minCount change deletes PodGroup→Workload in dependency order and
+    raises :class:`SchedulingRetry` so a later reconcile recreates them.
+    """
+
+    name = "kubernetes-was-v1alpha2"
+    PROTECTION_FINALIZER = "scheduling.k8s.io/podgroup-protection"
+    GROUP_VERSION = "scheduling.k8s.io/v1alpha2"
+    TEMPLATE_NAME = "cluster"
+
+    # -- naming -----------------------------------------------------------
+    def _workload_name(self, cluster) -> str:
+        return cluster.metadata.name
+
+    def _pod_group_name(self, cluster) -> str:
+        return names.check_name_63(f"{cluster.metadata.name}-{self.TEMPLATE_NAME}")
+
+    # -- gating -----------------------------------------------------------
+    def _skip_reason(self, cluster: RayCluster) -> Optional[str]:
+        labels = cluster.metadata.labels or {}
+        if labels.get(C.RAY_GANG_SCHEDULING_ENABLED, "").lower() != "true":
+            return "gang scheduling not enabled on RayCluster"
+        if cluster.spec.enable_in_tree_autoscaling:
+            return "autoscaling is not yet supported"
+        return None
+
+    @staticmethod
+    def _owned(obj: Optional[dict], cluster: RayCluster) -> bool:
+        if obj is None:
+            return False
+        owners = (obj.get("metadata") or {}).get("ownerReferences") or []
+        return any(o.get("uid") == cluster.metadata.uid and o.get("controller")
+                   for o in owners)
+
+    # -- desired state ----------------------------------------------------
+    def _gang_policy(self, cluster: RayCluster) -> dict:
+        return {"gang": {"minCount": _min_member(cluster)}}
+
+    def _build(self, cluster: RayCluster):
+        ns = cluster.metadata.namespace or "default"
+        meta = {
+            "namespace": ns,
+            "labels": {C.RAY_CLUSTER_LABEL_KEY: cluster.metadata.name},
+            "ownerReferences": [k8s.owner_reference_for(cluster).to_dict()],
+        }
+        policy = self._gang_policy(cluster)
+        workload = {
+            "apiVersion": self.GROUP_VERSION, "kind": "Workload",
+            "metadata": {**meta, "name": self._workload_name(cluster)},
+            "spec": {
+                "controllerRef": {"apiGroup": C.GROUP,
+                                  "kind": "RayCluster",
+                                  "name": cluster.metadata.name},
+                "podGroupTemplates": [{"name": self.TEMPLATE_NAME,
+                                       "schedulingPolicy": policy}],
+            },
+        }
+        pod_group = {
+            "apiVersion": self.GROUP_VERSION, "kind": "PodGroup",
+            "metadata": {**meta, "name": self._pod_group_name(cluster)},
+            "spec": {
+                "podGroupTemplateRef": {"workload": {
+                    "workloadName": self._workload_name(cluster),
+                    "podGroupTemplateName": self.TEMPLATE_NAME}},
+                "schedulingPolicy": policy,
+            },
+        }
+        return workload, pod_group
+
+    # -- immutable-resource sync ------------------------------------------
+    @staticmethod
+    def _min_count_of(obj: Optional[dict], *path) -> Optional[int]:
+        node = (obj or {}).get("spec") or {}
+        for key in path:
+            if isinstance(key, int):
+                node = node[key] if isinstance(node, list) and \
+                    len(node) > key else None
+            else:
+                node = node.get(key) if isinstance(node, dict) else None
+            if node is None:
+                return None
+        return (node.get("gang") or {}).get("minCount") \
+            if isinstance(node, dict) else None
+
+    def _delete_pod_group(self, server, pg: dict) -> None:
+        """Strip the kubelet-style protection finalizer, then delete."""
+        meta = pg.setdefault("metadata", {})
+        fins = meta.get("finalizers") or []
+        if self.PROTECTION_FINALIZER in fins:
+            meta["finalizers"] = [f for f in fins
+                                  if f != self.PROTECTION_FINALIZER]
+            try:
+                server.update(pg)
+            except ApiError:
+                pass
+        ns = meta.get("namespace") or "default"
+        try:
+            server.delete("PodGroup", ns, meta["name"])
+        except ApiError:
+            pass
+
+    def do_batch_scheduling_on_submission(self, client, cluster) -> None:
+        server = getattr(client, "server", None)
+        if server is None:
+            return
+        if self._skip_reason(cluster) is not None:
+            self.cleanup_on_completion(client, cluster)
+            return
+        ns = cluster.metadata.namespace or "default"
+        workload, pod_group = self._build(cluster)
+
+        existing = server.try_get("Workload", ns, workload["metadata"]["name"])
+        if existing is None:
+            server.create(workload)
+        elif not self._owned(existing, cluster):
+            raise RuntimeError(
+                f"Workload {ns}/{workload['metadata']['name']} exists and is "
+                "not owned by this RayCluster; rename to avoid the collision")
+        elif (existing["metadata"].get("deletionTimestamp")
+              or self._min_count_of(existing, "podGroupTemplates", 0,
+                                    "schedulingPolicy")
+              != _min_member(cluster)):
+            # stale/terminating Workload: drop dependent PodGroup first
+            pg = server.try_get("PodGroup", ns, pod_group["metadata"]["name"])
+            if self._owned(pg, cluster):
+                self._delete_pod_group(server, pg)
+            try:
+                server.delete("Workload", ns, workload["metadata"]["name"])
+            except ApiError:
+                pass
+            raise SchedulingRetry(
+                f"replaced stale Workload {ns}/{workload['metadata']['name']}"
+                "; retrying after deletion completes")
+
+        pg = server.try_get("PodGroup", ns, pod_group["metadata"]["name"])
+        if pg is None:
+            server.create(pod_group)
+        elif not self._owned(pg, cluster):
+            raise RuntimeError(
+                f"PodGroup {ns}/{pod_group['metadata']['name']} exists and is "
+                "not owned by this RayCluster; rename to avoid the collision")
+        elif (pg["metadata"].get("deletionTimestamp")
+              or self._min_count_of(pg, "schedulingPolicy")
+              != _min_member(cluster)):
+            self._delete_pod_group(server, pg)
+            raise SchedulingRetry(
+                f"replaced stale PodGroup {ns}/{pod_group['metadata']['name']}"
+                "; retrying after deletion completes")
+
+    def add_metadata_to_pod(self, client, cluster, group_name, pod) -> None:
+        if self._skip_reason(cluster) is not None:
+            return
+        # gang members go to the default scheduler, which understands
+        # spec.schedulingGroup in WAS-enabled clusters
+        pod.spec.scheduler_name = "default-scheduler"
+        pod.spec.schedulingGroup = {"podGroupName": self._pod_group_name(cluster)}
+
+    def cleanup_on_completion(self, client, cluster) -> None:
+        server = getattr(client, "server", None)
+        if server is None:
+            return
+        ns = cluster.metadata.namespace or "default"
+        pg = server.try_get("PodGroup", ns, self._pod_group_name(cluster))
+        if self._owned(pg, cluster):
+            self._delete_pod_group(server, pg)
+        wl = server.try_get("Workload", ns, self._workload_name(cluster))
+        if self._owned(wl, cluster):
+            try:
+                server.delete("Workload", ns, self._workload_name(cluster))
+            except ApiError:
+                pass
+
+
 SCHEDULERS = {
     VolcanoBatchScheduler.name: VolcanoBatchScheduler,
     YunikornBatchScheduler.name: YunikornBatchScheduler,
     SchedulerPluginsBatchScheduler.name: SchedulerPluginsBatchScheduler,
     KaiBatchScheduler.name: KaiBatchScheduler,
+    KubernetesWASBatchScheduler.name: KubernetesWASBatchScheduler,
     XgmiGangScheduler.name: XgmiGangScheduler,
 }
 

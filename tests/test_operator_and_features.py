@@ -244,3 +244,123 @@ class TestKaiScheduler:
                                                 "default-group", pod)
         assert pod.spec.scheduler_name == "kai-scheduler"
         assert pod.metadata.labels["kai.scheduler/queue"] == "team-a"
+
+
+class TestKubernetesWASScheduler:
+    """Parity: batchscheduler/kubernetes-was/v1alpha2 (Workload + PodGroup
+    gang scheduling the whole cluster, immutable-resource replacement)."""
+
+    def _cluster(self, workers=3):
+        cluster = simple_raycluster("was-demo", workers=workers,
+                                    gpus_per_worker=1)
+        cluster.metadata.labels = {"ray.io/gang-scheduling-enabled": "true"}
+        return cluster
+
+    def _sched(self):
+        from kuberay_amd.parallel.batchscheduler import (
+            KubernetesWASBatchScheduler)
+        return KubernetesWASBatchScheduler()
+
+    def test_creates_workload_and_podgroup(self):
+        client = InMemoryClient()
+        cluster = self._cluster(workers=3)
+        cluster = client.create(cluster)
+        sched = self._sched()
+        sched.do_batch_scheduling_on_submission(client, cluster)
+        wl = client.server.get("Workload", "default", "was-demo")
+        tpl = wl["spec"]["podGroupTemplates"][0]
+        assert tpl["name"] == "cluster"
+        assert tpl["schedulingPolicy"]["gang"]["minCount"] == 4  # head+3
+        pg = client.server.get("PodGroup", "default", "was-demo-cluster")
+        assert pg["spec"]["podGroupTemplateRef"]["workload"] == {
+            "workloadName": "was-demo", "podGroupTemplateName": "cluster"}
+        assert pg["spec"]["schedulingPolicy"]["gang"]["minCount"] == 4
+        # owned by the cluster for GC
+        assert wl["metadata"]["ownerReferences"][0]["uid"] == \
+            cluster.metadata.uid
+        # idempotent second pass
+        sched.do_batch_scheduling_on_submission(client, cluster)
+        assert client.server.count("Workload") == 1
+
+    def test_pod_metadata_sets_scheduling_group(self):
+        client = InMemoryClient()
+        cluster = self._cluster()
+        cluster = client.create(cluster)
+        sched = self._sched()
+        group = cluster.spec.worker_group_specs[0]
+        from kuberay_amd.common import pod as podlib
+        fqdn = "x.default.svc.cluster.local"
+        t = podlib.default_worker_pod_template(cluster, group, "p-", fqdn,
+                                               "6379")
+        pod = podlib.build_pod(t, "worker", group.ray_start_params, "6379",
+                               False, None, fqdn)
+        sched.add_metadata_to_pod(client, cluster, "default-group", pod)
+        assert pod.spec.scheduler_name == "default-scheduler"
+        assert pod.to_dict()["spec"]["schedulingGroup"] == {
+            "podGroupName": "was-demo-cluster"}
+
+    def test_min_count_change_replaces_immutable_resources(self):
+        from kuberay_amd.parallel.batchscheduler import SchedulingRetry
+        client = InMemoryClient()
+        cluster = self._cluster(workers=3)
+        cluster = client.create(cluster)
+        sched = self._sched()
+        sched.do_batch_scheduling_on_submission(client, cluster)
+        # scale: minCount 4 -> 6; immutable resources must be recreated
+        cluster.spec.worker_group_specs[0].replicas = 5
+        client.update(cluster)
+        with pytest.raises(SchedulingRetry):
+            sched.do_batch_scheduling_on_submission(client, cluster)
+        assert client.server.try_get("Workload", "default", "was-demo") is None
+        # next reconcile recreates both at the new size
+        sched.do_batch_scheduling_on_submission(client, cluster)
+        pg = client.server.get("PodGroup", "default", "was-demo-cluster")
+        assert pg["spec"]["schedulingPolicy"]["gang"]["minCount"] == 6
+
+    def test_protection_finalizer_removed_before_delete(self):
+        client = InMemoryClient()
+        cluster = self._cluster()
+        cluster = client.create(cluster)
+        sched = self._sched()
+        sched.do_batch_scheduling_on_submission(client, cluster)
+        pg = client.server.get("PodGroup", "default", "was-demo-cluster")
+        pg["metadata"]["finalizers"] = ["scheduling.k8s.io/podgroup-protection"]
+        client.server.update(pg)
+        sched.cleanup_on_completion(client, cluster)
+        assert client.server.try_get("PodGroup", "default",
+                                     "was-demo-cluster") is None
+        assert client.server.try_get("Workload", "default", "was-demo") is None
+
+    def test_skips_and_cleans_up_without_gang_label(self):
+        client = InMemoryClient()
+        cluster = self._cluster()
+        cluster = client.create(cluster)
+        sched = self._sched()
+        sched.do_batch_scheduling_on_submission(client, cluster)
+        assert client.server.count("Workload") == 1
+        # label removed -> skip + stale resources cleaned up
+        cluster.metadata.labels = {}
+        client.update(cluster)
+        sched.do_batch_scheduling_on_submission(client, cluster)
+        assert client.server.count("Workload") == 0
+        assert client.server.count("PodGroup") == 0
+
+    def test_skips_when_autoscaling_enabled(self):
+        client = InMemoryClient()
+        cluster = self._cluster()
+        cluster.spec.enable_in_tree_autoscaling = True
+        cluster = client.create(cluster)
+        sched = self._sched()
+        sched.do_batch_scheduling_on_submission(client, cluster)
+        assert client.server.count("Workload") == 0
+
+    def test_foreign_workload_collision_fails_loudly(self):
+        client = InMemoryClient()
+        cluster = self._cluster()
+        cluster = client.create(cluster)
+        client.server.create({"apiVersion": "scheduling.k8s.io/v1alpha2",
+                              "kind": "Workload",
+                              "metadata": {"name": "was-demo",
+                                           "namespace": "default"}})
+        with pytest.raises(RuntimeError, match="not owned"):
+            self._sched().do_batch_scheduling_on_submission(client, cluster)
