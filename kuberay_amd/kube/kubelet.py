@@ -46,12 +46,13 @@ class SimKubelet:
         self._watcher = None
 
     # -- lifecycle -----------------------------------------------------
+    WATCHED_KINDS = ("Pod", "Job", "Gateway", "HTTPRoute")
+
     def start(self) -> None:
-        self._watcher = self.server.watch({"Pod", "Job"})
-        for obj in self.server.list("Pod"):
-            self._on_event("ADDED", obj)
-        for obj in self.server.list("Job"):
-            self._on_event("ADDED", obj)
+        self._watcher = self.server.watch(set(self.WATCHED_KINDS))
+        for kind in self.WATCHED_KINDS:
+            for obj in self.server.list(kind):
+                self._on_event("ADDED", obj)
         t1 = threading.Thread(target=self._watch_loop, name="sim-kubelet-watch", daemon=True)
         t2 = threading.Thread(target=self._timer_loop, name="sim-kubelet-timer", daemon=True)
         t1.start(); t2.start()
@@ -109,6 +110,12 @@ class SimKubelet:
             self._schedule(self.startup_delay, lambda: self._start_pod(ns, name))
         elif kind == "Job":
             self._schedule(self.job_runtime, lambda: self._complete_job(ns, name))
+        elif kind == "Gateway":
+            self._schedule(self.startup_delay,
+                           lambda: self._program_gateway(ns, name))
+        elif kind == "HTTPRoute":
+            self._schedule(self.startup_delay,
+                           lambda: self._accept_route(ns, name))
 
     def _start_pod(self, namespace: str, name: str) -> None:
         pod = self.server.try_get("Pod", namespace, name)
@@ -153,6 +160,40 @@ class SimKubelet:
                 if C.AMD_GPU_RESOURCE_NAME in block:
                     return True
         return False
+
+    def _program_gateway(self, namespace: str, name: str) -> None:
+        """Stand-in gateway controller: accept + program every Gateway, so
+        the incremental-upgrade readiness gate (Gateway API conditions) is
+        genuinely exercised in envtest-style runs."""
+        gw = self.server.try_get("Gateway", namespace, name)
+        if gw is None or gw["metadata"].get("deletionTimestamp"):
+            return
+        self.server.patch_merge("Gateway", namespace, name, {
+            "status": {"conditions": [
+                {"type": "Accepted", "status": "True",
+                 "lastTransitionTime": now_iso()},
+                {"type": "Programmed", "status": "True",
+                 "lastTransitionTime": now_iso()},
+            ]}}, subresource="status")
+
+    def _accept_route(self, namespace: str, name: str) -> None:
+        route = self.server.try_get("HTTPRoute", namespace, name)
+        if route is None or route["metadata"].get("deletionTimestamp"):
+            return
+        parents = []
+        for ref in (route.get("spec") or {}).get("parentRefs") or []:
+            parents.append({
+                "parentRef": dict(ref),
+                "controllerName": "kuberay-amd.sim/gateway-controller",
+                "conditions": [
+                    {"type": "Accepted", "status": "True",
+                     "lastTransitionTime": now_iso()},
+                    {"type": "ResolvedRefs", "status": "True",
+                     "lastTransitionTime": now_iso()},
+                ]})
+        self.server.patch_merge("HTTPRoute", namespace, name,
+                                {"status": {"parents": parents}},
+                                subresource="status")
 
     def _complete_job(self, namespace: str, name: str) -> None:
         job = self.server.try_get("Job", namespace, name)

@@ -50,6 +50,40 @@ def upgrade_options(svc: RayService):
     return us.cluster_upgrade_options if us else None
 
 
+def _condition_true(conditions, cond_type: str) -> bool:
+    return any(c.get("type") == cond_type and c.get("status") == "True"
+               for c in conditions or [])
+
+
+def is_gateway_ready(gw: Optional[Dict[str, Any]]) -> bool:
+    """util.go:874-883 — Gateway must be Accepted AND Programmed."""
+    if gw is None:
+        return False
+    conditions = (gw.get("status") or {}).get("conditions")
+    return (_condition_true(conditions, "Accepted")
+            and _condition_true(conditions, "Programmed"))
+
+
+def is_http_route_ready(gw: Optional[Dict[str, Any]],
+                        route: Optional[Dict[str, Any]]) -> bool:
+    """util.go:896-915 — the route's parent-status entry for OUR Gateway
+    must be Accepted with ResolvedRefs before traffic weights advance."""
+    if gw is None or route is None:
+        return False
+    gw_name = (gw.get("metadata") or {}).get("name")
+    gw_ns = (gw.get("metadata") or {}).get("namespace")
+    for parent in (route.get("status") or {}).get("parents") or []:
+        ref = parent.get("parentRef") or {}
+        if ref.get("name") != gw_name:
+            continue
+        if ref.get("namespace") and ref["namespace"] != gw_ns:
+            continue
+        if (_condition_true(parent.get("conditions"), "Accepted")
+                and _condition_true(parent.get("conditions"), "ResolvedRefs")):
+            return True
+    return False
+
+
 class _RawObjects:
     """Dict-object verbs over either backend: the in-memory server directly,
     or a RestClient's raw_* methods (Gateway/HTTPRoute have no typed model)."""
@@ -166,6 +200,14 @@ class IncrementalUpgrader:
                      pending: RayCluster) -> bool:
         """Advance the weighted migration. Returns True when the pending
         cluster carries 100% and the service should promote."""
+        namespace = svc.metadata.namespace or "default"
+        gw = self.raw.try_get("Gateway", namespace, gateway_name(svc))
+        route = self.raw.try_get("HTTPRoute", namespace, route_name(svc))
+        if not is_gateway_ready(gw) or not is_http_route_ready(gw, route):
+            # hold until the gateway controller accepts + programs the route
+            # (rayservice_controller.go:1657-1667)
+            return False
+
         opts = upgrade_options(svc)
         step = (opts.step_size_percent if opts and opts.step_size_percent
                 else 25)

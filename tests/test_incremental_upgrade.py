@@ -137,3 +137,70 @@ class TestIncrementalUpgrade:
             assert plane.server.count("Gateway") == 0  # no gateway infra
         finally:
             plane.stop()
+
+
+class TestGatewayReadinessGate:
+    """Parity: util.go:874-915 + rayservice_controller.go:1657-1667 —
+    traffic weights only advance once the Gateway is Programmed and the
+    HTTPRoute's parent status is Accepted + ResolvedRefs."""
+
+    def test_readiness_helpers(self):
+        from kuberay_amd.ops.incremental import (is_gateway_ready,
+                                                 is_http_route_ready)
+        gw = {"metadata": {"name": "g", "namespace": "default"},
+              "status": {"conditions": [
+                  {"type": "Accepted", "status": "True"},
+                  {"type": "Programmed", "status": "True"}]}}
+        assert is_gateway_ready(gw)
+        assert not is_gateway_ready({"metadata": {"name": "g"},
+                                     "status": {"conditions": [
+                                         {"type": "Accepted",
+                                          "status": "True"}]}})
+        assert not is_gateway_ready(None)
+        route = {"status": {"parents": [{
+            "parentRef": {"name": "g"},
+            "conditions": [{"type": "Accepted", "status": "True"},
+                           {"type": "ResolvedRefs", "status": "True"}]}]}}
+        assert is_http_route_ready(gw, route)
+        # parent entry for a DIFFERENT gateway doesn't count
+        other = {"status": {"parents": [{
+            "parentRef": {"name": "not-ours"},
+            "conditions": [{"type": "Accepted", "status": "True"},
+                           {"type": "ResolvedRefs", "status": "True"}]}]}}
+        assert not is_http_route_ready(gw, other)
+        assert not is_http_route_ready(gw, None)
+
+    def test_step_held_until_route_accepted(self):
+        """Without any gateway controller writing status, weights stay 0."""
+        from kuberay_amd.kube.client import InMemoryClient
+        from kuberay_amd.kube.events import StoreRecorder
+        from kuberay_amd.ops.incremental import IncrementalUpgrader
+        client = InMemoryClient()
+        svc = make_service()
+        svc = client.create(svc)
+        active = client.create(simple_raycluster("active-c", workers=1))
+        pending = client.create(simple_raycluster("pending-c", workers=1))
+        up = IncrementalUpgrader(client, StoreRecorder(client.server))
+        up.ensure_gateway_infra(svc, active, pending)
+        # no sim kubelet: Gateway/HTTPRoute status never set -> hold
+        assert up.step_traffic(svc, active, pending) is False
+        assert (svc.status.pending_service_status.traffic_routed_percent
+                or 0) == 0
+        # hand-program the gateway + route like a gateway controller would
+        ns = "default"
+        client.server.patch_merge(
+            "Gateway", ns, f"{svc.metadata.name}-gateway",
+            {"status": {"conditions": [
+                {"type": "Accepted", "status": "True"},
+                {"type": "Programmed", "status": "True"}]}},
+            subresource="status")
+        client.server.patch_merge(
+            "HTTPRoute", ns, f"{svc.metadata.name}-route",
+            {"status": {"parents": [{
+                "parentRef": {"name": f"{svc.metadata.name}-gateway"},
+                "conditions": [
+                    {"type": "Accepted", "status": "True"},
+                    {"type": "ResolvedRefs", "status": "True"}]}]}},
+            subresource="status")
+        assert up.step_traffic(svc, active, pending) is False  # 50%
+        assert svc.status.pending_service_status.traffic_routed_percent == 50
