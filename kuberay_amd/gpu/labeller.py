@@ -13,11 +13,11 @@ from __future__ import annotations
 import os
 from typing import Dict, Optional
 
-from ..parallel.batchscheduler import XGMI_ISLAND_NODE_LABEL
+from ..utils.constants import (AMD_GPU_COUNT_LABEL as GPU_COUNT_LABEL,
+                               XGMI_FULLY_CONNECTED_LABEL,
+                               XGMI_ISLAND_NODE_LABEL,
+                               XGMI_LARGEST_ISLAND_LABEL)
 from . import topology
-
-GPU_COUNT_LABEL = "amd.com/gpu.count"
-XGMI_FULLY_CONNECTED_LABEL = "amd.com/xgmi-fully-connected"
 
 
 def compute_node_labels(topo: Optional[topology.XgmiTopology],
@@ -33,7 +33,7 @@ def compute_node_labels(topo: Optional[topology.XgmiTopology],
         XGMI_ISLAND_NODE_LABEL: f"{node_name}-island0",
         GPU_COUNT_LABEL: str(topo.num_gpus),
         XGMI_FULLY_CONNECTED_LABEL: "true" if topo.fully_connected() else "false",
-        "amd.com/xgmi-largest-island": str(largest),
+        XGMI_LARGEST_ISLAND_LABEL: str(largest),
     }
 
 

@@ -27,7 +27,7 @@ from ..utils.quantity import add_quantities, format_quantity, parse_quantity
 from ..utils.resources import pod_gpu_count, worker_group_desired_replicas
 
 # node label published by the node labeller / our topology discovery
-XGMI_ISLAND_NODE_LABEL = "amd.com/xgmi-island"
+XGMI_ISLAND_NODE_LABEL = C.XGMI_ISLAND_NODE_LABEL
 POD_GROUP_ANNOTATION = "scheduling.k8s.io/group-name"
 VOLCANO_POD_GROUP_ANNOTATION = "scheduling.k8s.io/group-name"
 VOLCANO_QUEUE_NAME_LABEL = "volcano.sh/queue-name"
@@ -209,6 +209,54 @@ class XgmiGangScheduler(BatchScheduler):
         if any(True for _ in self._gpu_groups(cluster)):
             self.inner.do_batch_scheduling_on_submission(client, cluster)
 
+    # -- island scoring ----------------------------------------------------
+    @staticmethod
+    def _island_capacities(client) -> Dict[str, int]:
+        """Per-island usable GPU capacity from the node labeller's labels.
+
+        A split-xGMI node (``amd.com/xgmi-fully-connected: "false"``) only
+        contributes its LARGEST fully-connected island: a gang ring spanning
+        the split would cross PCIe, exactly what this scheduler exists to
+        avoid.
+        """
+        server = getattr(client, "server", None)
+        if server is None:
+            return {}
+        capacities: Dict[str, int] = {}
+        for node in server.list("Node"):
+            labels = (node.get("metadata") or {}).get("labels") or {}
+            island = labels.get(C.XGMI_ISLAND_NODE_LABEL)
+            if not island:
+                continue
+            try:
+                count = int(labels.get(C.AMD_GPU_COUNT_LABEL) or 0)
+                if labels.get(C.XGMI_FULLY_CONNECTED_LABEL) == "false":
+                    count = int(labels.get(C.XGMI_LARGEST_ISLAND_LABEL)
+                                or count)
+            except ValueError:
+                continue
+            capacities[island] = capacities.get(island, 0) + count
+        return capacities
+
+    def _best_island(self, client, cluster: RayCluster,
+                     group_name: str) -> Optional[str]:
+        """Best-fit: the smallest island whose usable GPU capacity covers the
+        whole gang's demand, minimizing fragmentation of big islands."""
+        group = next((g for g in cluster.spec.worker_group_specs
+                      if g.group_name == group_name), None)
+        if group is None:
+            return None
+        try:
+            demand = pod_gpu_count(group.template) * \
+                worker_group_desired_replicas(group) * max(group.num_of_hosts, 1)
+        except (IndexError, AttributeError):
+            return None
+        if demand <= 0:
+            return None
+        fits = [(cap, name) for name, cap in
+                self._island_capacities(client).items() if cap >= demand]
+        return min(fits)[1] if fits else None
+
     def add_metadata_to_pod(self, client, cluster, group_name, pod) -> None:
         is_gpu = pod_gpu_count(pod) > 0
         if not (is_gpu or self.gang_cpu_groups):
@@ -228,6 +276,20 @@ class XgmiGangScheduler(BatchScheduler):
                 "topologyKey": XGMI_ISLAND_NODE_LABEL,
             },
         })
+        # Island scoring: when labelled Nodes are visible, pin the gang to the
+        # best-fit island outright (the kube scheduler then only bin-packs
+        # within it); without Node data the preferred affinity above still
+        # keeps gang members together.
+        island = self._best_island(client, cluster, group_name)
+        if island is not None:
+            node_affinity = affinity.setdefault("nodeAffinity", {})
+            required = node_affinity.setdefault(
+                "requiredDuringSchedulingIgnoredDuringExecution",
+                {"nodeSelectorTerms": []})
+            required["nodeSelectorTerms"].append({
+                "matchExpressions": [{"key": XGMI_ISLAND_NODE_LABEL,
+                                      "operator": "In",
+                                      "values": [island]}]})
         pod.spec.affinity = affinity
 
 

@@ -206,6 +206,12 @@ RAY_CONTAINER_INDEX = 0
 #   common/pod.go:40-49      (custom accelerators) -> dropped
 # ---------------------------------------------------------------------------
 AMD_GPU_RESOURCE_NAME = "amd.com/gpu"
+# node labels published by kuberay_amd.gpu.labeller, consumed by the
+# XgmiGangScheduler for topology-aware gang placement
+XGMI_ISLAND_NODE_LABEL = "amd.com/xgmi-island"
+AMD_GPU_COUNT_LABEL = "amd.com/gpu.count"
+XGMI_FULLY_CONNECTED_LABEL = "amd.com/xgmi-fully-connected"
+XGMI_LARGEST_ISLAND_LABEL = "amd.com/xgmi-largest-island"
 
 # Device nodes every ROCm container needs (AMD k8s device plugin mounts these
 # automatically when amd.com/gpu is requested; we also support explicit

@@ -364,3 +364,75 @@ class TestKubernetesWASScheduler:
                                            "namespace": "default"}})
         with pytest.raises(RuntimeError, match="not owned"):
             self._sched().do_batch_scheduling_on_submission(client, cluster)
+
+
+class TestXgmiIslandScoring:
+    """Best-fit island selection from node-labeller labels: gangs pin to the
+    smallest island that fits; split-xGMI nodes only count their largest
+    fully-connected island."""
+
+    def _node(self, name, island, gpus, fully="true", largest=None):
+        n = {"apiVersion": "v1", "kind": "Node",
+             "metadata": {"name": name, "namespace": "default",
+                          "labels": {
+                              "amd.com/xgmi-island": island,
+                              "amd.com/gpu.count": str(gpus),
+                              "amd.com/xgmi-fully-connected": fully}}}
+        if largest is not None:
+            n["metadata"]["labels"]["amd.com/xgmi-largest-island"] = \
+                str(largest)
+        return n
+
+    def test_best_fit_island_pinned(self):
+        client = InMemoryClient()
+        client.server.create(self._node("n1", "n1-island0", 8))
+        client.server.create(self._node("n2", "n2-island0", 4))
+        cluster = simple_raycluster("demo", workers=2, gpus_per_worker=2)
+        cluster = client.create(cluster)
+        sched = XgmiGangScheduler()
+        pod = TestGangScheduling._pod(TestGangScheduling(), cluster)
+        sched.add_metadata_to_pod(client, cluster, "default-group", pod)
+        # demand = 2 pods x 2 GPUs = 4 -> best fit is the 4-GPU island
+        terms = pod.spec.affinity["nodeAffinity"][
+            "requiredDuringSchedulingIgnoredDuringExecution"][
+            "nodeSelectorTerms"]
+        assert terms[0]["matchExpressions"][0] == {
+            "key": "amd.com/xgmi-island", "operator": "In",
+            "values": ["n2-island0"]}
+
+    def test_split_node_counts_largest_island_only(self):
+        client = InMemoryClient()
+        # split node: 8 GPUs but largest xGMI island is 4 -> cannot host a
+        # 6-GPU gang; the healthy 8-GPU node must win
+        client.server.create(self._node("split", "split-island0", 8,
+                                        fully="false", largest=4))
+        client.server.create(self._node("full", "full-island0", 8))
+        cluster = simple_raycluster("demo", workers=3, gpus_per_worker=2)
+        cluster = client.create(cluster)
+        sched = XgmiGangScheduler()
+        pod = TestGangScheduling._pod(TestGangScheduling(), cluster)
+        sched.add_metadata_to_pod(client, cluster, "default-group", pod)
+        terms = pod.spec.affinity["nodeAffinity"][
+            "requiredDuringSchedulingIgnoredDuringExecution"][
+            "nodeSelectorTerms"]
+        assert terms[0]["matchExpressions"][0]["values"] == ["full-island0"]
+
+    def test_no_fit_keeps_preferred_affinity_only(self):
+        client = InMemoryClient()
+        client.server.create(self._node("n1", "n1-island0", 2))
+        cluster = simple_raycluster("demo", workers=4, gpus_per_worker=2)
+        cluster = client.create(cluster)
+        sched = XgmiGangScheduler()
+        pod = TestGangScheduling._pod(TestGangScheduling(), cluster)
+        sched.add_metadata_to_pod(client, cluster, "default-group", pod)
+        assert "nodeAffinity" not in pod.spec.affinity
+        assert pod.spec.affinity["podAffinity"]  # co-location still preferred
+
+    def test_no_nodes_no_pin(self):
+        client = InMemoryClient()
+        cluster = simple_raycluster("demo", workers=2, gpus_per_worker=1)
+        cluster = client.create(cluster)
+        pod = TestGangScheduling._pod(TestGangScheduling(), cluster)
+        XgmiGangScheduler().add_metadata_to_pod(client, cluster,
+                                                "default-group", pod)
+        assert "nodeAffinity" not in pod.spec.affinity
