@@ -387,9 +387,18 @@ def job_submit(ctx, name, entrypoint, image, worker_replicas, worker_gpu,
 # ---------------------------------------------------------------------------
 @cli.command()
 @click.argument("cluster_name")
+@click.option("--forward", "-f", is_flag=True,
+              help="forward dashboard/client/serve ports to localhost and "
+                   "block (Ctrl-C to stop)")
+@click.option("--port", "ports", multiple=True,
+              help="local:remote pair to forward (repeatable; default "
+                   "8265:8265 10001:10001 8000:8000)")
+@click.option("--target", default=None,
+              help="override the forward target host (defaults to the "
+                   "head pod IP from cluster status)")
 @click.pass_context
-def session(ctx, cluster_name):
-    """Print connection endpoints for a cluster (port-forward analog)."""
+def session(ctx, cluster_name, forward, ports, target):
+    """Connect to a cluster: print endpoints, or --forward its ports."""
     client = client_of(ctx)
     rc = client.get(RayCluster, ctx.obj["namespace"], cluster_name)
     head_svc = f"{cluster_name}-head-svc.{ctx.obj['namespace']}.svc"
@@ -397,8 +406,26 @@ def session(ctx, cluster_name):
     click.echo(f"dashboard: http://{head_svc}:8265")
     click.echo(f"client:    ray://{head_svc}:10001")
     click.echo(f"serve:     http://{head_svc}:8000")
-    click.echo("kubectl port-forward "
-               f"svc/{cluster_name}-head-svc 8265:8265 10001:10001 8000:8000")
+    if not forward:
+        click.echo("kubectl port-forward "
+                   f"svc/{cluster_name}-head-svc 8265:8265 10001:10001 "
+                   "8000:8000   (or: kray session --forward)")
+        return
+
+    from .portforward import PortForwarder
+    host = target or rc.status.head.pod_ip
+    if not host:
+        raise click.ClickException(
+            f"cluster {cluster_name} has no head pod IP yet (state="
+            f"{rc.status.state}); pass --target to forward anyway")
+    mappings = []
+    for pair in ports or ("8265:8265", "10001:10001", "8000:8000"):
+        local, _, remote = pair.partition(":")
+        mappings.append((int(local), int(remote or local)))
+    fwd = PortForwarder(host, mappings).start()
+    for (local, remote), bound in zip(mappings, fwd.local_ports):
+        click.echo(f"forwarding 127.0.0.1:{bound} -> {host}:{remote}")
+    fwd.wait()
 
 
 @cli.command()

@@ -367,3 +367,116 @@ class TestKrayNodeToken:
         finally:
             climod.make_client = original
             cp.stop()
+
+
+class TestPortForward:
+    """kray session --forward: user-space TCP splice (kubectl-plugin
+    session analog)."""
+
+    def test_forward_round_trip(self):
+        import socket
+        import threading
+        from kuberay_amd.cli.portforward import PortForwarder
+
+        # a tiny upstream echo server standing in for the Ray dashboard
+        upstream = socket.socket()
+        upstream.bind(("127.0.0.1", 0))
+        upstream.listen(4)
+        uport = upstream.getsockname()[1]
+
+        def echo_once():
+            conn, _ = upstream.accept()
+            data = conn.recv(1024)
+            conn.sendall(b"echo:" + data)
+            conn.close()
+
+        threading.Thread(target=echo_once, daemon=True).start()
+
+        fwd = PortForwarder("127.0.0.1", [(0, uport)]).start()
+        try:
+            local = fwd.local_ports[0]
+            c = socket.create_connection(("127.0.0.1", local), timeout=5)
+            c.sendall(b"ping")
+            assert c.recv(1024) == b"echo:ping"
+            c.close()
+        finally:
+            fwd.stop()
+            upstream.close()
+
+    def test_connection_refused_upstream_is_tolerated(self):
+        import socket
+        from kuberay_amd.cli.portforward import PortForwarder
+        # remote port nobody listens on: local connect succeeds, then closes
+        fwd = PortForwarder("127.0.0.1", [(0, 1)]).start()
+        try:
+            c = socket.create_connection(("127.0.0.1", fwd.local_ports[0]),
+                                         timeout=5)
+            c.settimeout(5)
+            assert c.recv(1024) == b""  # closed cleanly, no hang
+            c.close()
+        finally:
+            fwd.stop()
+
+    def test_session_forward_requires_head_ip(self, kray):
+        runner, client = kray
+        from kuberay_amd.testing import simple_raycluster
+        client.create(_ns(simple_raycluster("pfc1"), "ns1"))
+        r = runner.invoke(cli, ["-n", "ns1", "session", "pfc1", "--forward"])
+        assert r.exit_code != 0
+        assert "no head pod IP" in r.output
+
+    def test_session_forward_uses_head_ip(self, kray):
+        import socket
+        import threading
+        runner, client = kray
+        from kuberay_amd.models import RayCluster
+        from kuberay_amd.testing import simple_raycluster
+        client.create(_ns(simple_raycluster("pfc2"), "ns1"))
+        rc = client.get(RayCluster, "ns1", "pfc2")
+        rc.status.head.pod_ip = "127.0.0.1"
+        client.update_status(rc)
+
+        # upstream "dashboard" on an ephemeral port
+        upstream = socket.socket()
+        upstream.bind(("127.0.0.1", 0))
+        upstream.listen(1)
+        uport = upstream.getsockname()[1]
+
+        def serve_once():
+            conn, _ = upstream.accept()
+            conn.sendall(b"dash")
+            conn.close()
+
+        threading.Thread(target=serve_once, daemon=True).start()
+
+        # patch PortForwarder.wait so the command returns immediately,
+        # keeping the forwarder alive for the assertion below
+        from kuberay_amd.cli import portforward as pfmod
+        held = {}
+        orig_start = pfmod.PortForwarder.start
+
+        def capture_start(self):
+            held["fwd"] = self
+            return orig_start(self)
+
+        pfmod.PortForwarder.start = capture_start
+        pfmod.PortForwarder.wait = lambda self: None
+        try:
+            r = runner.invoke(cli, ["-n", "ns1", "session", "pfc2",
+                                    "--forward", "--port", f"0:{uport}"])
+            assert r.exit_code == 0, r.output
+            assert "forwarding 127.0.0.1:" in r.output
+            local = held["fwd"].local_ports[0]
+            c = socket.create_connection(("127.0.0.1", local), timeout=5)
+            assert c.recv(1024) == b"dash"
+            c.close()
+        finally:
+            held["fwd"].stop()
+            upstream.close()
+            pfmod.PortForwarder.start = orig_start
+            del pfmod.PortForwarder.wait
+
+
+def _ns(obj, namespace):
+    obj.metadata.namespace = namespace
+    return obj
