@@ -3,7 +3,7 @@
 read-only API over stored sessions)."""
 from __future__ import annotations
 
-from typing import Dict, Optional
+from typing import Dict, List, Optional
 
 from fastapi import FastAPI, HTTPException
 
@@ -46,33 +46,158 @@ def create_history_app(storage: Optional[StorageReader] = None) -> FastAPI:
     def _state(ns: str, cluster: str, session: str) -> SessionState:
         return loader.load(ns, cluster, session)
 
+    def _list_options(filter_keys, filter_predicates, filter_values,
+                      limit) -> tuple:
+        """Ray state-API list options (router.go:1814-1896 ParseOptionsFromReq
+        analog): parallel filter_keys/filter_predicates/filter_values triples
+        with '=' / '!=' predicates, plus a result limit."""
+        keys = filter_keys or []
+        preds = filter_predicates or ["="] * len(keys)
+        vals = filter_values or []
+        if not (len(keys) == len(vals) and len(preds) == len(keys)):
+            raise HTTPException(
+                400, "filter_keys, filter_predicates and filter_values "
+                     "must have equal lengths")
+        for p in preds:
+            if p not in ("=", "!="):
+                raise HTTPException(400, f"unsupported predicate {p!r}")
+        return list(zip(keys, preds, vals)), limit
+
+    def _apply_filters(items, filters, limit):
+        """Returns (page, num_after_truncation, num_filtered)."""
+        total = len(items)
+        if filters:
+            def keep(it):
+                for key, pred, val in filters:
+                    have = str(it.get(key, ""))
+                    if (pred == "=") != (have == val):
+                        return False
+                return True
+            items = [it for it in items if keep(it)]
+        filtered = len(items)
+        return items[:limit], total, filtered
+
+    def _envelope(key, page, total, filtered):
+        return {"result": True, "msg": f"{key} fetched.",
+                "data": {key: page,
+                         "num_after_truncation": total,
+                         "num_filtered": filtered}}
+
+    def _listing(key, items, filter_keys, filter_predicates, filter_values,
+                 limit):
+        filters, limit = _list_options(filter_keys, filter_predicates,
+                                       filter_values, limit)
+        page, total, filtered = _apply_filters(items, filters, limit)
+        return _envelope(key, page, total, filtered)
+
+    from fastapi import Query
+
+    _FK = Query(default=None, alias="filter_keys")
+    _FP = Query(default=None, alias="filter_predicates")
+    _FV = Query(default=None, alias="filter_values")
+
     @app.get("/api/sessions/{ns}/{cluster}/{session}/jobs")
-    def jobs(ns: str, cluster: str, session: str):
-        return {"data": {"jobs": list(_state(ns, cluster, session).jobs.values())}}
+    def jobs(ns: str, cluster: str, session: str,
+             filter_keys: Optional[List[str]] = _FK,
+             filter_predicates: Optional[List[str]] = _FP,
+             filter_values: Optional[List[str]] = _FV,
+             limit: int = 10000):
+        return _listing("jobs", list(_state(ns, cluster, session).jobs.values()),
+                        filter_keys, filter_predicates, filter_values, limit)
+
+    @app.get("/api/sessions/{ns}/{cluster}/{session}/jobs/{job_id}")
+    def job_detail(ns: str, cluster: str, session: str, job_id: str):
+        job = _state(ns, cluster, session).jobs.get(job_id)
+        if job is None:
+            raise HTTPException(404, f"job {job_id} not found")
+        return {"result": True, "data": {"detail": job}}
 
     @app.get("/api/sessions/{ns}/{cluster}/{session}/tasks")
-    def tasks(ns: str, cluster: str, session: str):
-        return {"data": {"tasks": list(_state(ns, cluster, session).tasks.values())}}
+    def tasks(ns: str, cluster: str, session: str,
+              filter_keys: Optional[List[str]] = _FK,
+              filter_predicates: Optional[List[str]] = _FP,
+              filter_values: Optional[List[str]] = _FV,
+              limit: int = 10000):
+        return _listing("tasks",
+                        list(_state(ns, cluster, session).tasks.values()),
+                        filter_keys, filter_predicates, filter_values, limit)
+
+    @app.get("/api/sessions/{ns}/{cluster}/{session}/tasks/summarize")
+    def tasks_summarize(ns: str, cluster: str, session: str):
+        """router.go:184-190 — per-function aggregate of task states."""
+        summary: Dict[str, Dict] = {}
+        for t in _state(ns, cluster, session).tasks.values():
+            name = t.get("func_or_class_name") or t.get("name") or "unknown"
+            entry = summary.setdefault(name, {
+                "func_or_class_name": name, "type": t.get("type"),
+                "state_counts": {}})
+            state = t.get("state") or "UNKNOWN"
+            entry["state_counts"][state] = \
+                entry["state_counts"].get(state, 0) + 1
+        return {"result": True, "data": {
+            "summary": sorted(summary.values(),
+                              key=lambda e: e["func_or_class_name"]),
+            "total_tasks": len(_state(ns, cluster, session).tasks)}}
+
+    @app.get("/api/sessions/{ns}/{cluster}/{session}/tasks/{task_id}")
+    def task_detail(ns: str, cluster: str, session: str, task_id: str):
+        task = _state(ns, cluster, session).tasks.get(task_id)
+        if task is None:
+            raise HTTPException(404, f"task {task_id} not found")
+        return {"result": True, "data": {"detail": task}}
 
     @app.get("/api/sessions/{ns}/{cluster}/{session}/actors")
-    def actors(ns: str, cluster: str, session: str):
-        return {"data": {"actors": list(_state(ns, cluster, session).actors.values())}}
+    def actors(ns: str, cluster: str, session: str,
+               filter_keys: Optional[List[str]] = _FK,
+               filter_predicates: Optional[List[str]] = _FP,
+               filter_values: Optional[List[str]] = _FV,
+               limit: int = 10000):
+        return _listing("actors",
+                        list(_state(ns, cluster, session).actors.values()),
+                        filter_keys, filter_predicates, filter_values, limit)
 
     @app.get("/api/sessions/{ns}/{cluster}/{session}/nodes")
     def nodes(ns: str, cluster: str, session: str):
         return {"data": {"nodes": list(_state(ns, cluster, session).nodes.values())}}
 
+    @app.get("/api/sessions/{ns}/{cluster}/{session}/nodes/{node_id}")
+    def node_detail(ns: str, cluster: str, session: str, node_id: str):
+        node = _state(ns, cluster, session).nodes.get(node_id)
+        if node is None:
+            raise HTTPException(404, f"node {node_id} not found")
+        return {"result": True, "data": {"detail": node}}
+
     @app.get("/api/sessions/{ns}/{cluster}/{session}/timeline")
     def timeline(ns: str, cluster: str, session: str):
         return _state(ns, cluster, session).timeline()
 
+    @app.get("/api/sessions/{ns}/{cluster}/{session}/logs")
+    def list_logs(ns: str, cluster: str, session: str):
+        """router.go:144-149 getNodeLogs analog: enumerate stored log files."""
+        prefix = f"{ns}/{cluster}/{session}/logs/"
+        names = sorted(p[len(prefix):].removesuffix(".gz")
+                       for p in storage.list(prefix))
+        return {"result": True, "data": {"logs": names}}
+
     @app.get("/api/sessions/{ns}/{cluster}/{session}/logs/{log_name}")
-    def logs(ns: str, cluster: str, session: str, log_name: str):
+    def logs(ns: str, cluster: str, session: str, log_name: str,
+             lines: int = 0, offset: int = 0):
+        """``lines`` > 0 tails that many lines (Ray /api/v0/logs semantics);
+        ``offset`` skips leading lines first — together they paginate."""
         from .storage import decompress
         path = f"{ns}/{cluster}/{session}/logs/{log_name}.gz"
         if not storage.exists(path):
             raise HTTPException(404, f"log {log_name} not found")
-        return {"logs": decompress(storage.read(path)).decode()}
+        text = decompress(storage.read(path)).decode()
+        total = text.count("\n") + (0 if text.endswith("\n") or not text
+                                    else 1)
+        if offset or lines:
+            split = text.splitlines(keepends=True)
+            split = split[offset:]
+            if lines > 0:
+                split = split[-lines:] if offset == 0 else split[:lines]
+            text = "".join(split)
+        return {"logs": text, "total_lines": total}
 
     @app.get("/healthz")
     def healthz():

@@ -151,3 +151,82 @@ class TestHistoryServerApi:
         r = app.get("/api/sessions/ns1/c1/session-1/logs/raylet.out")
         assert r.json()["logs"] == "hello log"
         assert app.get("/api/sessions/ns1/c1/session-1/logs/nope").status_code == 404
+
+
+class TestHistoryServerListOptions:
+    """Ray state-API list options (router.go getTasks/getTaskSummarize):
+    filter triples, limit, detail endpoints, log listing/pagination."""
+
+    @pytest.fixture()
+    def app(self):
+        storage = MemoryStorage()
+        events = list(EVENTS) + [
+            {"event_type": "TASK_DEFINITION_EVENT",
+             "data": {"task_id": "t2", "name": "g", "job_id": "j1"}},
+            {"event_type": "TASK_LIFECYCLE_EVENT",
+             "data": {"task_id": "t2", "state": "RUNNING",
+                      "timestamp": 106}},
+            {"event_type": "TASK_DEFINITION_EVENT",
+             "data": {"task_id": "t3", "name": "f", "job_id": "j2"}},
+        ]
+        collector = Collector(storage, "c1", namespace="ns1",
+                              fetch_events=lambda: events,
+                              fetch_logs=lambda: {
+                                  "raylet.out": "l1\nl2\nl3\nl4\nl5\n",
+                                  "gcs.out": "g"})
+        collector.push_once()
+        return TestClient(create_history_app(storage))
+
+    BASE = "/api/sessions/ns1/c1/session-1"
+
+    def test_filter_equals(self, app):
+        r = app.get(f"{self.BASE}/tasks", params={
+            "filter_keys": "job_id", "filter_values": "j1"})
+        data = r.json()["data"]
+        assert {t["task_id"] for t in data["tasks"]} == {"t1", "t2"}
+        assert data["num_after_truncation"] == 3
+        assert data["num_filtered"] == 2
+
+    def test_filter_not_equals_and_limit(self, app):
+        r = app.get(f"{self.BASE}/tasks", params=[
+            ("filter_keys", "state"), ("filter_predicates", "!="),
+            ("filter_values", "FINISHED"), ("limit", "1")])
+        data = r.json()["data"]
+        assert len(data["tasks"]) == 1
+        assert data["num_filtered"] == 2  # t2 RUNNING + t3 (no state)
+
+    def test_bad_filter_rejected(self, app):
+        assert app.get(f"{self.BASE}/tasks", params=[
+            ("filter_keys", "a"), ("filter_keys", "b"),
+            ("filter_values", "1")]).status_code == 400
+        assert app.get(f"{self.BASE}/tasks", params={
+            "filter_keys": "a", "filter_predicates": ">",
+            "filter_values": "1"}).status_code == 400
+
+    def test_task_summarize(self, app):
+        r = app.get(f"{self.BASE}/tasks/summarize").json()
+        by_name = {e["func_or_class_name"]: e for e in
+                   r["data"]["summary"]}
+        assert by_name["f"]["state_counts"]["FINISHED"] == 1
+        assert by_name["g"]["state_counts"]["RUNNING"] == 1
+        assert r["data"]["total_tasks"] == 3
+
+    def test_detail_endpoints(self, app):
+        assert app.get(f"{self.BASE}/tasks/t1").json()["data"]["detail"][
+            "task_id"] == "t1"
+        assert app.get(f"{self.BASE}/jobs/j1").json()["data"]["detail"][
+            "job_id"] == "j1"
+        assert app.get(f"{self.BASE}/nodes/n1").json()["data"]["detail"][
+            "node_id"] == "n1"
+        assert app.get(f"{self.BASE}/tasks/zz").status_code == 404
+
+    def test_log_listing_and_pagination(self, app):
+        names = app.get(f"{self.BASE}/logs").json()["data"]["logs"]
+        assert names == ["gcs.out", "raylet.out"]
+        # tail
+        r = app.get(f"{self.BASE}/logs/raylet.out", params={"lines": 2}).json()
+        assert r["logs"] == "l4\nl5\n" and r["total_lines"] == 5
+        # window
+        r = app.get(f"{self.BASE}/logs/raylet.out",
+                    params={"offset": 1, "lines": 2}).json()
+        assert r["logs"] == "l2\nl3\n"
