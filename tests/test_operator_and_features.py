@@ -436,3 +436,66 @@ class TestXgmiIslandScoring:
         XgmiGangScheduler().add_metadata_to_pod(client, cluster,
                                                 "default-group", pod)
         assert "nodeAffinity" not in pod.spec.affinity
+
+
+class TestKubernetesWASEndToEnd:
+    """e2ekuberneteswas analog: WAS scheduler wired into the real RayCluster
+    reconciler; the immutable-replace retry path must converge under the
+    workqueue's backoff."""
+
+    def test_cluster_lifecycle_with_was_scheduling(self):
+        import time as _time
+
+        from kuberay_amd.models import RayCluster
+        from kuberay_amd.parallel.batchscheduler import (
+            KubernetesWASBatchScheduler)
+        from kuberay_amd.testing import ControlPlane
+
+        cp = ControlPlane(kubelet_delay=0.0, poll_seconds=0.05)
+        cp.raycluster_reconciler.batch_scheduler = \
+            KubernetesWASBatchScheduler()
+        cp.start()
+        try:
+            rc = simple_raycluster("wase2e", workers=2, gpus_per_worker=1)
+            rc.metadata.labels = {"ray.io/gang-scheduling-enabled": "true"}
+            cp.client.create(rc)
+
+            def wait(cond, timeout=20):
+                deadline = _time.monotonic() + timeout
+                while _time.monotonic() < deadline:
+                    if cond():
+                        return True
+                    _time.sleep(0.05)
+                return False
+
+            assert wait(lambda: (cp.client.try_get(
+                RayCluster, "default", "wase2e") or RayCluster()
+                ).status.state == "ready")
+            pg = cp.server.try_get("PodGroup", "default", "wase2e-cluster")
+            assert pg["spec"]["schedulingPolicy"]["gang"]["minCount"] == 3
+            # every pod joined the cluster PodGroup via spec.schedulingGroup
+            pods = cp.server.list("Pod", "default")
+            assert pods and all(
+                p["spec"].get("schedulingGroup", {}).get("podGroupName")
+                == "wase2e-cluster" for p in pods)
+
+            # scale up: immutable resources get replaced (retry path), then
+            # the cluster converges at the new size
+            rc = cp.client.get(RayCluster, "default", "wase2e")
+            rc.spec.worker_group_specs[0].replicas = 4
+            cp.client.update(rc)
+            assert wait(lambda: (pg := cp.server.try_get(
+                "PodGroup", "default", "wase2e-cluster")) is not None
+                and pg["spec"]["schedulingPolicy"]["gang"]["minCount"] == 5)
+            assert wait(lambda: cp.client.get(
+                RayCluster, "default", "wase2e"
+                ).status.available_worker_replicas == 4)
+
+            # deletion: owner GC removes Workload + PodGroup with the cluster
+            cp.client.delete(cp.client.get(RayCluster, "default", "wase2e"))
+            assert wait(lambda: cp.server.try_get(
+                "Workload", "default", "wase2e") is None
+                and cp.server.try_get("PodGroup", "default",
+                                      "wase2e-cluster") is None)
+        finally:
+            cp.stop()
