@@ -375,6 +375,22 @@ class InMemoryApiServer:
                 merge(merged, patch)
         return self.update(merged, subresource=subresource)
 
+    def mutate_status(self, kind: str, namespace: str, name: str,
+                      fn) -> Optional[Dict[str, Any]]:
+        """Single-fetch status mutation: ``fn(obj)`` edits ``obj['status']``
+        in place (or returns False to skip the write). One backend
+        fetch/parse instead of the try_get + patch_merge pair — the sim
+        kubelet's pod-start transition is hot enough at 2000-pod bursts for
+        the double parse to show up in profiles. Returns the stored object,
+        or None when skipped/missing."""
+        with self._lock:
+            obj = self._backend.fetch((kind, namespace, name))
+            if obj is None:
+                return None
+            if fn(obj) is False:
+                return None
+        return self.update(obj, subresource="status")
+
     def delete(self, kind: str, namespace: str, name: str) -> None:
         key = (kind, namespace, name)
         with self._lock:
