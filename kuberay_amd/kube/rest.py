@@ -121,7 +121,8 @@ class RestClient(KubeClient):
 
     def __init__(self, kubeconfig: Optional[str] = None,
                  http_client: Optional[httpx.Client] = None,
-                 base_url: Optional[str] = None):
+                 base_url: Optional[str] = None,
+                 metrics=None):
         if http_client is not None:
             self._http = http_client
         else:
@@ -132,6 +133,26 @@ class RestClient(KubeClient):
             else:
                 server, kwargs = _load_kubeconfig(kubeconfig)
             self._http = httpx.Client(base_url=server, timeout=30.0, **kwargs)
+        if metrics is not None:
+            self.instrument(metrics)
+
+    def instrument(self, metrics) -> None:
+        """Attach request-latency observation (client_go_metrics.go:27-76
+        analog: per-verb/status histograms on every kube API call)."""
+        import time as _time
+
+        def on_request(request: httpx.Request) -> None:
+            request.extensions["kuberay_t0"] = _time.monotonic()
+
+        def on_response(response: httpx.Response) -> None:
+            t0 = response.request.extensions.get("kuberay_t0")
+            if t0 is not None:
+                metrics.api_request_duration.labels(
+                    response.request.method,
+                    str(response.status_code)).observe(_time.monotonic() - t0)
+
+        self._http.event_hooks["request"].append(on_request)
+        self._http.event_hooks["response"].append(on_response)
 
     # -- paths ---------------------------------------------------------
     @staticmethod
@@ -299,8 +320,11 @@ class RestApiServerAdapter:
     seeding + a RestClient with an informer-backed pod-view cache."""
 
     def __init__(self, kubeconfig: Optional[str] = None,
-                 rest_client: Optional[RestClient] = None):
+                 rest_client: Optional[RestClient] = None,
+                 metrics=None):
         self._client = rest_client or RestClient(kubeconfig=kubeconfig)
+        if metrics is not None:
+            self._client.instrument(metrics)
         self._view_cache = _PodViewCache()
         self._client.list_pod_views = self._view_cache.list  # type: ignore[attr-defined]
         self._client.pod_cache_contains = self._view_cache.contains  # type: ignore[attr-defined]

@@ -83,6 +83,13 @@ class OperatorMetrics:
             "kuberay_workqueue_depth", "Work queue depth", ["controller"],
             registry=r)
 
+        # client request latency (reference: metrics/client_go_metrics.go
+        # rest-client duration histograms)
+        self.api_request_duration = Histogram(
+            "kuberay_api_request_duration_seconds",
+            "Kube API request latency by HTTP verb and status code",
+            ["verb", "code"], registry=r)
+
         # -- MI355X-native ----------------------------------------------
         self.gpu_utilization = Gauge(
             "kuberay_mi355x_gpu_utilization_pct",

@@ -39,17 +39,18 @@ def build_manager(cfg: Configuration, server=None, client=None):
 
     features.parse_feature_gates(cfg.feature_gates)
 
+    metrics = OperatorMetrics() if cfg.enable_metrics else None
     if server is None:
         if cfg.backend == "memory":
             server = InMemoryApiServer()
         else:
             from .kube.rest import RestApiServerAdapter
-            server = RestApiServerAdapter(kubeconfig=cfg.kubeconfig)
+            server = RestApiServerAdapter(kubeconfig=cfg.kubeconfig,
+                                          metrics=metrics)
     if client is None:
         client = InMemoryClient(server) if isinstance(server, InMemoryApiServer) else server.client()
 
     recorder = StoreRecorder(server) if isinstance(server, InMemoryApiServer) else None
-    metrics = OperatorMetrics() if cfg.enable_metrics else None
 
     scheduler = None
     if cfg.enable_batch_scheduler:

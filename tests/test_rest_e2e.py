@@ -121,3 +121,29 @@ class TestOperatorOverRest:
             kubelet.stop()
             adapter.stop()
             manager.stop()
+
+
+@pytest.mark.timeout(60)
+def test_api_request_metrics_observed():
+    """client_go_metrics analog: REST verbs land in the per-verb/status
+    latency histogram."""
+    from kuberay_amd.kube.httpserver import KubeApiFacade
+    from kuberay_amd.kube.rest import RestClient
+    from kuberay_amd.kube.store import InMemoryApiServer
+    from kuberay_amd.metrics import OperatorMetrics
+    from kuberay_amd.models import RayCluster
+    from kuberay_amd.testing import simple_raycluster
+
+    server = InMemoryApiServer()
+    facade = KubeApiFacade(server, port=0)
+    facade.start()
+    try:
+        metrics = OperatorMetrics()
+        client = RestClient(base_url=facade.url, metrics=metrics)
+        client.create(simple_raycluster("m1"))
+        client.get(RayCluster, "default", "m1")
+        text = metrics.exposition().decode()
+        assert 'kuberay_api_request_duration_seconds_count{code="200",verb="GET"}' in text
+        assert 'verb="POST"' in text
+    finally:
+        facade.stop()
