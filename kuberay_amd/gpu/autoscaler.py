@@ -59,13 +59,16 @@ class MI355XAutoscaler:
                  telemetry: Optional[Callable[[], Dict]] = None,
                  policy: Optional[AutoscalerPolicy] = None,
                  clock: Callable[[], float] = time.monotonic,
-                 recorder=None):
+                 recorder=None, metrics=None, node_name: str = ""):
+        import os
         from .rocm_smi import node_gpu_summary
         self.client = client
         self.telemetry = telemetry or node_gpu_summary
         self.policy = policy or AutoscalerPolicy()
         self.clock = clock
         self.recorder = recorder
+        self.metrics = metrics
+        self.node_name = node_name or os.environ.get("NODE_NAME", "local")
         self._states: Dict[Tuple[str, str, str], GroupState] = {}
         self._stop = threading.Event()
         self._thread: Optional[threading.Thread] = None
@@ -77,7 +80,11 @@ class MI355XAutoscaler:
         decisions: List[str] = []
         try:
             summary = self.telemetry()
+            if self.metrics is not None:
+                self.metrics.observe_gpu_health(self.node_name, True)
         except Exception:
+            if self.metrics is not None:
+                self.metrics.observe_gpu_health(self.node_name, False)
             return decisions
         from ..kube.store import ApiError
         for cluster in self.client.list(RayCluster):

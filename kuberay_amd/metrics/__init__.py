@@ -101,6 +101,11 @@ class OperatorMetrics:
         self.autoscaler_decisions = Counter(
             "kuberay_mi355x_autoscaler_decisions_total",
             "rocm-smi-driven autoscaler decisions", ["direction"], registry=r)
+        self.gpu_health = Gauge(
+            "kuberay_mi355x_gpu_health",
+            "1 when the node's GPU telemetry/health probe succeeds, 0 when "
+            "it fails (feeds the MI355XGpuUnhealthy alert rule)", ["node"],
+            registry=r)
 
     # -- hooks used by the reconcilers ----------------------------------
     def observe_cluster_ready(self, cluster) -> None:
@@ -145,6 +150,9 @@ class OperatorMetrics:
         for s in stats:
             self.gpu_utilization.labels(str(s.index)).set(s.utilization_pct)
             self.gpu_hbm_used_fraction.labels(str(s.index)).set(s.vram_used_fraction)
+
+    def observe_gpu_health(self, node: str, healthy: bool) -> None:
+        self.gpu_health.labels(node).set(1 if healthy else 0)
 
     def exposition(self) -> bytes:
         return generate_latest(self.registry)

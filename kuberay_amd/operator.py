@@ -109,7 +109,8 @@ def build_manager(cfg: Configuration, server=None, client=None):
     autoscaler = None
     if cfg.enable_mi355x_autoscaler and features.enabled("MI355XAutoscaler"):
         from .gpu.autoscaler import MI355XAutoscaler
-        autoscaler = MI355XAutoscaler(client, recorder=recorder)
+        autoscaler = MI355XAutoscaler(client, recorder=recorder,
+                                      metrics=metrics)
 
     return manager, client, metrics, autoscaler
 
