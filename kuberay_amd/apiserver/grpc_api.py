@@ -10,6 +10,7 @@ Services (reference proto/cluster.proto:26, config.proto:26, job.proto:28):
   ClusterService          Create/Get/List/Delete
   ComputeTemplateService  Create/Get/List/Delete
   RayJobService           Create/Get/List/Delete
+  RayServeService         Create/Get/List/Delete (serve.proto:25 analog)
 
 Complex cluster specs travel as a JSON payload field (`spec_json`) — the
 simplified scalar fields match the HTTP v1 surface.
@@ -66,6 +67,12 @@ _msg("RayJob", [("name", "string"), ("namespace", "string"),
                 ("entrypoint", "string"), ("job_status", "string"),
                 ("job_deployment_status", "string"),
                 ("ray_cluster_name", "string"), ("spec_json", "string")])
+_msg("RayServiceMsg", [("name", "string"), ("namespace", "string"),
+                       ("serve_config_v2", "string"),
+                       ("service_status", "string"),
+                       ("active_ray_cluster_name", "string"),
+                       ("pending_ray_cluster_name", "string"),
+                       ("spec_json", "string")])
 _msg("GetRequest", [("name", "string"), ("namespace", "string")])
 _msg("ListRequest", [("namespace", "string")])
 _msg("DeleteRequest", [("name", "string"), ("namespace", "string")])
@@ -74,6 +81,7 @@ _msg("ListClusterResponse", [("clusters", "repeated:Cluster")])
 _msg("ListComputeTemplateResponse",
      [("compute_templates", "repeated:ComputeTemplate")])
 _msg("ListRayJobResponse", [("jobs", "repeated:RayJob")])
+_msg("ListRayServiceResponse", [("services", "repeated:RayServiceMsg")])
 
 _pool = descriptor_pool.DescriptorPool()
 _file_desc = _pool.Add(_FDP)
@@ -94,6 +102,8 @@ Empty = _cls("Empty")
 ListClusterResponse = _cls("ListClusterResponse")
 ListComputeTemplateResponse = _cls("ListComputeTemplateResponse")
 ListRayJobResponse = _cls("ListRayJobResponse")
+RayServiceMsg = _cls("RayServiceMsg")
+ListRayServiceResponse = _cls("ListRayServiceResponse")
 
 
 # ---------------------------------------------------------------------------
@@ -243,6 +253,55 @@ class _Service:
             context.abort(grpc.StatusCode.NOT_FOUND, "not found")
         return Empty()
 
+    # -- services (reference proto/serve.proto RayServeService) ---------
+    def create_ray_service(self, request, context):
+        from . import converters as conv
+        ns = request.namespace or "default"
+        body = {"name": request.name,
+                "serveConfigV2": request.serve_config_v2 or None}
+        if request.spec_json:
+            body.update(json.loads(request.spec_json))
+        svc = conv.api_service_to_rayservice(ns, body, self._templates(ns))
+        created = self.client.create(svc)
+        return self._service_msg(created)
+
+    def _service_msg(self, svc):
+        return RayServiceMsg(
+            name=svc.metadata.name or "",
+            namespace=svc.metadata.namespace or "",
+            serve_config_v2=svc.spec.serve_config_v2 or "",
+            service_status=svc.status.service_status or "",
+            active_ray_cluster_name=
+                svc.status.active_service_status.ray_cluster_name or "",
+            pending_ray_cluster_name=
+                svc.status.pending_service_status.ray_cluster_name or "")
+
+    def get_ray_service(self, request, context):
+        from ..models import RayService
+        svc = self.client.try_get(RayService, request.namespace or "default",
+                                  request.name)
+        if svc is None:
+            context.abort(grpc.StatusCode.NOT_FOUND,
+                          f"service {request.name} not found")
+        return self._service_msg(svc)
+
+    def list_ray_services(self, request, context):
+        from ..models import RayService
+        out = ListRayServiceResponse()
+        for svc in self.client.list(RayService, request.namespace or "default"):
+            out.services.append(self._service_msg(svc))
+        return out
+
+    def delete_ray_service(self, request, context):
+        from ..kube.store import NotFoundError
+        from ..models import RayService
+        try:
+            self.client.delete(RayService, request.namespace or "default",
+                               request.name)
+        except NotFoundError:
+            context.abort(grpc.StatusCode.NOT_FOUND, "not found")
+        return Empty()
+
 
 def _unary(handler, req_cls, resp_cls):
     return grpc.unary_unary_rpc_method_handler(
@@ -272,6 +331,16 @@ def create_grpc_server(client=None, port: int = 8887,
                                           ListComputeTemplateResponse),
             "DeleteComputeTemplate": _unary(svc.delete_compute_template,
                                             DeleteRequest, Empty),
+        },
+        "kuberayamd.v1.RayServeService": {
+            "CreateRayService": _unary(svc.create_ray_service, RayServiceMsg,
+                                       RayServiceMsg),
+            "GetRayService": _unary(svc.get_ray_service, GetRequest,
+                                    RayServiceMsg),
+            "ListRayServices": _unary(svc.list_ray_services, ListRequest,
+                                      ListRayServiceResponse),
+            "DeleteRayService": _unary(svc.delete_ray_service, DeleteRequest,
+                                       Empty),
         },
         "kuberayamd.v1.RayJobService": {
             "CreateRayJob": _unary(svc.create_ray_job, RayJobMsg, RayJobMsg),

@@ -480,3 +480,56 @@ class TestPortForward:
 def _ns(obj, namespace):
     obj.metadata.namespace = namespace
     return obj
+
+
+class TestGrpcRayServeService:
+    """serve.proto analog over the dynamic-proto gRPC server."""
+
+    def test_service_crud(self):
+        import grpc as grpclib
+        from kuberay_amd.apiserver.grpc_api import (
+            DeleteRequest, Empty, GetRequest, ListRequest,
+            ListRayServiceResponse, RayServiceMsg, create_grpc_server)
+        from kuberay_amd.kube.client import InMemoryClient
+        from kuberay_amd.models import RayService
+        client = InMemoryClient()
+        server = create_grpc_server(client, port=0)
+        port = server.add_insecure_port("127.0.0.1:0")
+        server.start()
+        try:
+            channel = grpclib.insecure_channel(f"127.0.0.1:{port}")
+
+            def call(method, request, resp_cls):
+                fn = channel.unary_unary(
+                    f"/kuberayamd.v1.RayServeService/{method}",
+                    request_serializer=lambda m: m.SerializeToString(),
+                    response_deserializer=resp_cls.FromString)
+                return fn(request)
+
+            import json as _json
+            spec = {"rayClusterConfig": {
+                "headGroupSpec": {"rayStartParams": {}, "template": {"spec": {
+                    "containers": [{"name": "ray-head",
+                                    "image": "rocm/ray:2.46.0"}]}}},
+            }}
+            created = call("CreateRayService", RayServiceMsg(
+                name="s1", namespace="default",
+                serve_config_v2="applications: []",
+                spec_json=_json.dumps(spec)), RayServiceMsg)
+            assert created.name == "s1"
+            assert client.try_get(RayService, "default", "s1") is not None
+
+            got = call("GetRayService", GetRequest(name="s1",
+                                                   namespace="default"),
+                       RayServiceMsg)
+            assert got.serve_config_v2 == "applications: []"
+
+            listed = call("ListRayServices", ListRequest(namespace="default"),
+                          ListRayServiceResponse)
+            assert [s.name for s in listed.services] == ["s1"]
+
+            call("DeleteRayService", DeleteRequest(name="s1",
+                                                   namespace="default"), Empty)
+            assert client.try_get(RayService, "default", "s1") is None
+        finally:
+            server.stop(0)
