@@ -160,7 +160,6 @@ def get_token(ctx, cluster_name):
 @click.pass_context
 def get_events(ctx, name):
     """List operator events (optionally for one object)."""
-    import httpx
     client = client_of(ctx)
     ns = ctx.obj["namespace"]
     # events are core objects — fetch through the raw surface when available
