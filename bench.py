@@ -136,6 +136,11 @@ def main() -> int:
     from kuberay_amd.testing import ControlPlane
     kubelet_executors = int(os.environ.get("KUBERAY_BENCH_KUBELET_EXECUTORS",
                                            "1"))
+    switch_interval = os.environ.get("KUBERAY_BENCH_SWITCH_INTERVAL")
+    if switch_interval:
+        # longer GIL switch interval reduces convoying between the
+        # controller workers and the sim kubelet on many-core boxes
+        sys.setswitchinterval(float(switch_interval))
     cp = ControlPlane(kubelet_delay=0.0, workers=workers,
                       record_events=False, requeue_seconds=3600,
                       poll_seconds=1.0, kubelet_executors=kubelet_executors)
