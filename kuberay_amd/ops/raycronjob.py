@@ -50,6 +50,25 @@ class RayCronJobReconciler(Reconciler):
         schedule = parse_cron(cron.spec.schedule)
         now = self.now_fn()
 
+        # spec.timeZone: the schedule's wall-clock fields are interpreted in
+        # that IANA zone (CronJob semantics); stored timestamps stay UTC
+        tz = None
+        if cron.spec.time_zone:
+            from zoneinfo import ZoneInfo
+            tz = ZoneInfo(cron.spec.time_zone)
+
+        def to_local(t_utc: dt.datetime) -> dt.datetime:
+            if tz is None:
+                return t_utc
+            return t_utc.replace(tzinfo=dt.timezone.utc).astimezone(tz) \
+                .replace(tzinfo=None)
+
+        def to_utc(t_local: dt.datetime) -> dt.datetime:
+            if tz is None:
+                return t_local
+            return t_local.replace(tzinfo=tz).astimezone(dt.timezone.utc) \
+                .replace(tzinfo=None)
+
         last = None
         if cron.status.last_schedule_time:
             try:
@@ -61,20 +80,22 @@ class RayCronJobReconciler(Reconciler):
                                               "%Y-%m-%dT%H:%M:%SZ")
                          if cron.metadata.creation_timestamp else now)
 
-        fire = schedule.next_after(basis)
-        if fire is None:
+        fire_local = schedule.next_after(to_local(basis))
+        if fire_local is None:
             return Result()
+        fire = to_utc(fire_local)
         if fire > now:
             return Result(requeue_after=min((fire - now).total_seconds(), 300))
 
         # fire (catch up at most the most recent missed tick, like CronJob
         # with startingDeadline unbounded collapsed to latest)
-        latest = fire
+        latest_local = fire_local
         while True:
-            nxt = schedule.next_after(latest)
-            if nxt is None or nxt > now:
+            nxt = schedule.next_after(latest_local)
+            if nxt is None or to_utc(nxt) > now:
                 break
-            latest = nxt
+            latest_local = nxt
+        latest = to_utc(latest_local)
 
         job_name = f"{cron.metadata.name}-{_minute_hash(latest)}"
         rayjob = RayJob(

@@ -272,6 +272,12 @@ def validate_raycronjob_spec(cronjob: RayCronJob) -> List[str]:
             parse_cron(cronjob.spec.schedule)
         except ValueError as e:
             errs.append(f"invalid schedule: {e}")
+    if cronjob.spec.time_zone:
+        try:
+            from zoneinfo import ZoneInfo
+            ZoneInfo(cronjob.spec.time_zone)
+        except Exception:
+            errs.append(f"invalid timeZone {cronjob.spec.time_zone!r}")
     job = RayJob(metadata=cronjob.metadata, spec=cronjob.spec.job_template)
     errs += validate_rayjob_spec(job)
     return errs

@@ -243,3 +243,60 @@ class TestCronEndToEnd:
         from kuberay_amd.models import RayCronJob as RCJ
         cron = control_plane.client.get(RCJ, "default", "nightly")
         assert cron.status.last_schedule_time  # recorded the fire time
+
+
+class TestCronTimeZone:
+    """spec.timeZone: cron wall-clock fields interpret in the IANA zone
+    (CronJob semantics) while stored timestamps stay UTC."""
+
+    def _reconcile(self, now_utc, tz, last=None):
+        import datetime as dt
+        from kuberay_amd.kube.client import InMemoryClient
+        from kuberay_amd.models import RayCronJob
+        from kuberay_amd.ops.raycronjob import RayCronJobReconciler
+        client = InMemoryClient()
+        cron = client.create(RayCronJob.from_dict({
+            "apiVersion": "ray.io/v1", "kind": "RayCronJob",
+            "metadata": {"name": "tzcron", "namespace": "default"},
+            "spec": {"schedule": "0 9 * * *", "timeZone": tz,
+                     "jobTemplate": make_rayjob().spec.to_dict()}}))
+        if last:
+            cron.status.last_schedule_time = last
+            client.update_status(cron)
+        rec = RayCronJobReconciler(client, now_fn=lambda: now_utc)
+        result = rec.reconcile(("default", "tzcron"))
+        return client, result
+
+    def test_9am_new_york_fires_at_13_utc_in_summer(self):
+        import datetime as dt
+        # EDT (UTC-4): 9:00 America/New_York == 13:00 UTC. At 12:59 UTC the
+        # job must NOT fire; at 13:01 it must.
+        client, result = self._reconcile(
+            dt.datetime(2026, 7, 1, 12, 59), "America/New_York",
+            last="2026-06-30T13:00:00Z")
+        assert client.server.count("RayJob") == 0
+        assert result.requeue_after is not None
+        client, _ = self._reconcile(
+            dt.datetime(2026, 7, 1, 13, 1), "America/New_York",
+            last="2026-06-30T13:00:00Z")
+        assert client.server.count("RayJob") == 1
+        # the recorded lastScheduleTime is the UTC fire instant
+        cron = client.server.get("RayCronJob", "default", "tzcron")
+        assert cron["status"]["lastScheduleTime"] == "2026-07-01T13:00:00Z"
+
+    def test_utc_default_unchanged(self):
+        import datetime as dt
+        client, _ = self._reconcile(dt.datetime(2026, 7, 1, 9, 1), None,
+                                    last="2026-06-30T09:00:00Z")
+        assert client.server.count("RayJob") == 1
+
+    def test_invalid_timezone_rejected(self):
+        from kuberay_amd.models import RayCronJob
+        from kuberay_amd.utils.validation import validate_raycronjob_spec
+        cron = RayCronJob.from_dict({
+            "apiVersion": "ray.io/v1", "kind": "RayCronJob",
+            "metadata": {"name": "bad", "namespace": "default"},
+            "spec": {"schedule": "0 9 * * *", "timeZone": "Mars/Olympus",
+                     "jobTemplate": make_rayjob().spec.to_dict()}})
+        errs = validate_raycronjob_spec(cron)
+        assert any("timeZone" in e for e in errs)
