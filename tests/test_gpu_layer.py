@@ -329,3 +329,88 @@ class TestAutoscalerConflictTolerance:
         client.update = original_update
         clock["t"] = 2.0
         assert any("scale-up" in d for d in a.step())
+
+
+@pytest.mark.gpu
+class TestBaselineConfigsOnDevice:
+    """BASELINE.json configs #3 and #4 exercised against the real MI355X
+    health gate: every GPU pod only turns Ready after the on-device
+    MFMA/HBM probe passes."""
+
+    def _plane(self):
+        from kuberay_amd.gpu.health import sim_kubelet_gpu_gate
+        from kuberay_amd.testing import ControlPlane
+        return ControlPlane(kubelet_delay=0.01, gpu_gate=sim_kubelet_gpu_gate,
+                            poll_seconds=0.05)
+
+    def test_rayjob_ephemeral_8gpu_cluster_auto_deletes(self):
+        """Config #3: RayJob submit -> ephemeral RayCluster with an
+        amd.com/gpu=8 worker group, auto-delete on completion."""
+        from kuberay_amd.models import RayJob
+        from kuberay_amd.testing import simple_raycluster
+        cp = self._plane()
+        cp.start()
+        try:
+            spec = simple_raycluster("x", workers=1,
+                                     gpus_per_worker=8).spec.to_dict()
+            job = RayJob.from_dict({
+                "apiVersion": "ray.io/v1", "kind": "RayJob",
+                "metadata": {"name": "gpujob", "namespace": "default"},
+                "spec": {"entrypoint": "python train.py",
+                         "shutdownAfterJobFinishes": True,
+                         "ttlSecondsAfterFinished": 0,
+                         "rayClusterSpec": spec}})
+            cp.client.create(job)
+            assert cp.wait_for(
+                lambda: cp.client.get(RayJob, "default", "gpujob")
+                .status.job_deployment_status == "Complete", timeout=120)
+            # ephemeral cluster auto-deleted after completion
+            assert cp.wait_for(
+                lambda: not cp.server.list("RayCluster", "default"),
+                timeout=60)
+        finally:
+            cp.stop()
+
+    def test_rayservice_zero_downtime_rollout_two_gpu_groups(self):
+        """Config #4: RayService zero-downtime rollout across two MI355X
+        worker groups (spec change -> pending cluster -> promote)."""
+        from kuberay_amd.models import RayService
+        from kuberay_amd.testing import simple_raycluster
+        cp = self._plane()
+        cp.start()
+        try:
+            cluster_spec = simple_raycluster(
+                "x", workers=1, gpus_per_worker=1).spec.to_dict()
+            cluster_spec["workerGroupSpecs"].append({
+                **cluster_spec["workerGroupSpecs"][0],
+                "groupName": "mi355x-b"})
+            svc = RayService.from_dict({
+                "apiVersion": "ray.io/v1", "kind": "RayService",
+                "metadata": {"name": "gpusvc", "namespace": "default"},
+                "spec": {"serveConfigV2":
+                         "applications:\n- name: a\n  import_path: m.g\n",
+                         "rayClusterConfig": cluster_spec}})
+            cp.client.create(svc)
+
+            def ready():
+                s = cp.client.get(RayService, "default", "gpusvc")
+                return s.status.service_status == "Running" and \
+                    s.status.active_service_status.ray_cluster_name
+            assert cp.wait_for(ready, timeout=120)
+            first = cp.client.get(RayService, "default", "gpusvc") \
+                .status.active_service_status.ray_cluster_name
+
+            # zero-downtime upgrade: worker image change -> new cluster
+            s = cp.client.get(RayService, "default", "gpusvc")
+            s.spec.ray_cluster_spec.worker_group_specs[0].template.spec \
+                .containers[0].image = "rocm/ray:2.47.0"
+            cp.client.update(s)
+
+            def promoted():
+                cur = cp.client.get(RayService, "default", "gpusvc")
+                active = cur.status.active_service_status.ray_cluster_name
+                return active and active != first and \
+                    cur.status.service_status == "Running"
+            assert cp.wait_for(promoted, timeout=120)
+        finally:
+            cp.stop()
