@@ -383,13 +383,19 @@ class InMemoryApiServer:
         kubelet's pod-start transition is hot enough at 2000-pod bursts for
         the double parse to show up in profiles. Returns the stored object,
         or None when skipped/missing."""
-        with self._lock:
-            obj = self._backend.fetch((kind, namespace, name))
-            if obj is None:
-                return None
-            if fn(obj) is False:
-                return None
-        return self.update(obj, subresource="status")
+        for attempt in (0, 1):
+            with self._lock:
+                obj = self._backend.fetch((kind, namespace, name))
+                if obj is None:
+                    return None
+                if fn(obj) is False:
+                    return None
+            try:
+                return self.update(obj, subresource="status")
+            except ConflictError:
+                if attempt:
+                    raise
+        return None  # unreachable; loop always returns or raises
 
     def delete(self, kind: str, namespace: str, name: str) -> None:
         key = (kind, namespace, name)
