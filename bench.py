@@ -135,7 +135,7 @@ def main() -> int:
 
     from kuberay_amd.testing import ControlPlane
     kubelet_executors = int(os.environ.get("KUBERAY_BENCH_KUBELET_EXECUTORS",
-                                           "1"))
+                                           "2"))
     switch_interval = os.environ.get("KUBERAY_BENCH_SWITCH_INTERVAL")
     if switch_interval:
         # longer GIL switch interval reduces convoying between the
