@@ -533,3 +533,45 @@ class TestGrpcRayServeService:
             assert client.try_get(RayService, "default", "s1") is None
         finally:
             server.stop(0)
+
+
+class TestDashboardFormContract:
+    """Pins the exact request shapes the static dashboard's JS emits
+    (createCluster/submitJob in apiserver/dashboard.py)."""
+
+    def test_dashboard_cluster_create_shape(self, api):
+        t, client, _ = api
+        t.post("/apis/v1/namespaces/ns1/compute_templates",
+               json={"name": "ui-tpl", "cpu": 4, "memory": 8, "gpu": 2})
+        r = t.post("/apis/v1/namespaces/ns1/clusters", json={
+            "name": "ui-c1", "version": "2.46.0", "clusterSpec": {
+                "headGroupSpec": {"computeTemplate": "ui-tpl"},
+                "workerGroupSpec": [{"groupName": "default-group",
+                                     "computeTemplate": "ui-tpl",
+                                     "replicas": 2, "minReplicas": 0,
+                                     "maxReplicas": 8}]}})
+        assert r.status_code == 200, r.text
+        rc = client.get(RayCluster, "ns1", "ui-c1")
+        limits = rc.spec.worker_group_specs[0].template.spec.containers[0] \
+            .resources.limits
+        assert limits["amd.com/gpu"] == "2"
+
+    def test_dashboard_job_submit_shape(self, api):
+        t, client, _ = api
+        t.post("/apis/v1/namespaces/ns1/compute_templates",
+               json={"name": "uj-tpl", "cpu": 4, "memory": 8, "gpu": 1})
+        r = t.post("/apis/v1/namespaces/ns1/jobs", json={
+            "name": "ui-j1", "entrypoint": "python train.py",
+            "clusterSpec": {
+                "headGroupSpec": {"computeTemplate": "uj-tpl"},
+                "workerGroupSpec": [{"groupName": "default-group",
+                                     "computeTemplate": "uj-tpl",
+                                     "replicas": 1, "minReplicas": 0,
+                                     "maxReplicas": 4}]}})
+        assert r.status_code == 200, r.text
+        from kuberay_amd.models import RayJob
+        job = client.get(RayJob, "ns1", "ui-j1")
+        assert job.spec.entrypoint == "python train.py"
+        worker = job.spec.ray_cluster_spec.worker_group_specs[0]
+        assert worker.template.spec.containers[0].resources.limits[
+            "amd.com/gpu"] == "1"
