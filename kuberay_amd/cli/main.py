@@ -445,16 +445,34 @@ def suspend(ctx, kind, name, resume):
 
 @cli.command()
 @click.argument("cluster_name")
+@click.option("--pods", is_flag=True,
+              help="also fetch each pod's container logs (REST backend)")
+@click.option("--tail", type=int, default=100, help="lines per pod with --pods")
 @click.pass_context
-def log(ctx, cluster_name):
-    """Show recent events for a cluster (log-download analog)."""
+def log(ctx, cluster_name, pods, tail):
+    """Show cluster state/conditions; --pods downloads pod logs."""
     client = client_of(ctx)
-    rc = client.get(RayCluster, ctx.obj["namespace"], cluster_name)
+    ns = ctx.obj["namespace"]
+    rc = client.get(RayCluster, ns, cluster_name)
     click.echo(f"state={rc.status.state} head={rc.status.head.pod_name} "
                f"ready={rc.status.ready_worker_replicas}/"
                f"{rc.status.desired_worker_replicas}")
     for cond in rc.status.conditions or []:
         click.echo(f"  {cond.type}={cond.status} ({cond.reason})")
+    if not pods:
+        return
+    fetch = getattr(client, "pod_logs", None)
+    if fetch is None:
+        raise click.ClickException(
+            "pod logs need the kubernetes REST backend (this server has no "
+            "kubelet, so there are no container logs to download)")
+    for view in client.list_pod_views(ns, {C.RAY_CLUSTER_LABEL_KEY:
+                                           cluster_name}):
+        click.echo(f"----- {view.name} -----")
+        try:
+            click.echo(fetch(ns, view.name, tail_lines=tail))
+        except Exception as exc:  # keep going across pods
+            click.echo(f"(failed to fetch logs: {exc})")
 
 
 def _table(headers, rows):

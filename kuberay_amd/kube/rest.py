@@ -266,6 +266,19 @@ class RestClient(KubeClient):
             o.setdefault("kind", kind)
         return items
 
+    def pod_logs(self, namespace: str, name: str,
+                 container: Optional[str] = None,
+                 tail_lines: Optional[int] = None) -> str:
+        """GET .../pods/{name}/log (kubectl-plugin log-download analog)."""
+        params: Dict[str, Any] = {}
+        if container:
+            params["container"] = container
+        if tail_lines:
+            params["tailLines"] = str(tail_lines)
+        resp = self._check(self._http.get(
+            self._path("Pod", namespace, name, "log"), params=params))
+        return resp.text
+
     def raw_watch_stream(self, kind: str, resource_version: Optional[str] = None):
         """Generator of (event_type, obj) from a K8s watch request."""
         prefix, plural = RESOURCES[kind]

@@ -169,3 +169,20 @@ class TestKubeconfig:
         server, kwargs = _load_kubeconfig(str(p))
         assert isinstance(kwargs["verify"], str)
         assert open(kwargs["verify"], "rb").read() == b"CERT"
+
+
+def test_pod_logs_via_rest():
+    """kubectl-plugin log-download analog: GET pods/{name}/log."""
+    import httpx
+    from kuberay_amd.kube.rest import RestClient
+
+    def handler(request):
+        if request.url.path.endswith("/pods/p1/log"):
+            assert request.url.params["tailLines"] == "5"
+            return httpx.Response(200, text="line1\nline2\n")
+        return httpx.Response(404, text="not found")
+
+    http = httpx.Client(transport=httpx.MockTransport(handler),
+                        base_url="http://kube")
+    client = RestClient(http_client=http)
+    assert client.pod_logs("ns1", "p1", tail_lines=5) == "line1\nline2\n"
