@@ -428,7 +428,8 @@ def session(ctx, cluster_name, forward, ports, target):
 
 
 @cli.command()
-@click.argument("kind", type=click.Choice(["cluster", "job", "service"]))
+@click.argument("kind", type=click.Choice(["cluster", "job", "service",
+                                           "cronjob"]))
 @click.argument("name")
 @click.option("--resume/--suspend", "resume", default=False)
 @click.pass_context
@@ -436,7 +437,8 @@ def suspend(ctx, kind, name, resume):
     """Suspend (or --resume) a Ray resource."""
     client = client_of(ctx)
     ns = ctx.obj["namespace"]
-    model = {"cluster": RayCluster, "job": RayJob, "service": RayService}[kind]
+    model = {"cluster": RayCluster, "job": RayJob, "service": RayService,
+             "cronjob": RayCronJob}[kind]
     obj = client.get(model, ns, name)
     obj.spec.suspend = not resume
     client.update(obj)
