@@ -70,3 +70,20 @@ def test_helm_clusterrole_covers_operator_surface():
 
 def test_manifest_rbac_covers_operator_surface():
     _check(os.path.join(HERE, "deploy", "manifests", "operator.yaml"))
+
+
+def test_committed_crds_match_generator():
+    """Drift guard (generate-crd-schema.sh analog): deploy/crds must equal
+    what kuberay_amd.crds generates from the pydantic models."""
+    import tempfile
+
+    from kuberay_amd import crds as crdgen
+
+    with tempfile.TemporaryDirectory() as tmp:
+        crdgen.write_crds(tmp)
+        for name in os.listdir(tmp):
+            generated = open(os.path.join(tmp, name)).read()
+            committed_path = os.path.join(HERE, "deploy", "crds", name)
+            assert os.path.exists(committed_path), f"{name} not committed"
+            assert open(committed_path).read() == generated, \
+                f"{name} drifted — run: python -m kuberay_amd.crds deploy/crds"
