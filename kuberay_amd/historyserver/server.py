@@ -3,6 +3,7 @@
 read-only API over stored sessions)."""
 from __future__ import annotations
 
+import json
 from typing import Dict, List, Optional
 
 from fastapi import FastAPI, HTTPException
@@ -166,6 +167,49 @@ def create_history_app(storage: Optional[StorageReader] = None) -> FastAPI:
         if node is None:
             raise HTTPException(404, f"node {node_id} not found")
         return {"result": True, "data": {"detail": node}}
+
+    @app.get("/api/sessions/{ns}/{cluster}/{session}/cluster_status")
+    def cluster_status(ns: str, cluster: str, session: str):
+        """cluster_status.go analog: autoscaler-style summary built from the
+        replayed node/task/actor state."""
+        state = _state(ns, cluster, session)
+        alive, failed = [], []
+        for node in state.nodes.values():
+            (failed if (node.get("state") or "").upper() == "DEAD"
+             else alive).append(node)
+        demands: Dict[str, Dict] = {}
+        pending_states = {"", "PENDING", "PENDING_NODE_ASSIGNMENT",
+                          "PENDING_ARGS_AVAIL", "PENDING_CREATION",
+                          "DEPENDENCIES_UNREADY"}
+        for item in list(state.tasks.values()) + list(state.actors.values()):
+            if (item.get("state") or "").upper() not in pending_states:
+                continue
+            res = item.get("required_resources") or {"CPU": 1}
+            key = json.dumps(res, sort_keys=True)
+            entry = demands.setdefault(key, {"resources": res, "count": 0})
+            entry["count"] += 1
+        lines = ["======== Cluster status (replayed) ========",
+                 f"Active: {len(alive)} node(s)"]
+        for node in failed[:20]:
+            lines.append(f"Failed: {node.get('node_ip') or node['node_id']}")
+        if demands:
+            lines.append("Pending demands:")
+            for d in demands.values():
+                lines.append(f"  {d['resources']}: {d['count']}+")
+        return {"result": True, "data": {"clusterStatus": {
+            "activeNodes": len(alive),
+            "failedNodes": [n["node_id"] for n in failed],
+            "pendingDemands": list(demands.values()),
+            "text": "\n".join(lines)}}}
+
+    @app.get("/api/grafana_health")
+    def grafana_health():
+        """router.go:124 analog — no Grafana bundled in this offline image."""
+        return {"result": False, "msg": "grafana is not configured"}
+
+    @app.get("/api/prometheus_health")
+    def prometheus_health():
+        return {"result": False, "msg": "prometheus is not configured"}
 
     @app.get("/api/sessions/{ns}/{cluster}/{session}/timeline")
     def timeline(ns: str, cluster: str, session: str):

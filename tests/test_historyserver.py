@@ -230,3 +230,40 @@ class TestHistoryServerListOptions:
         r = app.get(f"{self.BASE}/logs/raylet.out",
                     params={"offset": 1, "lines": 2}).json()
         assert r["logs"] == "l2\nl3\n"
+
+
+class TestClusterStatus:
+    """cluster_status.go analog: autoscaler-style replayed summary."""
+
+    @pytest.fixture()
+    def app(self):
+        storage = MemoryStorage()
+        events = list(EVENTS) + [
+            {"event_type": "NODE_DEFINITION_EVENT",
+             "data": {"node_id": "n2", "node_ip": "10.0.0.2"}},
+            {"event_type": "NODE_LIFECYCLE_EVENT",
+             "data": {"node_id": "n2", "state": "DEAD"}},
+            {"event_type": "TASK_DEFINITION_EVENT",
+             "data": {"task_id": "tp", "name": "pending_f", "job_id": "j1"}},
+            {"event_type": "TASK_LIFECYCLE_EVENT",
+             "data": {"task_id": "tp", "state": "PENDING_NODE_ASSIGNMENT"}},
+        ]
+        collector = Collector(storage, "c1", namespace="ns1",
+                              fetch_events=lambda: events,
+                              fetch_logs=lambda: {})
+        collector.push_once()
+        return TestClient(create_history_app(storage))
+
+    def test_status_counts_and_text(self, app):
+        r = app.get("/api/sessions/ns1/c1/session-1/cluster_status").json()
+        cs = r["data"]["clusterStatus"]
+        assert cs["activeNodes"] == 1          # n1 alive, n2 dead
+        assert cs["failedNodes"] == ["n2"]
+        assert cs["pendingDemands"] == [{"resources": {"CPU": 1},
+                                         "count": 1}]
+        assert "Active: 1 node(s)" in cs["text"]
+        assert "10.0.0.2" in cs["text"]
+
+    def test_monitoring_health_stubs(self, app):
+        assert app.get("/api/grafana_health").json()["result"] is False
+        assert app.get("/api/prometheus_health").json()["result"] is False
