@@ -167,6 +167,7 @@ def main():
         errors = sum(c.error_count for c in cp.manager.controllers)
         reconciles = sum(c.reconcile_count for c in cp.manager.controllers)
         rss_end = rss_mb()
+        store_bytes = getattr(cp.server._backend, "total_bytes", lambda: -1)()
         result = {
             "suite": "500-raycluster-soak",
             "minutes": args.minutes,
@@ -179,6 +180,8 @@ def main():
             "rss_mb_baseline": round(rss_baseline, 1),
             "rss_mb_end": round(rss_end, 1),
             "rss_growth_mb": round(rss_end - rss_baseline, 1),
+            "store_bytes_mb": round(store_bytes / 1e6, 2)
+                              if store_bytes >= 0 else None,
         }
         print(json.dumps(result, indent=2))
         here = os.path.dirname(os.path.abspath(__file__))
