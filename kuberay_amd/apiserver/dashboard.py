@@ -51,6 +51,14 @@ DASHBOARD_HTML = """<!doctype html>
 </tr></thead><tbody></tbody></table>
 
 <h2>RayServices</h2>
+<p>
+  <input id="sname" placeholder="service name">
+  <input id="sgpus" type="number" value="1" min="0" max="8" style="width:4rem"
+         title="amd.com/gpu per worker">
+  <button onclick="createService()">create service</button><br>
+  <textarea id="serveconfig" rows="4" cols="60"
+            placeholder="serveConfigV2 YAML (applications: ...)"></textarea>
+</p>
 <table id="services"><thead><tr>
   <th>name</th><th>status</th><th>endpoints</th><th>active cluster</th><th></th>
 </tr></thead><tbody></tbody></table>
@@ -127,6 +135,24 @@ async function submitJob() {
   await fetch(`/apis/v1/namespaces/${ns()}/jobs`, {
     method: 'POST', headers: {'Content-Type': 'application/json'},
     body: JSON.stringify({name, entrypoint, clusterSpec: {
+      headGroupSpec: {computeTemplate: `${name}-tpl`},
+      workerGroupSpec: [{groupName: 'default-group',
+        computeTemplate: `${name}-tpl`, replicas: 1, minReplicas: 0,
+        maxReplicas: 4}]}})});
+  refresh();
+}
+
+async function createService() {
+  const name = document.getElementById('sname').value;
+  const cfg = document.getElementById('serveconfig').value;
+  if (!name || !cfg) return alert('name and serveConfigV2 required');
+  const gpus = +document.getElementById('sgpus').value;
+  await fetch(`/apis/v1/namespaces/${ns()}/compute_templates`, {
+    method: 'POST', headers: {'Content-Type': 'application/json'},
+    body: JSON.stringify({name: `${name}-tpl`, cpu: 4, memory: 8, gpu: gpus})});
+  await fetch(`/apis/v1/namespaces/${ns()}/services`, {
+    method: 'POST', headers: {'Content-Type': 'application/json'},
+    body: JSON.stringify({name, serveConfig_V2: cfg, clusterSpec: {
       headGroupSpec: {computeTemplate: `${name}-tpl`},
       workerGroupSpec: [{groupName: 'default-group',
         computeTemplate: `${name}-tpl`, replicas: 1, minReplicas: 0,

@@ -575,3 +575,22 @@ class TestDashboardFormContract:
         worker = job.spec.ray_cluster_spec.worker_group_specs[0]
         assert worker.template.spec.containers[0].resources.limits[
             "amd.com/gpu"] == "1"
+
+
+def test_dashboard_service_create_shape(api):
+    """Pins the RayService create request shape the dashboard JS emits."""
+    t, client, _ = api
+    t.post("/apis/v1/namespaces/ns1/compute_templates",
+           json={"name": "us-tpl", "cpu": 4, "memory": 8, "gpu": 1})
+    r = t.post("/apis/v1/namespaces/ns1/services", json={
+        "name": "ui-s1",
+        "serveConfig_V2": "applications:\n- name: a\n  import_path: m.g\n",
+        "clusterSpec": {
+            "headGroupSpec": {"computeTemplate": "us-tpl"},
+            "workerGroupSpec": [{"groupName": "default-group",
+                                 "computeTemplate": "us-tpl", "replicas": 1,
+                                 "minReplicas": 0, "maxReplicas": 4}]}})
+    assert r.status_code == 200, r.text
+    from kuberay_amd.models import RayService
+    svc = client.get(RayService, "ns1", "ui-s1")
+    assert svc.spec.serve_config_v2.startswith("applications:")
