@@ -338,3 +338,52 @@ class TestEventRecorder:
         rec.eventf(obj, "Normal", "CreatedWorkerPod", "Created worker Pod %s", "a")
         rec.eventf(obj, "Normal", "CreatedWorkerPod", "Created worker Pod %s", "b")
         assert len(server.list("Event")) == 2
+
+
+class TestPodPool:
+    """Warm pod pool (reference: podpool/ virtual-kubelet warm pods)."""
+
+    TEMPLATE = {"apiVersion": "v1", "kind": "Pod",
+                "metadata": {},
+                "spec": {"containers": [{
+                    "name": "ray-worker", "image": "rocm/ray:2.46.0",
+                    "resources": {"limits": {"amd.com/gpu": "1"}}}]}}
+
+    def _mgr(self):
+        from kuberay_amd.kube.client import InMemoryClient
+        from kuberay_amd.ops.podpool import PodPoolManager
+        client = InMemoryClient()
+        return PodPoolManager(client), client
+
+    def test_reconcile_tops_up_to_target(self):
+        mgr, client = self._mgr()
+        mgr.define_pool("mi355x", self.TEMPLATE, size=3)
+        mgr.reconcile()
+        pods = client.server.list("Pod", "default")
+        assert len(pods) == 3
+        assert all(p["metadata"]["labels"]["ray.io/warm-pod"] == "true"
+                   for p in pods)
+        mgr.reconcile()  # idempotent
+        assert len(client.server.list("Pod", "default")) == 3
+
+    def test_adopt_relabels_and_pool_refills(self):
+        mgr, client = self._mgr()
+        mgr.define_pool("mi355x", self.TEMPLATE, size=2)
+        mgr.reconcile()
+        name = mgr.adopt("mi355x", {"ray.io/cluster": "c1",
+                                    "ray.io/node-type": "worker"})
+        assert name is not None
+        pod = client.server.get("Pod", "default", name)
+        assert pod["metadata"]["labels"]["ray.io/warm-pod"] == "adopted"
+        assert pod["metadata"]["labels"]["ray.io/cluster"] == "c1"
+        # top-up replaces the adopted pod
+        mgr.reconcile()
+        warm = [p for p in client.server.list("Pod", "default")
+                if p["metadata"]["labels"].get("ray.io/warm-pod") == "true"]
+        assert len(warm) == 2
+
+    def test_adopt_from_dry_pool_returns_none(self):
+        mgr, _ = self._mgr()
+        mgr.define_pool("empty", self.TEMPLATE, size=0)
+        mgr.reconcile()
+        assert mgr.adopt("empty", {"ray.io/cluster": "c1"}) is None
