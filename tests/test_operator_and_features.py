@@ -499,3 +499,43 @@ class TestKubernetesWASEndToEnd:
                                       "wase2e-cluster") is None)
         finally:
             cp.stop()
+
+
+class TestIngressAndRoute:
+    """common/ingress.go + common/openshift.go builder behavior."""
+
+    def test_default_ingress_shape(self):
+        from kuberay_amd.common.ingress import build_ingress_for_head_service
+        cluster = simple_raycluster("demo")
+        ing = build_ingress_for_head_service(cluster)
+        rule = ing.spec["rules"][0]
+        path = rule["http"]["paths"][0]
+        assert path["path"] == "/demo/(.*)"
+        assert path["pathType"] == "ImplementationSpecific"
+        assert path["backend"]["service"]["port"]["number"] == 8265
+        assert ing.metadata.annotations[
+            "nginx.ingress.kubernetes.io/rewrite-target"] == "/$1"
+
+    def test_ingress_options_override(self):
+        from kuberay_amd.common.ingress import build_ingress_for_head_service
+        cluster = simple_raycluster("demo")
+        cluster.spec.head_group_spec.ingress_options = {
+            "host": "ray.example.com", "path": "/dash",
+            "ingressClassName": "nginx",
+            "tls": [{"hosts": ["ray.example.com"]}]}
+        ing = build_ingress_for_head_service(cluster)
+        rule = ing.spec["rules"][0]
+        assert rule["host"] == "ray.example.com"
+        assert rule["http"]["paths"][0]["path"] == "/dash"
+        assert rule["http"]["paths"][0]["pathType"] == "Exact"
+        assert ing.spec["ingressClassName"] == "nginx"
+        assert ing.spec["tls"]
+        assert ing.metadata.annotations is None  # no rewrite with user path
+
+    def test_openshift_route_shape(self):
+        from kuberay_amd.common.openshift import build_route_for_head_service
+        cluster = simple_raycluster("demo")
+        route = build_route_for_head_service(cluster)
+        assert route.kind == "Route"
+        assert route.spec["to"]["name"].startswith("demo-head")
+        assert route.spec["port"]["targetPort"] == 8265
