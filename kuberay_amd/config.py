@@ -21,6 +21,11 @@ class Configuration(K8sModel):
     probe_addr: str = ":8082"
     enable_leader_election: bool = True
     leader_election_namespace: str = ""
+    # horizontal sharding: run N operator processes against one cluster,
+    # each owning CRs whose crc32(ns/name) %% shards == shard_index; every
+    # shard elects its own leader Lease (kuberay-amd-operator-shard-I)
+    shards: int = 1
+    shard_index: int = 0
     reconcile_concurrency: int = 4          # reference default 1; Python
                                             # reconcilers are pure cache fns
     watch_namespaces: Optional[List[str]] = None
@@ -57,6 +62,8 @@ def load_config(argv: Optional[List[str]] = None) -> Configuration:
     parser.add_argument("--probe-addr")
     parser.add_argument("--reconcile-concurrency", type=int)
     parser.add_argument("--watch-namespace", action="append", dest="watch_namespaces")
+    parser.add_argument("--shards", type=int)
+    parser.add_argument("--shard-index", type=int)
     parser.add_argument("--batch-scheduler")
     parser.add_argument("--enable-batch-scheduler", action="store_true", default=None)
     parser.add_argument("--feature-gates", default=None)
@@ -77,6 +84,7 @@ def load_config(argv: Optional[List[str]] = None) -> Configuration:
     for flag, attr in [
         ("metrics_addr", "metrics_addr"), ("probe_addr", "probe_addr"),
         ("reconcile_concurrency", "reconcile_concurrency"),
+        ("shards", "shards"), ("shard_index", "shard_index"),
         ("watch_namespaces", "watch_namespaces"),
         ("batch_scheduler", "batch_scheduler"),
         ("enable_batch_scheduler", "enable_batch_scheduler"),

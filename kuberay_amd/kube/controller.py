@@ -67,6 +67,7 @@ class Controller:
         watch_namespaces: Optional[List[str]] = None,
         metrics=None,
         owned_event_coalesce_s: float = 0.0,
+        shard: Optional[Tuple[int, int]] = None,
     ):
         self.name = name
         self.primary_kind = primary_kind
@@ -87,6 +88,23 @@ class Controller:
         # reconcile; 0 (default) measured fastest — the dedup queue already
         # coalesces while workers are busy
         self.owned_event_coalesce_s = owned_event_coalesce_s
+        # horizontal sharding: (index, total) — this controller only handles
+        # CRs whose stable name-hash lands on its shard, so N operator
+        # processes can split one cluster's load (each shard elects its own
+        # leader Lease). None = unsharded.
+        self.shard = shard
+        if shard is not None:
+            index, total = shard
+            if not (total >= 1 and 0 <= index < total):
+                raise ValueError(f"invalid shard {shard}")
+
+    def _owns(self, namespace: str, name: str) -> bool:
+        if self.shard is None:
+            return True
+        index, total = self.shard
+        # crc32: stable across processes (python str hash is randomized)
+        import zlib
+        return zlib.crc32(f"{namespace}/{name}".encode()) % total == index
 
     # -- event routing -------------------------------------------------
     def observe(self, event_type: str, obj: Dict[str, Any]) -> None:
@@ -97,6 +115,8 @@ class Controller:
             return
         if kind == self.primary_kind:
             key: Request = (meta.get("namespace", "default"), meta.get("name", ""))
+            if not self._owns(*key):
+                return
             if event_type == "MODIFIED" and self.use_predicates:
                 fp = _meta_fingerprint(obj)
                 if self._fingerprints.get(key) == fp:
@@ -111,6 +131,8 @@ class Controller:
             for ref in meta.get("ownerReferences") or []:
                 if ref.get("kind") == self.primary_kind:
                     key = (meta.get("namespace", "default"), ref.get("name", ""))
+                    if not self._owns(*key):
+                        continue
                     if self.owned_event_coalesce_s > 0:
                         self.queue.add_after(key, self.owned_event_coalesce_s)
                     else:

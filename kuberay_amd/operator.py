@@ -68,17 +68,19 @@ def build_manager(cfg: Configuration, server=None, client=None):
     manager = Manager(server)
     workers = cfg.reconcile_concurrency
     ns_scope = cfg.watch_namespaces or None
+    shard = (cfg.shard_index, cfg.shards) if cfg.shards > 1 else None
     manager.add_controller(Controller(
         "raycluster", "RayCluster",
         RayClusterReconciler(client, recorder=recorder, batch_scheduler=scheduler,
                              options=options, metrics=metrics),
         owned_kinds=["Pod", "Service", "Secret", "PersistentVolumeClaim", "Job"],
-        workers=workers, watch_namespaces=ns_scope, metrics=metrics))
+        workers=workers, watch_namespaces=ns_scope, metrics=metrics,
+        shard=shard))
     manager.add_controller(Controller(
         "rayjob", "RayJob",
         RayJobReconciler(client, recorder=recorder, metrics=metrics),
         owned_kinds=["RayCluster", "Job"], workers=workers,
-        watch_namespaces=ns_scope, metrics=metrics))
+        watch_namespaces=ns_scope, metrics=metrics, shard=shard))
     # real clusters: serve-proxy healthz drives the head pod's serve label
     http_proxy = None
     if cfg.backend == "kubernetes":
@@ -89,22 +91,22 @@ def build_manager(cfg: Configuration, server=None, client=None):
         RayServiceReconciler(client, recorder=recorder, metrics=metrics,
                              http_proxy_client=http_proxy),
         owned_kinds=["RayCluster", "Service"], workers=workers,
-        watch_namespaces=ns_scope, metrics=metrics))
+        watch_namespaces=ns_scope, metrics=metrics, shard=shard))
     if features.enabled("RayCronJob"):
         manager.add_controller(Controller(
             "raycronjob", "RayCronJob",
             RayCronJobReconciler(client, recorder=recorder),
-            owned_kinds=["RayJob"], workers=1))
+            owned_kinds=["RayJob"], workers=1, shard=shard))
     if features.enabled("RayClusterNetworkPolicy"):
         manager.add_controller(Controller(
             "networkpolicy", "RayCluster",
             NetworkPolicyReconciler(client, recorder=recorder),
-            owned_kinds=["NetworkPolicy"], workers=1))
+            owned_kinds=["NetworkPolicy"], workers=1, shard=shard))
     if features.enabled("RayClusterMTLS"):
         from .ops.mtls import MTLSReconciler
         manager.add_controller(Controller(
             "mtls", "RayCluster", MTLSReconciler(client, recorder=recorder),
-            owned_kinds=["Secret", "Pod"], workers=1))
+            owned_kinds=["Secret", "Pod"], workers=1, shard=shard))
 
     autoscaler = None
     if cfg.enable_mi355x_autoscaler and features.enabled("MI355XAutoscaler"):
@@ -204,8 +206,11 @@ def main(argv=None) -> int:
         from .kube.leaderelection import LeaderElector
         import socket
         started = threading.Event()
+        lease = "kuberay-amd-operator" if cfg.shards <= 1 else \
+            f"kuberay-amd-operator-shard-{cfg.shard_index}"
         elector = LeaderElector(
             client,
+            lease_name=lease,
             namespace=cfg.leader_election_namespace or "ray-system",
             identity=f"{socket.gethostname()}-{os.getpid()}",
             on_started_leading=started.set)
