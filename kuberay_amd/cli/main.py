@@ -184,6 +184,23 @@ def get_events(ctx, name):
     _table(["LAST SEEN", "TYPE", "REASON", "OBJECT", "COUNT", "MESSAGE"], rows)
 
 
+@get.command("cronjob")
+@click.argument("name", required=False)
+@click.pass_context
+def get_cronjob(ctx, name):
+    client = client_of(ctx)
+    ns = ctx.obj["namespace"]
+    crons = ([client.get(RayCronJob, ns, name)] if name
+             else client.list(RayCronJob, ns))
+    rows = [(c.metadata.name, c.spec.schedule,
+             c.spec.time_zone or "UTC",
+             "yes" if c.spec.suspend else "no",
+             c.status.last_schedule_time or "-")
+            for c in crons]
+    _table(["NAME", "SCHEDULE", "TIMEZONE", "SUSPENDED", "LAST SCHEDULE"],
+           rows)
+
+
 @get.command("service")
 @click.argument("name", required=False)
 @click.pass_context

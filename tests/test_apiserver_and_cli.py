@@ -594,3 +594,19 @@ def test_dashboard_service_create_shape(api):
     from kuberay_amd.models import RayService
     svc = client.get(RayService, "ns1", "ui-s1")
     assert svc.spec.serve_config_v2.startswith("applications:")
+
+
+def test_kray_get_cronjob(kray):
+    runner, client = kray
+    from kuberay_amd.models import RayCronJob
+    from kuberay_amd.testing import simple_raycluster
+    spec = {"schedule": "0 9 * * *", "timeZone": "America/New_York",
+            "jobTemplate": {"entrypoint": "python t.py",
+                            "rayClusterSpec":
+                                simple_raycluster("x").spec.to_dict()}}
+    client.create(RayCronJob.from_dict({
+        "apiVersion": "ray.io/v1", "kind": "RayCronJob",
+        "metadata": {"name": "nightly", "namespace": "ns1"}, "spec": spec}))
+    r = runner.invoke(cli, ["-n", "ns1", "get", "cronjob"])
+    assert r.exit_code == 0, r.output
+    assert "nightly" in r.output and "America/New_York" in r.output
