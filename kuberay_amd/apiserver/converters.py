@@ -85,14 +85,17 @@ def api_cluster_to_raycluster(namespace: str, cluster: Dict[str, Any],
     spec = cluster.get("clusterSpec", {})
     head = spec.get("headGroupSpec", {})
     head_template = templates.get(head.get("computeTemplate", ""), {"cpu": 1, "memory": 2})
+    head_pod_spec: Dict[str, Any] = {"containers": [
+        _container_from_template("ray-head", head.get("image")
+                                 or C.DEFAULT_RAY_ROCM_IMAGE,
+                                 head_template,
+                                 head.get("environment"))]}
+    if head_template.get("tolerations"):
+        head_pod_spec["tolerations"] = head_template["tolerations"]
     head_group = {
         "serviceType": head.get("serviceType"),
         "rayStartParams": dict(head.get("rayStartParams") or {}),
-        "template": {"spec": {"containers": [
-            _container_from_template("ray-head", head.get("image")
-                                     or C.DEFAULT_RAY_ROCM_IMAGE,
-                                     head_template,
-                                     head.get("environment"))]}},
+        "template": {"spec": head_pod_spec},
     }
     worker_groups = []
     for wg in spec.get("workerGroupSpec", []) or []:
@@ -103,10 +106,14 @@ def api_cluster_to_raycluster(namespace: str, cluster: Dict[str, Any],
             "minReplicas": wg.get("minReplicas", 0),
             "maxReplicas": wg.get("maxReplicas", wg.get("replicas", 1)),
             "rayStartParams": dict(wg.get("rayStartParams") or {}),
-            "template": {"spec": {"containers": [
-                _container_from_template("ray-worker", wg.get("image")
-                                         or C.DEFAULT_RAY_ROCM_IMAGE, t,
-                                         wg.get("environment"))]}},
+            "template": {"spec": {
+                "containers": [
+                    _container_from_template("ray-worker", wg.get("image")
+                                             or C.DEFAULT_RAY_ROCM_IMAGE, t,
+                                             wg.get("environment"))],
+                **({"tolerations": t["tolerations"]}
+                   if t.get("tolerations") else {}),
+            }},
         })
     annotations = dict(cluster.get("annotations") or {})
     labels = dict(cluster.get("labels") or {})

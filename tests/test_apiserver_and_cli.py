@@ -610,3 +610,23 @@ def test_kray_get_cronjob(kray):
     r = runner.invoke(cli, ["-n", "ns1", "get", "cronjob"])
     assert r.exit_code == 0, r.output
     assert "nightly" in r.output and "America/New_York" in r.output
+
+
+def test_compute_template_tolerations_applied(api):
+    """cluster.go: compute-template tolerations land on generated pods."""
+    t, client, _ = api
+    t.post("/apis/v1/namespaces/ns1/compute_templates", json={
+        "name": "tol-tpl", "cpu": 2, "memory": 4, "gpu": 1,
+        "tolerations": [{"key": "amd.com/gpu", "operator": "Exists",
+                         "effect": "NoSchedule"}]})
+    r = t.post("/apis/v1/namespaces/ns1/clusters", json={
+        "name": "tol-c1", "clusterSpec": {
+            "headGroupSpec": {"computeTemplate": "tol-tpl"},
+            "workerGroupSpec": [{"groupName": "g", "computeTemplate":
+                                 "tol-tpl", "replicas": 1}]}})
+    assert r.status_code == 200, r.text
+    rc = client.get(RayCluster, "ns1", "tol-c1")
+    for template in (rc.spec.head_group_spec.template,
+                     rc.spec.worker_group_specs[0].template):
+        tol = template.spec.to_dict().get("tolerations")
+        assert tol and tol[0]["key"] == "amd.com/gpu"
