@@ -630,3 +630,62 @@ def test_compute_template_tolerations_applied(api):
                      rc.spec.worker_group_specs[0].template):
         tol = template.spec.to_dict().get("tolerations")
         assert tol and tol[0]["key"] == "amd.com/gpu"
+
+
+class TestHttpClientRetry:
+    """apiserversdk proxy.go:106-208 retry round-tripper analog."""
+
+    def test_transient_5xx_retried_then_succeeds(self):
+        import httpx as hx
+
+        from kuberay_amd.kube.httpclient import HttpKubeClient
+        calls = {"n": 0}
+
+        def handler(request):
+            calls["n"] += 1
+            if calls["n"] < 3:
+                return hx.Response(503, text="apiserver hiccup")
+            return hx.Response(200, json={
+                "apiVersion": "ray.io/v1", "kind": "RayCluster",
+                "metadata": {"name": "r1", "namespace": "ns1"},
+                "spec": {"headGroupSpec": {"rayStartParams": {},
+                                           "template": {"spec": {
+                                               "containers": []}}}}})
+
+        client = HttpKubeClient(
+            "http://x", http_client=hx.Client(
+                transport=hx.MockTransport(handler), base_url="http://x"))
+        rc = client.get(RayCluster, "ns1", "r1")
+        assert rc.metadata.name == "r1"
+        assert calls["n"] == 3
+
+    def test_persistent_5xx_raises_after_retries(self):
+        import httpx as hx
+
+        from kuberay_amd.kube.httpclient import HttpKubeClient
+        from kuberay_amd.kube.store import ApiError
+        client = HttpKubeClient(
+            "http://x", http_client=hx.Client(
+                transport=hx.MockTransport(
+                    lambda r: hx.Response(500, text="boom")),
+                base_url="http://x"))
+        with pytest.raises(ApiError):
+            client.get(RayCluster, "ns1", "r1")
+
+    def test_4xx_not_retried(self):
+        import httpx as hx
+
+        from kuberay_amd.kube.httpclient import HttpKubeClient
+        from kuberay_amd.kube.store import NotFoundError
+        calls = {"n": 0}
+
+        def handler(request):
+            calls["n"] += 1
+            return hx.Response(404, text="nope")
+
+        client = HttpKubeClient(
+            "http://x", http_client=hx.Client(
+                transport=hx.MockTransport(handler), base_url="http://x"))
+        with pytest.raises(NotFoundError):
+            client.get(RayCluster, "ns1", "r1")
+        assert calls["n"] == 1
