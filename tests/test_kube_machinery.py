@@ -387,3 +387,21 @@ class TestPodPool:
         mgr.define_pool("empty", self.TEMPLATE, size=0)
         mgr.reconcile()
         assert mgr.adopt("empty", {"ray.io/cluster": "c1"}) is None
+
+
+class TestShardedLeases:
+    def test_shards_hold_distinct_leases_concurrently(self):
+        """Two operator shards elect leaders on independent Lease names."""
+        from kuberay_amd.kube.client import InMemoryClient
+        from kuberay_amd.kube.leaderelection import LeaderElector
+        client = InMemoryClient()
+        s0 = LeaderElector(client, identity="proc-0",
+                           lease_name="kuberay-amd-operator-shard-0")
+        s1 = LeaderElector(client, identity="proc-1",
+                           lease_name="kuberay-amd-operator-shard-1")
+        assert s0.try_acquire_or_renew() is True
+        assert s1.try_acquire_or_renew() is True  # no contention across shards
+        # a standby for shard 0 still blocks on that shard's lease
+        standby = LeaderElector(client, identity="proc-0b",
+                                lease_name="kuberay-amd-operator-shard-0")
+        assert standby.try_acquire_or_renew() is False
