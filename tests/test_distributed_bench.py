@@ -49,3 +49,21 @@ class TestBenchDistributed:
         # whole-job aggregate: 2 ranks x 15 clusters
         assert d["config"]["global_batch"] == 30
         assert d["config"]["parallelism"] == "dp2"
+
+    def test_world4_gloo(self):
+        """Four ranks — same shape the driver uses for the N=4 scaling
+        point; validates rendezvous + MAX/SUM aggregation at higher fan-out."""
+        out = subprocess.run(
+            [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+             "--nproc-per-node", "4", "--master-addr", "127.0.0.1",
+             "--master-port", "29519", "bench.py", "--clusters", "8",
+             "--steps", "1", "--warmup", "0", "--workers-per-cluster", "1"],
+            capture_output=True, text=True, cwd=REPO, timeout=540,
+            env={**os.environ, "MASTER_ADDR": "127.0.0.1"})
+        assert out.returncode == 0, (out.stdout[-1500:], out.stderr[-1500:])
+        lines = [l for l in out.stdout.splitlines() if l.startswith("{")]
+        assert len(lines) == 1
+        d = json.loads(lines[0])
+        assert d["n_gpus"] == 4
+        assert d["config"]["global_batch"] == 32  # 4 ranks x 8 clusters
+        assert d["config"]["parallelism"] == "dp4"
