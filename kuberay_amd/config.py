@@ -62,6 +62,8 @@ def load_config(argv: Optional[List[str]] = None) -> Configuration:
     parser.add_argument("--probe-addr")
     parser.add_argument("--reconcile-concurrency", type=int)
     parser.add_argument("--watch-namespace", action="append", dest="watch_namespaces")
+    parser.add_argument("--snapshot-interval", type=float,
+                        dest="state_snapshot_interval_s")
     parser.add_argument("--shards", type=int)
     parser.add_argument("--shard-index", type=int)
     parser.add_argument("--batch-scheduler")
@@ -85,6 +87,7 @@ def load_config(argv: Optional[List[str]] = None) -> Configuration:
         ("metrics_addr", "metrics_addr"), ("probe_addr", "probe_addr"),
         ("reconcile_concurrency", "reconcile_concurrency"),
         ("shards", "shards"), ("shard_index", "shard_index"),
+        ("state_snapshot_interval_s", "state_snapshot_interval_s"),
         ("watch_namespaces", "watch_namespaces"),
         ("batch_scheduler", "batch_scheduler"),
         ("enable_batch_scheduler", "enable_batch_scheduler"),

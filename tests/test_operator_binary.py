@@ -150,3 +150,89 @@ class TestOperatorBinaryRestBackend:
                 proc.wait(timeout=10)
             kubelet.stop()
             facade.stop()
+
+
+@pytest.mark.timeout(150)
+class TestOperatorCrashRecovery:
+    """SIGKILL mid-flight, restart from the periodic snapshot, converge."""
+
+    def test_kill9_then_restart_from_snapshot(self, tmp_path):
+        import httpx
+
+        state_file = str(tmp_path / "state.jsonl")
+
+        def spawn(api_port):
+            return subprocess.Popen(
+                [sys.executable, "-m", "kuberay_amd.operator",
+                 "--api-port", str(api_port), "--metrics-addr", "",
+                 "--state-file", state_file, "--snapshot-interval", "0.2"],
+                cwd=REPO, stdout=subprocess.PIPE,
+                stderr=subprocess.STDOUT, text=True)
+
+        def wait_up(base):
+            deadline = time.monotonic() + 30
+            while time.monotonic() < deadline:
+                try:
+                    if httpx.get(f"{base}/apis/ray.io/v1/namespaces/default/"
+                                 "rayclusters", timeout=2).status_code == 200:
+                        return True
+                except httpx.HTTPError:
+                    time.sleep(0.3)
+            return False
+
+        def wait_ready(base, name, timeout=45):
+            deadline = time.monotonic() + timeout
+            while time.monotonic() < deadline:
+                try:
+                    r = httpx.get(f"{base}/apis/ray.io/v1/namespaces/default/"
+                                  f"rayclusters/{name}", timeout=3)
+                    if ((r.json().get("status") or {}).get("state")
+                            == "ready"):
+                        return True
+                except httpx.HTTPError:
+                    pass
+                time.sleep(0.3)
+            return False
+
+        from kuberay_amd.testing import simple_raycluster
+        port1 = _free_port()
+        proc = spawn(port1)
+        base = f"http://127.0.0.1:{port1}"
+        try:
+            assert wait_up(base)
+            for i in range(3):
+                body = simple_raycluster(f"crash-{i}", workers=1).to_dict()
+                assert httpx.post(
+                    f"{base}/apis/ray.io/v1/namespaces/default/rayclusters",
+                    json=body, timeout=5).status_code in (200, 201)
+            for i in range(3):
+                assert wait_ready(base, f"crash-{i}")
+            time.sleep(0.6)  # let a snapshot tick capture the ready state
+
+            proc.kill()  # SIGKILL: no graceful snapshot
+            proc.wait(timeout=10)
+
+            port2 = _free_port()
+            proc = spawn(port2)
+            base = f"http://127.0.0.1:{port2}"
+            assert wait_up(base)
+            # restored clusters converge again after the crash
+            for i in range(3):
+                assert wait_ready(base, f"crash-{i}")
+            # and the control plane still takes new work
+            body = simple_raycluster("crash-new", workers=1).to_dict()
+            assert httpx.post(
+                f"{base}/apis/ray.io/v1/namespaces/default/rayclusters",
+                json=body, timeout=5).status_code in (200, 201)
+            assert wait_ready(base, "crash-new")
+            # no duplicate pods for a restored cluster: 1 head + 1 worker
+            r = httpx.get(f"{base}/api/v1/namespaces/default/pods", timeout=5)
+            pods = [p for p in r.json().get("items", [])
+                    if p["metadata"]["labels"].get("ray.io/cluster")
+                    == "crash-0"
+                    and not p["metadata"].get("deletionTimestamp")]
+            assert len(pods) == 2, [p["metadata"]["name"] for p in pods]
+        finally:
+            if proc.poll() is None:
+                proc.kill()
+                proc.wait(timeout=10)
