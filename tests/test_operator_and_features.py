@@ -539,3 +539,14 @@ class TestIngressAndRoute:
         assert route.kind == "Route"
         assert route.spec["to"]["name"].startswith("demo-head")
         assert route.spec["port"]["targetPort"] == 8265
+
+
+def test_sharding_and_snapshot_flags_parse():
+    from kuberay_amd.config import load_config
+    cfg = load_config(["--shards", "4", "--shard-index", "2",
+                       "--snapshot-interval", "5"])
+    assert cfg.shards == 4
+    assert cfg.shard_index == 2
+    assert cfg.state_snapshot_interval_s == 5.0
+    # defaults: unsharded
+    assert load_config([]).shards == 1
