@@ -198,6 +198,7 @@ class KubeApiFacade:
     def stop(self) -> None:
         self._httpd.stopping = True  # type: ignore[attr-defined]
         self._httpd.shutdown()
+        self._httpd.server_close()  # release the listening socket/port
         if self._thread:
             self._thread.join(timeout=3)
 
