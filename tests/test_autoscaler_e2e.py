@@ -114,3 +114,42 @@ class TestAutoscalerEndToEnd:
             "default", {C.RAY_CLUSTER_LABEL_KEY: "demo",
                         C.RAY_NODE_TYPE_LABEL_KEY: "worker"})]
         assert victim not in names
+
+
+class TestZeroToEight:
+    def test_full_0_to_8_scale_up_then_drain(self, stack):
+        """BASELINE config #5 shape: 0 -> 8 MI355X workers under sustained
+        synthetic actor load, then a full drain back to min when idle."""
+        cp, autoscaler, telemetry = stack
+        cluster = simple_raycluster("demo", workers=0, gpus_per_worker=1,
+                                    enableInTreeAutoscaling=True)
+        cluster.spec.worker_group_specs[0].min_replicas = 0
+        cluster.spec.worker_group_specs[0].max_replicas = 8
+        cluster.metadata.annotations = {AMD_AUTOSCALER_ANNOTATION: "true"}
+        cp.client.create(cluster)
+        assert cp.wait_for(lambda: cluster_of(cp).status.state == "ready")
+
+        telemetry["avg_utilization_pct"] = 95.0
+        telemetry["max_vram_used_fraction"] = 0.9
+        deadline = time.monotonic() + 40
+        while time.monotonic() < deadline:
+            autoscaler.step()
+            if cluster_of(cp).status.available_worker_replicas == 8:
+                break
+            time.sleep(0.08)
+        assert cluster_of(cp).status.available_worker_replicas == 8
+        # never exceeds maxReplicas
+        autoscaler.step()
+        assert cluster_of(cp).status.desired_worker_replicas == 8
+
+        # load drops: drain back toward min with NAMED victims each step
+        telemetry["avg_utilization_pct"] = 2.0
+        telemetry["max_vram_used_fraction"] = 0.02
+        deadline = time.monotonic() + 40
+        while time.monotonic() < deadline:
+            autoscaler.step()
+            if cluster_of(cp).status.available_worker_replicas == 0:
+                break
+            time.sleep(0.12)
+        assert cluster_of(cp).status.available_worker_replicas == 0
+        assert cluster_of(cp).status.desired_worker_replicas == 0
