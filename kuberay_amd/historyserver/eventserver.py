@@ -7,9 +7,12 @@ historyserver/pkg/historyserver/timeline.go).
 from __future__ import annotations
 
 import json
+import logging
 from typing import Any, Dict, List
 
 from .storage import StorageReader, decompress
+
+logger = logging.getLogger("kuberay.historyserver")
 
 # Ray event types (reference DEFAULT_RAY_EXPOSABLE_EVENT_TYPES)
 TASK_DEFINITION = "TASK_DEFINITION_EVENT"
@@ -140,10 +143,17 @@ def load_session(storage: StorageReader, prefix: str) -> SessionState:
     """Replay every stored event batch under {prefix}/events/."""
     state = SessionState()
     for path in storage.list(f"{prefix}/events"):
-        raw = storage.read(path)
-        if path.endswith(".gz"):
-            raw = decompress(raw)
-        for line in raw.decode().splitlines():
+        try:
+            raw = storage.read(path)
+            if path.endswith(".gz"):
+                raw = decompress(raw)
+            text = raw.decode()
+        except Exception:
+            # one corrupt batch must not kill post-mortem browsing of the
+            # rest of the session
+            logger.warning("skipping unreadable event batch %s", path)
+            continue
+        for line in text.splitlines():
             line = line.strip()
             if not line:
                 continue

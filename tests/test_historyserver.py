@@ -267,3 +267,17 @@ class TestClusterStatus:
     def test_monitoring_health_stubs(self, app):
         assert app.get("/api/grafana_health").json()["result"] is False
         assert app.get("/api/prometheus_health").json()["result"] is False
+
+
+def test_corrupt_event_batch_skipped():
+    """A corrupt gzip batch must not abort replay of the whole session."""
+    from kuberay_amd.historyserver.eventserver import load_session
+    from kuberay_amd.historyserver.storage import MemoryStorage, compress
+    storage = MemoryStorage()
+    good = b'{"event_type": "DRIVER_JOB_DEFINITION_EVENT", "data": {"job_id": "j1"}}\n'
+    storage.write("ns1/c1/s1/events/000.jsonl.gz", compress(good))
+    storage.write("ns1/c1/s1/events/001.jsonl.gz", b"\x1f\x8bnot-gzip-data")
+    storage.write("ns1/c1/s1/events/002.jsonl.gz", compress(
+        b'{"event_type": "DRIVER_JOB_DEFINITION_EVENT", "data": {"job_id": "j2"}}\n'))
+    state = load_session(storage, "ns1/c1/s1")
+    assert set(state.jobs) == {"j1", "j2"}  # batch 001 skipped, not fatal
