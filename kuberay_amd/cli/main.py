@@ -284,6 +284,42 @@ def create_cluster(ctx, name, image, head_cpu, head_memory, worker_replicas,
     click.echo(f"raycluster.ray.io/{name} created")
 
 
+@create.command("service")
+@click.argument("name")
+@click.option("--serve-config", "serve_config", required=True,
+              type=click.Path(exists=True),
+              help="path to a serveConfigV2 YAML file")
+@click.option("--image", default=C.DEFAULT_RAY_ROCM_IMAGE, show_default=True)
+@click.option("--worker-replicas", default=1, type=int)
+@click.option("--worker-gpu", default=1, type=int,
+              help="amd.com/gpu per worker (MI355X)")
+@click.option("--upgrade-strategy", default=None,
+              type=click.Choice(["NewCluster",
+                                 "NewClusterWithIncrementalUpgrade", "None"]))
+@click.option("--dry-run", is_flag=True)
+@click.pass_context
+def create_service(ctx, name, serve_config, image, worker_replicas,
+                   worker_gpu, upgrade_strategy, dry_run):
+    """Create a RayService from a serve-config file."""
+    with open(serve_config) as f:
+        cfg = f.read()
+    spec = {
+        "serveConfigV2": cfg,
+        "rayClusterConfig": _cluster_spec(image, "2", "4Gi", worker_replicas,
+                                          "4", "8Gi", worker_gpu, False),
+    }
+    if upgrade_strategy:
+        spec["upgradeStrategy"] = {"type": upgrade_strategy}
+    obj = {"apiVersion": "ray.io/v1", "kind": "RayService",
+           "metadata": {"name": name, "namespace": ctx.obj["namespace"]},
+           "spec": spec}
+    if dry_run:
+        click.echo(yaml.safe_dump(obj, sort_keys=False))
+        return
+    client_of(ctx).create(RayService.from_dict(obj))
+    click.echo(f"rayservice.ray.io/{name} created")
+
+
 @create.command("workergroup")
 @click.argument("cluster_name")
 @click.option("--group-name", required=True)

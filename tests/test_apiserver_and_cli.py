@@ -689,3 +689,18 @@ class TestHttpClientRetry:
         with pytest.raises(NotFoundError):
             client.get(RayCluster, "ns1", "r1")
         assert calls["n"] == 1
+
+
+def test_kray_create_service(kray, tmp_path):
+    runner, client = kray
+    cfg = tmp_path / "serve.yaml"
+    cfg.write_text("applications:\n- name: a\n  import_path: m.g\n")
+    r = runner.invoke(cli, ["-n", "ns1", "create", "service", "svc-cli",
+                            "--serve-config", str(cfg), "--worker-gpu", "2"])
+    assert r.exit_code == 0, r.output
+    from kuberay_amd.models import RayService
+    svc = client.get(RayService, "ns1", "svc-cli")
+    assert svc.spec.serve_config_v2.startswith("applications:")
+    limits = svc.spec.ray_cluster_spec.worker_group_specs[0].template.spec \
+        .containers[0].resources.limits
+    assert limits["amd.com/gpu"] == "2"
