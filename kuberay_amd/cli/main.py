@@ -320,6 +320,42 @@ def create_service(ctx, name, serve_config, image, worker_replicas,
     click.echo(f"rayservice.ray.io/{name} created")
 
 
+@create.command("cronjob")
+@click.argument("name")
+@click.option("--schedule", required=True, help='5-field cron, e.g. "0 9 * * *"')
+@click.option("--timezone", "time_zone", default=None,
+              help="IANA zone for the schedule (default UTC)")
+@click.option("--entrypoint", required=True)
+@click.option("--image", default=C.DEFAULT_RAY_ROCM_IMAGE, show_default=True)
+@click.option("--worker-replicas", default=1, type=int)
+@click.option("--worker-gpu", default=0, type=int)
+@click.option("--dry-run", is_flag=True)
+@click.pass_context
+def create_cronjob(ctx, name, schedule, time_zone, entrypoint, image,
+                   worker_replicas, worker_gpu, dry_run):
+    """Create a RayCronJob that submits a RayJob on a schedule."""
+    spec = {
+        "schedule": schedule,
+        "jobTemplate": {
+            "entrypoint": entrypoint,
+            "shutdownAfterJobFinishes": True,
+            "rayClusterSpec": _cluster_spec(image, "2", "4Gi",
+                                            worker_replicas, "4", "8Gi",
+                                            worker_gpu, False),
+        },
+    }
+    if time_zone:
+        spec["timeZone"] = time_zone
+    obj = {"apiVersion": "ray.io/v1", "kind": "RayCronJob",
+           "metadata": {"name": name, "namespace": ctx.obj["namespace"]},
+           "spec": spec}
+    if dry_run:
+        click.echo(yaml.safe_dump(obj, sort_keys=False))
+        return
+    client_of(ctx).create(RayCronJob.from_dict(obj))
+    click.echo(f"raycronjob.ray.io/{name} created")
+
+
 @create.command("workergroup")
 @click.argument("cluster_name")
 @click.option("--group-name", required=True)

@@ -704,3 +704,18 @@ def test_kray_create_service(kray, tmp_path):
     limits = svc.spec.ray_cluster_spec.worker_group_specs[0].template.spec \
         .containers[0].resources.limits
     assert limits["amd.com/gpu"] == "2"
+
+
+def test_kray_create_cronjob(kray):
+    runner, client = kray
+    r = runner.invoke(cli, ["-n", "ns1", "create", "cronjob", "nightly2",
+                            "--schedule", "0 3 * * *",
+                            "--timezone", "America/New_York",
+                            "--entrypoint", "python train.py",
+                            "--worker-gpu", "1"])
+    assert r.exit_code == 0, r.output
+    from kuberay_amd.models import RayCronJob
+    cron = client.get(RayCronJob, "ns1", "nightly2")
+    assert cron.spec.schedule == "0 3 * * *"
+    assert cron.spec.time_zone == "America/New_York"
+    assert cron.spec.job_template.entrypoint == "python train.py"
