@@ -26,10 +26,17 @@ def main():
     parser = argparse.ArgumentParser()
     parser.add_argument("--services", type=int, default=50)
     parser.add_argument("--timeout", type=float, default=300.0)
+    parser.add_argument("--incremental", action="store_true",
+                        help="NewClusterWithIncrementalUpgrade via Gateway "
+                             "API weighted migration (25%% steps)")
     args = parser.parse_args()
 
     from kuberay_amd.models import RayService
     from kuberay_amd.testing import ControlPlane, simple_raycluster
+
+    if args.incremental:
+        import kuberay_amd.features as features
+        features.set_gate("RayServiceIncrementalUpgrade", True)
 
     cp = ControlPlane(kubelet_delay=0.0, record_events=False,
                       poll_seconds=0.2, workers=4)
@@ -38,6 +45,13 @@ def main():
     t0 = time.monotonic()
     try:
         names = [f"upsvc-{i:03d}" for i in range(args.services)]
+        upgrade_spec = ({"upgradeStrategy": {
+            "type": "NewClusterWithIncrementalUpgrade",
+            "clusterUpgradeOptions": {"gatewayClassName": "istio",
+                                      "stepSizePercent": 25,
+                                      "intervalSeconds": 0,
+                                      "maxSurgePercent": 100}}}
+            if args.incremental else {})
         for name in names:
             cp.client.create(RayService.from_dict({
                 "apiVersion": "ray.io/v1", "kind": "RayService",
@@ -45,7 +59,8 @@ def main():
                 "spec": {"serveConfigV2": SERVE_CONFIG,
                          "rayClusterConfig": simple_raycluster(
                              "x", workers=1,
-                             gpus_per_worker=1).spec.to_dict()}}))
+                             gpus_per_worker=1).spec.to_dict(),
+                         **upgrade_spec}}))
 
         def state_of(name):
             svc = cp.client.try_get(RayService, "default", name)
@@ -108,7 +123,8 @@ def main():
 
         errors = sum(c.error_count for c in cp.manager.controllers)
         result = {
-            "suite": f"{args.services}-rayservice-upgrade-storm",
+            "suite": f"{args.services}-rayservice-upgrade-storm"
+                     + ("-incremental" if args.incremental else ""),
             "provision_s": round(provision_s, 2),
             "upgrade_storm_s": round(upgrade_s, 2),
             "all_promoted": promoted,
