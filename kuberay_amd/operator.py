@@ -53,7 +53,16 @@ def build_manager(cfg: Configuration, server=None, client=None):
     recorder = StoreRecorder(server) if isinstance(server, InMemoryApiServer) else None
 
     scheduler = None
-    if cfg.enable_batch_scheduler:
+    if features.enabled("KubernetesWAS"):
+        # the KubernetesWAS gate SELECTS the WAS scheduler and must not be
+        # combined with --batch-scheduler/--enable-batch-scheduler
+        # (reference: schedulermanager.go:56-60 + ValidateBatchSchedulerConfig)
+        if cfg.enable_batch_scheduler or cfg.batch_scheduler:
+            raise ValueError(
+                "feature gate KubernetesWAS cannot be combined with "
+                "--enable-batch-scheduler/--batch-scheduler")
+        scheduler = scheduler_for("kubernetes-was-v1alpha2")
+    elif cfg.enable_batch_scheduler:
         name = cfg.batch_scheduler or (
             "xgmi-gang" if features.enabled("XgmiGangScheduling") else None)
         scheduler = scheduler_for(name)

@@ -550,3 +550,24 @@ def test_sharding_and_snapshot_flags_parse():
     assert cfg.state_snapshot_interval_s == 5.0
     # defaults: unsharded
     assert load_config([]).shards == 1
+
+
+def test_kubernetes_was_gate_selects_scheduler():
+    """schedulermanager.go:56-60: the gate selects WAS; combining it with
+    --batch-scheduler is a config error."""
+    import kuberay_amd.features as features
+    from kuberay_amd.config import Configuration
+    from kuberay_amd.operator import build_manager
+    from kuberay_amd.parallel.batchscheduler import KubernetesWASBatchScheduler
+    features.set_gate("KubernetesWAS", True)
+    try:
+        m, _, _, _ = build_manager(Configuration(enable_metrics=False))
+        rc = next(c for c in m.controllers if c.name == "raycluster")
+        assert isinstance(rc.reconciler.batch_scheduler,
+                          KubernetesWASBatchScheduler)
+        with pytest.raises(ValueError, match="cannot be combined"):
+            build_manager(Configuration(enable_metrics=False,
+                                        enable_batch_scheduler=True,
+                                        batch_scheduler="volcano"))
+    finally:
+        features.reset()
