@@ -719,3 +719,37 @@ def test_kray_create_cronjob(kray):
     assert cron.spec.schedule == "0 3 * * *"
     assert cron.spec.time_zone == "America/New_York"
     assert cron.spec.job_template.entrypoint == "python train.py"
+
+
+class TestApiServerPythonClient:
+    """python-apiserver-client analog: typed client over v1 endpoints."""
+
+    def test_full_surface(self, api):
+        t, backing, _ = api
+        from kuberay_amd.apiserver.client import (ApiServerClient,
+                                                  ApiServerError)
+        c = ApiServerClient(http_client=t)
+        c.create_compute_template("ns1", {"name": "pyc-tpl", "cpu": 2,
+                                          "memory": 4, "gpu": 1})
+        assert any(x["name"] == "pyc-tpl"
+                   for x in c.list_compute_templates("ns1"))
+        spec = {"headGroupSpec": {"computeTemplate": "pyc-tpl"},
+                "workerGroupSpec": [{"groupName": "g",
+                                     "computeTemplate": "pyc-tpl",
+                                     "replicas": 1}]}
+        c.create_cluster("ns1", {"name": "pyc-c1", "clusterSpec": spec})
+        assert c.get_cluster("ns1", "pyc-c1")["name"] == "pyc-c1"
+        c.create_job("ns1", {"name": "pyc-j1", "entrypoint": "python x.py",
+                             "clusterSpec": spec})
+        assert [j["name"] for j in c.list_jobs("ns1")] == ["pyc-j1"]
+        c.create_service("ns1", {"name": "pyc-s1",
+                                 "serveConfig_V2": "applications: []",
+                                 "clusterSpec": spec})
+        assert c.get_service("ns1", "pyc-s1")["name"] == "pyc-s1"
+        for deleter, name in ((c.delete_service, "pyc-s1"),
+                              (c.delete_job, "pyc-j1"),
+                              (c.delete_cluster, "pyc-c1"),
+                              (c.delete_compute_template, "pyc-tpl")):
+            deleter("ns1", name)
+        with pytest.raises(ApiServerError):
+            c.get_cluster("ns1", "pyc-c1")
