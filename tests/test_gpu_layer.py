@@ -414,3 +414,19 @@ class TestBaselineConfigsOnDevice:
             assert cp.wait_for(promoted, timeout=120)
         finally:
             cp.stop()
+
+
+def test_probe_cli_fails_loudly_without_gpu():
+    """On a non-GPU node the readiness probe must fail LOUDLY (exit != 0),
+    never silently pass — a silent fallback would mark GPU pods Ready on
+    nodes that can't run them."""
+    import subprocess
+    import sys
+    out = subprocess.run(
+        [sys.executable, "-m", "kuberay_amd.gpu.probe", "--quick", "--json"],
+        capture_output=True, text=True, timeout=60)
+    if out.returncode == 0:
+        import pytest
+        pytest.skip("running on a GPU node")
+    assert out.returncode != 0
+    assert "not a GPU node" in out.stdout + out.stderr
