@@ -68,6 +68,41 @@ def validate_raycluster_spec(cluster: RayCluster) -> List[str]:
     errs += _validate_gcs_ft(cluster)
     errs += _validate_auth(spec)
     errs += _validate_network_policy(spec)
+    errs += _validate_tls(spec)
+    return errs
+
+
+def _validate_tls(spec: RayClusterSpec) -> List[str]:
+    """validation.go:446-480 validateTLSOptions — with tlsOptions enabled,
+    the operator owns the TLS env vars and cert mount; user-set values
+    would silently fight it."""
+    errs: List[str] = []
+    opts = spec.tls_options
+    enabled = (opts.get("enabled") if isinstance(opts, dict)
+               else getattr(opts, "enabled", None)) if opts else None
+    if not enabled:
+        return errs
+    forbidden = {C.RAY_USE_TLS, C.RAY_TLS_SERVER_CERT, C.RAY_TLS_SERVER_KEY,
+                 C.RAY_TLS_CA_CERT}
+
+    def check(template, where):
+        for container in template.spec.containers or []:
+            for env in container.env or []:
+                if env.name in forbidden:
+                    errs.append(
+                        f"cannot set {env.name} env in {where} when "
+                        "tlsOptions is enabled — the operator manages TLS")
+            for vm in container.volume_mounts or []:
+                if vm.mount_path == C.RAY_TLS_CERT_MOUNT_PATH or \
+                        vm.name == C.RAY_TLS_VOLUME_NAME:
+                    errs.append(
+                        f"cannot mount {vm.mount_path} in {where} when "
+                        "tlsOptions is enabled — the operator manages the "
+                        "cert mount")
+
+    check(spec.head_group_spec.template, "head Pod")
+    for group in spec.worker_group_specs:
+        check(group.template, f"worker group '{group.group_name}'")
     return errs
 
 

@@ -312,3 +312,43 @@ class TestDeletionStrategyValidation:
              "condition": {"jobDeploymentStatus": "Failed"}}]},
             shutdownAfterJobFinishes=False)
         assert validate_rayjob_spec(job) == []
+
+
+class TestTLSValidation:
+    """validation.go:446-480: user-set TLS env/mounts conflict with
+    operator-managed TLS."""
+
+    def _cluster(self, env=None, mount=None):
+        from kuberay_amd.testing import simple_raycluster
+        rc = simple_raycluster("tlsv")
+        rc.spec.tls_options = {"enabled": True}
+        head = rc.spec.head_group_spec.template.spec.containers[0]
+        if env:
+            from kuberay_amd.kube.objects import EnvVar
+            head.env = (head.env or []) + [EnvVar(name=env, value="1")]
+        if mount:
+            from kuberay_amd.kube.objects import VolumeMount
+            head.volume_mounts = (head.volume_mounts or []) + [
+                VolumeMount(name="user-tls", mount_path=mount)]
+        return rc
+
+    def test_user_tls_env_rejected(self):
+        from kuberay_amd.utils.validation import validate_raycluster_spec
+        errs = validate_raycluster_spec(self._cluster(env="RAY_USE_TLS"))
+        assert any("RAY_USE_TLS" in e for e in errs)
+
+    def test_user_cert_mount_rejected(self):
+        from kuberay_amd.utils.validation import validate_raycluster_spec
+        errs = validate_raycluster_spec(
+            self._cluster(mount="/etc/ray/tls"))
+        assert any("cert mount" in e for e in errs)
+
+    def test_clean_tls_cluster_passes(self):
+        from kuberay_amd.utils.validation import validate_raycluster_spec
+        assert validate_raycluster_spec(self._cluster()) == []
+
+    def test_env_fine_when_tls_disabled(self):
+        from kuberay_amd.utils.validation import validate_raycluster_spec
+        rc = self._cluster(env="RAY_USE_TLS")
+        rc.spec.tls_options = None
+        assert validate_raycluster_spec(rc) == []
