@@ -65,6 +65,13 @@ def validate_raycluster_spec(cluster: RayCluster) -> List[str]:
             errs.append(
                 f"worker group '{group.group_name}': idleTimeoutSeconds requires "
                 "autoscaler v2 (spec.autoscalerOptions.version: v2)")
+        errs += _validate_group_resources_and_labels(
+            group.group_name or "worker", group.ray_start_params,
+            getattr(group, "resources", None), getattr(group, "labels", None))
+    errs += _validate_group_resources_and_labels(
+        "head", spec.head_group_spec.ray_start_params,
+        getattr(spec.head_group_spec, "resources", None),
+        getattr(spec.head_group_spec, "labels", None))
     errs += _validate_gcs_ft(cluster)
     errs += _validate_auth(spec)
     errs += _validate_network_policy(spec)
@@ -103,6 +110,37 @@ def _validate_tls(spec: RayClusterSpec) -> List[str]:
     check(spec.head_group_spec.template, "head Pod")
     for group in spec.worker_group_specs:
         check(group.template, f"worker group '{group.group_name}'")
+    return errs
+
+
+_LABEL_KEY_RE = re.compile(
+    r"^([a-z0-9]([-a-z0-9.]*[a-z0-9])?/)?[A-Za-z0-9]([-A-Za-z0-9_.]*[A-Za-z0-9])?$")
+_LABEL_VALUE_RE = re.compile(r"^([A-Za-z0-9]([-A-Za-z0-9_.]*[A-Za-z0-9])?)?$")
+
+
+def _validate_group_resources_and_labels(group_name: str, ray_start_params,
+                                         resources, labels) -> List[str]:
+    """validation.go:60-101 — the top-level group `resources`/`labels`
+    fields own their rayStartParams keys; manual duplicates conflict."""
+    errs: List[str] = []
+    params = ray_start_params or {}
+    has_param_resources = any(params.get(k) for k in
+                              ("num-cpus", "num-gpus", "memory", "resources"))
+    if has_param_resources and resources:
+        errs.append(
+            f"resource fields should not be set in both rayStartParams and "
+            f"resources for {group_name} group; please use only one")
+    if "labels" in params:
+        errs.append(
+            f"rayStartParams['labels'] is not supported for {group_name} "
+            "group; please use the top-level labels field instead")
+    for key, val in (labels or {}).items():
+        if len(key) > 253 or not _LABEL_KEY_RE.match(key):
+            errs.append(f"invalid label key for {group_name} group: '{key}'")
+        if len(str(val)) > 63 or not _LABEL_VALUE_RE.match(str(val)):
+            errs.append(
+                f"invalid label value for key '{key}' in {group_name} "
+                f"group: '{val}'")
     return errs
 
 

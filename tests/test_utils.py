@@ -352,3 +352,46 @@ class TestTLSValidation:
         rc = self._cluster(env="RAY_USE_TLS")
         rc.spec.tls_options = None
         assert validate_raycluster_spec(rc) == []
+
+
+class TestGroupResourcesLabelsValidation:
+    """validation.go:60-101: top-level resources/labels vs rayStartParams."""
+
+    def _cluster(self, params=None, resources=None, labels=None):
+        from kuberay_amd.testing import simple_raycluster
+        rc = simple_raycluster("grl")
+        g = rc.spec.worker_group_specs[0]
+        if params:
+            g.ray_start_params.update(params)
+        if resources is not None:
+            g.resources = resources
+        if labels is not None:
+            g.labels = labels
+        return rc
+
+    def test_conflicting_resources_rejected(self):
+        from kuberay_amd.utils.validation import validate_raycluster_spec
+        errs = validate_raycluster_spec(self._cluster(
+            params={"num-gpus": "4"}, resources={"CPU": "8"}))
+        assert any("use only one" in e for e in errs)
+
+    def test_ray_start_labels_param_rejected(self):
+        from kuberay_amd.utils.validation import validate_raycluster_spec
+        errs = validate_raycluster_spec(self._cluster(
+            params={"labels": "a=b"}))
+        assert any("top-level labels field" in e for e in errs)
+
+    def test_bad_label_syntax_rejected(self):
+        from kuberay_amd.utils.validation import validate_raycluster_spec
+        errs = validate_raycluster_spec(self._cluster(
+            labels={"-bad-key": "x"}))
+        assert any("invalid label key" in e for e in errs)
+        errs = validate_raycluster_spec(self._cluster(
+            labels={"ok": "bad value!"}))
+        assert any("invalid label value" in e for e in errs)
+
+    def test_clean_group_labels_pass(self):
+        from kuberay_amd.utils.validation import validate_raycluster_spec
+        assert validate_raycluster_spec(self._cluster(
+            resources={"TPUAnalog": "0"},
+            labels={"ray.io/market-type": "spot"})) == []
