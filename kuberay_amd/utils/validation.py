@@ -76,6 +76,58 @@ def validate_raycluster_spec(cluster: RayCluster) -> List[str]:
     errs += _validate_auth(spec)
     errs += _validate_network_policy(spec)
     errs += _validate_tls(spec)
+    errs += _validate_cluster_upgrade(cluster)
+    for group in spec.worker_group_specs:
+        errs += _validate_worker_priority(group, spec)
+    return errs
+
+
+def _validate_cluster_upgrade(cluster: RayCluster) -> List[str]:
+    """validation.go:42-58 ValidateRayClusterUpgradeOptions."""
+    errs: List[str] = []
+    us = cluster.spec.upgrade_strategy
+    if us is None or us.type is None:
+        return errs
+    if us.type not in ("Recreate", "None"):
+        errs.append(f"upgradeStrategy.type '{us.type}' is invalid; valid "
+                    "options are Recreate or None")
+    creator = (cluster.metadata.labels or {}).get(
+        C.RAY_ORIGINATED_FROM_CRD_LABEL_KEY)
+    if creator in ("RayJob", "RayService"):
+        errs.append(f"upgradeStrategy cannot be set when RayCluster is "
+                    f"created by {creator}")
+    return errs
+
+
+def _validate_worker_priority(group, spec: RayClusterSpec) -> List[str]:
+    """validation.go:1113-1142 — priority needs Ray >= 2.56 + autoscaler v2."""
+    errs: List[str] = []
+    if not group.priority:
+        return errs
+    name = group.group_name or "worker"
+    if not spec.ray_version:
+        errs.append(f"worker group {name}: priority is set, but rayVersion "
+                    "was not specified (Ray >= 2.56.0 required)")
+        return errs
+    try:
+        parts = tuple(int(x) for x in
+                      re.match(r"(\d+)\.(\d+)", spec.ray_version).groups())
+    except (AttributeError, ValueError):
+        errs.append(f"worker group {name}: priority is set, but rayVersion "
+                    f"format is invalid: {spec.ray_version}")
+        return errs
+    if parts < (2, 56):
+        errs.append(f"worker group {name}: priority requires Ray >= 2.56.0, "
+                    f"got {spec.ray_version}")
+    if not _autoscaler_v2(spec):
+        head_env = []
+        if spec.head_group_spec.template.spec.containers:
+            head_env = spec.head_group_spec.template.spec.containers[0].env or []
+        if not any(e.name == C.RAY_ENABLE_AUTOSCALER_V2
+                   and str(e.value).lower() in ("1", "true")
+                   for e in head_env):
+            errs.append(f"worker group {name}: priority is only supported "
+                        "with autoscaler v2 enabled")
     return errs
 
 
@@ -294,9 +346,28 @@ def validate_rayservice_metadata(meta: ObjectMeta) -> List[str]:
     return errs
 
 
+def _validate_initializing_timeout(annotations) -> List[str]:
+    """validation.go:651-678 — ray.io/initializing-timeout accepts a Go-style
+    duration ("30m", "1h") or positive integer seconds."""
+    value = (annotations or {}).get(
+        C.RAY_SERVICE_INITIALIZING_TIMEOUT_ANNOTATION)
+    if not value:
+        return []
+    m = re.fullmatch(r"(\d+(\.\d+)?)(s|m|h)?", value)
+    if m is None:
+        return [f"annotation {C.RAY_SERVICE_INITIALIZING_TIMEOUT_ANNOTATION} "
+                f"has invalid format: {value!r} (expected a duration like "
+                "'5m'/'1h' or positive integer seconds)"]
+    if float(m.group(1)) <= 0:
+        return [f"annotation {C.RAY_SERVICE_INITIALIZING_TIMEOUT_ANNOTATION} "
+                f"must be positive, got: {value!r}"]
+    return []
+
+
 def validate_rayservice_spec(rayservice: RayService) -> List[str]:
     errs = []
     spec = rayservice.spec
+    errs += _validate_initializing_timeout(rayservice.metadata.annotations)
     sub = RayCluster(metadata=rayservice.metadata, spec=spec.ray_cluster_spec)
     errs += validate_raycluster_spec(sub)
     us = spec.upgrade_strategy

@@ -395,3 +395,63 @@ class TestGroupResourcesLabelsValidation:
         assert validate_raycluster_spec(self._cluster(
             resources={"TPUAnalog": "0"},
             labels={"ray.io/market-type": "spot"})) == []
+
+
+class TestClusterUpgradeAndPriorityValidation:
+    """validation.go:42-58 + :1113-1142 + :651-678."""
+
+    def test_invalid_upgrade_type_rejected(self):
+        from kuberay_amd.testing import simple_raycluster
+        from kuberay_amd.utils.validation import validate_raycluster_spec
+        rc = simple_raycluster("u1")
+        from kuberay_amd.models.raycluster import RayClusterUpgradeStrategy
+        rc.spec.upgrade_strategy = RayClusterUpgradeStrategy(type="BlueGreen")
+        assert any("upgradeStrategy.type" in e
+                   for e in validate_raycluster_spec(rc))
+
+    def test_upgrade_on_child_cluster_rejected(self):
+        from kuberay_amd.testing import simple_raycluster
+        from kuberay_amd.utils.validation import validate_raycluster_spec
+        rc = simple_raycluster("u2")
+        from kuberay_amd.models.raycluster import RayClusterUpgradeStrategy
+        rc.spec.upgrade_strategy = RayClusterUpgradeStrategy(type="Recreate")
+        rc.metadata.labels = {"ray.io/originated-from-crd": "RayJob"}
+        assert any("created by RayJob" in e
+                   for e in validate_raycluster_spec(rc))
+
+    def test_priority_needs_new_ray_and_autoscaler_v2(self):
+        from kuberay_amd.testing import simple_raycluster
+        from kuberay_amd.utils.validation import validate_raycluster_spec
+        rc = simple_raycluster("p1")
+        rc.spec.worker_group_specs[0].priority = 5
+        rc.spec.ray_version = "2.46.0"
+        errs = validate_raycluster_spec(rc)
+        assert any("requires Ray >= 2.56.0" in e for e in errs)
+        rc.spec.ray_version = "2.56.0"
+        errs = validate_raycluster_spec(rc)
+        assert any("autoscaler v2" in e for e in errs)
+        from kuberay_amd.models.raycluster import AutoscalerOptions
+        rc.spec.autoscaler_options = AutoscalerOptions(version="v2")
+        assert validate_raycluster_spec(rc) == []
+
+    def test_initializing_timeout_annotation(self):
+        from kuberay_amd.models import RayService
+        from kuberay_amd.testing import simple_raycluster
+        from kuberay_amd.utils.validation import validate_rayservice_spec
+
+        def svc(value):
+            return RayService.from_dict({
+                "apiVersion": "ray.io/v1", "kind": "RayService",
+                "metadata": {"name": "s", "namespace": "default",
+                             "annotations": {
+                                 "ray.io/initializing-timeout": value}},
+                "spec": {"serveConfigV2": "applications: []",
+                         "rayClusterConfig":
+                             simple_raycluster("x").spec.to_dict()}})
+
+        assert validate_rayservice_spec(svc("30m")) == []
+        assert validate_rayservice_spec(svc("600")) == []
+        assert any("invalid format" in e
+                   for e in validate_rayservice_spec(svc("soon")))
+        assert any("must be positive" in e
+                   for e in validate_rayservice_spec(svc("0")))
