@@ -77,8 +77,40 @@ def validate_raycluster_spec(cluster: RayCluster) -> List[str]:
     errs += _validate_network_policy(spec)
     errs += _validate_tls(spec)
     errs += _validate_cluster_upgrade(cluster)
+    errs += _validate_collector_options(spec)
     for group in spec.worker_group_specs:
         errs += _validate_worker_priority(group, spec)
+    return errs
+
+
+# env names injected into the collector by the operator; user overrides
+# would silently fight the injection (validation.go:1029-1064 analog)
+_COLLECTOR_MANAGED_ENV = {"POD_IP", "RAY_ROLE", "OWNER_KIND", "OWNER_NAME",
+                          C.RAY_CLUSTER_NAMESPACE, "EVENTS_PORT"}
+_COLLECTOR_BACKEND_REQUIRES = {
+    "s3": "S3_REGION", "gcs": "GCS_BUCKET", "azure": "AZURE_STORAGE_ACCOUNT",
+    "oss": "OSS_ENDPOINT"}
+
+
+def _validate_collector_options(spec: RayClusterSpec) -> List[str]:
+    errs: List[str] = []
+    hso = getattr(spec, "history_server_options", None)
+    opts = getattr(hso, "collector_options", None) if hso else None
+    if opts is None:
+        return errs
+    env = {e.name: e for e in (getattr(opts, "env", None) or [])}
+    for name in sorted(_COLLECTOR_MANAGED_ENV & set(env)):
+        errs.append(
+            f"historyServerOptions.collectorOptions.env must not contain "
+            f"{name}: it is injected by the operator")
+    backend = env.get("STORAGE_BACKEND")
+    if backend is not None and backend.value:
+        needed = _COLLECTOR_BACKEND_REQUIRES.get(backend.value.lower())
+        if needed and not (env.get(needed) and
+                           (env[needed].value or env[needed].value_from)):
+            errs.append(
+                f"{needed} env must be set when STORAGE_BACKEND is "
+                f"{backend.value}")
     return errs
 
 

@@ -455,3 +455,33 @@ class TestClusterUpgradeAndPriorityValidation:
                    for e in validate_rayservice_spec(svc("soon")))
         assert any("must be positive" in e
                    for e in validate_rayservice_spec(svc("0")))
+
+
+class TestCollectorOptionsValidation:
+    """validation.go:1029-1064: collector env hygiene."""
+
+    def _cluster(self, env):
+        from kuberay_amd.testing import simple_raycluster
+        return simple_raycluster("col", historyServerOptions={
+            "collectorOptions": {"env": env}})
+
+    def test_managed_env_rejected(self):
+        from kuberay_amd.utils.validation import validate_raycluster_spec
+        errs = validate_raycluster_spec(self._cluster(
+            [{"name": "POD_IP", "value": "1.2.3.4"}]))
+        assert any("must not contain POD_IP" in e for e in errs)
+
+    def test_cloud_backend_requires_companion_env(self):
+        from kuberay_amd.utils.validation import validate_raycluster_spec
+        errs = validate_raycluster_spec(self._cluster(
+            [{"name": "STORAGE_BACKEND", "value": "s3"}]))
+        assert any("S3_REGION" in e for e in errs)
+        errs = validate_raycluster_spec(self._cluster(
+            [{"name": "STORAGE_BACKEND", "value": "s3"},
+             {"name": "S3_REGION", "value": "us-east-1"}]))
+        assert errs == []
+
+    def test_local_backend_passes(self):
+        from kuberay_amd.utils.validation import validate_raycluster_spec
+        assert validate_raycluster_spec(self._cluster(
+            [{"name": "STORAGE_BACKEND", "value": "local"}])) == []
