@@ -31,6 +31,21 @@ def make_rayservice(name="svc1", **spec_overrides):
     })
 
 
+def update_svc_with_retry(cp, mutate, attempts=20):
+    """get-mutate-update with optimistic-concurrency retry: the reconciler
+    can bump resourceVersion between our get and update."""
+    from kuberay_amd.kube.store import ConflictError
+    for _ in range(attempts):
+        svc = svc_of(cp)
+        mutate(svc)
+        try:
+            cp.client.update(svc)
+            return
+        except ConflictError:
+            time.sleep(0.05)
+    raise AssertionError("update kept conflicting")
+
+
 def svc_of(cp, name="svc1"):
     return cp.client.get(RayService, "default", name)
 
@@ -93,10 +108,10 @@ class TestZeroDowntimeUpgrade:
         assert wait_ready(cp)
         old_active = svc_of(cp).status.active_service_status.ray_cluster_name
 
-        svc = svc_of(cp)
-        svc.spec.ray_cluster_spec.worker_group_specs[0].template.spec \
-            .containers[0].image = "rayproject/ray:2.47.0-rocm"
-        cp.client.update(svc)
+        def set_image(svc):
+            svc.spec.ray_cluster_spec.worker_group_specs[0].template.spec \
+                .containers[0].image = "rayproject/ray:2.47.0-rocm"
+        update_svc_with_retry(cp, set_image)
 
         def promoted():
             s = svc_of(cp)
@@ -113,9 +128,9 @@ class TestZeroDowntimeUpgrade:
         cp.client.create(make_rayservice())
         assert wait_ready(cp)
         active = svc_of(cp).status.active_service_status.ray_cluster_name
-        svc = svc_of(cp)
-        svc.spec.ray_cluster_spec.worker_group_specs[0].replicas = 3
-        cp.client.update(svc)
+        def set_replicas(svc):
+            svc.spec.ray_cluster_spec.worker_group_specs[0].replicas = 3
+        update_svc_with_retry(cp, set_replicas)
         def scaled():
             rc = cp.client.try_get(RayCluster, "default", active)
             return rc is not None and rc.status.available_worker_replicas == 3
@@ -128,10 +143,10 @@ class TestZeroDowntimeUpgrade:
         cp.client.create(make_rayservice(upgradeStrategy={"type": "None"}))
         assert wait_ready(cp)
         active = svc_of(cp).status.active_service_status.ray_cluster_name
-        svc = svc_of(cp)
-        svc.spec.ray_cluster_spec.worker_group_specs[0].template.spec \
-            .containers[0].image = "rayproject/ray:2.47.0-rocm"
-        cp.client.update(svc)
+        def set_image(svc):
+            svc.spec.ray_cluster_spec.worker_group_specs[0].template.spec \
+                .containers[0].image = "rayproject/ray:2.47.0-rocm"
+        update_svc_with_retry(cp, set_image)
         time.sleep(0.5)
         s = svc_of(cp)
         assert s.status.active_service_status.ray_cluster_name == active
@@ -143,16 +158,16 @@ class TestSuspend:
         cp = control_plane
         cp.client.create(make_rayservice())
         assert wait_ready(cp)
-        svc = svc_of(cp)
-        svc.spec.suspend = True
-        cp.client.update(svc)
+        def suspend_on(svc):
+            svc.spec.suspend = True
+        update_svc_with_retry(cp, suspend_on)
         assert cp.wait_for(lambda: svc_of(cp).condition_true("Suspended"), timeout=15)
         assert cp.wait_for(lambda: cp.server.count("RayCluster") == 0)
         assert svc_of(cp).status.num_serve_endpoints == 0
 
-        svc = svc_of(cp)
-        svc.spec.suspend = False
-        cp.client.update(svc)
+        def suspend_off(svc):
+            svc.spec.suspend = False
+        update_svc_with_retry(cp, suspend_off)
         assert cp.wait_for(lambda: wait_ready(cp), timeout=25)
 
 
