@@ -53,11 +53,20 @@ class _LeaseStore:
         return self.client.raw_create(obj)
 
     def update(self, obj):
+        """Full-object PUT carrying the resourceVersion read by the caller.
+
+        Both backends enforce the optimistic-concurrency precondition and
+        raise ConflictError on a stale write, so two standbys racing to take
+        over an expired lease cannot both win (client-go leaderelection
+        semantics; a merge patch here would allow split-brain).
+        """
         if self.server is not None:
             return self.server.update(obj)
-        ns = obj["metadata"]["namespace"]
-        return self.client.raw_patch(LEASE_KIND, ns, obj["metadata"]["name"],
-                                     {"spec": obj["spec"]})
+        fn = getattr(self.client, "raw_update", None)
+        if fn is not None:
+            return fn(obj)
+        raise ConflictError("lease backend lacks raw_update; refusing "
+                            "non-atomic lease takeover")
 
 
 class LeaderElector:

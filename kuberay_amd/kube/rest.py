@@ -242,6 +242,18 @@ class RestClient(KubeClient):
         resp = self._check(self._http.post(self._path(kind, ns), json=obj))
         return resp.json()
 
+    def raw_update(self, obj: Dict[str, Any]) -> Dict[str, Any]:
+        """PUT the full object. With ``metadata.resourceVersion`` set, the
+        apiserver enforces the optimistic-concurrency precondition and a stale
+        write surfaces as :class:`ConflictError` (client-go leaderelection
+        semantics — see ``kube/leaderelection.py``)."""
+        kind = obj.get("kind", "")
+        md = obj.get("metadata", {})
+        ns = md.get("namespace", "default")
+        resp = self._check(self._http.put(
+            self._path(kind, ns, md.get("name")), json=obj))
+        return resp.json()
+
     def raw_patch(self, kind: str, namespace: str, name: str,
                   patch: Dict[str, Any]) -> Dict[str, Any]:
         resp = self._check(self._http.patch(

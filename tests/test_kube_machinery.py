@@ -314,6 +314,103 @@ class TestLeaderElection:
         assert b.try_acquire_or_renew() is True  # released lease is free
 
 
+class TestLeaderElectionOverRest:
+    """The production '--backend kubernetes' lease path. Regression for the
+    round-1 split-brain: lease takeover used a merge patch with no
+    resourceVersion precondition, so two standbys racing on an expired lease
+    could both become leader."""
+
+    def _facade(self):
+        from kuberay_amd.kube.httpserver import KubeApiFacade
+        return KubeApiFacade().start()
+
+    def test_acquire_renew_and_blocked_standby(self):
+        from kuberay_amd.kube.leaderelection import LeaderElector
+        from kuberay_amd.kube.rest import RestClient
+        facade = self._facade()
+        try:
+            a = LeaderElector(RestClient(base_url=facade.url), identity="a")
+            b = LeaderElector(RestClient(base_url=facade.url), identity="b")
+            assert a.try_acquire_or_renew() is True
+            assert a.try_acquire_or_renew() is True  # renewal
+            assert b.try_acquire_or_renew() is False
+        finally:
+            facade.stop()
+
+    def test_expired_lease_takeover_race_elects_exactly_one(self):
+        """Both standbys read the same expired lease revision; only the first
+        write may win — the second must get 409 and stand down."""
+        import pytest as _pytest
+
+        from kuberay_amd.kube.leaderelection import LEASE_KIND, LeaderElector, _LeaseStore
+        from kuberay_amd.kube.rest import RestClient
+        from kuberay_amd.kube.store import ConflictError
+        facade = self._facade()
+        try:
+            # dead leader: lease exists but renewTime is ancient
+            facade.store.create({
+                "apiVersion": "coordination.k8s.io/v1", "kind": LEASE_KIND,
+                "metadata": {"name": "op-lease", "namespace": "ray-system"},
+                "spec": {"holderIdentity": "dead",
+                         "leaseDurationSeconds": 1,
+                         "renewTime": "2000-01-01T00:00:00.000000Z",
+                         "leaseTransitions": 1}})
+            sa = _LeaseStore(RestClient(base_url=facade.url))
+            sb = _LeaseStore(RestClient(base_url=facade.url))
+            lease_a = sa.get("ray-system", "op-lease")
+            lease_b = sb.get("ray-system", "op-lease")  # same resourceVersion
+            lease_a["spec"]["holderIdentity"] = "standby-a"
+            sa.update(lease_a)  # first write wins
+            lease_b["spec"]["holderIdentity"] = "standby-b"
+            with _pytest.raises(ConflictError):
+                sb.update(lease_b)  # stale RV must NOT take the lease
+            # and the full elector treats that conflict as a lost election
+            b = LeaderElector(RestClient(base_url=facade.url),
+                              identity="standby-b", lease_name="op-lease",
+                              namespace="ray-system")
+            got = facade.store.get(LEASE_KIND, "ray-system", "op-lease")
+            assert got["spec"]["holderIdentity"] == "standby-a"
+        finally:
+            facade.stop()
+
+    def test_elector_conflict_on_takeover_stands_down(self):
+        """End-to-end through try_acquire_or_renew: a racing write between a
+        standby's read and write yields is_leader False, not split-brain."""
+        from kuberay_amd.kube.leaderelection import LEASE_KIND, LeaderElector
+        from kuberay_amd.kube.rest import RestClient
+        facade = self._facade()
+        try:
+            facade.store.create({
+                "apiVersion": "coordination.k8s.io/v1", "kind": LEASE_KIND,
+                "metadata": {"name": "op-lease", "namespace": "ray-system"},
+                "spec": {"holderIdentity": "dead",
+                         "leaseDurationSeconds": 1,
+                         "renewTime": "2000-01-01T00:00:00.000000Z",
+                         "leaseTransitions": 1}})
+            b = LeaderElector(RestClient(base_url=facade.url),
+                              identity="standby-b", lease_name="op-lease",
+                              namespace="ray-system")
+            real_get = b.store.get
+
+            def racing_get(ns, name):
+                lease = real_get(ns, name)
+                # another standby sneaks its write in between read and write
+                fresh = dict(facade.store.get(LEASE_KIND, ns, name))
+                fresh["spec"] = dict(fresh["spec"],
+                                     holderIdentity="standby-a",
+                                     renewTime="2099-01-01T00:00:00.000000Z")
+                facade.store.update(fresh)
+                return lease
+
+            b.store.get = racing_get
+            assert b.try_acquire_or_renew() is False
+            assert b.is_leader is False
+            got = facade.store.get(LEASE_KIND, "ray-system", "op-lease")
+            assert got["spec"]["holderIdentity"] == "standby-a"
+        finally:
+            facade.stop()
+
+
 class TestEventRecorder:
     def test_events_aggregate_by_reason(self):
         from kuberay_amd.kube.events import StoreRecorder
