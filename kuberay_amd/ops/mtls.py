@@ -144,21 +144,29 @@ class MTLSReconciler(Reconciler):
         fqdn = names.fqdn_service_name(cluster, namespace)
         dns_sans = [head_svc, fqdn, "localhost"]
 
-        # pod IP SANs (reference :279-448 tracks pod IPs)
+        # Per-role pod IP SANs (reference reconcileHeadCertificate /
+        # reconcileWorkerCertificate, raycluster_mtls_controller.go:279-448):
+        # GCS dials WORKERS by pod IP, so worker certs must carry worker pod
+        # IPs; the head cert carries only head pod IPs so a worker scale
+        # event never forces a head cert reissue.
+        head_ips: List[str] = ["127.0.0.1"]
+        worker_ips: List[str] = ["127.0.0.1"]
         views = getattr(self.client, "list_pod_views", None)
-        ips: List[str] = ["127.0.0.1"]
         if views is not None:
             for v in views(namespace,
-                           association.cluster_all_pods_selector(cname)):
+                           association.cluster_head_pod_selector(cname)):
                 if v.pod_ip:
-                    ips.append(v.pod_ip)
+                    head_ips.append(v.pod_ip)
+            for v in views(namespace,
+                           association.cluster_worker_pods_selector(cname)):
+                if v.pod_ip:
+                    worker_ips.append(v.pod_ip)
 
         changed = False
-        for role, secret_name in (("head", f"ray-head-secret-{cname}"),
-                                  ("worker", f"ray-worker-secret-{cname}")):
+        for role, secret_name, want_ips in (
+                ("head", f"ray-head-secret-{cname}", sorted(set(head_ips))),
+                ("worker", f"ray-worker-secret-{cname}", sorted(set(worker_ips)))):
             existing = self._secret(namespace, secret_name)
-            need_ips = role == "head"
-            want_ips = sorted(set(ips)) if need_ips else ["127.0.0.1"]
             if existing is not None:
                 covered = self._sans_cover(existing, want_ips)
                 if covered:
@@ -176,12 +184,21 @@ class MTLSReconciler(Reconciler):
 
     @staticmethod
     def _sans_cover(secret: k8s.Secret, want_ips: List[str]) -> bool:
+        """Exact-entry SAN membership: substring matching would let
+        10.0.0.11 falsely 'cover' 10.0.0.1 and skip a needed reissue."""
         try:
             cert = base64.b64decode(secret.data["tls.crt"])
-            sans = cert_sans(cert)
+            sans_text = cert_sans(cert)
         except Exception:
             return False
-        return all(ip in sans for ip in want_ips)
+        entries = set()
+        for part in sans_text.replace("\n", ",").split(","):
+            part = part.strip()
+            if part.startswith("IP Address:"):
+                entries.add(part[len("IP Address:"):].strip())
+            elif part.startswith("IP:"):
+                entries.add(part[len("IP:"):].strip())
+        return all(ip in entries for ip in want_ips)
 
     # ------------------------------------------------------------------
     def _reconcile_cert_manager(self, cluster: RayCluster) -> None:
@@ -192,6 +209,21 @@ class MTLSReconciler(Reconciler):
         if server is None:
             return
         owner = [k8s.owner_reference_for(cluster).to_dict()]
+        # Per-role IP SANs, mirroring reconcileHeadCertificate /
+        # reconcileWorkerCertificate: worker cert tracks worker pod IPs
+        # (GCS dials workers by IP), head cert tracks head pod IPs only.
+        head_ips: List[str] = ["127.0.0.1"]
+        worker_ips: List[str] = ["127.0.0.1"]
+        views = getattr(self.client, "list_pod_views", None)
+        if views is not None:
+            for v in views(namespace,
+                           association.cluster_head_pod_selector(cname)):
+                if v.pod_ip:
+                    head_ips.append(v.pod_ip)
+            for v in views(namespace,
+                           association.cluster_worker_pods_selector(cname)):
+                if v.pod_ip:
+                    worker_ips.append(v.pod_ip)
         objs = [
             {"apiVersion": "cert-manager.io/v1", "kind": "Issuer",
              "metadata": {"name": f"ray-selfsigned-issuer-{cname}",
@@ -213,14 +245,14 @@ class MTLSReconciler(Reconciler):
                           "namespace": namespace, "ownerReferences": owner},
              "spec": {"secretName": f"ray-head-secret-{cname}",
                       "dnsNames": [names.fqdn_service_name(cluster, namespace)],
-                      "ipAddresses": ["127.0.0.1"],
+                      "ipAddresses": sorted(set(head_ips)),
                       "issuerRef": {"name": f"ray-ca-issuer-{cname}",
                                     "kind": "Issuer"}}},
             {"apiVersion": "cert-manager.io/v1", "kind": "Certificate",
              "metadata": {"name": f"ray-worker-cert-{cname}",
                           "namespace": namespace, "ownerReferences": owner},
              "spec": {"secretName": f"ray-worker-secret-{cname}",
-                      "ipAddresses": ["127.0.0.1"],
+                      "ipAddresses": sorted(set(worker_ips)),
                       "issuerRef": {"name": f"ray-ca-issuer-{cname}",
                                     "kind": "Issuer"}}},
         ]
@@ -228,4 +260,13 @@ class MTLSReconciler(Reconciler):
             try:
                 server.create(obj)
             except AlreadyExistsError:
-                pass
+                # Certificates: keep ipAddresses tracking the live pod IPs
+                # (the reference updates the Certificate spec in place).
+                if obj["kind"] == "Certificate" and "ipAddresses" in obj["spec"]:
+                    md = obj["metadata"]
+                    cur = server.try_get("Certificate", md["namespace"],
+                                         md["name"])
+                    if cur is not None and (cur.get("spec", {}).get("ipAddresses")
+                                            != obj["spec"]["ipAddresses"]):
+                        cur["spec"]["ipAddresses"] = obj["spec"]["ipAddresses"]
+                        server.update(cur)

@@ -63,6 +63,65 @@ class TestSelfSignedReconciler:
         head = client.get(k8s.Secret, "default", "ray-head-secret-demo")
         assert "10.99.1.2" in cert_sans(base64.b64decode(head.data["tls.crt"]))
 
+    def _add_pod(self, client, name, node_type, ip):
+        client.server.create({
+            "kind": "Pod",
+            "metadata": {"name": name, "namespace": "default",
+                         "labels": {"ray.io/cluster": "demo",
+                                    "ray.io/node-type": node_type}},
+            "spec": {"containers": [{"name": "ray"}]},
+            "status": {"phase": "Running", "podIP": ip}})
+
+    def test_worker_cert_tracks_worker_pod_ips(self):
+        """GCS dials workers by pod IP — the WORKER cert must carry worker
+        pod IPs (reference reconcileWorkerCertificate)."""
+        client = InMemoryClient()
+        client.create(self._cluster())
+        r = MTLSReconciler(client)
+        r.reconcile(("default", "demo"))
+        self._add_pod(client, "demo-worker-a", "worker", "10.99.2.7")
+        r.reconcile(("default", "demo"))
+        worker = client.get(k8s.Secret, "default", "ray-worker-secret-demo")
+        assert "10.99.2.7" in cert_sans(
+            base64.b64decode(worker.data["tls.crt"]))
+
+    def test_head_cert_not_reissued_on_worker_scale(self):
+        """Worker churn must not force a head cert reissue: the head cert
+        carries head pod IPs only (reference reconcileHeadCertificate)."""
+        client = InMemoryClient()
+        client.create(self._cluster())
+        r = MTLSReconciler(client)
+        self._add_pod(client, "demo-head-x", "head", "10.99.1.2")
+        r.reconcile(("default", "demo"))
+        head1 = client.get(k8s.Secret, "default", "ray-head-secret-demo")
+        sans1 = cert_sans(base64.b64decode(head1.data["tls.crt"]))
+        assert "10.99.1.2" in sans1
+        # worker scale event
+        self._add_pod(client, "demo-worker-a", "worker", "10.99.2.7")
+        r.reconcile(("default", "demo"))
+        head2 = client.get(k8s.Secret, "default", "ray-head-secret-demo")
+        assert head2.metadata.resource_version == head1.metadata.resource_version
+        assert "10.99.2.7" not in cert_sans(
+            base64.b64decode(head2.data["tls.crt"]))
+
+    def test_sans_cover_is_exact_not_substring(self):
+        """10.0.0.11 in the cert must NOT 'cover' a required 10.0.0.1."""
+        client = InMemoryClient()
+        client.create(self._cluster())
+        r = MTLSReconciler(client)
+        self._add_pod(client, "demo-head-x", "head", "10.0.0.11")
+        r.reconcile(("default", "demo"))
+        head1 = client.get(k8s.Secret, "default", "ray-head-secret-demo")
+        assert "10.0.0.11" in cert_sans(base64.b64decode(head1.data["tls.crt"]))
+        # second head IP that is a prefix of the existing SAN entry
+        self._add_pod(client, "demo-head-y", "head", "10.0.0.1")
+        r.reconcile(("default", "demo"))
+        head2 = client.get(k8s.Secret, "default", "ray-head-secret-demo")
+        sans = cert_sans(base64.b64decode(head2.data["tls.crt"]))
+        assert "IP Address:10.0.0.1," in sans + "," or \
+            sans.rstrip().endswith("10.0.0.1")
+        assert head2.metadata.resource_version != head1.metadata.resource_version
+
     def test_noop_without_tls(self):
         client = InMemoryClient()
         client.create(simple_raycluster("demo"))
@@ -94,3 +153,24 @@ class TestCertManagerMode:
         ca_cert = next(c for c in certs
                        if c["metadata"]["name"] == "ray-ca-certificate-demo")
         assert ca_cert["spec"]["isCA"] is True
+
+    def test_certificate_ip_sans_track_per_role_pods(self):
+        client = InMemoryClient()
+        client.create(simple_raycluster("demo", tlsOptions={"enabled": True}))
+        r = MTLSReconciler(client, mode="cert-manager")
+        r.reconcile(("default", "demo"))
+        for name, node_type, ip in (("demo-head-x", "head", "10.99.1.2"),
+                                    ("demo-worker-a", "worker", "10.99.2.7")):
+            client.server.create({
+                "kind": "Pod",
+                "metadata": {"name": name, "namespace": "default",
+                             "labels": {"ray.io/cluster": "demo",
+                                        "ray.io/node-type": node_type}},
+                "spec": {"containers": [{"name": "ray"}]},
+                "status": {"phase": "Running", "podIP": ip}})
+        r.reconcile(("default", "demo"))
+        head = client.server.get("Certificate", "default", "ray-head-cert-demo")
+        worker = client.server.get("Certificate", "default",
+                                   "ray-worker-cert-demo")
+        assert head["spec"]["ipAddresses"] == ["10.99.1.2", "127.0.0.1"]
+        assert worker["spec"]["ipAddresses"] == ["10.99.2.7", "127.0.0.1"]
