@@ -132,11 +132,8 @@ class RayJobReconciler(Reconciler):
                                     or names.ray_job_id(rayjob.metadata.name))
         if not rayjob.status.ray_cluster_name:
             if rayjob.spec.cluster_selector:
-                selected = self.client.list(RayCluster,
-                                            rayjob.metadata.namespace or "default",
-                                            rayjob.spec.cluster_selector)
-                if selected:
-                    rayjob.status.ray_cluster_name = selected[0].metadata.name
+                rayjob.status.ray_cluster_name = \
+                    self._resolve_selected_cluster_name(rayjob) or ""
             else:
                 rayjob.status.ray_cluster_name = names.ray_cluster_name_for(
                     rayjob.metadata.name)
@@ -439,7 +436,32 @@ class RayJobReconciler(Reconciler):
     # ------------------------------------------------------------------
     # cluster & submitter management
     # ------------------------------------------------------------------
+    def _resolve_selected_cluster_name(self, rayjob: RayJob) -> Optional[str]:
+        """Name of the cluster a clusterSelector points at.
+
+        The ``ray.io/cluster`` selector key IS the cluster name — resolve it
+        directly and independently of whether that RayCluster exists yet
+        (reference getRayClusterNameFromSelector semantics), falling back to
+        a live selector lookup for arbitrary label selectors.
+        """
+        selector = rayjob.spec.cluster_selector or {}
+        direct = selector.get(C.RAY_CLUSTER_LABEL_KEY)
+        if direct:
+            return direct
+        matches = self.client.list(RayCluster,
+                                   rayjob.metadata.namespace or "default",
+                                   selector)
+        return matches[0].metadata.name if matches else None
+
     def _owned_cluster(self, rayjob: RayJob) -> Optional[RayCluster]:
+        if not rayjob.status.ray_cluster_name and rayjob.spec.cluster_selector:
+            # Re-resolve each reconcile: the selected cluster may have been
+            # created after the RayJob (round-1 bug left the job stuck in
+            # Initializing forever when the selector matched nothing at
+            # creation time).
+            name = self._resolve_selected_cluster_name(rayjob)
+            if name:
+                rayjob.status.ray_cluster_name = name
         if not rayjob.status.ray_cluster_name:
             return None
         return self.client.try_get(RayCluster, rayjob.metadata.namespace or "default",
@@ -452,7 +474,7 @@ class RayJobReconciler(Reconciler):
         if cluster is not None:
             return cluster
         if rayjob.spec.cluster_selector:
-            return None  # selected cluster vanished; wait
+            return None  # selected cluster vanished (or not created yet); wait
         spec = rayjob.spec.ray_cluster_spec.clone()
         if rayjob.spec.submission_mode == Mode.SIDECAR:
             self._inject_sidecar_submitter(rayjob, spec)

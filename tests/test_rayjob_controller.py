@@ -195,6 +195,40 @@ class TestClusterSelector:
         assert control_plane.server.try_get("RayCluster", "default", "shared")
 
 
+    def test_selector_created_before_cluster_exists(self, control_plane):
+        """A RayJob whose clusterSelector matches nothing at creation time
+        must pick up the cluster once it appears — not sit in Initializing
+        forever (round-1 bug: the selector was resolved exactly once)."""
+        control_plane.client.create(make_rayjob(
+            clusterSelector={C.RAY_CLUSTER_LABEL_KEY: "late"},
+            rayClusterSpec=None))
+        assert wait_deployment_status(control_plane, "job1", "Initializing")
+        # ray.io/cluster key resolves directly even before the cluster exists
+        assert control_plane.wait_for(
+            lambda: job_of(control_plane).status.ray_cluster_name == "late")
+        time.sleep(0.3)
+        assert job_of(control_plane).status.job_deployment_status == "Initializing"
+        # now the cluster arrives
+        control_plane.client.create(simple_raycluster("late", workers=1))
+        assert wait_deployment_status(control_plane, "job1", "Complete",
+                                      timeout=30)
+
+    def test_label_selector_rechecked_each_reconcile(self, control_plane):
+        """Arbitrary label selectors (no ray.io/cluster key) are re-run every
+        reconcile until a cluster matches."""
+        control_plane.client.create(make_rayjob(
+            clusterSelector={"pool": "b"}, rayClusterSpec=None))
+        assert wait_deployment_status(control_plane, "job1", "Initializing")
+        time.sleep(0.3)
+        assert not job_of(control_plane).status.ray_cluster_name
+        late = simple_raycluster("pool-b", workers=1)
+        late.metadata.labels = {"pool": "b"}
+        control_plane.client.create(late)
+        assert wait_deployment_status(control_plane, "job1", "Complete",
+                                      timeout=30)
+        assert job_of(control_plane).status.ray_cluster_name == "pool-b"
+
+
 class TestInteractiveMode:
     def test_waits_for_submission_id_annotation(self, control_plane):
         job = make_rayjob(submissionMode="InteractiveMode", entrypoint=None)
