@@ -94,6 +94,7 @@ class RayServiceReconciler(Reconciler):
             return Result()
 
         self._gc_old_clusters(namespace)
+        self._sweep_orphan_clusters(svc)
 
         errs = validate_rayservice_metadata(svc.metadata) + validate_rayservice_spec(svc)
         if errs:
@@ -280,6 +281,36 @@ class RayServiceReconciler(Reconciler):
         key = (cluster.metadata.namespace or "default", cluster.metadata.name)
         with self._lock:
             self._pending_deletions.setdefault(key, time.monotonic() + delay)
+
+    def _sweep_orphan_clusters(self, svc: RayService) -> None:
+        """Schedule deletion of owned RayClusters that are neither active nor
+        pending (reference reconcileRayCluster's cleanUpRayClusterInstance).
+
+        The delayed-deletion timer lives in operator memory; if the operator
+        restarts inside the delay window the replaced cluster would otherwise
+        leak forever. This list-based sweep rediscovers orphans from the API
+        server on every reconcile, so restart loses only the remaining delay,
+        never the deletion.
+        """
+        namespace = svc.metadata.namespace or "default"
+        keep = {svc.status.active_service_status.ray_cluster_name,
+                svc.status.pending_service_status.ray_cluster_name}
+        owned = self.client.list(
+            RayCluster, namespace,
+            association.originated_from_selector(svc.metadata.name,
+                                                 C.KIND_RAYSERVICE))
+        for cluster in owned:
+            if cluster.metadata.name in keep or cluster.metadata.deletion_timestamp:
+                continue
+            key = (namespace, cluster.metadata.name)
+            with self._lock:
+                already = key in self._pending_deletions
+            if not already:
+                self._delete_cluster_later(cluster)
+                self.recorder.eventf(
+                    svc, "Normal", "ScheduledOrphanClusterDeletion",
+                    "RayCluster %s is neither active nor pending; scheduled "
+                    "for deletion", cluster.metadata.name)
 
     def _gc_old_clusters(self, namespace: str) -> None:
         now = time.monotonic()

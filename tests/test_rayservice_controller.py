@@ -153,6 +153,56 @@ class TestZeroDowntimeUpgrade:
         assert not s.status.pending_service_status.ray_cluster_name
 
 
+class TestOrphanClusterSweep:
+    def test_replaced_cluster_survives_operator_restart(self):
+        """Operator restart inside the deletion-delay window must not leak
+        the replaced active cluster: the orphan sweep rediscovers it from the
+        API server (reference reconcileRayCluster GC of non-active/pending
+        owned clusters)."""
+        from kuberay_amd.testing import ControlPlane
+        cp1 = ControlPlane(kubelet_delay=0.01, job_runtime=0.2,
+                           poll_seconds=0.05)
+        # long delay: the old cluster is still pending deletion at "crash"
+        cp1.rayservice_reconciler.cluster_deletion_delay_s = 300
+        cp1.start()
+        try:
+            cp1.client.create(make_rayservice())
+            assert wait_ready(cp1)
+            old_active = svc_of(cp1).status.active_service_status.ray_cluster_name
+
+            def set_image(svc):
+                svc.spec.ray_cluster_spec.worker_group_specs[0].template.spec \
+                    .containers[0].image = "rayproject/ray:2.47.0-rocm"
+            update_svc_with_retry(cp1, set_image)
+
+            def promoted():
+                s = svc_of(cp1)
+                return (s.status.active_service_status.ray_cluster_name
+                        not in (None, old_active) and s.condition_true("Ready"))
+            assert cp1.wait_for(promoted, timeout=25)
+            # old cluster still exists — deletion is pending in cp1's memory
+            assert cp1.server.try_get("RayCluster", "default", old_active)
+        finally:
+            cp1.stop()  # operator "crash" loses the in-memory timer
+
+        cp2 = ControlPlane(kubelet_delay=0.01, job_runtime=0.2,
+                           poll_seconds=0.05, server=cp1.server)
+        cp2.rayservice_reconciler.cluster_deletion_delay_s = 0.2
+        cp2.start()
+        try:
+            assert cp2.wait_for(
+                lambda: cp2.server.try_get("RayCluster", "default",
+                                           old_active) is None,
+                timeout=15), "replaced cluster leaked across operator restart"
+            # the promoted active cluster is untouched
+            s = svc_of(cp2)
+            assert cp2.server.try_get(
+                "RayCluster", "default",
+                s.status.active_service_status.ray_cluster_name)
+        finally:
+            cp2.stop()
+
+
 class TestSuspend:
     def test_suspend_and_resume(self, control_plane):
         cp = control_plane
