@@ -368,21 +368,27 @@ def create_cronjob(ctx, name, schedule, time_zone, entrypoint, image,
 def create_workergroup(ctx, cluster_name, group_name, image, worker_replicas,
                        worker_cpu, worker_memory, worker_gpu):
     client = client_of(ctx)
-    rc = client.get(RayCluster, ctx.obj["namespace"], cluster_name)
     limits = {"cpu": worker_cpu, "memory": worker_memory}
     if worker_gpu:
         limits[C.AMD_GPU_RESOURCE_NAME] = str(worker_gpu)
-    head_image = rc.spec.head_group_spec.template.spec.containers[0].image
     from ..models.raycluster import WorkerGroupSpec
-    rc.spec.worker_group_specs.append(WorkerGroupSpec.from_dict({
-        "groupName": group_name, "replicas": worker_replicas,
-        "minReplicas": 0, "maxReplicas": max(worker_replicas, 8),
-        "rayStartParams": {},
-        "template": {"spec": {"containers": [{
-            "name": "ray-worker", "image": image or head_image,
-            "resources": {"limits": limits, "requests": dict(limits)}}]}},
-    }))
-    client.update(rc)
+
+    def add_group(rc):
+        if any(g.group_name == group_name for g in rc.spec.worker_group_specs):
+            raise click.ClickException(
+                f"worker group '{group_name}' already exists")
+        head_image = rc.spec.head_group_spec.template.spec.containers[0].image
+        rc.spec.worker_group_specs.append(WorkerGroupSpec.from_dict({
+            "groupName": group_name, "replicas": worker_replicas,
+            "minReplicas": 0, "maxReplicas": max(worker_replicas, 8),
+            "rayStartParams": {},
+            "template": {"spec": {"containers": [{
+                "name": "ray-worker", "image": image or head_image,
+                "resources": {"limits": limits, "requests": dict(limits)}}]}},
+        }))
+
+    client.update_with_retry(RayCluster, ctx.obj["namespace"], cluster_name,
+                             add_group)
     click.echo(f"worker group {group_name} added to {cluster_name}")
 
 
@@ -413,22 +419,28 @@ def scale():
 @click.pass_context
 def scale_cluster(ctx, name, worker_group, replicas):
     client = client_of(ctx)
-    rc = client.get(RayCluster, ctx.obj["namespace"], name)
-    groups = rc.spec.worker_group_specs
-    target = None
-    if worker_group:
-        target = next((g for g in groups if g.group_name == worker_group), None)
-        if target is None:
+    scaled_group = {}
+
+    def set_replicas(rc):
+        groups = rc.spec.worker_group_specs
+        if worker_group:
+            target = next((g for g in groups
+                           if g.group_name == worker_group), None)
+            if target is None:
+                raise click.ClickException(
+                    f"worker group '{worker_group}' not found "
+                    f"(have: {[g.group_name for g in groups]})")
+        elif len(groups) == 1:
+            target = groups[0]
+        else:
             raise click.ClickException(
-                f"worker group '{worker_group}' not found "
-                f"(have: {[g.group_name for g in groups]})")
-    elif len(groups) == 1:
-        target = groups[0]
-    else:
-        raise click.ClickException("--worker-group required (multiple groups)")
-    target.replicas = replicas
-    client.update(rc)
-    click.echo(f"cluster {name}/{target.group_name} scaled to {replicas}")
+                "--worker-group required (multiple groups)")
+        target.replicas = replicas
+        scaled_group["name"] = target.group_name
+
+    client.update_with_retry(RayCluster, ctx.obj["namespace"], name,
+                             set_replicas)
+    click.echo(f"cluster {name}/{scaled_group['name']} scaled to {replicas}")
 
 
 # ---------------------------------------------------------------------------
@@ -528,9 +540,10 @@ def suspend(ctx, kind, name, resume):
     ns = ctx.obj["namespace"]
     model = {"cluster": RayCluster, "job": RayJob, "service": RayService,
              "cronjob": RayCronJob}[kind]
-    obj = client.get(model, ns, name)
-    obj.spec.suspend = not resume
-    client.update(obj)
+    def set_suspend(obj):
+        obj.spec.suspend = not resume
+
+    client.update_with_retry(model, ns, name, set_suspend)
     click.echo(f"{kind} {name} {'resumed' if resume else 'suspended'}")
 
 

@@ -13,7 +13,7 @@ _DEFAULTS: Dict[str, bool] = {
     "RayClusterStatusConditions": True,       # beta, on
     "RayJobDeletionPolicy": True,             # beta, on
     "RayMultiHostIndexing": True,             # beta, on
-    "RayServiceIncrementalUpgrade": False,    # needs Gateway API; off in v1
+    "RayServiceIncrementalUpgrade": True,     # beta, on (features.go:105-117)
     "RayCronJob": True,
     "SidecarSubmitterRestart": False,
     "RayClusterNetworkPolicy": False,

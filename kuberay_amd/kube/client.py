@@ -82,6 +82,32 @@ class KubeClient:
                name: Optional[str] = None) -> None:
         raise NotImplementedError
 
+    def update_with_retry(self, model: Type[T], namespace: str, name: str,
+                          mutate, attempts: int = 10,
+                          backoff_s: float = 0.02) -> T:
+        """Get-mutate-update with optimistic-concurrency retry
+        (client-go ``retry.RetryOnConflict`` analog).
+
+        ``mutate(obj)`` is re-applied to a FRESH read on every attempt, so a
+        409 from a racing writer (the reconciler, an autoscaler) never
+        surfaces to the caller as a failure. All user-facing read-modify-write
+        (CLI scale/suspend, python clients, upgrade spec edits) must go
+        through this instead of naked ``update``.
+        """
+        import time as _time
+
+        from .store import ConflictError
+        last: Optional[Exception] = None
+        for attempt in range(attempts):
+            obj = self.get(model, namespace, name)
+            mutate(obj)
+            try:
+                return self.update(obj)
+            except ConflictError as e:
+                last = e
+                _time.sleep(backoff_s * (attempt + 1))
+        raise last if last is not None else RuntimeError("unreachable")
+
 
 def _kind_of(model_or_obj) -> str:
     if isinstance(model_or_obj, type):

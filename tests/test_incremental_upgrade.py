@@ -49,11 +49,15 @@ def wait_ready(cp, name="svc1", timeout=25):
     return cp.wait_for(lambda: svc_of(cp, name).condition_true("Ready"), timeout)
 
 
+def set_image_with_retry(cp, image, name="svc1"):
+    def mutate(svc):
+        svc.spec.ray_cluster_spec.worker_group_specs[0].template.spec \
+            .containers[0].image = image
+    cp.client.update_with_retry(RayService, "default", name, mutate)
+
+
 def trigger_upgrade(cp):
-    svc = svc_of(cp)
-    svc.spec.ray_cluster_spec.worker_group_specs[0].template.spec \
-        .containers[0].image = "rayproject/ray:2.47.0-rocm"
-    cp.client.update(svc)
+    set_image_with_retry(cp, "rayproject/ray:2.47.0-rocm")
 
 
 class TestIncrementalUpgrade:
@@ -104,10 +108,7 @@ class TestIncrementalUpgrade:
             lambda: svc_of(cp).status.pending_service_status.ray_cluster_name,
             timeout=20)
         # revert the spec to the running cluster's shape mid-upgrade
-        svc = svc_of(cp)
-        svc.spec.ray_cluster_spec.worker_group_specs[0].template.spec \
-            .containers[0].image = original_image
-        cp.client.update(svc)
+        set_image_with_retry(cp, original_image)
 
         def rolled_back():
             s = svc_of(cp)
@@ -116,6 +117,7 @@ class TestIncrementalUpgrade:
         assert cp.wait_for(rolled_back, timeout=25)
 
     def test_gate_off_promotes_instantly(self):
+        features.set_gate("RayServiceIncrementalUpgrade", False)
         plane = ControlPlane(kubelet_delay=0.01, poll_seconds=0.05)
         plane.rayservice_reconciler.cluster_deletion_delay_s = 0.2
         plane.start()
@@ -126,10 +128,12 @@ class TestIncrementalUpgrade:
                 .condition_true("Ready"), timeout=25)
             old = plane.client.get(RayService, "default", "svc1") \
                 .status.active_service_status.ray_cluster_name
-            svc = plane.client.get(RayService, "default", "svc1")
-            svc.spec.ray_cluster_spec.worker_group_specs[0].template.spec \
-                .containers[0].image = "rayproject/ray:2.47.0-rocm"
-            plane.client.update(svc)
+
+            def set_image(svc):
+                svc.spec.ray_cluster_spec.worker_group_specs[0].template.spec \
+                    .containers[0].image = "rayproject/ray:2.47.0-rocm"
+            plane.client.update_with_retry(RayService, "default", "svc1",
+                                           set_image)
             assert plane.wait_for(
                 lambda: plane.client.get(RayService, "default", "svc1")
                 .status.active_service_status.ray_cluster_name not in (None, old),
@@ -137,6 +141,7 @@ class TestIncrementalUpgrade:
             assert plane.server.count("Gateway") == 0  # no gateway infra
         finally:
             plane.stop()
+            features.reset()
 
 
 class TestGatewayReadinessGate:
