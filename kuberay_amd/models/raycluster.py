@@ -177,7 +177,8 @@ class GcsFaultToleranceOptions(K8sModel):
 class AuthOptions(K8sModel):
     """raycluster_types.go:112-137."""
 
-    enable_k8s_token_auth: Optional[bool] = None
+    enable_k8s_token_auth: Optional[bool] = Field(
+        default=None, alias="enableK8sTokenAuth")
     secret_name: Optional[str] = None
     mode: Optional[str] = None  # token | disabled
 

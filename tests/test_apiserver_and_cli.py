@@ -358,8 +358,9 @@ class TestKrayNodeToken:
                                          "amd.com/xgmi-island": "n1-island0"}}})
             r = CliRunner().invoke(cli, ["get", "node"])
             assert r.exit_code == 0 and "n1-island0" in r.output
-            cp.client.create(simple_raycluster("authy",
-                                               authOptions={"mode": "token"}))
+            cp.client.create(simple_raycluster(
+                "authy", authOptions={"mode": "token"},
+                rayVersion="2.53.0"))  # token auth needs Ray >= 2.52
             assert cp.wait_for(lambda: cp.server.try_get(
                 "Secret", "default", "authy-auth-token"))
             r = CliRunner().invoke(cli, ["get", "token", "authy"])

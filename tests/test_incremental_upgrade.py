@@ -11,13 +11,16 @@ from kuberay_amd.testing import ControlPlane, simple_raycluster
 SERVE_CONFIG = "applications:\n- name: app1\n  import_path: m.g\n"
 
 
-def make_service(name="svc1", interval=0, step=50):
+def make_service(name="svc1", interval=1, step=50):
+    # incremental upgrade requires the autoscaler (validation.go:719-721)
+    cluster_spec = simple_raycluster("x", workers=1).spec.to_dict()
+    cluster_spec["enableInTreeAutoscaling"] = True
     return RayService.from_dict({
         "apiVersion": "ray.io/v1", "kind": "RayService",
         "metadata": {"name": name, "namespace": "default"},
         "spec": {
             "serveConfigV2": SERVE_CONFIG,
-            "rayClusterConfig": simple_raycluster("x", workers=1).spec.to_dict(),
+            "rayClusterConfig": cluster_spec,
             "upgradeStrategy": {
                 "type": "NewClusterWithIncrementalUpgrade",
                 "clusterUpgradeOptions": {

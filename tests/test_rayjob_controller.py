@@ -125,7 +125,8 @@ class TestFailure:
 
 class TestSuspend:
     def test_created_suspended(self, control_plane):
-        control_plane.client.create(make_rayjob(suspend=True))
+        control_plane.client.create(make_rayjob(
+            suspend=True, shutdownAfterJobFinishes=True))
         assert wait_deployment_status(control_plane, "job1", "Suspended")
         assert control_plane.server.count("RayCluster") == 0
 
@@ -142,7 +143,8 @@ class TestSuspend:
             lambda: control_plane.server.count("RayCluster") == 0)
 
     def test_resume_creates_fresh_cluster(self, control_plane):
-        control_plane.client.create(make_rayjob(suspend=True))
+        control_plane.client.create(make_rayjob(
+            suspend=True, shutdownAfterJobFinishes=True))
         assert wait_deployment_status(control_plane, "job1", "Suspended")
         job = job_of(control_plane)
         job.spec.suspend = False

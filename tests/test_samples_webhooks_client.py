@@ -26,6 +26,16 @@ def load_samples():
 
 
 class TestSampleYamls:
+    @pytest.fixture(autouse=True)
+    def _gates(self):
+        # samples exercising gated features name the gate in their header
+        # comment (tls-auth -> RayClusterMTLS); enable them like a deployed
+        # operator would via --feature-gates
+        import kuberay_amd.features as features
+        features.set_gate("RayClusterMTLS", True)
+        yield
+        features.reset()
+
     @pytest.mark.parametrize("fname,doc", load_samples(),
                              ids=[f for f, _ in load_samples()])
     def test_sample_applies_and_reconciles(self, control_plane, fname, doc):

@@ -184,7 +184,7 @@ class TestValidation:
                          "onFailure": {"policy": "DeleteNone"},
                          "deletionRules": [{"policy": "DeleteSelf",
                                             "condition": {"jobStatus": "FAILED"}}]}}})
-        assert any("mutually exclusive" in e for e in validate_rayjob_spec(job))
+        assert any("simultaneously" in e for e in validate_rayjob_spec(job))
 
     def test_rayservice_incremental_requires_gateway(self):
         from kuberay_amd.models import RayService
@@ -196,7 +196,7 @@ class TestValidation:
                 "rayClusterConfig": simple_raycluster("x").spec.to_dict(),
                 "upgradeStrategy": {"type": "NewClusterWithIncrementalUpgrade"}}})
         errs = validate_rayservice_spec(svc)
-        assert any("clusterUpgradeOptions" in e for e in errs)
+        assert any("ClusterUpgradeOptions are required" in e for e in errs)
 
 
 class TestStructuralSchemas:
@@ -271,13 +271,13 @@ class TestDeletionStrategyValidation:
                          "deletionRules": [{"policy": "DeleteCluster",
                                             "condition": {"jobStatus":
                                                           "SUCCEEDED"}}]})
-        assert any("mutually exclusive" in e for e in
+        assert any("simultaneously" in e for e in
                    validate_rayjob_spec(job))
 
     def test_legacy_requires_both_blocks(self):
         from kuberay_amd.utils.validation import validate_rayjob_spec
         job = self._job({"onSuccess": {"policy": "DeleteCluster"}})
-        assert any("both onSuccess and onFailure" in e for e in
+        assert any("OnSuccess and DeletionStrategy.OnFailure" in e for e in
                    validate_rayjob_spec(job))
 
     def test_non_terminal_job_status_rejected(self):
@@ -317,6 +317,13 @@ class TestDeletionStrategyValidation:
 class TestTLSValidation:
     """validation.go:446-480: user-set TLS env/mounts conflict with
     operator-managed TLS."""
+
+    @pytest.fixture(autouse=True)
+    def _gate(self):
+        import kuberay_amd.features as features
+        features.set_gate("RayClusterMTLS", True)
+        yield
+        features.reset()
 
     def _cluster(self, env=None, mount=None):
         from kuberay_amd.testing import simple_raycluster
@@ -460,10 +467,18 @@ class TestClusterUpgradeAndPriorityValidation:
 class TestCollectorOptionsValidation:
     """validation.go:1029-1064: collector env hygiene."""
 
+    @pytest.fixture(autouse=True)
+    def _gate(self):
+        import kuberay_amd.features as features
+        features.set_gate("RayClusterHistoryServer", True)
+        yield
+        features.reset()
+
     def _cluster(self, env):
         from kuberay_amd.testing import simple_raycluster
         return simple_raycluster("col", historyServerOptions={
-            "collectorOptions": {"env": env}})
+            "collectorOptions": {"image": "rocm/history-collector:1.0",
+                                 "env": env}})
 
     def test_managed_env_rejected(self):
         from kuberay_amd.utils.validation import validate_raycluster_spec
