@@ -60,11 +60,13 @@ def _parse_ts(ts) -> Optional[float]:
 
 class RayJobReconciler(Reconciler):
     def __init__(self, client: KubeClient, recorder: Optional[EventRecorder] = None,
-                 dashboard_factory: Optional[Callable] = None, metrics=None):
+                 dashboard_factory: Optional[Callable] = None, metrics=None,
+                 batch_scheduler=None):
         self.client = client
         self.recorder = recorder or NullRecorder()
         self.dashboard_factory = dashboard_factory or self._default_dashboard
         self.metrics = metrics
+        self.batch_scheduler = batch_scheduler
         self.requeue_seconds = REQUEUE_SECONDS
 
     @staticmethod
@@ -340,6 +342,12 @@ class RayJobReconciler(Reconciler):
     # ------------------------------------------------------------------
     def _handle_terminal(self, rayjob: RayJob) -> Result:
         """Deletion policies (rayjob_controller.go:1413-1560)."""
+        if self.batch_scheduler is not None:
+            # gang-scheduling artifacts (PodGroup) are no longer needed once
+            # the job is terminal (BatchScheduler.CleanupOnCompletion)
+            cluster = self._owned_cluster(rayjob)
+            if cluster is not None:
+                self.batch_scheduler.cleanup_on_completion(self.client, cluster)
         ds = rayjob.spec.deletion_strategy
         succeeded = rayjob.status.job_deployment_status == JDS.COMPLETE
 

@@ -19,7 +19,7 @@ from typing import Dict, Optional
 
 from ..kube import objects as k8s
 from ..kube.client import KubeClient
-from ..kube.store import AlreadyExistsError, ApiError
+from ..kube.store import AlreadyExistsError, ApiError, NotFoundError
 from ..models.raycluster import RayCluster
 from ..utils import constants as C
 from ..utils import names
@@ -164,6 +164,18 @@ class SchedulerPluginsBatchScheduler(BatchScheduler):
     def add_metadata_to_pod(self, client, cluster, group_name, pod) -> None:
         pod.metadata.ensure_labels()[POD_GROUP_ANNOTATION] = pod_group_name(cluster)
 
+    def cleanup_on_completion(self, client, cluster) -> None:
+        """Delete the PodGroup when the owning workload completes
+        (volcano_scheduler.go CleanupOnCompletion behavior)."""
+        server = getattr(client, "server", None)
+        if server is None:
+            return
+        try:
+            server.delete("PodGroup", cluster.metadata.namespace or "default",
+                          pod_group_name(cluster))
+        except NotFoundError:
+            pass
+
 
 class KaiBatchScheduler(BatchScheduler):
     """batchscheduler/kai analog: KAI's pod-grouper gang-groups pods by
@@ -209,6 +221,9 @@ class XgmiGangScheduler(BatchScheduler):
         if any(True for _ in self._gpu_groups(cluster)):
             self.inner.do_batch_scheduling_on_submission(client, cluster)
 
+    def cleanup_on_completion(self, client, cluster) -> None:
+        self.inner.cleanup_on_completion(client, cluster)
+
     # -- island scoring ----------------------------------------------------
     @staticmethod
     def _island_capacities(client) -> Dict[str, int]:
@@ -238,10 +253,21 @@ class XgmiGangScheduler(BatchScheduler):
             capacities[island] = capacities.get(island, 0) + count
         return capacities
 
+    # sentinel island value no node carries: a gang that fits no island is
+    # pinned to it so ALL its pods hold Pending — refused as a unit, never
+    # split across islands (a split ring would cross PCIe)
+    UNSCHEDULABLE_ISLAND = "xgmi.kuberay.amd/unschedulable"
+
     def _best_island(self, client, cluster: RayCluster,
                      group_name: str) -> Optional[str]:
         """Best-fit: the smallest island whose usable GPU capacity covers the
-        whole gang's demand, minimizing fragmentation of big islands."""
+        whole gang's demand, minimizing fragmentation of big islands.
+
+        Returns None when no island data is visible (single-node dev: the
+        preferred affinity still keeps gangs together), and
+        UNSCHEDULABLE_ISLAND when islands ARE known but none can hold the
+        gang — the gang must refuse, not split.
+        """
         group = next((g for g in cluster.spec.worker_group_specs
                       if g.group_name == group_name), None)
         if group is None:
@@ -253,9 +279,12 @@ class XgmiGangScheduler(BatchScheduler):
             return None
         if demand <= 0:
             return None
-        fits = [(cap, name) for name, cap in
-                self._island_capacities(client).items() if cap >= demand]
-        return min(fits)[1] if fits else None
+        capacities = self._island_capacities(client)
+        if not capacities:
+            return None
+        fits = [(cap, name) for name, cap in capacities.items()
+                if cap >= demand]
+        return min(fits)[1] if fits else self.UNSCHEDULABLE_ISLAND
 
     def add_metadata_to_pod(self, client, cluster, group_name, pod) -> None:
         is_gpu = pod_gpu_count(pod) > 0

@@ -50,6 +50,26 @@ class TestBenchDistributed:
         assert d["config"]["global_batch"] == 30
         assert d["config"]["parallelism"] == "dp2"
 
+    def test_world8_gloo(self):
+        """Eight ranks with tiny per-rank work — pre-proves rendezvous and
+        MAX/SUM aggregation at the driver's full N=8 scaling point so
+        `bench.py --gpus 8` just works when the driver gets an 8-GPU node."""
+        out = subprocess.run(
+            [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+             "--nproc-per-node", "8", "--master-addr", "127.0.0.1",
+             "--master-port", "29521", "bench.py", "--clusters", "4",
+             "--steps", "1", "--warmup", "0", "--workers-per-cluster", "1",
+             "--controller-workers", "2"],
+            capture_output=True, text=True, cwd=REPO, timeout=540,
+            env={**os.environ, "MASTER_ADDR": "127.0.0.1"})
+        assert out.returncode == 0, (out.stdout[-1500:], out.stderr[-1500:])
+        lines = [l for l in out.stdout.splitlines() if l.startswith("{")]
+        assert len(lines) == 1
+        d = json.loads(lines[0])
+        assert d["n_gpus"] == 8
+        assert d["config"]["global_batch"] == 32  # 8 ranks x 4 clusters
+        assert d["config"]["parallelism"] == "dp8"
+
     def test_world4_gloo(self):
         """Four ranks — same shape the driver uses for the N=4 scaling
         point; validates rendezvous + MAX/SUM aggregation at higher fan-out."""

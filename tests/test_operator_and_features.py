@@ -417,7 +417,11 @@ class TestXgmiIslandScoring:
             "nodeSelectorTerms"]
         assert terms[0]["matchExpressions"][0]["values"] == ["full-island0"]
 
-    def test_no_fit_keeps_preferred_affinity_only(self):
+    def test_gang_larger_than_every_island_refuses_not_splits(self):
+        """Islands are known but none can hold the gang: every gang pod is
+        pinned to the unschedulable sentinel so the gang holds Pending as a
+        unit — splitting the RCCL ring across islands (PCIe) is never an
+        outcome."""
         client = InMemoryClient()
         client.server.create(self._node("n1", "n1-island0", 2))
         cluster = simple_raycluster("demo", workers=4, gpus_per_worker=2)
@@ -425,8 +429,12 @@ class TestXgmiIslandScoring:
         sched = XgmiGangScheduler()
         pod = TestGangScheduling._pod(TestGangScheduling(), cluster)
         sched.add_metadata_to_pod(client, cluster, "default-group", pod)
-        assert "nodeAffinity" not in pod.spec.affinity
-        assert pod.spec.affinity["podAffinity"]  # co-location still preferred
+        terms = pod.spec.affinity["nodeAffinity"][
+            "requiredDuringSchedulingIgnoredDuringExecution"][
+            "nodeSelectorTerms"]
+        assert terms[0]["matchExpressions"][0]["values"] == \
+            [XgmiGangScheduler.UNSCHEDULABLE_ISLAND]
+        assert pod.spec.affinity["podAffinity"]  # co-location still declared
 
     def test_no_nodes_no_pin(self):
         client = InMemoryClient()
@@ -435,7 +443,84 @@ class TestXgmiIslandScoring:
         pod = TestGangScheduling._pod(TestGangScheduling(), cluster)
         XgmiGangScheduler().add_metadata_to_pod(client, cluster,
                                                 "default-group", pod)
+        # no island data at all (single-node dev): no hard pin, preferred
+        # affinity alone keeps the gang together
         assert "nodeAffinity" not in pod.spec.affinity
+
+    def test_mixed_cpu_gpu_groups_only_gpu_gets_gang(self):
+        """A cluster with one CPU and one GPU worker group: only the GPU
+        group's pods carry gang metadata; CPU pods schedule freely."""
+        from kuberay_amd.models.raycluster import WorkerGroupSpec
+        client = InMemoryClient()
+        client.server.create(self._node("n1", "n1-island0", 8))
+        cluster = simple_raycluster("demo", workers=2, gpus_per_worker=2)
+        cluster.spec.worker_group_specs.append(WorkerGroupSpec.from_dict({
+            "groupName": "cpu-group", "replicas": 2, "minReplicas": 0,
+            "maxReplicas": 4, "rayStartParams": {},
+            "template": {"spec": {"containers": [{
+                "name": "ray-worker", "image": "img",
+                "resources": {"limits": {"cpu": "4", "memory": "8Gi"}}}]}}}))
+        cluster = client.create(cluster)
+        sched = XgmiGangScheduler()
+        gpu_pod = TestGangScheduling._pod(TestGangScheduling(), cluster)
+        sched.add_metadata_to_pod(client, cluster, "default-group", gpu_pod)
+        assert gpu_pod.metadata.labels.get("ray.io/xgmi-gang")
+
+        from kuberay_amd.kube import objects as k8s
+        cpu_pod = k8s.Pod(
+            metadata=k8s.ObjectMeta(name="cpu-w", namespace="default"),
+            spec=k8s.PodSpec(containers=[k8s.Container(
+                name="ray-worker", image="img",
+                resources=k8s.ResourceRequirements(
+                    limits={"cpu": "4", "memory": "8Gi"}))]))
+        sched.add_metadata_to_pod(client, cluster, "cpu-group", cpu_pod)
+        assert not (cpu_pod.metadata.labels or {}).get("ray.io/xgmi-gang")
+        assert cpu_pod.spec.affinity is None
+
+    def test_pod_group_cleanup_on_completion(self):
+        """cleanup_on_completion deletes the gang's PodGroup (volcano
+        CleanupOnCompletion behavior), and is idempotent."""
+        client = InMemoryClient()
+        cluster = simple_raycluster("demo", workers=2, gpus_per_worker=1)
+        cluster = client.create(cluster)
+        sched = XgmiGangScheduler()
+        sched.do_batch_scheduling_on_submission(client, cluster)
+        assert client.server.count("PodGroup") == 1
+        sched.cleanup_on_completion(client, cluster)
+        assert client.server.count("PodGroup") == 0
+        sched.cleanup_on_completion(client, cluster)  # idempotent
+
+    def test_rayjob_terminal_cleans_pod_group(self):
+        """End-to-end: a RayJob whose cluster was gang-scheduled removes the
+        PodGroup once the job is terminal."""
+        import time as _time
+
+        from kuberay_amd.models import RayJob
+        from kuberay_amd.testing import ControlPlane
+        cp = ControlPlane(kubelet_delay=0.01, job_runtime=0.1,
+                          poll_seconds=0.05)
+        sched = XgmiGangScheduler()
+        cp.raycluster_reconciler.batch_scheduler = sched
+        cp.rayjob_reconciler.batch_scheduler = sched
+        cp.start()
+        try:
+            cluster_spec = simple_raycluster(
+                "x", workers=1, gpus_per_worker=1).spec.to_dict()
+            cp.client.create(RayJob.from_dict({
+                "apiVersion": "ray.io/v1", "kind": "RayJob",
+                "metadata": {"name": "gangjob", "namespace": "default"},
+                "spec": {"entrypoint": "python x.py",
+                         "rayClusterSpec": cluster_spec}}))
+            assert cp.wait_for(lambda: cp.server.count("PodGroup") == 1,
+                               timeout=15)
+            assert cp.wait_for(
+                lambda: (cp.client.get(RayJob, "default", "gangjob")
+                         .status.job_deployment_status == "Complete"),
+                timeout=25)
+            assert cp.wait_for(lambda: cp.server.count("PodGroup") == 0,
+                               timeout=15)
+        finally:
+            cp.stop()
 
 
 class TestKubernetesWASEndToEnd:
