@@ -128,9 +128,9 @@ def create_history_app(storage: Optional[StorageReader] = None) -> FastAPI:
         """router.go:184-190 — per-function aggregate of task states."""
         summary: Dict[str, Dict] = {}
         for t in _state(ns, cluster, session).tasks.values():
-            name = t.get("func_or_class_name") or t.get("name") or "unknown"
+            name = t.get("funcOrClassName") or t.get("name") or "unknown"
             entry = summary.setdefault(name, {
-                "func_or_class_name": name, "type": t.get("type"),
+                "func_or_class_name": name, "type": t.get("taskType"),
                 "state_counts": {}})
             state = t.get("state") or "UNKNOWN"
             entry["state_counts"][state] = \
@@ -140,12 +140,36 @@ def create_history_app(storage: Optional[StorageReader] = None) -> FastAPI:
                               key=lambda e: e["func_or_class_name"]),
             "total_tasks": len(_state(ns, cluster, session).tasks)}}
 
+    @app.get("/api/sessions/{ns}/{cluster}/{session}/tasks/timeline")
+    def tasks_timeline(ns: str, cluster: str, session: str,
+                       job_id: Optional[str] = None):
+        """Ray /api/v0/tasks/timeline analog (timeline.go:13)."""
+        return _state(ns, cluster, session).timeline(job_id)
+
     @app.get("/api/sessions/{ns}/{cluster}/{session}/tasks/{task_id}")
     def task_detail(ns: str, cluster: str, session: str, task_id: str):
-        task = _state(ns, cluster, session).tasks.get(task_id)
-        if task is None:
+        """Latest attempt for the task, with every attempt inline (the
+        reference lists one row per attempt; detail groups them)."""
+        state = _state(ns, cluster, session)
+        attempts = sorted(
+            (t for t in state.tasks.values() if t.get("taskId") == task_id),
+            key=lambda t: t.get("taskAttempt", 0))
+        if not attempts:
             raise HTTPException(404, f"task {task_id} not found")
-        return {"result": True, "data": {"detail": task}}
+        detail = dict(attempts[-1])
+        detail["attempts"] = attempts
+        return {"result": True, "data": {"detail": detail}}
+
+    @app.get("/api/sessions/{ns}/{cluster}/{session}/events")
+    def log_events(ns: str, cluster: str, session: str,
+                   job_id: Optional[str] = None):
+        """Dashboard /events analog (log_event_reader.go): log events
+        grouped by job."""
+        ev = _state(ns, cluster, session).log_events
+        if job_id is not None:
+            return {"result": True,
+                    "data": {"events": {job_id: ev.get(job_id, [])}}}
+        return {"result": True, "data": {"events": ev}}
 
     @app.get("/api/sessions/{ns}/{cluster}/{session}/actors")
     def actors(ns: str, cluster: str, session: str,
@@ -168,6 +192,13 @@ def create_history_app(storage: Optional[StorageReader] = None) -> FastAPI:
             raise HTTPException(404, f"node {node_id} not found")
         return {"result": True, "data": {"detail": node}}
 
+    @app.get("/api/sessions/{ns}/{cluster}/{session}/actors/{actor_id}")
+    def actor_detail(ns: str, cluster: str, session: str, actor_id: str):
+        actor = _state(ns, cluster, session).actors.get(actor_id)
+        if actor is None:
+            raise HTTPException(404, f"actor {actor_id} not found")
+        return {"result": True, "data": {"detail": actor}}
+
     @app.get("/api/sessions/{ns}/{cluster}/{session}/cluster_status")
     def cluster_status(ns: str, cluster: str, session: str):
         """cluster_status.go analog: autoscaler-style summary built from the
@@ -184,21 +215,22 @@ def create_history_app(storage: Optional[StorageReader] = None) -> FastAPI:
         for item in list(state.tasks.values()) + list(state.actors.values()):
             if (item.get("state") or "").upper() not in pending_states:
                 continue
-            res = item.get("required_resources") or {"CPU": 1}
+            res = item.get("requiredResources") or {"CPU": 1}
             key = json.dumps(res, sort_keys=True)
             entry = demands.setdefault(key, {"resources": res, "count": 0})
             entry["count"] += 1
         lines = ["======== Cluster status (replayed) ========",
                  f"Active: {len(alive)} node(s)"]
         for node in failed[:20]:
-            lines.append(f"Failed: {node.get('node_ip') or node['node_id']}")
+            lines.append(
+                f"Failed: {node.get('nodeIpAddress') or node['nodeId']}")
         if demands:
             lines.append("Pending demands:")
             for d in demands.values():
                 lines.append(f"  {d['resources']}: {d['count']}+")
         return {"result": True, "data": {"clusterStatus": {
             "activeNodes": len(alive),
-            "failedNodes": [n["node_id"] for n in failed],
+            "failedNodes": [n["nodeId"] for n in failed],
             "pendingDemands": list(demands.values()),
             "text": "\n".join(lines)}}}
 
