@@ -77,6 +77,37 @@ _msg("GetRequest", [("name", "string"), ("namespace", "string")])
 _msg("ListRequest", [("namespace", "string")])
 _msg("DeleteRequest", [("name", "string"), ("namespace", "string")])
 _msg("Empty", [])
+# error.proto Status analog (error codes ride gRPC status; this shape is for
+# HTTP-gateway error bodies)
+_msg("Status", [("error", "string"), ("code", "int32")])
+# job_submission.proto:73-179 shapes
+_msg("RayJobSubmission", [("entrypoint", "string"),
+                          ("submission_id", "string"),
+                          ("runtime_env", "string"),
+                          ("metadata_json", "string"),
+                          ("entrypoint_num_cpus", "int32"),
+                          ("entrypoint_num_gpus", "int32")])
+_msg("SubmitRayJobRequest", [("namespace", "string"),
+                             ("clustername", "string"),
+                             ("jobsubmission", "RayJobSubmission")])
+_msg("SubmitRayJobReply", [("submission_id", "string")])
+_msg("JobSubmissionQuery", [("namespace", "string"),
+                            ("clustername", "string"),
+                            ("submissionid", "string")])
+_msg("JobSubmissionInfo", [("submission_id", "string"),
+                           ("status", "string"), ("entrypoint", "string"),
+                           ("message", "string"), ("error_type", "string"),
+                           ("start_time", "string"), ("end_time", "string")])
+_msg("ListJobSubmissionInfo",
+     [("submissions", "repeated:JobSubmissionInfo")])
+_msg("GetJobLogReply", [("log", "string")])
+# config.proto ImageTemplateService shapes (:138-231)
+_msg("ImageTemplate", [("name", "string"), ("namespace", "string"),
+                       ("base_image", "string"),
+                       ("pip_packages", "repeated:string"),
+                       ("environment_variables_json", "string")])
+_msg("ListImageTemplateResponse",
+     [("image_templates", "repeated:ImageTemplate")])
 _msg("ListClusterResponse", [("clusters", "repeated:Cluster")])
 _msg("ListComputeTemplateResponse",
      [("compute_templates", "repeated:ComputeTemplate")])
@@ -104,6 +135,16 @@ ListComputeTemplateResponse = _cls("ListComputeTemplateResponse")
 ListRayJobResponse = _cls("ListRayJobResponse")
 RayServiceMsg = _cls("RayServiceMsg")
 ListRayServiceResponse = _cls("ListRayServiceResponse")
+Status = _cls("Status")
+RayJobSubmission = _cls("RayJobSubmission")
+SubmitRayJobRequest = _cls("SubmitRayJobRequest")
+SubmitRayJobReply = _cls("SubmitRayJobReply")
+JobSubmissionQuery = _cls("JobSubmissionQuery")
+JobSubmissionInfo = _cls("JobSubmissionInfo")
+ListJobSubmissionInfo = _cls("ListJobSubmissionInfo")
+GetJobLogReply = _cls("GetJobLogReply")
+ImageTemplate = _cls("ImageTemplate")
+ListImageTemplateResponse = _cls("ListImageTemplateResponse")
 
 
 # ---------------------------------------------------------------------------
@@ -111,11 +152,34 @@ ListRayServiceResponse = _cls("ListRayServiceResponse")
 # ---------------------------------------------------------------------------
 
 class _Service:
-    """Shared backend for the three gRPC services."""
+    """Shared backend for the six gRPC services."""
 
-    def __init__(self, client):
+    def __init__(self, client, dashboard_factory=None):
         from ..kube.client import InMemoryClient
         self.client = client or InMemoryClient()
+        self.dashboard_factory = dashboard_factory
+
+    def _dashboard_for_cluster(self, context, namespace: str,
+                               cluster_name: str):
+        """Resolve a RayCluster's dashboard (same seam as the HTTP
+        job-submission proxy in app.py)."""
+        from ..models import RayCluster
+        from ..utils import constants as C
+        from ..utils import names
+        rc = self.client.try_get(RayCluster, namespace or "default",
+                                 cluster_name)
+        if rc is None:
+            context.abort(grpc.StatusCode.NOT_FOUND,
+                          f"cluster {cluster_name} not found")
+        if self.dashboard_factory is not None:
+            return self.dashboard_factory(
+                f"{cluster_name}.{namespace}:{C.DEFAULT_DASHBOARD_PORT}")
+        from ..utils.dashboard_client import RayDashboardClient
+        head_svc = names.head_service_name(C.KIND_RAYCLUSTER, rc.spec,
+                                           cluster_name)
+        return RayDashboardClient(
+            f"{head_svc}.{namespace}.svc.{names.cluster_domain_name()}:"
+            f"{C.DEFAULT_DASHBOARD_PORT}")
 
     def _templates(self, namespace: str) -> Dict[str, Dict[str, Any]]:
         from ..kube import objects as k8s
@@ -302,6 +366,203 @@ class _Service:
             context.abort(grpc.StatusCode.NOT_FOUND, "not found")
         return Empty()
 
+    def update_ray_service(self, request, context):
+        """serve.proto:39 UpdateRayService — replace serveConfigV2/spec."""
+        from ..models import RayService
+        ns = request.namespace or "default"
+
+        def mutate(svc):
+            if request.serve_config_v2:
+                svc.spec.serve_config_v2 = request.serve_config_v2
+            if request.spec_json:
+                spec = json.loads(request.spec_json)
+                fresh = RayService.from_dict(
+                    {"apiVersion": "ray.io/v1", "kind": "RayService",
+                     "metadata": {"name": request.name, "namespace": ns},
+                     "spec": spec})
+                svc.spec = fresh.spec
+
+        from ..kube.store import NotFoundError
+        try:
+            updated = self.client.update_with_retry(RayService, ns,
+                                                    request.name, mutate)
+        except NotFoundError:
+            context.abort(grpc.StatusCode.NOT_FOUND,
+                          f"service {request.name} not found")
+        return self._service_msg(updated)
+
+    # -- ListAll* variants (cluster.proto:50, config.proto:50, job.proto:52,
+    # serve.proto:60 — cross-namespace listings) -------------------------
+    def list_all_clusters(self, request, context):
+        from ..models import RayCluster
+        out = ListClusterResponse()
+        for rc in self.client.list(RayCluster, None):
+            out.clusters.append(self._cluster_msg(rc))
+        return out
+
+    def list_all_ray_jobs(self, request, context):
+        from ..models import RayJob
+        out = ListRayJobResponse()
+        for job in self.client.list(RayJob, None):
+            out.jobs.append(self._job_msg(job))
+        return out
+
+    def list_all_ray_services(self, request, context):
+        from ..models import RayService
+        out = ListRayServiceResponse()
+        for svc in self.client.list(RayService, None):
+            out.services.append(self._service_msg(svc))
+        return out
+
+    def get_compute_template(self, request, context):
+        t = self._templates(request.namespace or "default").get(request.name)
+        if t is None:
+            context.abort(grpc.StatusCode.NOT_FOUND,
+                          f"compute template {request.name} not found")
+        return ComputeTemplate(
+            name=t["name"], namespace=t.get("namespace") or "",
+            cpu=t["cpu"], memory=t["memory"], gpu=t["gpu"],
+            gpu_accelerator=t["gpuAccelerator"])
+
+    def list_all_compute_templates(self, request, context):
+        from ..kube import objects as k8s
+        from . import converters as conv
+        out = ListComputeTemplateResponse()
+        for cm in self.client.list(k8s.ConfigMap, None):
+            if (cm.metadata.labels or {}).get(conv.COMPUTE_TEMPLATE_LABEL):
+                t = conv.configmap_to_compute_template(cm)
+                out.compute_templates.append(ComputeTemplate(
+                    name=t["name"], namespace=t.get("namespace") or "",
+                    cpu=t["cpu"], memory=t["memory"], gpu=t["gpu"],
+                    gpu_accelerator=t["gpuAccelerator"]))
+        return out
+
+    # -- image templates (config.proto:138-231 ImageTemplateService) -----
+    IMAGE_TEMPLATE_LABEL = "ray.io/image-template"
+
+    def create_image_template(self, request, context):
+        from ..kube import objects as k8s
+        ns = request.namespace or "default"
+        cm = k8s.ConfigMap(
+            metadata=k8s.ObjectMeta(
+                name=f"imagetpl-{request.name}", namespace=ns,
+                labels={self.IMAGE_TEMPLATE_LABEL: request.name}),
+            data={"name": request.name,
+                  "baseImage": request.base_image,
+                  "pipPackages": json.dumps(list(request.pip_packages)),
+                  "environmentVariables":
+                      request.environment_variables_json or "{}"})
+        self.client.create(cm)
+        return request
+
+    def _image_template_msg(self, cm):
+        data = cm.data or {}
+        msg = ImageTemplate(
+            name=data.get("name", ""),
+            namespace=cm.metadata.namespace or "",
+            base_image=data.get("baseImage", ""),
+            environment_variables_json=data.get("environmentVariables", ""))
+        for p in json.loads(data.get("pipPackages", "[]")):
+            msg.pip_packages.append(p)
+        return msg
+
+    def get_image_template(self, request, context):
+        from ..kube import objects as k8s
+        cm = self.client.try_get(k8s.ConfigMap, request.namespace or "default",
+                                 f"imagetpl-{request.name}")
+        if cm is None:
+            context.abort(grpc.StatusCode.NOT_FOUND,
+                          f"image template {request.name} not found")
+        return self._image_template_msg(cm)
+
+    def list_image_templates(self, request, context):
+        from ..kube import objects as k8s
+        out = ListImageTemplateResponse()
+        for cm in self.client.list(k8s.ConfigMap,
+                                   request.namespace or "default"):
+            if (cm.metadata.labels or {}).get(self.IMAGE_TEMPLATE_LABEL):
+                out.image_templates.append(self._image_template_msg(cm))
+        return out
+
+    def delete_image_template(self, request, context):
+        from ..kube import objects as k8s
+        from ..kube.store import NotFoundError
+        try:
+            self.client.delete(k8s.ConfigMap, request.namespace or "default",
+                               f"imagetpl-{request.name}")
+        except NotFoundError:
+            context.abort(grpc.StatusCode.NOT_FOUND, "not found")
+        return Empty()
+
+    # -- job submissions (job_submission.proto:26-70) ---------------------
+    def submit_ray_job(self, request, context):
+        import yaml
+        dashboard = self._dashboard_for_cluster(
+            context, request.namespace, request.clustername)
+        sub = request.jobsubmission
+        body = {"entrypoint": sub.entrypoint}
+        if sub.submission_id:
+            body["submission_id"] = sub.submission_id
+        if sub.runtime_env:
+            body["runtime_env"] = yaml.safe_load(sub.runtime_env)
+        if sub.metadata_json:
+            body["metadata"] = json.loads(sub.metadata_json)
+        if sub.entrypoint_num_cpus:
+            body["entrypoint_num_cpus"] = sub.entrypoint_num_cpus
+        if sub.entrypoint_num_gpus:
+            body["entrypoint_num_gpus"] = sub.entrypoint_num_gpus
+        submission_id = dashboard.submit_job(body)
+        return SubmitRayJobReply(submission_id=submission_id)
+
+    def _submission_msg(self, info: Dict[str, Any]):
+        return JobSubmissionInfo(
+            submission_id=str(info.get("submission_id") or ""),
+            status=str(info.get("status") or ""),
+            entrypoint=str(info.get("entrypoint") or ""),
+            message=str(info.get("message") or ""),
+            error_type=str(info.get("error_type") or ""),
+            start_time=str(info.get("start_time") or ""),
+            end_time=str(info.get("end_time") or ""))
+
+    def get_job_details(self, request, context):
+        dashboard = self._dashboard_for_cluster(
+            context, request.namespace, request.clustername)
+        info = dashboard.get_job_info(request.submissionid)
+        if info is None:
+            context.abort(grpc.StatusCode.NOT_FOUND,
+                          f"submission {request.submissionid} not found")
+        return self._submission_msg(info)
+
+    def list_job_details(self, request, context):
+        dashboard = self._dashboard_for_cluster(
+            context, request.namespace, request.clustername)
+        out = ListJobSubmissionInfo()
+        for info in dashboard.list_jobs():
+            out.submissions.append(self._submission_msg(info))
+        return out
+
+    def get_job_log(self, request, context):
+        dashboard = self._dashboard_for_cluster(
+            context, request.namespace, request.clustername)
+        log = dashboard.get_job_log(request.submissionid)
+        return GetJobLogReply(log=log or "")
+
+    def stop_ray_job_submission(self, request, context):
+        dashboard = self._dashboard_for_cluster(
+            context, request.namespace, request.clustername)
+        dashboard.stop_job(request.submissionid)
+        return Empty()
+
+    def delete_ray_job_submission(self, request, context):
+        dashboard = self._dashboard_for_cluster(
+            context, request.namespace, request.clustername)
+        delete = getattr(dashboard, "delete_job", None)
+        if delete is not None:
+            delete(request.submissionid)
+        else:
+            dashboard.stop_job(request.submissionid)
+        return Empty()
+
 
 def _unary(handler, req_cls, resp_cls):
     return grpc.unary_unary_rpc_method_handler(
@@ -311,9 +572,13 @@ def _unary(handler, req_cls, resp_cls):
 
 
 def create_grpc_server(client=None, port: int = 8887,
-                       max_workers: int = 8) -> grpc.Server:
-    """Build (not start) the gRPC server with the three services."""
-    svc = _Service(client)
+                       max_workers: int = 8,
+                       dashboard_factory=None) -> grpc.Server:
+    """Build (not start) the gRPC server with the six v1 services
+    (cluster.proto, config.proto incl. ImageTemplateService, job.proto,
+    job_submission.proto, serve.proto; error.proto shapes ride gRPC
+    status)."""
+    svc = _Service(client, dashboard_factory=dashboard_factory)
     server = grpc.server(futures.ThreadPoolExecutor(max_workers=max_workers))
     handlers = {
         "kuberayamd.v1.ClusterService": {
@@ -321,24 +586,46 @@ def create_grpc_server(client=None, port: int = 8887,
             "GetCluster": _unary(svc.get_cluster, GetRequest, Cluster),
             "ListCluster": _unary(svc.list_clusters, ListRequest,
                                   ListClusterResponse),
+            "ListAllClusters": _unary(svc.list_all_clusters, Empty,
+                                      ListClusterResponse),
             "DeleteCluster": _unary(svc.delete_cluster, DeleteRequest, Empty),
         },
         "kuberayamd.v1.ComputeTemplateService": {
             "CreateComputeTemplate": _unary(svc.create_compute_template,
                                             ComputeTemplate, ComputeTemplate),
+            "GetComputeTemplate": _unary(svc.get_compute_template,
+                                         GetRequest, ComputeTemplate),
             "ListComputeTemplate": _unary(svc.list_compute_templates,
                                           ListRequest,
                                           ListComputeTemplateResponse),
+            "ListAllComputeTemplates": _unary(svc.list_all_compute_templates,
+                                              Empty,
+                                              ListComputeTemplateResponse),
             "DeleteComputeTemplate": _unary(svc.delete_compute_template,
                                             DeleteRequest, Empty),
         },
+        "kuberayamd.v1.ImageTemplateService": {
+            "CreateImageTemplate": _unary(svc.create_image_template,
+                                          ImageTemplate, ImageTemplate),
+            "GetImageTemplate": _unary(svc.get_image_template, GetRequest,
+                                       ImageTemplate),
+            "ListImageTemplates": _unary(svc.list_image_templates,
+                                         ListRequest,
+                                         ListImageTemplateResponse),
+            "DeleteImageTemplate": _unary(svc.delete_image_template,
+                                          DeleteRequest, Empty),
+        },
         "kuberayamd.v1.RayServeService": {
             "CreateRayService": _unary(svc.create_ray_service, RayServiceMsg,
+                                       RayServiceMsg),
+            "UpdateRayService": _unary(svc.update_ray_service, RayServiceMsg,
                                        RayServiceMsg),
             "GetRayService": _unary(svc.get_ray_service, GetRequest,
                                     RayServiceMsg),
             "ListRayServices": _unary(svc.list_ray_services, ListRequest,
                                       ListRayServiceResponse),
+            "ListAllRayServices": _unary(svc.list_all_ray_services, Empty,
+                                         ListRayServiceResponse),
             "DeleteRayService": _unary(svc.delete_ray_service, DeleteRequest,
                                        Empty),
         },
@@ -347,7 +634,24 @@ def create_grpc_server(client=None, port: int = 8887,
             "GetRayJob": _unary(svc.get_ray_job, GetRequest, RayJobMsg),
             "ListRayJob": _unary(svc.list_ray_jobs, ListRequest,
                                  ListRayJobResponse),
+            "ListAllRayJobs": _unary(svc.list_all_ray_jobs, Empty,
+                                     ListRayJobResponse),
             "DeleteRayJob": _unary(svc.delete_ray_job, DeleteRequest, Empty),
+        },
+        "kuberayamd.v1.RayJobSubmissionService": {
+            "SubmitRayJob": _unary(svc.submit_ray_job, SubmitRayJobRequest,
+                                   SubmitRayJobReply),
+            "GetJobDetails": _unary(svc.get_job_details, JobSubmissionQuery,
+                                    JobSubmissionInfo),
+            "GetJobLog": _unary(svc.get_job_log, JobSubmissionQuery,
+                                GetJobLogReply),
+            "ListJobDetails": _unary(svc.list_job_details,
+                                     JobSubmissionQuery,
+                                     ListJobSubmissionInfo),
+            "StopRayJob": _unary(svc.stop_ray_job_submission,
+                                 JobSubmissionQuery, Empty),
+            "DeleteRayJob": _unary(svc.delete_ray_job_submission,
+                                   JobSubmissionQuery, Empty),
         },
     }
     for service_name, methods in handlers.items():
