@@ -112,6 +112,17 @@ class RayServiceReconciler(Reconciler):
         active = self._get_cluster(svc, svc.status.active_service_status.ray_cluster_name)
         pending = self._get_cluster(svc, svc.status.pending_service_status.ray_cluster_name)
 
+        # prune serve-config cache entries for clusters that no longer exist
+        # (cleanUpServeConfigCache analog — bounds memory and guarantees a
+        # recreated cluster is treated as config-less)
+        live_uids = {c.metadata.uid or c.metadata.name
+                     for c in (active, pending) if c is not None}
+        with self._lock:
+            for key in [k for k in self._serve_config_cache
+                        if k[0] == namespace and k[1] == name
+                        and k[2] not in live_uids]:
+                self._serve_config_cache.pop(key, None)
+
         # serve config goes to the pending cluster first, else the active one
         target, is_pending = (pending, True) if pending is not None else (active, False)
         ready = False
@@ -341,8 +352,19 @@ class RayServiceReconciler(Reconciler):
         namespace = svc.metadata.namespace or "default"
         config = parse_serve_config_v2(svc.spec.serve_config_v2 or "")
         config_hash = json_hash(config)
-        cache_key = (namespace, svc.metadata.name, cluster.metadata.name)
+        # Keyed by cluster UID, not name: a cluster deleted and recreated
+        # under the same name is a FRESH serve controller that has never
+        # seen the config — a name-keyed cache would silently skip
+        # resubmission (reference cleanUpServeConfigCache,
+        # rayservice_controller.go:1896-1925).
+        cache_key = (namespace, svc.metadata.name,
+                     cluster.metadata.uid or cluster.metadata.name)
         dashboard = self._dashboard_for(svc, cluster)
+        # Submission failures invalidate the cache (retry next reconcile);
+        # STATUS-read failures must NOT — resubmitting a config the serve
+        # controller already has resets deploy progress for nothing
+        # (reference keeps the cache across getAndCheckServeStatus errors,
+        # rayservice_controller.go:1551-1642).
         try:
             if self._serve_config_cache.get(cache_key) != config_hash:
                 dashboard.update_serve_applications(config)
@@ -350,10 +372,16 @@ class RayServiceReconciler(Reconciler):
                 self.recorder.eventf(svc, "Normal", "SubmittedServeConfig",
                                      "Submitted serve config to RayCluster %s",
                                      cluster.metadata.name)
+        except DashboardClientError as e:
+            self.recorder.eventf(svc, "Warning", "FailedServeConfigSubmission",
+                                 str(e))
+            self._serve_config_cache.pop(cache_key, None)
+            return False
+        try:
             details = dashboard.get_serve_applications()
         except DashboardClientError as e:
-            self.recorder.eventf(svc, "Warning", "ServeConfigOrStatusFailed", str(e))
-            self._serve_config_cache.pop(cache_key, None)
+            self.recorder.eventf(svc, "Warning", "FailedServeStatusCheck",
+                                 str(e))
             return False
 
         target_status = (svc.status.pending_service_status if is_pending
