@@ -9,11 +9,13 @@ endpoint (``python -m kuberay_amd.kube.httpserver``).
 """
 from __future__ import annotations
 
+import base64
+import binascii
 import json
 import re
 import threading
 from http.server import BaseHTTPRequestHandler, ThreadingHTTPServer
-from typing import Optional, Tuple
+from typing import Any, Dict, List, Optional, Tuple
 from urllib.parse import parse_qs, urlparse
 
 from .rest import RESOURCES
@@ -42,6 +44,134 @@ def _parse_path(path: str) -> Optional[Tuple[str, Optional[str], Optional[str], 
     if name == "status":  # /plural/name/status matched oddly
         name, sub = None, "status"
     return kind, m.group("ns"), name, sub
+
+
+class JsonPatchTestFailed(Exception):
+    """RFC 6902 `test` op failed → 409 like the real apiserver."""
+
+
+def _resolve_pointer(doc: Any, path: str, *, parent: bool = False):
+    """JSON-pointer walk; returns (container, last_token)."""
+    if path == "":
+        return None, None
+    tokens = [t.replace("~1", "/").replace("~0", "~")
+              for t in path.lstrip("/").split("/")]
+    node = doc
+    walk = tokens[:-1] if parent else tokens
+    for token in walk:
+        if isinstance(node, list):
+            node = node[int(token)]
+        elif isinstance(node, dict):
+            if token not in node:
+                raise ValueError(f"path {path} not found")
+            node = node[token]
+        else:
+            raise ValueError(f"path {path} not traversable")
+    return node, tokens[-1]
+
+
+def apply_json_patch(doc: Dict[str, Any],
+                     ops: List[Dict[str, Any]]) -> Dict[str, Any]:
+    """RFC 6902 add/remove/replace/test/copy/move on a deep copy."""
+    out = json.loads(json.dumps(doc))
+    if not isinstance(ops, list):
+        raise ValueError("json-patch body must be an array of operations")
+    for op in ops:
+        kind, path = op.get("op"), op.get("path", "")
+        container, last = _resolve_pointer(out, path, parent=True)
+        if kind == "add":
+            if isinstance(container, list):
+                idx = len(container) if last == "-" else int(last)
+                container.insert(idx, op.get("value"))
+            else:
+                container[last] = op.get("value")
+        elif kind == "replace":
+            if isinstance(container, list):
+                container[int(last)] = op.get("value")
+            else:
+                if last not in container:
+                    raise ValueError(f"replace path {path} not found")
+                container[last] = op.get("value")
+        elif kind == "remove":
+            if isinstance(container, list):
+                container.pop(int(last))
+            else:
+                if last not in container:
+                    raise ValueError(f"remove path {path} not found")
+                del container[last]
+        elif kind == "test":
+            node, _ = _resolve_pointer(out, path)
+            if node != op.get("value"):
+                raise JsonPatchTestFailed(
+                    f"test failed at {path}: {node!r} != {op.get('value')!r}")
+        elif kind in ("copy", "move"):
+            src_node, _ = _resolve_pointer(out, op.get("from", ""))
+            value = json.loads(json.dumps(src_node))
+            if kind == "move":
+                src_parent, src_last = _resolve_pointer(
+                    out, op.get("from", ""), parent=True)
+                if isinstance(src_parent, list):
+                    src_parent.pop(int(src_last))
+                else:
+                    del src_parent[src_last]
+            if isinstance(container, list):
+                idx = len(container) if last == "-" else int(last)
+                container.insert(idx, value)
+            else:
+                container[last] = value
+        else:
+            raise ValueError(f"unsupported json-patch op {kind!r}")
+    return out
+
+
+# strategic-merge list merge keys (subset of the k8s patchMergeKey tags
+# that matter for Ray pod specs)
+_MERGE_KEYS = {"containers": "name", "initContainers": "name",
+               "volumes": "name", "env": "name", "volumeMounts": "name",
+               "ports": "containerPort", "imagePullSecrets": "name",
+               "workerGroupSpecs": "groupName"}
+
+
+def strategic_merge(current: Any, patch: Any, field: str = "") -> Any:
+    """Strategic merge patch: dict-recursive like merge patch, but lists
+    with a known merge key merge element-wise instead of replacing, and
+    ``$patch: delete`` directives remove elements."""
+    if isinstance(patch, dict) and isinstance(current, dict):
+        out = dict(current)
+        for key, val in patch.items():
+            if val is None:
+                out.pop(key, None)
+            elif key in out:
+                out[key] = strategic_merge(out[key], val, key)
+            else:
+                out[key] = val
+        return out
+    if isinstance(patch, list) and isinstance(current, list) and \
+            field in _MERGE_KEYS:
+        merge_key = _MERGE_KEYS[field]
+        out = list(current)
+        index = {item.get(merge_key): i for i, item in enumerate(out)
+                 if isinstance(item, dict)}
+        for elem in patch:
+            if not isinstance(elem, dict):
+                out.append(elem)
+                continue
+            key_val = elem.get(merge_key)
+            if elem.get("$patch") == "delete":
+                out = [o for o in out
+                       if not (isinstance(o, dict)
+                               and o.get(merge_key) == key_val)]
+                index = {item.get(merge_key): i for i, item in enumerate(out)
+                         if isinstance(item, dict)}
+                continue
+            if key_val in index:
+                out[index[key_val]] = strategic_merge(out[index[key_val]],
+                                                      elem)
+            else:
+                out.append(elem)
+                index[key_val] = len(out) - 1
+        return out
+    return patch
 
 
 class _Handler(BaseHTTPRequestHandler):
@@ -90,31 +220,101 @@ class _Handler(BaseHTTPRequestHandler):
                 return self._send_error_status(NotFoundError(f"{kind} {name}"))
             return self._send_json(200, obj)
         if params.get("watch", ["false"])[0] == "true":
-            return self._stream_watch(kind)
+            rv = params.get("resourceVersion", [None])[0]
+            bookmarks = params.get("allowWatchBookmarks",
+                                   ["false"])[0] == "true"
+            return self._stream_watch(kind, rv, bookmarks)
         selector = None
         if "labelSelector" in params:
             selector = dict(kv.split("=", 1)
                             for kv in params["labelSelector"][0].split(","))
         items = self.store.list(kind, ns, selector)
+        # chunked lists: ?limit=N&continue=token (apiserver pagination)
+        list_meta = {"resourceVersion": str(self.store.current_rv)}
+        limit = int(params.get("limit", ["0"])[0] or 0)
+        if limit > 0:
+            offset = 0
+            token = params.get("continue", [None])[0]
+            if token:
+                try:
+                    offset = int(base64.b64decode(token).decode())
+                except (ValueError, binascii.Error):
+                    return self._send_json(400, {
+                        "kind": "Status", "status": "Failure",
+                        "reason": "BadRequest",
+                        "message": "invalid continue token", "code": 400})
+            page = items[offset:offset + limit]
+            if offset + limit < len(items):
+                list_meta["continue"] = base64.b64encode(
+                    str(offset + limit).encode()).decode()
+            items = page
         self._send_json(200, {"kind": f"{kind}List", "apiVersion": "v1",
-                              "items": items})
+                              "metadata": list_meta, "items": items})
 
-    def _stream_watch(self, kind: str) -> None:
+    def _stream_watch(self, kind: str, resource_version: Optional[str],
+                      bookmarks: bool) -> None:
+        """Watch with kube-apiserver semantics: resume from rv (replay from
+        the event history), 410 Gone when the rv predates retained history,
+        optional BOOKMARK events carrying the current rv."""
+        start_rv = 0
+        replay = []
+        if resource_version:
+            try:
+                start_rv = int(resource_version)
+            except ValueError:
+                start_rv = 0
+            replay = self.store.events_since(start_rv, {kind})
+            if replay is None:
+                # too old — client must re-list (Expired)
+                return self._send_json(410, {
+                    "kind": "Status", "apiVersion": "v1",
+                    "status": "Failure", "reason": "Expired",
+                    "message": f"too old resource version: {start_rv}",
+                    "code": 410})
         watcher = self.store.watch({kind})
         try:
             self.send_response(200)
             self.send_header("Content-Type", "application/json")
             self.send_header("Transfer-Encoding", "chunked")
             self.end_headers()
-            while not self.server.stopping:  # type: ignore[attr-defined]
-                ev = watcher.next(timeout=0.5)
-                if ev is None:
-                    continue
-                event_type, obj = ev
+
+            def emit(event_type, obj):
                 line = json.dumps({"type": event_type, "object": obj}) + "\n"
                 data = line.encode()
                 self.wfile.write(f"{len(data):x}\r\n".encode() + data + b"\r\n")
                 self.wfile.flush()
+
+            seen_rv = start_rv
+            for event_type, obj in replay:
+                emit(event_type, obj)
+                try:
+                    seen_rv = max(seen_rv, int(
+                        obj.get("metadata", {}).get("resourceVersion") or 0))
+                except ValueError:
+                    pass
+            idle = 0.0
+            while not self.server.stopping:  # type: ignore[attr-defined]
+                ev = watcher.next(timeout=0.5)
+                if ev is None:
+                    idle += 0.5
+                    if bookmarks and idle >= 1.0:
+                        idle = 0.0
+                        emit("BOOKMARK", {
+                            "kind": kind, "apiVersion": "v1",
+                            "metadata": {"resourceVersion":
+                                         str(self.store.current_rv)}})
+                    continue
+                idle = 0.0
+                event_type, obj = ev
+                # skip events already delivered via replay
+                try:
+                    rv = int(obj.get("metadata", {})
+                             .get("resourceVersion") or 0)
+                except ValueError:
+                    rv = 0
+                if event_type != "DELETED" and 0 < rv <= seen_rv:
+                    continue
+                emit(event_type, obj)
             self.wfile.write(b"0\r\n\r\n")
         except (BrokenPipeError, ConnectionResetError, OSError):
             pass
@@ -150,13 +350,38 @@ class _Handler(BaseHTTPRequestHandler):
             self._send_error_status(e)
 
     def do_PATCH(self):  # noqa: N802
+        """Dispatch on Content-Type like the real apiserver:
+        merge-patch (RFC 7386), json-patch (RFC 6902) and
+        strategic-merge-patch (list merge by the `name` merge key)."""
         parsed = _parse_path(urlparse(self.path).path)
         if parsed is None:
             return self._send_json(404, {"message": "unknown path"})
         kind, ns, name, sub = parsed
+        content_type = (self.headers.get("Content-Type") or "").split(";")[0]
+        body = self._body()
         try:
-            self._send_json(200, self.store.patch_merge(
-                kind, ns or "default", name, self._body(), subresource=sub))
+            if content_type == "application/json-patch+json":
+                current = self.store.get(kind, ns or "default", name)
+                patched = apply_json_patch(current, body)
+                # PUT with the read rv → optimistic concurrency holds
+                self._send_json(200, self.store.update(patched,
+                                                       subresource=sub))
+            elif content_type == "application/strategic-merge-patch+json":
+                current = self.store.get(kind, ns or "default", name)
+                patched = strategic_merge(current, body)
+                self._send_json(200, self.store.update(patched,
+                                                       subresource=sub))
+            else:  # merge patch (default)
+                self._send_json(200, self.store.patch_merge(
+                    kind, ns or "default", name, body, subresource=sub))
+        except JsonPatchTestFailed as e:
+            self._send_json(409, {
+                "kind": "Status", "apiVersion": "v1", "status": "Failure",
+                "reason": "Conflict", "message": str(e), "code": 409})
+        except ValueError as e:
+            self._send_json(422, {
+                "kind": "Status", "apiVersion": "v1", "status": "Failure",
+                "reason": "Invalid", "message": str(e), "code": 422})
         except ApiError as e:
             self._send_error_status(e)
 

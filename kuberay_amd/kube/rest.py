@@ -26,6 +26,7 @@ from .store import (
     AlreadyExistsError,
     ApiError,
     ConflictError,
+    GoneError,
     NotFoundError,
     PodView,
     compute_pod_view,
@@ -175,6 +176,8 @@ class RestClient(KubeClient):
             if "AlreadyExists" in body:
                 raise AlreadyExistsError(body[:300])
             raise ConflictError(body[:300])
+        if resp.status_code == 410:
+            raise GoneError(resp.text[:300])
         if resp.status_code >= 400:
             raise ApiError(resp.status_code, resp.text[:500])
         return resp
@@ -291,14 +294,39 @@ class RestClient(KubeClient):
             self._path("Pod", namespace, name, "log"), params=params))
         return resp.text
 
-    def raw_watch_stream(self, kind: str, resource_version: Optional[str] = None):
-        """Generator of (event_type, obj) from a K8s watch request."""
+    def raw_list_with_rv(self, kind: str, namespace: Optional[str] = None):
+        """(items, list resourceVersion) — the rv a watch should resume
+        from after a re-list."""
+        resp = self._check(self._http.get(self._path(kind, namespace)))
+        body = resp.json()
+        items = body.get("items", [])
+        for o in items:
+            o.setdefault("kind", kind)
+        return items, (body.get("metadata") or {}).get("resourceVersion")
+
+    def raw_watch_stream(self, kind: str, resource_version: Optional[str] = None,
+                         allow_bookmarks: bool = False):
+        """Generator of (event_type, obj) from a K8s watch request.
+
+        Raises :class:`GoneError` when the apiserver answers 410 (the
+        requested resourceVersion was compacted away) — the caller must
+        re-list and re-watch from the fresh list rv. BOOKMARK events are
+        yielded too (type ``"BOOKMARK"``); callers use them to advance
+        their rv without real traffic.
+        """
         prefix, plural = RESOURCES[kind]
         params = {"watch": "true"}
         if resource_version:
             params["resourceVersion"] = resource_version
+        if allow_bookmarks:
+            params["allowWatchBookmarks"] = "true"
         with self._http.stream("GET", f"{prefix}/{plural}", params=params,
                                timeout=None) as resp:
+            if resp.status_code == 410:
+                raise GoneError(f"watch {kind} from rv {resource_version}: "
+                                "too old resource version")
+            if resp.status_code >= 400:
+                raise ApiError(resp.status_code, f"watch {kind} failed")
             for line in resp.iter_lines():
                 if not line:
                     continue
@@ -387,20 +415,33 @@ class RestApiServerAdapter:
         return w
 
     def _watch_loop(self, kind: str, watcher: "_AdapterWatcher") -> None:
+        """client-go Reflector ListAndWatch analog: list (seed + rv), watch
+        from that rv with bookmarks; on 410 Gone re-list IMMEDIATELY (no
+        backoff — the server told us exactly what to do), on transport
+        errors back off briefly."""
         rv: Optional[str] = None
         while not self._stopped.is_set():
             try:
-                # list first (seed + resourceVersion)
-                items = self._client.raw_list(kind)
+                # list first (seed + resourceVersion to watch from)
+                items, rv = self._client.raw_list_with_rv(kind)
                 for obj in items:
                     self._view_cache.apply("ADDED", obj)
                     watcher.push("ADDED", obj)
-                for event_type, obj in self._client.raw_watch_stream(kind, rv):
+                for event_type, obj in self._client.raw_watch_stream(
+                        kind, rv, allow_bookmarks=True):
                     if self._stopped.is_set():
                         return
-                    rv = obj.get("metadata", {}).get("resourceVersion", rv)
+                    new_rv = obj.get("metadata", {}).get("resourceVersion")
+                    if new_rv:
+                        rv = new_rv
+                    if event_type == "BOOKMARK":
+                        continue  # rv advanced; nothing to deliver
                     self._view_cache.apply(event_type, obj)
                     watcher.push(event_type, obj)
+            except GoneError:
+                logger.info("watch %s: rv %s expired (410); re-listing",
+                            kind, rv)
+                rv = None
             except Exception as e:
                 logger.warning("watch %s dropped (%s); reconnecting", kind, e)
                 self._stopped.wait(2.0)

@@ -65,6 +65,13 @@ class AlreadyExistsError(ApiError):
         super().__init__(409, message)
 
 
+class GoneError(ApiError):
+    """410 — watch resourceVersion too old; re-list and re-watch."""
+
+    def __init__(self, message: str = "resource version too old"):
+        super().__init__(410, message)
+
+
 def now_iso() -> str:
     return time.strftime("%Y-%m-%dT%H:%M:%SZ", time.gmtime())
 
@@ -240,11 +247,17 @@ def default_backend():
 class InMemoryApiServer:
     """Thread-safe object store with Kubernetes verb semantics."""
 
+    # events retained for resourceVersion-based watch resumption; a client
+    # asking for an rv older than this window gets 410 Gone (real-apiserver
+    # etcd-compaction semantics)
+    EVENT_HISTORY_LIMIT = 1024
+
     def __init__(self, backend=None) -> None:
         self._lock = threading.RLock()
         self._backend = backend if backend is not None else default_backend()
         self._rv = 0
         self._watchers: List["Watcher"] = []
+        self._event_history: deque = deque(maxlen=self.EVENT_HISTORY_LIMIT)
 
     @property
     def backend_name(self) -> str:
@@ -256,8 +269,36 @@ class InMemoryApiServer:
         return str(self._rv)
 
     def _notify(self, event_type: str, obj: Dict[str, Any]) -> None:
+        try:
+            rv = int(obj.get("metadata", {}).get("resourceVersion") or self._rv)
+        except (TypeError, ValueError):
+            rv = self._rv
+        if event_type == "DELETED":
+            # real apiserver: a delete is itself a new revision
+            rv = self._rv
+        self._event_history.append((rv, event_type, obj))
         for w in list(self._watchers):
             w.push(event_type, obj)
+
+    @property
+    def current_rv(self) -> int:
+        with self._lock:
+            return self._rv
+
+    def events_since(self, rv: int, kinds: Optional[set] = None):
+        """Events with resourceVersion > rv, oldest first.
+
+        Returns None when the requested rv predates the retained history —
+        the watch must be answered with 410 Gone and the client re-lists
+        (kube-apiserver 'too old resource version' semantics).
+        """
+        with self._lock:
+            hist = self._event_history
+            if len(hist) == hist.maxlen and hist[0][0] > rv + 1:
+                return None
+            return [(etype, jsoncopy(obj)) for (erv, etype, obj) in hist
+                    if erv > rv and (kinds is None
+                                     or obj.get("kind") in kinds)]
 
     @staticmethod
     def _key_of(obj: Dict[str, Any]) -> Key:
