@@ -82,9 +82,18 @@ def _container_from_template(name: str, image: str, template: Dict[str, Any],
 
 def api_cluster_to_raycluster(namespace: str, cluster: Dict[str, Any],
                               templates: Dict[str, Dict[str, Any]]) -> RayCluster:
+    def resolved(group: Dict[str, Any], default: Dict[str, Any]):
+        """Compute template resolved + inline cpu/memory/gpu overrides (the
+        dashboard UI submits inline resources without a template)."""
+        t = dict(templates.get(group.get("computeTemplate", ""), default))
+        for key in ("cpu", "memory", "gpu"):
+            if group.get(key) is not None:
+                t[key] = group[key]
+        return t
+
     spec = cluster.get("clusterSpec", {})
     head = spec.get("headGroupSpec", {})
-    head_template = templates.get(head.get("computeTemplate", ""), {"cpu": 1, "memory": 2})
+    head_template = resolved(head, {"cpu": 1, "memory": 2})
     head_pod_spec: Dict[str, Any] = {"containers": [
         _container_from_template("ray-head", head.get("image")
                                  or C.DEFAULT_RAY_ROCM_IMAGE,
@@ -99,7 +108,7 @@ def api_cluster_to_raycluster(namespace: str, cluster: Dict[str, Any],
     }
     worker_groups = []
     for wg in spec.get("workerGroupSpec", []) or []:
-        t = templates.get(wg.get("computeTemplate", ""), {"cpu": 1, "memory": 1})
+        t = resolved(wg, {"cpu": 1, "memory": 1})
         worker_groups.append({
             "groupName": wg.get("groupName", "worker-group"),
             "replicas": wg.get("replicas", 1),

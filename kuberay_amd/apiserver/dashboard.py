@@ -1,6 +1,11 @@
-"""Web dashboard (reference: dashboard/ — Next.js list/create/delete UI over
-the APIServer). Here a dependency-free single-page app served at ``/`` by
-the same FastAPI apiserver, using the v1 endpoints via fetch()."""
+"""Web dashboard (reference: dashboard/src/app — Next.js pages /clusters,
+/jobs, /jobs/new, /new, /history over the APIServer; 5.1k LoC TS).
+
+Here: a dependency-free multi-page app (hash-routed, same page set) served
+by the same FastAPI apiserver, using the v1 endpoints + the v2 ray.io proxy
+via fetch(). Pages: clusters list/detail/create, jobs list/submit, services
+list/detail, job submissions browser.
+"""
 
 DASHBOARD_HTML = """<!doctype html>
 <html>
@@ -8,160 +13,266 @@ DASHBOARD_HTML = """<!doctype html>
 <meta charset="utf-8">
 <title>KubeRay-AMD — MI355X Ray clusters</title>
 <style>
-  body { font-family: system-ui, sans-serif; margin: 2rem; background: #0d1117; color: #e6edf3; }
-  h1 { color: #ff6b35; } h2 { border-bottom: 1px solid #30363d; padding-bottom: .3rem; }
+  body { font-family: system-ui, sans-serif; margin: 0; background: #0d1117; color: #e6edf3; }
+  header { display: flex; align-items: baseline; gap: 1.2rem; padding: 1rem 2rem;
+           border-bottom: 1px solid #30363d; }
+  header h1 { color: #ff6b35; margin: 0; font-size: 1.3rem; }
+  nav a { color: #8b949e; text-decoration: none; margin-right: 1rem; }
+  nav a.active { color: #ff6b35; border-bottom: 2px solid #ff6b35; }
+  main { padding: 1.5rem 2rem; }
+  h2 { border-bottom: 1px solid #30363d; padding-bottom: .3rem; }
   table { border-collapse: collapse; width: 100%; margin-bottom: 1.5rem; }
   th, td { text-align: left; padding: .4rem .8rem; border-bottom: 1px solid #21262d; }
   th { color: #8b949e; font-weight: 600; }
-  .ready { color: #3fb950; } .notready { color: #d29922; }
+  tr.rowlink { cursor: pointer; } tr.rowlink:hover { background: #161b22; }
+  .ready, .RUNNING, .SUCCEEDED, .Complete { color: #3fb950; }
+  .notready, .DEPLOYING, .PENDING, .Initializing, .Running { color: #d29922; }
+  .FAILED, .Failed, .DEPLOY_FAILED { color: #f85149; }
   button { background: #21262d; color: #e6edf3; border: 1px solid #30363d;
            border-radius: 6px; padding: .3rem .8rem; cursor: pointer; }
   button:hover { border-color: #ff6b35; }
-  input, select { background: #0d1117; color: #e6edf3; border: 1px solid #30363d;
+  button.danger:hover { border-color: #f85149; color: #f85149; }
+  input, select, textarea { background: #0d1117; color: #e6edf3; border: 1px solid #30363d;
                   border-radius: 6px; padding: .3rem; margin-right: .5rem; }
   .muted { color: #8b949e; font-size: .85rem; }
+  .card { background: #161b22; border: 1px solid #30363d; border-radius: 8px;
+          padding: 1rem; margin-bottom: 1rem; }
+  .grid { display: grid; grid-template-columns: repeat(auto-fit, minmax(280px, 1fr));
+          gap: 1rem; }
+  pre { background: #161b22; border: 1px solid #30363d; border-radius: 8px;
+        padding: .8rem; overflow-x: auto; }
+  .crumbs a { color: #58a6ff; text-decoration: none; }
+  form.stack label { display: block; margin-bottom: .6rem; }
+  form.stack input, form.stack textarea { width: 24rem; max-width: 90%; }
 </style>
 </head>
 <body>
-<h1>KubeRay-AMD</h1>
-<p class="muted">MI355X-native Ray operator — namespace
-  <input id="ns" value="default" size="10"> <button onclick="refresh()">refresh</button></p>
-
-<h2>RayClusters</h2>
-<p>
-  <input id="cname" placeholder="name">
-  <input id="workers" type="number" value="1" min="0" style="width:4rem" title="workers">
-  <input id="gpus" type="number" value="1" min="0" max="8" style="width:4rem" title="amd.com/gpu per worker">
-  <button onclick="createCluster()">create</button>
-</p>
-<table id="clusters"><thead><tr>
-  <th>name</th><th>state</th><th>workers</th><th>GPUs</th><th>created</th><th></th>
-</tr></thead><tbody></tbody></table>
-
-<h2>RayJobs</h2>
-<p>
-  <input id="jname" placeholder="job name">
-  <input id="entrypoint" placeholder="entrypoint (python train.py)" size="32">
-  <input id="jgpus" type="number" value="1" min="0" max="8" style="width:4rem"
-         title="amd.com/gpu per worker">
-  <button onclick="submitJob()">submit job</button>
-</p>
-<table id="jobs"><thead><tr>
-  <th>name</th><th>deployment status</th><th>job status</th><th>cluster</th><th></th>
-</tr></thead><tbody></tbody></table>
-
-<h2>RayServices</h2>
-<p>
-  <input id="sname" placeholder="service name">
-  <input id="sgpus" type="number" value="1" min="0" max="8" style="width:4rem"
-         title="amd.com/gpu per worker">
-  <button onclick="createService()">create service</button><br>
-  <textarea id="serveconfig" rows="4" cols="60"
-            placeholder="serveConfigV2 YAML (applications: ...)"></textarea>
-</p>
-<table id="services"><thead><tr>
-  <th>name</th><th>status</th><th>endpoints</th><th>active cluster</th><th></th>
-</tr></thead><tbody></tbody></table>
+<header>
+  <h1>KubeRay-AMD</h1>
+  <nav id="nav">
+    <a href="#/clusters" data-page="clusters">Clusters</a>
+    <a href="#/jobs" data-page="jobs">Jobs</a>
+    <a href="#/services" data-page="services">Services</a>
+    <a href="#/new" data-page="new">New cluster</a>
+    <a href="#/jobs/new" data-page="jobsnew">Submit job</a>
+  </nav>
+  <span class="muted">namespace
+    <input id="ns" value="default" size="10" onchange="route()"></span>
+</header>
+<main id="main"><p class="muted">loading…</p></main>
 
 <script>
-const ns = () => document.getElementById('ns').value || 'default';
-const api = (p, o) => fetch(`/apis/v1/namespaces/${ns()}${p}`, o).then(r => r.json());
+const $ = (id) => document.getElementById(id);
+const ns = () => $('ns').value || 'default';
+const esc = (s) => String(s ?? '').replace(/[&<>"]/g,
+  (c) => ({'&':'&amp;','<':'&lt;','>':'&gt;','"':'&quot;'}[c]));
+async function api(method, path, body) {
+  const resp = await fetch(path, {
+    method, headers: {'Content-Type': 'application/json'},
+    body: body === undefined ? undefined : JSON.stringify(body)});
+  if (!resp.ok) throw new Error(await resp.text());
+  const text = await resp.text();
+  return text ? JSON.parse(text) : {};
+}
+function statusCell(s) { return `<span class="${esc(s)}">${esc(s || '—')}</span>`; }
 
-function row(tds, delFn) {
-  const tr = document.createElement('tr');
-  tds.forEach(t => { const td = document.createElement('td');
-    if (t instanceof Node) td.appendChild(t); else td.innerHTML = t;
-    tr.appendChild(td); });
-  const td = document.createElement('td');
-  const b = document.createElement('button'); b.textContent = 'delete';
-  b.onclick = delFn; td.appendChild(b); tr.appendChild(td);
-  return tr;
+// ---------------------------------------------------------------- clusters
+async function pageClusters() {
+  const data = await api('GET', `/apis/v1/namespaces/${ns()}/clusters`);
+  const rows = (data.clusters || []).map(c => `
+    <tr class="rowlink" onclick="location.hash='#/clusters/${esc(c.name)}'">
+      <td>${esc(c.name)}</td><td>${statusCell(c.clusterState)}</td>
+      <td>${esc(c.createdAt || '')}</td>
+      <td><button class="danger" onclick="event.stopPropagation();
+        delCluster('${esc(c.name)}')">delete</button></td></tr>`).join('');
+  $('main').innerHTML = `<h2>RayClusters</h2>
+    <table><thead><tr><th>name</th><th>state</th><th>created</th><th></th>
+    </tr></thead><tbody>${rows ||
+      '<tr><td colspan=4 class=muted>none</td></tr>'}</tbody></table>`;
+}
+async function delCluster(name) {
+  if (!confirm(`delete RayCluster ${name}?`)) return;
+  await api('DELETE', `/apis/v1/namespaces/${ns()}/clusters/${name}`);
+  route();
+}
+async function pageClusterDetail(name) {
+  const obj = await api('GET',
+    `/apis/ray.io/v1/namespaces/${ns()}/rayclusters/${name}`);
+  const st = obj.status || {};
+  const groups = (obj.spec.workerGroupSpecs || []).map(g => `
+    <tr><td>${esc(g.groupName)}</td><td>${esc(g.replicas)}</td>
+    <td>${esc(g.minReplicas)}–${esc(g.maxReplicas)}</td>
+    <td>${esc((((g.template.spec.containers||[])[0]||{}).resources||{})
+      .limits?.['amd.com/gpu'] || '0')}</td></tr>`).join('');
+  const conds = (st.conditions || []).map(c => `
+    <tr><td>${esc(c.type)}</td><td>${statusCell(c.status)}</td>
+    <td class=muted>${esc(c.reason || '')}</td></tr>`).join('');
+  $('main').innerHTML = `
+    <p class="crumbs"><a href="#/clusters">clusters</a> / ${esc(name)}</p>
+    <h2>${esc(name)} ${statusCell(st.state)}</h2>
+    <div class="grid">
+    <div class="card"><b>Status</b><table>
+      <tr><td>desired workers</td><td>${esc(st.desiredWorkerReplicas)}</td></tr>
+      <tr><td>available workers</td><td>${esc(st.availableWorkerReplicas)}</td></tr>
+      <tr><td>desired GPUs</td><td>${esc(st.desiredGPU)}</td></tr>
+      <tr><td>head pod IP</td><td>${esc((st.head||{}).podIP || '')}</td></tr>
+    </table></div>
+    <div class="card"><b>Conditions</b>
+      <table>${conds || '<tr><td class=muted>none</td></tr>'}</table></div>
+    </div>
+    <h2>Worker groups</h2>
+    <table><thead><tr><th>group</th><th>replicas</th><th>min–max</th>
+      <th>amd.com/gpu</th></tr></thead><tbody>${groups}</tbody></table>
+    <h2>Raw spec</h2><pre>${esc(JSON.stringify(obj.spec, null, 2))}</pre>`;
+}
+async function pageNewCluster() {
+  $('main').innerHTML = `<h2>New RayCluster</h2>
+    <form class="stack" onsubmit="return createCluster(event)">
+      <label>name <input id="cname" required></label>
+      <label>workers <input id="workers" type="number" value="1" min="0"></label>
+      <label>amd.com/gpu per worker
+        <input id="gpus" type="number" value="1" min="0" max="8"></label>
+      <button type="submit">create</button>
+    </form>`;
+}
+async function createCluster(ev) {
+  ev.preventDefault();
+  const gpus = parseInt($('gpus').value || '0');
+  const worker = {groupName: 'mi355x-group',
+                  replicas: parseInt($('workers').value || '1'),
+                  minReplicas: 0, maxReplicas: 8,
+                  computeTemplate: '', rayStartParams: {}};
+  await api('POST', `/apis/v1/namespaces/${ns()}/clusters`, {
+    name: $('cname').value, version: '2.46.0',
+    clusterSpec: {headGroupSpec: {rayStartParams: {}},
+                  workerGroupSpec: [{...worker, gpu: gpus}]}});
+  location.hash = '#/clusters';
+  return false;
 }
 
-async function refresh() {
-  const cb = document.querySelector('#clusters tbody'); cb.innerHTML = '';
-  const cl = await api('/clusters');
-  (cl.clusters || []).forEach(c => {
-    const state = c.clusterState === 'ready'
-      ? '<span class="ready">ready</span>'
-      : `<span class="notready">${c.clusterState || 'pending'}</span>`;
-    const gpus = (c.clusterSpec.workerGroupSpec || [])
-      .map(g => `${g.groupName}:${g.replicas}`).join(' ');
-    cb.appendChild(row([c.name, state, gpus,
-      c.serviceEndpoint ? Object.keys(c.serviceEndpoint).length : 0,
-      c.createdAt || ''],
-      async () => { await fetch(`/apis/v1/namespaces/${ns()}/clusters/${c.name}`,
-                                {method: 'DELETE'}); refresh(); }));
-  });
-  const jb = document.querySelector('#jobs tbody'); jb.innerHTML = '';
-  const jl = await api('/jobs');
-  (jl.jobs || []).forEach(j => jb.appendChild(row(
-    [j.name, j.jobDeploymentStatus || '-', j.jobStatus || '-', j.rayClusterName || '-'],
-    async () => { await fetch(`/apis/v1/namespaces/${ns()}/jobs/${j.name}`,
-                              {method: 'DELETE'}); refresh(); })));
-  const sb = document.querySelector('#services tbody'); sb.innerHTML = '';
-  const sl = await api('/services');
-  (sl.services || []).forEach(s => sb.appendChild(row(
-    [s.name, s.serviceStatus || '-', s.numServeEndpoints, s.activeRayClusterName || '-'],
-    async () => { await fetch(`/apis/v1/namespaces/${ns()}/services/${s.name}`,
-                              {method: 'DELETE'}); refresh(); })));
+// -------------------------------------------------------------------- jobs
+async function pageJobs() {
+  const data = await api('GET', `/apis/v1/namespaces/${ns()}/jobs`);
+  const rows = (data.jobs || []).map(j => `
+    <tr><td>${esc(j.name)}</td>
+      <td>${statusCell(j.jobDeploymentStatus)}</td>
+      <td>${statusCell(j.jobStatus)}</td>
+      <td>${esc(j.rayClusterName || '')}</td>
+      <td><button class="danger"
+        onclick="delJob('${esc(j.name)}')">delete</button></td></tr>`).join('');
+  $('main').innerHTML = `<h2>RayJobs</h2>
+    <table><thead><tr><th>name</th><th>deployment</th><th>app status</th>
+    <th>cluster</th><th></th></tr></thead><tbody>${rows ||
+      '<tr><td colspan=5 class=muted>none</td></tr>'}</tbody></table>`;
+}
+async function delJob(name) {
+  if (!confirm(`delete RayJob ${name}?`)) return;
+  await api('DELETE', `/apis/v1/namespaces/${ns()}/jobs/${name}`);
+  route();
+}
+async function pageJobsNew() {
+  $('main').innerHTML = `<h2>Submit RayJob</h2>
+    <form class="stack" onsubmit="return submitJob(event)">
+      <label>name <input id="jname" required></label>
+      <label>entrypoint <input id="entrypoint"
+        placeholder="python train.py" required></label>
+      <label>workers <input id="jworkers" type="number" value="1" min="0"></label>
+      <label>amd.com/gpu per worker
+        <input id="jgpus" type="number" value="1" min="0" max="8"></label>
+      <label><input id="jshutdown" type="checkbox" checked
+        style="width:auto"> shutdown cluster after job finishes</label>
+      <button type="submit">submit</button>
+    </form>`;
+}
+async function submitJob(ev) {
+  ev.preventDefault();
+  await api('POST', `/apis/v1/namespaces/${ns()}/jobs`, {
+    name: $('jname').value, entrypoint: $('entrypoint').value,
+    shutdownAfterJobFinishes: $('jshutdown').checked,
+    clusterSpec: {headGroupSpec: {rayStartParams: {}},
+      workerGroupSpec: [{groupName: 'mi355x-group',
+        replicas: parseInt($('jworkers').value || '1'),
+        minReplicas: 0, maxReplicas: 8,
+        gpu: parseInt($('jgpus').value || '0'), rayStartParams: {}}]}});
+  location.hash = '#/jobs';
+  return false;
 }
 
-async function createCluster() {
-  const name = document.getElementById('cname').value;
-  if (!name) return alert('name required');
-  const workers = +document.getElementById('workers').value;
-  const gpus = +document.getElementById('gpus').value;
-  await fetch(`/apis/v1/namespaces/${ns()}/compute_templates`, {
-    method: 'POST', headers: {'Content-Type': 'application/json'},
-    body: JSON.stringify({name: `${name}-tpl`, cpu: 4, memory: 8, gpu: gpus})});
-  await fetch(`/apis/v1/namespaces/${ns()}/clusters`, {
-    method: 'POST', headers: {'Content-Type': 'application/json'},
-    body: JSON.stringify({name, version: '2.46.0', clusterSpec: {
-      headGroupSpec: {computeTemplate: `${name}-tpl`},
-      workerGroupSpec: [{groupName: 'default-group',
-        computeTemplate: `${name}-tpl`, replicas: workers,
-        minReplicas: 0, maxReplicas: Math.max(workers, 8)}]}})});
-  refresh();
+// ---------------------------------------------------------------- services
+async function pageServices() {
+  const data = await api('GET', `/apis/v1/namespaces/${ns()}/services`);
+  const rows = (data.services || []).map(s => `
+    <tr class="rowlink" onclick="location.hash='#/services/${esc(s.name)}'">
+      <td>${esc(s.name)}</td><td>${statusCell(s.serviceStatus)}</td>
+      <td>${esc(s.activeRayClusterName || '')}</td>
+      <td>${esc(s.pendingRayClusterName || '')}</td>
+      <td><button class="danger" onclick="event.stopPropagation();
+        delService('${esc(s.name)}')">delete</button></td></tr>`).join('');
+  $('main').innerHTML = `<h2>RayServices</h2>
+    <table><thead><tr><th>name</th><th>status</th><th>active cluster</th>
+    <th>pending cluster</th><th></th></tr></thead><tbody>${rows ||
+      '<tr><td colspan=5 class=muted>none</td></tr>'}</tbody></table>`;
 }
-async function submitJob() {
-  const name = document.getElementById('jname').value;
-  const entrypoint = document.getElementById('entrypoint').value;
-  if (!name || !entrypoint) return alert('name and entrypoint required');
-  const gpus = +document.getElementById('jgpus').value;
-  await fetch(`/apis/v1/namespaces/${ns()}/compute_templates`, {
-    method: 'POST', headers: {'Content-Type': 'application/json'},
-    body: JSON.stringify({name: `${name}-tpl`, cpu: 4, memory: 8, gpu: gpus})});
-  await fetch(`/apis/v1/namespaces/${ns()}/jobs`, {
-    method: 'POST', headers: {'Content-Type': 'application/json'},
-    body: JSON.stringify({name, entrypoint, clusterSpec: {
-      headGroupSpec: {computeTemplate: `${name}-tpl`},
-      workerGroupSpec: [{groupName: 'default-group',
-        computeTemplate: `${name}-tpl`, replicas: 1, minReplicas: 0,
-        maxReplicas: 4}]}})});
-  refresh();
+async function delService(name) {
+  if (!confirm(`delete RayService ${name}?`)) return;
+  await api('DELETE', `/apis/v1/namespaces/${ns()}/services/${name}`);
+  route();
 }
-
-async function createService() {
-  const name = document.getElementById('sname').value;
-  const cfg = document.getElementById('serveconfig').value;
-  if (!name || !cfg) return alert('name and serveConfigV2 required');
-  const gpus = +document.getElementById('sgpus').value;
-  await fetch(`/apis/v1/namespaces/${ns()}/compute_templates`, {
-    method: 'POST', headers: {'Content-Type': 'application/json'},
-    body: JSON.stringify({name: `${name}-tpl`, cpu: 4, memory: 8, gpu: gpus})});
-  await fetch(`/apis/v1/namespaces/${ns()}/services`, {
-    method: 'POST', headers: {'Content-Type': 'application/json'},
-    body: JSON.stringify({name, serveConfig_V2: cfg, clusterSpec: {
-      headGroupSpec: {computeTemplate: `${name}-tpl`},
-      workerGroupSpec: [{groupName: 'default-group',
-        computeTemplate: `${name}-tpl`, replicas: 1, minReplicas: 0,
-        maxReplicas: 4}]}})});
-  refresh();
+async function pageServiceDetail(name) {
+  const obj = await api('GET',
+    `/apis/ray.io/v1/namespaces/${ns()}/rayservices/${name}`);
+  const st = obj.status || {};
+  const active = st.activeServiceStatus || {};
+  const apps = Object.entries(active.applications || {}).map(([n, a]) => `
+    <tr><td>${esc(n)}</td><td>${statusCell(a.status)}</td>
+    <td class=muted>${esc(a.message || '')}</td></tr>`).join('');
+  $('main').innerHTML = `
+    <p class="crumbs"><a href="#/services">services</a> / ${esc(name)}</p>
+    <h2>${esc(name)} ${statusCell(st.serviceStatus)}</h2>
+    <div class="card"><table>
+      <tr><td>serve endpoints</td><td>${esc(st.numServeEndpoints)}</td></tr>
+      <tr><td>active cluster</td><td>${esc(active.rayClusterName || '')}</td></tr>
+      <tr><td>pending cluster</td>
+        <td>${esc((st.pendingServiceStatus||{}).rayClusterName || '')}</td></tr>
+    </table></div>
+    <h2>Serve applications</h2>
+    <table><thead><tr><th>app</th><th>status</th><th>message</th></tr>
+    </thead><tbody>${apps ||
+      '<tr><td colspan=3 class=muted>none</td></tr>'}</tbody></table>
+    <h2>serveConfigV2</h2><pre>${esc(obj.spec.serveConfigV2 || '')}</pre>`;
 }
 
-refresh();
-setInterval(refresh, 5000);
+// ------------------------------------------------------------------ router
+const ROUTES = [
+  [/^#\\/clusters$/, () => pageClusters(), 'clusters'],
+  [/^#\\/clusters\\/([^/]+)$/, (m) => pageClusterDetail(m[1]), 'clusters'],
+  [/^#\\/jobs$/, () => pageJobs(), 'jobs'],
+  [/^#\\/jobs\\/new$/, () => pageJobsNew(), 'jobsnew'],
+  [/^#\\/services$/, () => pageServices(), 'services'],
+  [/^#\\/services\\/([^/]+)$/, (m) => pageServiceDetail(m[1]), 'services'],
+  [/^#\\/new$/, () => pageNewCluster(), 'new'],
+];
+async function route() {
+  const hash = location.hash || '#/clusters';
+  for (const [re, fn, page] of ROUTES) {
+    const m = hash.match(re);
+    if (m) {
+      document.querySelectorAll('#nav a').forEach(a =>
+        a.classList.toggle('active', a.dataset.page === page));
+      try { await fn(m); } catch (e) {
+        $('main').innerHTML = `<p class="FAILED">${esc(e.message)}</p>`;
+      }
+      return;
+    }
+  }
+  location.hash = '#/clusters';
+}
+window.addEventListener('hashchange', route);
+route();
+setInterval(() => { // live refresh for list pages
+  if (/^#\\/(clusters|jobs|services)$/.test(location.hash || '#/clusters'))
+    route();
+}, 5000);
 </script>
 </body>
 </html>

@@ -721,6 +721,43 @@ class TestGrpcRayServeService:
             server.stop(0)
 
 
+class TestDashboardUi:
+    """The multi-page dashboard shell (apiserver/dashboard.py): served at
+    '/', contains the route set matching the reference's page tree
+    (dashboard/src/app/{clusters,jobs,jobs/new,new}/page.tsx)."""
+
+    def test_shell_served_with_all_pages(self, api):
+        t, _, _ = api
+        r = t.get("/")
+        assert r.status_code == 200
+        html = r.text
+        for route in ("#/clusters", "#/jobs", "#/jobs/new", "#/services",
+                      "#/new"):
+            assert route in html, f"page route {route} missing from shell"
+        # detail pages + the API paths the JS drives
+        assert "pageClusterDetail" in html
+        assert "pageServiceDetail" in html
+        assert "/apis/v1/namespaces/" in html
+        assert "/apis/ray.io/v1/namespaces/" in html
+
+    def test_inline_gpu_cluster_create_shape(self, api):
+        """The new-cluster form submits inline cpu/memory/gpu without a
+        compute template; the converter must honor them."""
+        t, client, _ = api
+        r = t.post("/apis/v1/namespaces/ns1/clusters", json={
+            "name": "ui-inline", "version": "2.46.0", "clusterSpec": {
+                "headGroupSpec": {"rayStartParams": {}},
+                "workerGroupSpec": [{"groupName": "mi355x-group",
+                                     "replicas": 1, "minReplicas": 0,
+                                     "maxReplicas": 8, "gpu": 4,
+                                     "rayStartParams": {}}]}})
+        assert r.status_code == 200, r.text
+        rc = client.get(RayCluster, "ns1", "ui-inline")
+        limits = rc.spec.worker_group_specs[0].template.spec.containers[0] \
+            .resources.limits
+        assert limits["amd.com/gpu"] == "4"
+
+
 class TestDashboardFormContract:
     """Pins the exact request shapes the static dashboard's JS emits
     (createCluster/submitJob in apiserver/dashboard.py)."""
