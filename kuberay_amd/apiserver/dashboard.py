@@ -64,7 +64,7 @@ DASHBOARD_HTML = """<!doctype html>
 <script>
 const $ = (id) => document.getElementById(id);
 const ns = () => $('ns').value || 'default';
-const esc = (s) => String(s ?? '').replace(/[&<>"]/g,
+const esc = (s) => String(s === undefined || s === null ? '' : s).replace(/[&<>"]/g,
   (c) => ({'&':'&amp;','<':'&lt;','>':'&gt;','"':'&quot;'}[c]));
 async function api(method, path, body) {
   const resp = await fetch(path, {
@@ -102,8 +102,8 @@ async function pageClusterDetail(name) {
   const groups = (obj.spec.workerGroupSpecs || []).map(g => `
     <tr><td>${esc(g.groupName)}</td><td>${esc(g.replicas)}</td>
     <td>${esc(g.minReplicas)}–${esc(g.maxReplicas)}</td>
-    <td>${esc((((g.template.spec.containers||[])[0]||{}).resources||{})
-      .limits?.['amd.com/gpu'] || '0')}</td></tr>`).join('');
+    <td>${esc((((((g.template.spec.containers||[])[0]||{}).resources||{})
+      .limits)||{})['amd.com/gpu'] || '0')}</td></tr>`).join('');
   const conds = (st.conditions || []).map(c => `
     <tr><td>${esc(c.type)}</td><td>${statusCell(c.status)}</td>
     <td class=muted>${esc(c.reason || '')}</td></tr>`).join('');
