@@ -656,3 +656,57 @@ def test_kubernetes_was_gate_selects_scheduler():
                                         batch_scheduler="volcano"))
     finally:
         features.reset()
+
+
+class TestMultiKueueInterop:
+    """managedBy interop (reference raycluster_controller.go:158,
+    rayjob_controller.go:107 + kueue multikueue semantics): a CR managed by
+    an external controller is left entirely alone until the field points
+    back at this operator."""
+
+    MULTIKUEUE = "kueue.x-k8s.io/multikueue"
+
+    def test_raycluster_managed_by_multikueue_not_reconciled(self,
+                                                             control_plane):
+        import time as _time
+        rc = simple_raycluster("kueued", workers=2)
+        rc.spec.managed_by = self.MULTIKUEUE
+        control_plane.client.create(rc)
+        _time.sleep(0.5)
+        # no pods, no services, no status written
+        assert control_plane.server.count("Pod") == 0
+        assert control_plane.server.count("Service") == 0
+        got = control_plane.client.get(RayCluster, "default", "kueued")
+        assert got.status.state is None
+
+    def test_rayjob_managed_by_multikueue_not_started(self, control_plane):
+        import time as _time
+        from kuberay_amd.models import RayJob
+        from kuberay_amd.testing import simple_raycluster as src
+        control_plane.client.create(RayJob.from_dict({
+            "apiVersion": "ray.io/v1", "kind": "RayJob",
+            "metadata": {"name": "kueued-job", "namespace": "default"},
+            "spec": {"entrypoint": "python t.py",
+                     "managedBy": self.MULTIKUEUE,
+                     "rayClusterSpec": src("x").spec.to_dict()}}))
+        _time.sleep(0.5)
+        assert control_plane.server.count("RayCluster") == 0
+        job = control_plane.client.get(RayJob, "default", "kueued-job")
+        assert job.status.job_deployment_status in (None, "", "New")
+
+    def test_handover_from_multikueue_starts_reconcile(self, control_plane):
+        """Kueue admits the workload by flipping managedBy to the operator
+        (multikueue handover): reconciliation must begin then."""
+        rc = simple_raycluster("handover", workers=1)
+        rc.spec.managed_by = self.MULTIKUEUE
+        control_plane.client.create(rc)
+        import time as _time
+        _time.sleep(0.3)
+        assert control_plane.server.count("Pod") == 0
+
+        def admit(obj):
+            obj.spec.managed_by = "ray.io/kuberay-operator"
+        control_plane.client.update_with_retry(RayCluster, "default",
+                                               "handover", admit)
+        assert control_plane.wait_cluster_state("default", "handover",
+                                                "ready", timeout=20)
