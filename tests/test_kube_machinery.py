@@ -486,6 +486,101 @@ class TestPodPool:
         assert mgr.adopt("empty", {"ray.io/cluster": "c1"}) is None
 
 
+class TestVirtualKubeletPodPool:
+    """Virtual-kubelet layer (reference podpool/cmd/main.go contract):
+    node registration + heartbeat + instant binding from the warm pool."""
+
+    TEMPLATE = TestPodPool.TEMPLATE
+
+    def _stack(self):
+        from kuberay_amd.kube.client import InMemoryClient
+        from kuberay_amd.ops.podpool import PodPoolManager, VirtualKubeletPodPool
+        client = InMemoryClient()
+        mgr = PodPoolManager(client)
+        mgr.define_pool("mi355x", self.TEMPLATE, size=2)
+        vk = VirtualKubeletPodPool(client, mgr, heartbeat_s=0.05)
+        return client, mgr, vk
+
+    def test_node_registered_with_taint_and_capacity(self):
+        client, mgr, vk = self._stack()
+        mgr.reconcile()
+        vk.register_node()
+        node = client.server.get("Node", "default", "kuberay-pod-pool")
+        assert node["metadata"]["labels"]["type"] == "virtual-kubelet"
+        assert node["spec"]["taints"][0]["key"] == \
+            "virtual-kubelet.io/provider"
+        assert node["status"]["capacity"]["pods"] == "2"
+
+    def test_heartbeat_refreshes_ready_condition(self):
+        import time as _t
+        client, _, vk = self._stack()
+        vk.register_node()
+        first = client.server.get("Node", "default", "kuberay-pod-pool")[
+            "status"]["conditions"][0]["lastHeartbeatTime"]
+        _t.sleep(1.1)  # now_iso has 1s granularity
+        vk.heartbeat()
+        second = client.server.get("Node", "default", "kuberay-pod-pool")[
+            "status"]["conditions"][0]["lastHeartbeatTime"]
+        assert second >= first
+
+    def test_pool_targeted_pod_binds_instantly(self):
+        client, mgr, vk = self._stack()
+        mgr.reconcile()
+        consumer = {
+            "apiVersion": "v1", "kind": "Pod",
+            "metadata": {"name": "worker-0", "namespace": "default",
+                         "labels": {"ray.io/warm-pod-pool": "mi355x",
+                                    "ray.io/cluster": "c1"}},
+            "spec": {"containers": [{"name": "ray-worker",
+                                     "image": "rocm/ray:2.46.0"}]}}
+        client.server.create(consumer)
+        assert vk.bind_pending_pods() == 1
+        pod = client.server.get("Pod", "default", "worker-0")
+        assert pod["status"]["phase"] == "Running"
+        assert pod["status"]["containerStatuses"][0]["ready"] is True
+        assert pod["metadata"]["annotations"]["ray.io/warm-pod-source"]
+        # consumed warm pod removed, pool refilled to target
+        warm = [p for p in client.server.list("Pod", "default")
+                if (p["metadata"].get("labels") or {})
+                .get("ray.io/warm-pod") == "true"]
+        assert len(warm) == 2
+
+    def test_dry_pool_leaves_pod_pending(self):
+        client, mgr, vk = self._stack()
+        mgr.define_pool("mi355x", self.TEMPLATE, size=0)
+        mgr.reconcile()
+        client.server.create({
+            "apiVersion": "v1", "kind": "Pod",
+            "metadata": {"name": "worker-0", "namespace": "default",
+                         "labels": {"ray.io/warm-pod-pool": "mi355x"}},
+            "spec": {"containers": [{"name": "w", "image": "i"}]}})
+        assert vk.bind_pending_pods() == 0
+        pod = client.server.get("Pod", "default", "worker-0")
+        assert pod.get("status", {}).get("phase") in (None, "Pending")
+
+    def test_watch_driven_bind_loop(self):
+        import time as _t
+        client, mgr, vk = self._stack()
+        vk.start()
+        try:
+            client.server.create({
+                "apiVersion": "v1", "kind": "Pod",
+                "metadata": {"name": "late-worker", "namespace": "default",
+                             "labels": {"ray.io/warm-pod-pool": "mi355x"}},
+                "spec": {"containers": [{"name": "w", "image": "i"}]}})
+            deadline = _t.monotonic() + 5
+            phase = None
+            while _t.monotonic() < deadline:
+                phase = client.server.get("Pod", "default", "late-worker") \
+                    .get("status", {}).get("phase")
+                if phase == "Running":
+                    break
+                _t.sleep(0.05)
+            assert phase == "Running"
+        finally:
+            vk.stop()
+
+
 class TestShardedLeases:
     def test_shards_hold_distinct_leases_concurrently(self):
         """Two operator shards elect leaders on independent Lease names."""
