@@ -21,6 +21,11 @@ class RateLimitingQueue:
         self._processing: Set[Hashable] = set()
         self._dirty: Set[Hashable] = set()
         self._delayed: List[Tuple[float, int, Hashable]] = []  # heap
+        # earliest pending deadline per item: a key requeued every reconcile
+        # (requeue_after) must hold ONE live heap entry, not one per
+        # reconcile — an unbounded heap was the main churn-RSS growth in the
+        # round-1 soak (~165k stale entries at 550 reconciles/s × 300 s)
+        self._delayed_next: Dict[Hashable, float] = {}
         self._failures: Dict[Hashable, int] = {}
         self._seq = 0
         self._shutdown = False
@@ -47,8 +52,13 @@ class RateLimitingQueue:
         with self._lock:
             if self._shutdown:
                 return
+            when = time.monotonic() + delay
+            pending = self._delayed_next.get(item)
+            if pending is not None and pending <= when:
+                return  # already scheduled at least as soon; entry coalesced
+            self._delayed_next[item] = when
             self._seq += 1
-            heapq.heappush(self._delayed, (time.monotonic() + delay, self._seq, item))
+            heapq.heappush(self._delayed, (when, self._seq, item))
             self._lock.notify()
 
     def add_rate_limited(self, item: Hashable) -> None:
@@ -70,6 +80,11 @@ class RateLimitingQueue:
             when, _, item = self._delayed[0]
             if when <= now:
                 heapq.heappop(self._delayed)
+                # stale entry: a sooner deadline was registered after this
+                # one was pushed (add_after coalescing) and already fired
+                if self._delayed_next.get(item) != when:
+                    continue
+                del self._delayed_next[item]
                 if item not in self._queued and item not in self._processing:
                     self._queued.add(item)
                     self._queue.append(item)
@@ -116,4 +131,4 @@ class RateLimitingQueue:
 
     def __len__(self) -> int:
         with self._lock:
-            return len(self._queue) + len(self._delayed)
+            return len(self._queue) + len(self._delayed_next)

@@ -152,6 +152,24 @@ class TestWorkqueue:
         assert q.get(timeout=0.02) is None
         assert q.get(timeout=1.0) == "a"
 
+    def test_add_after_coalesces_per_key(self):
+        """A key requeued every reconcile must hold ONE live heap entry —
+        an unbounded delayed heap was the round-1 soak's RSS growth."""
+        q = RateLimitingQueue()
+        for _ in range(10_000):
+            q.add_after("k", 300.0)
+        assert len(q._delayed) == 1
+        assert len(q) == 1
+
+    def test_add_after_earlier_deadline_wins(self):
+        q = RateLimitingQueue()
+        q.add_after("k", 300.0)
+        q.add_after("k", 0.05)  # sooner: must fire at the sooner deadline
+        assert q.get(timeout=1.0) == "k"
+        q.done("k")
+        # the stale 300 s entry must not fire the key a second time
+        assert q.get(timeout=0.3) is None
+
     def test_rate_limited_backoff_grows(self):
         q = RateLimitingQueue(base_delay=0.01, max_delay=1.0)
         t0 = time.monotonic()
