@@ -276,7 +276,13 @@ class InMemoryApiServer:
         if event_type == "DELETED":
             # real apiserver: a delete is itself a new revision
             rv = self._rv
-        self._event_history.append((rv, event_type, obj))
+        # history entries are stored SERIALIZED: one compact string instead
+        # of ~10² nested dicts per event keeps the 1024-entry window at
+        # ~1 MB instead of ~11 MB resident (replay decodes on demand)
+        import json as _json
+        self._event_history.append(
+            (rv, event_type, obj.get("kind", ""),
+             _json.dumps(obj, separators=(",", ":"))))
         for w in list(self._watchers):
             w.push(event_type, obj)
 
@@ -292,13 +298,14 @@ class InMemoryApiServer:
         the watch must be answered with 410 Gone and the client re-lists
         (kube-apiserver 'too old resource version' semantics).
         """
+        import json as _json
         with self._lock:
             hist = self._event_history
             if len(hist) == hist.maxlen and hist[0][0] > rv + 1:
                 return None
-            return [(etype, jsoncopy(obj)) for (erv, etype, obj) in hist
-                    if erv > rv and (kinds is None
-                                     or obj.get("kind") in kinds)]
+            return [(etype, _json.loads(payload))
+                    for (erv, etype, kind, payload) in hist
+                    if erv > rv and (kinds is None or kind in kinds)]
 
     @staticmethod
     def _key_of(obj: Dict[str, Any]) -> Key:
