@@ -59,15 +59,41 @@ def get():
     """Get Ray resources."""
 
 
+def _emit_objects(objs, output):
+    """kubectl-style -o yaml|json rendering; returns True when handled."""
+    import json as _json
+    if output == "yaml":
+        docs = [o.to_dict() | {"kind": o.kind} for o in objs]
+        click.echo(yaml.safe_dump_all(docs, sort_keys=False).rstrip())
+        return True
+    if output == "json":
+        items = [o.to_dict() | {"kind": o.kind} for o in objs]
+        out = items[0] if len(items) == 1 else {
+            "apiVersion": "v1", "kind": "List", "items": items}
+        click.echo(_json.dumps(out, indent=2))
+        return True
+    if output:
+        raise click.ClickException(f"unsupported output format '{output}' "
+                                   "(want yaml or json)")
+    return False
+
+
+_OUTPUT_OPT = click.option("-o", "--output", default=None,
+                           help="output format: yaml | json")
+
+
 @get.command("cluster")
 @click.argument("name", required=False)
+@_OUTPUT_OPT
 @click.pass_context
-def get_cluster(ctx, name):
+def get_cluster(ctx, name, output):
     client = client_of(ctx)
     ns = ctx.obj["namespace"]
     rows = []
     clusters = ([client.get(RayCluster, ns, name)] if name
                 else client.list(RayCluster, ns))
+    if _emit_objects(clusters, output):
+        return
     for rc in clusters:
         rows.append((rc.metadata.name, rc.status.state or "-",
                      rc.status.desired_worker_replicas,
@@ -90,11 +116,14 @@ def get_workergroup(ctx, cluster_name):
 
 @get.command("job")
 @click.argument("name", required=False)
+@_OUTPUT_OPT
 @click.pass_context
-def get_job(ctx, name):
+def get_job(ctx, name, output):
     client = client_of(ctx)
     ns = ctx.obj["namespace"]
     jobs = [client.get(RayJob, ns, name)] if name else client.list(RayJob, ns)
+    if _emit_objects(jobs, output):
+        return
     rows = [(j.metadata.name, j.status.job_deployment_status or "-",
              j.status.job_status or "-", j.status.ray_cluster_name or "-")
             for j in jobs]
@@ -203,12 +232,15 @@ def get_cronjob(ctx, name):
 
 @get.command("service")
 @click.argument("name", required=False)
+@_OUTPUT_OPT
 @click.pass_context
-def get_service(ctx, name):
+def get_service(ctx, name, output):
     client = client_of(ctx)
     ns = ctx.obj["namespace"]
     svcs = ([client.get(RayService, ns, name)] if name
             else client.list(RayService, ns))
+    if _emit_objects(svcs, output):
+        return
     rows = [(s.metadata.name, s.status.service_status or "-",
              s.status.num_serve_endpoints,
              s.status.active_service_status.ray_cluster_name or "-")

@@ -976,3 +976,41 @@ class TestApiServerPythonClient:
             deleter("ns1", name)
         with pytest.raises(ApiServerError):
             c.get_cluster("ns1", "pyc-c1")
+
+
+class TestKrayOutputFormats:
+    """kubectl -o yaml|json parity for kray get."""
+
+    def test_get_cluster_yaml_and_json(self, kray):
+        import json as _json
+
+        import yaml as _yaml
+        runner, backing = kray
+        runner.invoke(cli, ["-n", "ns1", "create", "cluster", "k1",
+                            "--worker-gpu", "2"])
+        r = runner.invoke(cli, ["-n", "ns1", "get", "cluster", "k1",
+                                "-o", "yaml"])
+        assert r.exit_code == 0, r.output
+        doc = _yaml.safe_load(r.output)
+        assert doc["kind"] == "RayCluster"
+        assert doc["spec"]["workerGroupSpecs"][0]["template"]["spec"][
+            "containers"][0]["resources"]["limits"]["amd.com/gpu"] == "2"
+        r = runner.invoke(cli, ["-n", "ns1", "get", "cluster", "k1",
+                                "-o", "json"])
+        assert r.exit_code == 0
+        assert _json.loads(r.output)["metadata"]["name"] == "k1"
+
+    def test_get_cluster_list_json_envelope(self, kray):
+        import json as _json
+        runner, _ = kray
+        runner.invoke(cli, ["-n", "ns1", "create", "cluster", "a1"])
+        runner.invoke(cli, ["-n", "ns1", "create", "cluster", "a2"])
+        r = runner.invoke(cli, ["-n", "ns1", "get", "cluster", "-o", "json"])
+        out = _json.loads(r.output)
+        assert out["kind"] == "List" and len(out["items"]) == 2
+
+    def test_bad_output_format_rejected(self, kray):
+        runner, _ = kray
+        r = runner.invoke(cli, ["-n", "ns1", "get", "cluster", "-o", "wide"])
+        assert r.exit_code != 0
+        assert "unsupported output format" in r.output
