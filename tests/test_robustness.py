@@ -207,3 +207,82 @@ class TestOperatorSharding:
             kubelet.stop()
             for manager, _ in managers:
                 manager.stop()
+
+
+class TestOperatorUpgradeAllKinds:
+    """e2eupgrade analog across every CRD kind: an operator 'version
+    upgrade' (old instance killed, fresh instance over the surviving
+    apiserver state) must resume RayJob/RayService/RayCluster state
+    machines mid-flight without losing or re-running work."""
+
+    def test_upgrade_resumes_job_service_and_suspended_cluster(self):
+        from kuberay_amd.models import RayJob, RayService
+        from kuberay_amd.testing import simple_raycluster
+
+        cp = ControlPlane(kubelet_delay=0.01, job_runtime=0.3,
+                          poll_seconds=0.05)
+        cp.start()
+        server = cp.server
+        # a job that will still be Running at the kill point
+        cp.dashboard.get_job_info_mock = lambda job_id: {
+            "submission_id": job_id, "status": "RUNNING"}
+        cp.client.create(RayJob.from_dict({
+            "apiVersion": "ray.io/v1", "kind": "RayJob",
+            "metadata": {"name": "upjob", "namespace": "default"},
+            "spec": {"entrypoint": "python t.py",
+                     "rayClusterSpec": simple_raycluster(
+                         "x", workers=1).spec.to_dict()}}))
+        # a service that is Ready
+        cp.client.create(RayService.from_dict({
+            "apiVersion": "ray.io/v1", "kind": "RayService",
+            "metadata": {"name": "upsvc", "namespace": "default"},
+            "spec": {"serveConfigV2": "applications:\n- name: a\n",
+                     "rayClusterConfig": simple_raycluster(
+                         "x", workers=1).spec.to_dict()}}))
+        # a suspended cluster
+        rc = simple_raycluster("upsleep", workers=1)
+        rc.spec.suspend = True
+        cp.client.create(rc)
+
+        assert cp.wait_for(
+            lambda: cp.client.get(RayJob, "default", "upjob")
+            .status.job_deployment_status == "Running", timeout=20)
+        assert cp.wait_for(
+            lambda: cp.client.get(RayService, "default", "upsvc")
+            .condition_true("Ready"), timeout=25)
+        cp.stop()  # "old operator version" goes away mid-flight
+
+        cp2 = ControlPlane(kubelet_delay=0.01, job_runtime=0.3,
+                           poll_seconds=0.05, server=server)
+        # new operator's dashboard sees the job finish
+        cp2.dashboard.get_job_info_mock = lambda job_id: {
+            "submission_id": job_id, "status": "SUCCEEDED"}
+        cp2.start()
+        try:
+            assert cp2.wait_for(
+                lambda: cp2.client.get(RayJob, "default", "upjob")
+                .status.job_deployment_status == "Complete", timeout=25)
+            assert cp2.wait_for(
+                lambda: cp2.client.get(RayService, "default", "upsvc")
+                .condition_true("Ready"), timeout=25)
+            # the suspended cluster stays suspended (no pods resurrected)
+            import time as _t
+            _t.sleep(0.3)
+            pods = server.list("Pod", "default",
+                               {"ray.io/cluster": "upsleep"})
+            assert pods == []
+            # and a post-upgrade zero-downtime service upgrade still works
+            def set_image(svc):
+                svc.spec.ray_cluster_spec.worker_group_specs[0].template \
+                    .spec.containers[0].image = "rayproject/ray:2.47.0-rocm"
+            cp2.rayservice_reconciler.cluster_deletion_delay_s = 0.2
+            old_active = cp2.client.get(RayService, "default", "upsvc") \
+                .status.active_service_status.ray_cluster_name
+            cp2.client.update_with_retry(RayService, "default", "upsvc",
+                                         set_image)
+            assert cp2.wait_for(
+                lambda: cp2.client.get(RayService, "default", "upsvc")
+                .status.active_service_status.ray_cluster_name
+                not in (None, old_active), timeout=30)
+        finally:
+            cp2.stop()
