@@ -29,10 +29,12 @@ class TestSampleYamls:
     @pytest.fixture(autouse=True)
     def _gates(self):
         # samples exercising gated features name the gate in their header
-        # comment (tls-auth -> RayClusterMTLS); enable them like a deployed
-        # operator would via --feature-gates
+        # comment; enable them like a deployed operator would via
+        # --feature-gates
         import kuberay_amd.features as features
         features.set_gate("RayClusterMTLS", True)
+        features.set_gate("RayClusterNetworkPolicy", True)
+        features.set_gate("RayClusterHistoryServer", True)
         yield
         features.reset()
 
