@@ -173,6 +173,29 @@ class RayClusterReconciler(Reconciler):
         self.options = options or RayClusterReconcilerOptions()
         self.metrics = metrics
         self._provision_start: Dict[str, float] = {}
+        # spec-hash memo: (uid -> (generation, hash)). The hash excludes
+        # replicas/workersToDelete so it is stable within a generation, and
+        # computing it costs ~150 µs — per-pod recomputation was ~15% of a
+        # 2000-pod bench step.
+        self._spec_hash_cache: Dict[str, tuple] = {}
+
+    def _spec_hash(self, cluster) -> str:
+        """Memoized hash_without_replicas_and_workers_to_delete[:63].
+        KUBERAY_SPEC_HASH_MEMO=0 disables the memo (A/B measurement)."""
+        key = cluster.metadata.uid or f"{cluster.metadata.namespace}/{cluster.metadata.name}"
+        generation = cluster.metadata.generation
+        if os.environ.get("KUBERAY_SPEC_HASH_MEMO", "1") != "1":
+            generation = None
+        if generation is not None:
+            cached = self._spec_hash_cache.get(key)
+            if cached is not None and cached[0] == generation:
+                return cached[1]
+        value = hash_without_replicas_and_workers_to_delete(cluster.spec)[:63]
+        if generation is not None:
+            if len(self._spec_hash_cache) > 8192:
+                self._spec_hash_cache.clear()
+            self._spec_hash_cache[key] = (generation, value)
+        return value
 
     # ------------------------------------------------------------------
     def reconcile(self, request: Request) -> Result:
@@ -453,8 +476,7 @@ class RayClusterReconciler(Reconciler):
         # stale (reference: UpgradeStrategyRecreateHashKey)
         if (cluster.spec.upgrade_strategy is not None
                 and cluster.spec.upgrade_strategy.type == "Recreate"):
-            current_hash = hash_without_replicas_and_workers_to_delete(
-                cluster.spec)[:63]
+            current_hash = self._spec_hash(cluster)
             stale = [p for p in self._active(pods)
                      if p.labels.get(
                          C.HASH_WITHOUT_REPLICAS_AND_WORKERS_TO_DELETE_KEY)
@@ -653,7 +675,7 @@ class RayClusterReconciler(Reconciler):
             fqdn, self.options.default_container_envs, cluster.spec.ray_version)
         pod.metadata.owner_references = [k8s.owner_reference_for(cluster)]
         pod.metadata.labels[C.HASH_WITHOUT_REPLICAS_AND_WORKERS_TO_DELETE_KEY] = \
-            hash_without_replicas_and_workers_to_delete(cluster.spec)[:63]
+            self._spec_hash(cluster)
         if self.batch_scheduler is not None:
             self.batch_scheduler.add_metadata_to_pod(self.client, cluster,
                                                      "headgroup", pod)
@@ -687,7 +709,7 @@ class RayClusterReconciler(Reconciler):
             fqdn, self.options.default_container_envs, cluster.spec.ray_version)
         pod.metadata.owner_references = [k8s.owner_reference_for(cluster)]
         pod.metadata.labels[C.HASH_WITHOUT_REPLICAS_AND_WORKERS_TO_DELETE_KEY] = \
-            hash_without_replicas_and_workers_to_delete(cluster.spec)[:63]
+            self._spec_hash(cluster)
         if self.batch_scheduler is not None:
             self.batch_scheduler.add_metadata_to_pod(self.client, cluster,
                                                      group.group_name, pod)

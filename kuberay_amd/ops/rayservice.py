@@ -68,6 +68,8 @@ class RayServiceReconciler(Reconciler):
         self._unhealthy_since: Dict[Tuple[str, str, str], float] = {}
         # delayed old-cluster GC: (namespace, cluster) -> not-before time
         self._pending_deletions: Dict[Tuple[str, str], float] = {}
+        # goal-hash memo: uid -> (generation, hash)
+        self._goal_hash_cache: Dict[str, Tuple[int, str]] = {}
         self._lock = threading.Lock()
 
     @staticmethod
@@ -176,8 +178,21 @@ class RayServiceReconciler(Reconciler):
                                    cluster_name)
 
     def _goal_hash(self, svc: RayService) -> str:
-        sub = RayCluster(spec=svc.spec.ray_cluster_spec)
-        return hash_without_replicas_and_workers_to_delete(sub.spec)
+        # memoized per (uid, generation): the hash costs ~150 µs and the
+        # 2 s requeue loop recomputes it every cycle per service
+        key = svc.metadata.uid or f"{svc.metadata.namespace}/{svc.metadata.name}"
+        generation = svc.metadata.generation
+        if generation is not None:
+            cached = self._goal_hash_cache.get(key)
+            if cached is not None and cached[0] == generation:
+                return cached[1]
+        value = hash_without_replicas_and_workers_to_delete(
+            svc.spec.ray_cluster_spec)
+        if generation is not None:
+            if len(self._goal_hash_cache) > 8192:
+                self._goal_hash_cache.clear()
+            self._goal_hash_cache[key] = (generation, value)
+        return value
 
     def _reconcile_ray_cluster(self, svc: RayService) -> None:
         """rayservice_controller.go:1200-1254 + decision helpers :1351-1440."""
