@@ -252,8 +252,12 @@ def create_app(client: Optional[KubeClient] = None,
         ``computeTemplate`` get its resources injected."""
         templates = _templates(ns)
         spec = body.get("spec") or {}
-        groups = [spec.get("headGroupSpec") or {}]
-        groups += list(spec.get("workerGroupSpecs") or [])
+        # RayJob nests the cluster under rayClusterSpec, RayService under
+        # rayClusterConfig (template.go:54-79)
+        cluster_spec = (spec.get("rayClusterSpec")
+                        or spec.get("rayClusterConfig") or spec)
+        groups = [cluster_spec.get("headGroupSpec") or {}]
+        groups += list(cluster_spec.get("workerGroupSpecs") or [])
         for g in groups:
             tpl_name = g.pop("computeTemplate", None)
             if not tpl_name:
@@ -281,8 +285,7 @@ def create_app(client: Optional[KubeClient] = None,
             raise HTTPException(404, f"resource {plural} is not proxied")
         model = model_for_kind(kind)
         body.setdefault("metadata", {})["namespace"] = ns
-        if kind == "RayCluster":
-            _expand_compute_templates(ns, body)
+        _expand_compute_templates(ns, body)
         obj = model.from_dict(body)
         return client.create(obj).to_dict()
 

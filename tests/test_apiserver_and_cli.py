@@ -1014,3 +1014,47 @@ class TestKrayOutputFormats:
         r = runner.invoke(cli, ["-n", "ns1", "get", "cluster", "-o", "wide"])
         assert r.exit_code != 0
         assert "unsupported output format" in r.output
+
+
+class TestV2ComputeTemplateNested:
+    """Round-2 extension of the v2 compute-template middleware: RayJob
+    (spec.rayClusterSpec) and RayService (spec.rayClusterConfig) shapes are
+    expanded too (apiserversdk/util/template.go:54-79)."""
+
+    def test_rayjob_nested_expansion(self, api):
+        t, _, _ = api
+        r = t.post("/apis/ray.io/v1/namespaces/ns1/rayjobs", json={
+            "apiVersion": "ray.io/v1", "kind": "RayJob",
+            "metadata": {"name": "v2job"},
+            "spec": {"entrypoint": "python x.py",
+                     "rayClusterSpec": {
+                         "headGroupSpec": {"computeTemplate": "tpl",
+                                           "rayStartParams": {}},
+                         "workerGroupSpecs": [{
+                             "groupName": "g", "replicas": 1,
+                             "maxReplicas": 2, "computeTemplate": "tpl",
+                             "rayStartParams": {}}]}}})
+        assert r.status_code == 200, r.text
+        got = t.get("/apis/ray.io/v1/namespaces/ns1/rayjobs/v2job").json()
+        limits = got["spec"]["rayClusterSpec"]["workerGroupSpecs"][0][
+            "template"]["spec"]["containers"][0]["resources"]["limits"]
+        assert limits["amd.com/gpu"] == "2"
+
+    def test_rayservice_nested_expansion(self, api):
+        t, _, _ = api
+        r = t.post("/apis/ray.io/v1/namespaces/ns1/rayservices", json={
+            "apiVersion": "ray.io/v1", "kind": "RayService",
+            "metadata": {"name": "v2svc"},
+            "spec": {"serveConfigV2": "applications:\n- name: a\n",
+                     "rayClusterConfig": {
+                         "headGroupSpec": {"computeTemplate": "tpl",
+                                           "rayStartParams": {}},
+                         "workerGroupSpecs": [{
+                             "groupName": "g", "replicas": 1,
+                             "maxReplicas": 2, "computeTemplate": "tpl",
+                             "rayStartParams": {}}]}}})
+        assert r.status_code == 200, r.text
+        got = t.get("/apis/ray.io/v1/namespaces/ns1/rayservices/v2svc").json()
+        limits = got["spec"]["rayClusterConfig"]["workerGroupSpecs"][0][
+            "template"]["spec"]["containers"][0]["resources"]["limits"]
+        assert limits["amd.com/gpu"] == "2"
