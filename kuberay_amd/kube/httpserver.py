@@ -257,21 +257,26 @@ class _Handler(BaseHTTPRequestHandler):
         the event history), 410 Gone when the rv predates retained history,
         optional BOOKMARK events carrying the current rv."""
         start_rv = 0
-        replay = []
         if resource_version:
             try:
                 start_rv = int(resource_version)
             except ValueError:
                 start_rv = 0
+        # live watcher FIRST, replay second: the reverse order loses every
+        # event that lands between the history read and the registration
+        # (the rv-dedup below drops the overlap instead)
+        watcher = self.store.watch({kind})
+        replay = []
+        if resource_version:
             replay = self.store.events_since(start_rv, {kind})
             if replay is None:
                 # too old — client must re-list (Expired)
+                self.store.stop_watch(watcher)
                 return self._send_json(410, {
                     "kind": "Status", "apiVersion": "v1",
                     "status": "Failure", "reason": "Expired",
                     "message": f"too old resource version: {start_rv}",
                     "code": 410})
-        watcher = self.store.watch({kind})
         try:
             self.send_response(200)
             self.send_header("Content-Type", "application/json")

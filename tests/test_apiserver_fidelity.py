@@ -406,3 +406,25 @@ class TestControllersOverRest:
             kubelet.stop()
             manager.stop()
             adapter.stop()
+
+
+@pytest.mark.timeout(180)
+def test_sharded_operator_topology_end_to_end():
+    """Two REAL operator processes (--shards 2) against the HTTP facade:
+    every shard elects its own lease, owns its hash-split of the CR space,
+    and the whole space converges (benchmark/perf-tests/sharded.py at CI
+    size). Also the regression test for the facade watch-registration race
+    (events between history read and watcher attach were lost)."""
+    import json
+    import os
+    import subprocess
+    import sys
+    here = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run(
+        [sys.executable, "benchmark/perf-tests/sharded.py",
+         "--shards", "2", "--clusters", "24", "--timeout", "90"],
+        capture_output=True, text=True, cwd=here, timeout=170)
+    assert out.returncode == 0, (out.stdout[-1500:], out.stderr[-800:])
+    result = json.loads(out.stdout)
+    assert result["all_ready"] is True
+    assert result["all_shards_active"] is True
