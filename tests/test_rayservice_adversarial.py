@@ -368,22 +368,24 @@ class TestPromoteRaces:
         set_image_with_retry(cp, "rayproject/ray:2.47.0-rocm")
         assert cp.wait_for(
             lambda: svc_of(cp).status.pending_service_status.ray_cluster_name,
-            timeout=15)
+            timeout=30)
 
         def suspend(svc):
             svc.spec.suspend = True
         cp.client.update_with_retry(RayService, "default", "svc1", suspend)
+        # generous timeouts: under a loaded full-suite run (xdist + other
+        # control planes) reconcile cycles stretch well past the poll
         assert cp.wait_for(lambda: svc_of(cp).condition_true("Suspended"),
-                           timeout=20)
+                           timeout=40)
         assert cp.wait_for(lambda: cp.server.count("RayCluster") == 0,
-                           timeout=20)
+                           timeout=40)
         held["on"] = False
 
         def resume(svc):
             svc.spec.suspend = False
         cp.client.update_with_retry(RayService, "default", "svc1", resume)
         assert cp.wait_for(lambda: svc_of(cp).condition_true("Ready"),
-                           timeout=30)
+                           timeout=60)
 
     def test_pending_never_promotes_with_empty_apps(self, cp):
         """An upgrade whose pending cluster reports no RUNNING apps must
