@@ -64,6 +64,8 @@ def main() -> int:
     parser.add_argument("--shards", type=int, default=2)
     parser.add_argument("--clusters", type=int, default=200)
     parser.add_argument("--timeout", type=float, default=300.0)
+    parser.add_argument("--churn-minutes", type=float, default=0.0,
+                        help="after ready: sustained scale churn over REST")
     args = parser.parse_args()
 
     from kuberay_amd.kube.httpserver import KubeApiFacade
@@ -136,6 +138,77 @@ def main() -> int:
         result["reconciles_per_shard"] = per_shard
         # every shard must have done real work (the CR space is hash-split)
         result["all_shards_active"] = all(c > 0 for c in per_shard)
+
+        if args.churn_minutes > 0:
+            # sustained scale churn over REST: read-modify-write with
+            # conflict retry against live sharded operators, then pin every
+            # cluster to a known shape and require full convergence
+            import random
+            import threading
+
+            from kuberay_amd.kube.store import NotFoundError
+            from kuberay_amd.models import RayCluster
+            ordered = sorted(names)
+            stop = threading.Event()
+            counts = {"ops": 0, "errors": 0}
+            lock = threading.Lock()
+
+            def churn(seed):
+                rng = random.Random(seed)
+                local = RestClient(base_url=facade.url)
+                while not stop.is_set():
+                    name = rng.choice(ordered)
+                    try:
+                        def mutate(rc):
+                            rc.spec.worker_group_specs[0].replicas = \
+                                rng.randrange(0, 5)
+                        local.update_with_retry(RayCluster, "default",
+                                                name, mutate, attempts=20)
+                        with lock:
+                            counts["ops"] += 1
+                    except NotFoundError:
+                        pass
+                    except Exception:  # noqa: BLE001
+                        with lock:
+                            counts["errors"] += 1
+                    time.sleep(0.01)
+
+            churners = [threading.Thread(target=churn, args=(s,),
+                                         daemon=True) for s in range(4)]
+            for t in churners:
+                t.start()
+            time.sleep(args.churn_minutes * 60)
+            stop.set()
+            for t in churners:
+                t.join(timeout=5)
+            # pin and require convergence
+            for name in ordered:
+                def pin(rc):
+                    rc.spec.worker_group_specs[0].replicas = 1
+                client.update_with_retry(RayCluster, "default", name, pin,
+                                         attempts=50)
+            deadline = time.monotonic() + 120
+
+            def all_converged():
+                for name in ordered:
+                    obj = store.try_get("RayCluster", "default", name)
+                    st = (obj or {}).get("status", {})
+                    if st.get("state") != "ready" or \
+                            st.get("availableWorkerReplicas") != 1:
+                        return False
+                return True
+            converged = False
+            while time.monotonic() < deadline:
+                if all_converged():
+                    converged = True
+                    break
+                time.sleep(0.5)
+            result["churn_minutes"] = args.churn_minutes
+            result["churn_ops"] = counts["ops"]
+            result["churn_client_errors"] = counts["errors"]
+            result["churn_converged"] = converged
+            result["reconciles_per_shard_after_churn"] = [
+                scrape_reconciles(p) for p in metric_ports]
     finally:
         for p in procs:
             p.terminate()
@@ -149,8 +222,11 @@ def main() -> int:
         os.unlink(kubeconfig)
 
     print(json.dumps(result, indent=2))
-    return 0 if result.get("all_ready") and result.get("all_shards_active") \
-        else 1
+    ok = result.get("all_ready") and result.get("all_shards_active")
+    if args.churn_minutes > 0:
+        ok = ok and result.get("churn_converged") and \
+            result.get("churn_client_errors") == 0
+    return 0 if ok else 1
 
 
 if __name__ == "__main__":
