@@ -19,3 +19,9 @@ soak:             ## 5-minute 500-cluster churn soak
 
 crds:             ## regenerate deploy/crds from the pydantic models
 	python -m kuberay_amd.crds deploy/crds
+
+sharded:          ## sharded-operator topology bench over the HTTP facade
+	python benchmark/perf-tests/sharded.py --shards 2 --clusters 200
+
+upgrade-storm:    ## 100 simultaneous zero-downtime RayService upgrades
+	python benchmark/perf-tests/rayservice_upgrade.py --services 100
