@@ -45,21 +45,25 @@ def main():
     t0 = time.monotonic()
     try:
         names = [f"upsvc-{i:03d}" for i in range(args.services)]
+        # incremental upgrades require the autoscaler + intervalSeconds > 0
+        # (ValidateClusterUpgradeOptions, validation.go:718-750)
         upgrade_spec = ({"upgradeStrategy": {
             "type": "NewClusterWithIncrementalUpgrade",
             "clusterUpgradeOptions": {"gatewayClassName": "istio",
                                       "stepSizePercent": 25,
-                                      "intervalSeconds": 0,
+                                      "intervalSeconds": 1,
                                       "maxSurgePercent": 100}}}
             if args.incremental else {})
         for name in names:
+            cluster_spec = simple_raycluster(
+                "x", workers=1, gpus_per_worker=1).spec.to_dict()
+            if args.incremental:
+                cluster_spec["enableInTreeAutoscaling"] = True
             cp.client.create(RayService.from_dict({
                 "apiVersion": "ray.io/v1", "kind": "RayService",
                 "metadata": {"name": name, "namespace": "default"},
                 "spec": {"serveConfigV2": SERVE_CONFIG,
-                         "rayClusterConfig": simple_raycluster(
-                             "x", workers=1,
-                             gpus_per_worker=1).spec.to_dict(),
+                         "rayClusterConfig": cluster_spec,
                          **upgrade_spec}}))
 
         def state_of(name):
