@@ -38,7 +38,11 @@ from ..utils import constants as C
 from ..utils import names
 from ..utils import resources as res
 from ..utils.hashing import hash_without_replicas_and_workers_to_delete
-from ..utils.validation import validate_raycluster_metadata, validate_raycluster_spec
+from ..utils.validation import (
+    validate_raycluster_metadata,
+    validate_raycluster_spec,
+    validate_raycluster_status,
+)
 
 logger = logging.getLogger("kuberay.raycluster")
 
@@ -210,6 +214,15 @@ class RayClusterReconciler(Reconciler):
 
         if cluster.metadata.deletion_timestamp:
             return self._handle_deletion(cluster)
+
+        # status invariants (raycluster_controller.go:212-217): an invalid
+        # status (Suspending+Suspended both true) means a racing writer —
+        # event + requeue, never reconcile on top of it
+        status_errs = validate_raycluster_status(cluster)
+        if status_errs:
+            self.recorder.eventf(cluster, "Warning", "InvalidRayClusterStatus",
+                                 "; ".join(status_errs))
+            return Result(requeue_after=2)
 
         # validation (invalid spec: record event, do not requeue hot)
         errs = validate_raycluster_metadata(cluster.metadata) + validate_raycluster_spec(cluster)

@@ -33,7 +33,11 @@ from ..models.raycluster import ClusterState
 from ..utils import constants as C
 from ..utils import names
 from ..utils.dashboard_client import DashboardClientError
-from ..utils.validation import validate_rayjob_metadata, validate_rayjob_spec
+from ..utils.validation import (
+    validate_rayjob_metadata,
+    validate_rayjob_spec,
+    validate_rayjob_status,
+)
 
 logger = logging.getLogger("kuberay.rayjob")
 
@@ -117,7 +121,9 @@ class RayJobReconciler(Reconciler):
 
     # ------------------------------------------------------------------
     def _handle_new(self, rayjob: RayJob) -> Result:
-        errs = validate_rayjob_metadata(rayjob.metadata) + validate_rayjob_spec(rayjob)
+        errs = (validate_rayjob_metadata(rayjob.metadata)
+                + validate_rayjob_spec(rayjob)
+                + validate_rayjob_status(rayjob))
         if errs:
             self.recorder.eventf(rayjob, "Warning", "InvalidRayJobSpec", "; ".join(errs))
             self._set_status(rayjob, JDS.VALIDATION_FAILED,
