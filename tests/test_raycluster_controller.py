@@ -413,3 +413,65 @@ class TestUpgradeStrategyRecreate:
             lambda: get_cluster(control_plane).status.available_worker_replicas == 2)
         current = {p["metadata"]["name"] for p in pods_of(control_plane)}
         assert old_names.issubset(current)  # hash mutes replica changes
+
+
+class TestGcsFtHeadRecovery:
+    """e2e/gcs_ft e2e analog: with GCS FT enabled, a head crash is
+    recovered by RECREATING ONLY THE HEAD — worker pods survive untouched
+    and reconnect to the restored GCS (the reference relies on the 600 s
+    RAY_gcs_rpc_server_reconnect_timeout_s it injects)."""
+
+    def test_head_crash_keeps_workers_alive(self, control_plane):
+        cp = control_plane
+        cp.client.create(simple_raycluster(
+            "ftdemo", workers=2,
+            gcsFaultToleranceOptions={"backend": "redis",
+                                      "redisAddress": "redis://r:6379"}))
+        assert cp.wait_cluster_state("default", "ftdemo", "ready")
+        pods = cp.server.list("Pod", "default", {"ray.io/cluster": "ftdemo"})
+        workers_before = sorted(
+            p["metadata"]["name"] for p in pods
+            if p["metadata"]["labels"][C.RAY_NODE_TYPE_LABEL_KEY] == "worker")
+        head = next(p for p in pods
+                    if p["metadata"]["labels"][C.RAY_NODE_TYPE_LABEL_KEY]
+                    == "head")
+        assert len(workers_before) == 2
+        # head crashes
+        cp.server.patch_merge("Pod", "default", head["metadata"]["name"],
+                              {"status": {"phase": "Failed"}},
+                              subresource="status")
+
+        def head_recreated():
+            heads = [p for p in cp.server.list(
+                "Pod", "default", {"ray.io/cluster": "ftdemo"})
+                if p["metadata"]["labels"][C.RAY_NODE_TYPE_LABEL_KEY]
+                == "head"]
+            return (len(heads) == 1 and heads[0]["metadata"]["name"]
+                    != head["metadata"]["name"]
+                    and heads[0].get("status", {}).get("phase") == "Running")
+        assert cp.wait_for(head_recreated)
+        assert cp.wait_cluster_state("default", "ftdemo", "ready")
+        # the original workers were never touched
+        workers_after = sorted(
+            p["metadata"]["name"] for p in cp.server.list(
+                "Pod", "default", {"ray.io/cluster": "ftdemo"})
+            if p["metadata"]["labels"][C.RAY_NODE_TYPE_LABEL_KEY]
+            == "worker")
+        assert workers_after == workers_before
+
+    def test_ft_env_reaches_worker_pods(self, control_plane):
+        """The reconnect-timeout env the recovery story depends on is
+        actually injected into worker pods (pod.go:94-102 analog)."""
+        cp = control_plane
+        cp.client.create(simple_raycluster(
+            "ftenv", workers=1,
+            gcsFaultToleranceOptions={"backend": "redis",
+                                      "redisAddress": "redis://r:6379"}))
+        assert cp.wait_cluster_state("default", "ftenv", "ready")
+        worker = next(p for p in cp.server.list(
+            "Pod", "default", {"ray.io/cluster": "ftenv"})
+            if p["metadata"]["labels"][C.RAY_NODE_TYPE_LABEL_KEY]
+            == "worker")
+        env = {e["name"]: e.get("value") for e in
+               worker["spec"]["containers"][0].get("env", [])}
+        assert "RAY_gcs_rpc_server_reconnect_timeout_s" in env
