@@ -336,3 +336,42 @@ class TestCronTimeZone:
                      "jobTemplate": make_rayjob().spec.to_dict()}})
         errs = validate_raycronjob_spec(cron)
         assert any("timeZone" in e for e in errs)
+
+
+class TestSubmitterPodTemplate:
+    """e2erayjobsubmitter analog: a user-supplied submitterPodTemplate is
+    honored — custom image kept, custom command NOT overwritten by the
+    generated `ray job submit` pipeline (job.go GetSubmitterTemplate)."""
+
+    def test_custom_template_used_and_command_preserved(self, control_plane):
+        cp = control_plane
+        cp.client.create(make_rayjob(submitterPodTemplate={
+            "spec": {"containers": [{
+                "name": "custom-submitter",
+                "image": "mycorp/ray-submitter:1.2",
+                "command": ["python", "/opt/submit.py"],
+                "resources": {"limits": {"cpu": "2", "memory": "2Gi"}}}],
+                "restartPolicy": "Never"}}))
+        assert control_plane.wait_for(
+            lambda: cp.server.try_get("Job", "default", "job1") is not None)
+        job = cp.server.get("Job", "default", "job1")
+        container = job["spec"]["template"]["spec"]["containers"][0]
+        assert container["image"] == "mycorp/ray-submitter:1.2"
+        assert container["command"] == ["python", "/opt/submit.py"]
+        # env the submitter needs is still injected
+        env = {e["name"] for e in container.get("env", [])}
+        assert "RAY_DASHBOARD_ADDRESS" in env
+        assert "RAY_JOB_SUBMISSION_ID" in env
+        assert wait_deployment_status(cp, "job1", "Complete", timeout=30)
+
+    def test_default_template_uses_cluster_image(self, control_plane):
+        cp = control_plane
+        cp.client.create(make_rayjob())
+        assert cp.wait_for(
+            lambda: cp.server.try_get("Job", "default", "job1") is not None)
+        job = cp.server.get("Job", "default", "job1")
+        container = job["spec"]["template"]["spec"]["containers"][0]
+        head_image = make_rayjob().spec.ray_cluster_spec.head_group_spec \
+            .template.spec.containers[0].image
+        assert container["image"] == head_image
+        assert "ray job submit" in " ".join(container.get("args", []))
