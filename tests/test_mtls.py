@@ -174,3 +174,122 @@ class TestCertManagerMode:
                                    "ray-worker-cert-demo")
         assert head["spec"]["ipAddresses"] == ["10.99.1.2", "127.0.0.1"]
         assert worker["spec"]["ipAddresses"] == ["10.99.2.7", "127.0.0.1"]
+
+
+class TestRealTlsHandshake:
+    """Crypto-level e2e (the sim can't run real pods, but the certs are
+    real): a server presenting the operator-issued HEAD cert and a client
+    trusting only the operator CA complete a mutual-TLS handshake with
+    hostname verification against the issued SANs."""
+
+    def test_mutual_handshake_with_issued_certs(self, tmp_path):
+        import socket
+        import ssl
+        import threading
+
+        client = InMemoryClient()
+        client.create(simple_raycluster("demo", tlsOptions={"enabled": True}))
+        client.server.create({
+            "kind": "Pod",
+            "metadata": {"name": "demo-head-x", "namespace": "default",
+                         "labels": {"ray.io/cluster": "demo",
+                                    "ray.io/node-type": "head"}},
+            "spec": {"containers": [{"name": "ray"}]},
+            "status": {"phase": "Running", "podIP": "127.0.0.1"}})
+        MTLSReconciler(client).reconcile(("default", "demo"))
+
+        def write(name, secret_name, key):
+            data = client.get(k8s.Secret, "default", secret_name).data[key]
+            p = tmp_path / name
+            p.write_bytes(base64.b64decode(data))
+            return str(p)
+
+        ca = write("ca.crt", "ca-secret-demo", "ca.crt")
+        head_crt = write("head.crt", "ray-head-secret-demo", "tls.crt")
+        head_key = write("head.key", "ray-head-secret-demo", "tls.key")
+        worker_crt = write("worker.crt", "ray-worker-secret-demo", "tls.crt")
+        worker_key = write("worker.key", "ray-worker-secret-demo", "tls.key")
+
+        server_ctx = ssl.SSLContext(ssl.PROTOCOL_TLS_SERVER)
+        server_ctx.load_cert_chain(head_crt, head_key)
+        server_ctx.load_verify_locations(ca)
+        server_ctx.verify_mode = ssl.CERT_REQUIRED  # mutual TLS
+
+        client_ctx = ssl.SSLContext(ssl.PROTOCOL_TLS_CLIENT)
+        client_ctx.load_verify_locations(ca)
+        client_ctx.load_cert_chain(worker_crt, worker_key)
+        client_ctx.check_hostname = True
+
+        listener = socket.socket()
+        listener.bind(("127.0.0.1", 0))
+        listener.listen(1)
+        port = listener.getsockname()[1]
+        server_err = []
+
+        def serve():
+            try:
+                conn, _ = listener.accept()
+                with server_ctx.wrap_socket(conn, server_side=True) as tls:
+                    assert tls.recv(5) == b"hello"
+                    tls.sendall(b"world")
+            except Exception as e:  # noqa: BLE001
+                server_err.append(e)
+
+        t = threading.Thread(target=serve, daemon=True)
+        t.start()
+        raw = socket.create_connection(("127.0.0.1", port), timeout=5)
+        # hostname verification against the head cert's IP SAN (127.0.0.1)
+        with client_ctx.wrap_socket(raw, server_hostname="127.0.0.1") as tls:
+            tls.sendall(b"hello")
+            assert tls.recv(5) == b"world"
+        t.join(timeout=5)
+        listener.close()
+        assert not server_err, server_err
+
+    def test_foreign_ca_rejected(self, tmp_path):
+        """A cert from a DIFFERENT operator CA must fail verification —
+        the trust boundary is per-cluster."""
+        import socket
+        import ssl
+        import threading
+
+        other_ca, other_key = generate_ca("other-ca")
+        crt, key = generate_leaf(other_ca, other_key, "imposter",
+                                 ["localhost"], ["127.0.0.1"])
+        (tmp_path / "i.crt").write_bytes(crt)
+        (tmp_path / "i.key").write_bytes(key)
+
+        client = InMemoryClient()
+        client.create(simple_raycluster("demo", tlsOptions={"enabled": True}))
+        MTLSReconciler(client).reconcile(("default", "demo"))
+        ca_pem = base64.b64decode(
+            client.get(k8s.Secret, "default", "ca-secret-demo").data["ca.crt"])
+        (tmp_path / "ca.crt").write_bytes(ca_pem)
+
+        server_ctx = ssl.SSLContext(ssl.PROTOCOL_TLS_SERVER)
+        server_ctx.load_cert_chain(str(tmp_path / "i.crt"),
+                                   str(tmp_path / "i.key"))
+        client_ctx = ssl.SSLContext(ssl.PROTOCOL_TLS_CLIENT)
+        client_ctx.load_verify_locations(str(tmp_path / "ca.crt"))
+
+        listener = socket.socket()
+        listener.bind(("127.0.0.1", 0))
+        listener.listen(1)
+        port = listener.getsockname()[1]
+
+        def serve():
+            try:
+                conn, _ = listener.accept()
+                with server_ctx.wrap_socket(conn, server_side=True):
+                    pass
+            except Exception:  # noqa: BLE001 — expected: client aborts
+                pass
+
+        t = threading.Thread(target=serve, daemon=True)
+        t.start()
+        raw = socket.create_connection(("127.0.0.1", port), timeout=5)
+        with pytest.raises(ssl.SSLError):
+            client_ctx.wrap_socket(raw, server_hostname="127.0.0.1")
+        raw.close()
+        t.join(timeout=5)
+        listener.close()
