@@ -375,3 +375,67 @@ class TestSubmitterPodTemplate:
             .template.spec.containers[0].image
         assert container["image"] == head_image
         assert "ray job submit" in " ".join(container.get("args", []))
+
+
+class TestCronSuspendAndCatchup:
+    """raycronjob_controller.go analogs not yet covered: suspend gates
+    firing; a long gap collapses to ONE catch-up job (latest missed tick),
+    not one per missed minute."""
+
+    def _cron(self, client, now_fn, **spec):
+        from kuberay_amd.models import RayCronJob
+        from kuberay_amd.ops.raycronjob import RayCronJobReconciler
+        body = {"schedule": "* * * * *",
+                "jobTemplate": make_rayjob().spec.to_dict()}
+        body.update(spec)
+        cron = client.create(RayCronJob.from_dict({
+            "apiVersion": "ray.io/v1", "kind": "RayCronJob",
+            "metadata": {"name": "cj", "namespace": "default"},
+            "spec": body}))
+        return cron, RayCronJobReconciler(client, now_fn=now_fn)
+
+    def test_suspended_cron_never_fires(self):
+        import datetime as dt
+
+        from kuberay_amd.kube.client import InMemoryClient
+        client = InMemoryClient()
+        now = dt.datetime.utcnow() + dt.timedelta(minutes=10)
+        _, rec = self._cron(client, lambda: now, suspend=True)
+        rec.reconcile(("default", "cj"))
+        assert client.server.count("RayJob") == 0
+
+    def test_missed_ticks_collapse_to_latest(self):
+        import datetime as dt
+
+        from kuberay_amd.kube.client import InMemoryClient
+        from kuberay_amd.models import RayCronJob
+        client = InMemoryClient()
+        # operator was down for 30 minutes: exactly ONE catch-up RayJob
+        now = dt.datetime.utcnow() + dt.timedelta(minutes=30)
+        _, rec = self._cron(client, lambda: now)
+        rec.reconcile(("default", "cj"))
+        assert client.server.count("RayJob") == 1
+        status = client.get(RayCronJob, "default", "cj").status
+        assert status.last_schedule_time
+        # the recorded tick is the LATEST missed one (within a minute of now)
+        fired = dt.datetime.strptime(status.last_schedule_time,
+                                     "%Y-%m-%dT%H:%M:%SZ")
+        assert (now - fired).total_seconds() <= 120
+
+    def test_resume_after_suspend_fires_fresh(self):
+        import datetime as dt
+
+        from kuberay_amd.kube.client import InMemoryClient
+        from kuberay_amd.models import RayCronJob
+        client = InMemoryClient()
+        now = {"t": dt.datetime.utcnow() + dt.timedelta(minutes=5)}
+        cron, rec = self._cron(client, lambda: now["t"], suspend=True)
+        rec.reconcile(("default", "cj"))
+        assert client.server.count("RayJob") == 0
+
+        def resume(obj):
+            obj.spec.suspend = False
+        client.update_with_retry(RayCronJob, "default", "cj", resume)
+        now["t"] += dt.timedelta(minutes=2)
+        rec.reconcile(("default", "cj"))
+        assert client.server.count("RayJob") == 1
