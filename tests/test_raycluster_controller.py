@@ -475,3 +475,30 @@ class TestGcsFtHeadRecovery:
         env = {e["name"]: e.get("value") for e in
                worker["spec"]["containers"][0].get("env", [])}
         assert "RAY_gcs_rpc_server_reconnect_timeout_s" in env
+
+
+class TestIngressReconcile:
+    """reconcileIngress e2e (raycluster_controller.go:496-606): the head
+    Ingress appears when enableIngress is set and is owned by the cluster."""
+
+    def test_ingress_created_when_enabled(self, control_plane):
+        cp = control_plane
+        rc = simple_raycluster("ingdemo", workers=0)
+        rc.spec.head_group_spec.enable_ingress = True
+        cp.client.create(rc)
+        assert cp.wait_cluster_state("default", "ingdemo", "ready")
+        assert cp.wait_for(
+            lambda: cp.server.count("Ingress") == 1, timeout=10)
+        ing = cp.server.list("Ingress")[0]
+        owner = ing["metadata"]["ownerReferences"][0]
+        assert owner["kind"] == "RayCluster" and owner["name"] == "ingdemo"
+        # routes to the head service's dashboard port
+        rule = ing["spec"]["rules"][0]
+        backend = rule["http"]["paths"][0]["backend"]["service"]
+        assert backend["name"] == "ingdemo-head-svc"
+
+    def test_no_ingress_by_default(self, control_plane):
+        cp = control_plane
+        cp.client.create(simple_raycluster("noing", workers=0))
+        assert cp.wait_cluster_state("default", "noing", "ready")
+        assert cp.server.count("Ingress") == 0
