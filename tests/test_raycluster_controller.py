@@ -305,6 +305,20 @@ class TestGcsPvcRetain:
         # Retain: the PVC is NOT owner-referenced, so it survives GC
         assert control_plane.server.try_get(
             "PersistentVolumeClaim", "default", "demo-gcs-pvc") is not None
+        # ADOPTION: recreating the cluster under the same name reuses the
+        # retained PVC — the GCS state survives the recreate
+        # (raycluster_controller.go:657-743 retain/adopt semantics)
+        pvc_uid = control_plane.server.get(
+            "PersistentVolumeClaim", "default", "demo-gcs-pvc"
+        )["metadata"]["uid"]
+        control_plane.client.create(simple_raycluster(
+            "demo", gcsFaultToleranceOptions={
+                "backend": "embedded",
+                "storage": {"size": "1Gi", "deletionPolicy": "Retain"}}))
+        assert control_plane.wait_cluster_state("default", "demo", "ready")
+        assert control_plane.server.get(
+            "PersistentVolumeClaim", "default", "demo-gcs-pvc"
+        )["metadata"]["uid"] == pvc_uid, "retained PVC was replaced"
 
     def test_delete_policy_pvc_garbage_collected(self, control_plane):
         cluster = simple_raycluster("demo", gcsFaultToleranceOptions={
