@@ -470,6 +470,10 @@ class InMemoryApiServer:
             current = self._backend.remove(key)
             if current is None:
                 return
+            # a delete is its own revision (real-apiserver semantics):
+            # without a fresh rv, a watcher resuming from the object's last
+            # rv would treat the DELETED event as already-seen and miss it
+            current["metadata"]["resourceVersion"] = self._next_rv()
             uid = current["metadata"]["uid"]
             dependents = self._backend.dependents(uid)
             self._backend.drop_owner(uid)

@@ -428,3 +428,20 @@ def test_sharded_operator_topology_end_to_end():
     result = json.loads(out.stdout)
     assert result["all_ready"] is True
     assert result["all_shards_active"] is True
+
+
+class TestDeleteRevision:
+    """A delete is its own revision: a watcher resuming from the rv at
+    which it saw the object's LAST update must still receive the DELETED
+    event (kube-apiserver semantics; regression for rv-reuse on delete)."""
+
+    def test_resumed_watch_sees_deletion(self, facade):
+        client = RestClient(base_url=facade.url)
+        client.raw_create(_cm("doomed"))
+        _, rv = client.raw_list_with_rv("ConfigMap")  # saw the object
+        facade.store.delete("ConfigMap", "default", "doomed")
+        got = []
+        for event_type, obj in client.raw_watch_stream("ConfigMap", rv):
+            got.append((event_type, obj["metadata"]["name"]))
+            break
+        assert got == [("DELETED", "doomed")]
