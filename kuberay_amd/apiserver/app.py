@@ -80,6 +80,57 @@ def create_app(client: Optional[KubeClient] = None,
         return {}
 
     # ------------------------------------------------------------------
+    # v1: image templates (config.proto ImageTemplateService :138-231,
+    # grpc-gateway HTTP mapping; stored as ConfigMaps like the gRPC layer)
+    # ------------------------------------------------------------------
+    IMAGE_TEMPLATE_LABEL = "ray.io/image-template"
+
+    def _image_template_of(cm: k8s.ConfigMap) -> Dict[str, Any]:
+        import json as _json
+        data = cm.data or {}
+        return {"name": data.get("name", ""),
+                "namespace": cm.metadata.namespace or "",
+                "baseImage": data.get("baseImage", ""),
+                "pipPackages": _json.loads(data.get("pipPackages", "[]")),
+                "environmentVariables": _json.loads(
+                    data.get("environmentVariables", "{}") or "{}")}
+
+    @app.post("/apis/v1/namespaces/{ns}/image_templates")
+    def create_image_template(ns: str, body: Dict[str, Any]):
+        import json as _json
+        if not body.get("name"):
+            raise HTTPException(400, "name is required")
+        cm = k8s.ConfigMap(
+            metadata=k8s.ObjectMeta(
+                name=f"imagetpl-{body['name']}", namespace=ns,
+                labels={IMAGE_TEMPLATE_LABEL: body["name"]}),
+            data={"name": body["name"],
+                  "baseImage": body.get("baseImage", ""),
+                  "pipPackages": _json.dumps(body.get("pipPackages") or []),
+                  "environmentVariables": _json.dumps(
+                      body.get("environmentVariables") or {})})
+        client.create(cm)
+        return body
+
+    @app.get("/apis/v1/namespaces/{ns}/image_templates")
+    def list_image_templates(ns: str):
+        out = [_image_template_of(cm) for cm in client.list(k8s.ConfigMap, ns)
+               if (cm.metadata.labels or {}).get(IMAGE_TEMPLATE_LABEL)]
+        return {"imageTemplates": out}
+
+    @app.get("/apis/v1/namespaces/{ns}/image_templates/{name}")
+    def get_image_template(ns: str, name: str):
+        cm = client.try_get(k8s.ConfigMap, ns, f"imagetpl-{name}")
+        if cm is None:
+            raise HTTPException(404, f"image template {name} not found")
+        return _image_template_of(cm)
+
+    @app.delete("/apis/v1/namespaces/{ns}/image_templates/{name}")
+    def delete_image_template(ns: str, name: str):
+        client.delete(k8s.ConfigMap, ns, f"imagetpl-{name}")
+        return {}
+
+    # ------------------------------------------------------------------
     # v1: clusters
     # ------------------------------------------------------------------
     @app.post("/apis/v1/namespaces/{ns}/clusters")

@@ -1058,3 +1058,50 @@ class TestV2ComputeTemplateNested:
         limits = got["spec"]["rayClusterConfig"]["workerGroupSpecs"][0][
             "template"]["spec"]["containers"][0]["resources"]["limits"]
         assert limits["amd.com/gpu"] == "2"
+
+
+class TestImageTemplatesHttp:
+    """HTTP mapping of ImageTemplateService (grpc-gateway parity)."""
+
+    def test_crud(self, api):
+        t, _, _ = api
+        r = t.post("/apis/v1/namespaces/ns1/image_templates", json={
+            "name": "rocm-base", "baseImage": "rocm/ray:2.46.0",
+            "pipPackages": ["numpy", "scipy"],
+            "environmentVariables": {"HSA_ENABLE_IPC_MODE_LEGACY": "0"}})
+        assert r.status_code == 200
+        got = t.get(
+            "/apis/v1/namespaces/ns1/image_templates/rocm-base").json()
+        assert got["baseImage"] == "rocm/ray:2.46.0"
+        assert got["pipPackages"] == ["numpy", "scipy"]
+        listed = t.get("/apis/v1/namespaces/ns1/image_templates").json()
+        assert len(listed["imageTemplates"]) == 1
+        assert t.delete(
+            "/apis/v1/namespaces/ns1/image_templates/rocm-base"
+        ).status_code == 200
+        assert t.get(
+            "/apis/v1/namespaces/ns1/image_templates/rocm-base"
+        ).status_code == 404
+
+    def test_grpc_and_http_share_storage(self, api):
+        """The gRPC ImageTemplateService and the HTTP routes read/write the
+        same ConfigMaps."""
+        t, client, _ = api
+        t.post("/apis/v1/namespaces/ns1/image_templates", json={
+            "name": "shared", "baseImage": "img"})
+        import grpc as _grpc
+        from kuberay_amd.apiserver.grpc_api import (
+            GetRequest, ImageTemplate, create_grpc_server)
+        server = create_grpc_server(client, port=0)
+        port = server.add_insecure_port("127.0.0.1:0")
+        server.start()
+        try:
+            channel = _grpc.insecure_channel(f"127.0.0.1:{port}")
+            fn = channel.unary_unary(
+                "/kuberayamd.v1.ImageTemplateService/GetImageTemplate",
+                request_serializer=lambda m: m.SerializeToString(),
+                response_deserializer=ImageTemplate.FromString)
+            got = fn(GetRequest(name="shared", namespace="ns1"), timeout=5)
+            assert got.base_image == "img"
+        finally:
+            server.stop(0)
