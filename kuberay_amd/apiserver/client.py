@@ -61,6 +61,25 @@ class ApiServerClient:
         self._check(self._http.delete(
             self._url(namespace, "compute_templates", name)))
 
+    # -- image templates (ImageTemplateService HTTP mapping) ------------
+    def create_image_template(self, namespace: str,
+                              template: Dict[str, Any]) -> Dict[str, Any]:
+        return self._check(self._http.post(
+            self._url(namespace, "image_templates"), json=template))
+
+    def get_image_template(self, namespace: str, name: str) -> Dict[str, Any]:
+        return self._check(self._http.get(
+            self._url(namespace, "image_templates", name)))
+
+    def list_image_templates(self, namespace: str) -> List[Dict[str, Any]]:
+        return self._check(self._http.get(
+            self._url(namespace, "image_templates"))).get(
+            "imageTemplates", [])
+
+    def delete_image_template(self, namespace: str, name: str) -> None:
+        self._check(self._http.delete(
+            self._url(namespace, "image_templates", name)))
+
     # -- clusters -------------------------------------------------------
     def create_cluster(self, namespace: str,
                        cluster: Dict[str, Any]) -> Dict[str, Any]:
@@ -120,3 +139,4 @@ class ApiServerClient:
 
     def delete_service(self, namespace: str, name: str) -> None:
         self._check(self._http.delete(self._url(namespace, "services", name)))
+
