@@ -14,10 +14,10 @@ _DEFAULTS: Dict[str, bool] = {
     "RayJobDeletionPolicy": True,             # beta, on
     "RayMultiHostIndexing": True,             # beta, on
     "RayServiceIncrementalUpgrade": True,     # beta, on (features.go:105-117)
-    "RayCronJob": True,
+    "RayCronJob": False,              # alpha, off (features.go:110)
     "SidecarSubmitterRestart": False,
     "RayClusterNetworkPolicy": False,
-    "GCSFaultToleranceEmbeddedStorage": True,
+    "GCSFaultToleranceEmbeddedStorage": False,  # alpha, off (features.go:113)
     "RayClusterMTLS": False,
     "RayClusterHistoryServer": False,
     "KubernetesWAS": False,

@@ -65,8 +65,14 @@ class TestOperatorWiring:
         cfg = Configuration()
         m, client, metrics, autoscaler = build_manager(cfg)
         names_ = [c.name for c in m.controllers]
-        assert names_ == ["raycluster", "rayjob", "rayservice", "raycronjob"]
+        # RayCronJob is alpha/off by default (features.go:110)
+        assert names_ == ["raycluster", "rayjob", "rayservice"]
         assert metrics is not None and autoscaler is not None
+
+    def test_raycronjob_controller_gated(self):
+        features.set_gate("RayCronJob", True)
+        m, *_ = build_manager(Configuration())
+        assert "raycronjob" in [c.name for c in m.controllers]
 
     def test_networkpolicy_controller_gated(self):
         features.set_gate("RayClusterNetworkPolicy", True)

@@ -214,6 +214,13 @@ class TestValidationRejection:
 
 
 class TestGcsFaultTolerance:
+    @pytest.fixture(autouse=True)
+    def _embedded_gate(self):
+        import kuberay_amd.features as features
+        features.set_gate("GCSFaultToleranceEmbeddedStorage", True)
+        yield
+        features.reset()
+
     def test_redis_cleanup_finalizer_machine(self, control_plane):
         cluster = simple_raycluster("demo", gcsFaultToleranceOptions={
             "backend": "redis", "redisAddress": "redis://r:6379"})
@@ -291,6 +298,13 @@ class TestWorkerGroupAddition:
 
 
 class TestGcsPvcRetain:
+    @pytest.fixture(autouse=True)
+    def _embedded_gate(self):
+        import kuberay_amd.features as features
+        features.set_gate("GCSFaultToleranceEmbeddedStorage", True)
+        yield
+        features.reset()
+
     def test_retain_policy_pvc_survives_cluster_deletion(self, control_plane):
         cluster = simple_raycluster("demo", gcsFaultToleranceOptions={
             "backend": "embedded",

@@ -35,6 +35,8 @@ class TestSampleYamls:
         features.set_gate("RayClusterMTLS", True)
         features.set_gate("RayClusterNetworkPolicy", True)
         features.set_gate("RayClusterHistoryServer", True)
+        features.set_gate("GCSFaultToleranceEmbeddedStorage", True)
+        features.set_gate("RayCronJob", True)
         yield
         features.reset()
 
