@@ -716,3 +716,26 @@ class TestMultiKueueInterop:
                                                "handover", admit)
         assert control_plane.wait_cluster_state("default", "handover",
                                                 "ready", timeout=20)
+
+
+class TestGateDefaultParity:
+    """Pins every gate default to the reference table
+    (pkg/features/features.go:104-117) so drift is caught at CI time."""
+
+    REFERENCE_DEFAULTS = {
+        "RayClusterStatusConditions": True,        # beta
+        "RayJobDeletionPolicy": True,              # beta
+        "RayMultiHostIndexing": True,              # beta
+        "RayServiceIncrementalUpgrade": True,      # beta
+        "RayCronJob": False,                       # alpha
+        "SidecarSubmitterRestart": False,          # alpha
+        "RayClusterNetworkPolicy": False,          # alpha
+        "GCSFaultToleranceEmbeddedStorage": False, # alpha
+        "RayClusterMTLS": False,                   # alpha
+        "RayClusterHistoryServer": False,          # alpha
+        "KubernetesWAS": False,                    # alpha
+    }
+
+    def test_defaults_match_reference(self):
+        for gate, want in self.REFERENCE_DEFAULTS.items():
+            assert features.enabled(gate) is want, gate
