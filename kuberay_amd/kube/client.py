@@ -109,6 +109,63 @@ class KubeClient:
         raise last if last is not None else RuntimeError("unreachable")
 
 
+class RawObjectClient:
+    """Dict-object verbs over either backend: the in-memory server
+    directly, or a RestClient's raw_* methods (for kinds with no typed
+    model — PodGroup, cert-manager CRs, Gateway API objects, Nodes).
+
+    On the REST backend the kind's path is derived from the object's
+    apiVersion when the kind isn't in the static RESOURCES map, so
+    schedulers and cert issuance work identically in real deployments.
+    """
+
+    def __init__(self, client):
+        self.server = getattr(client, "server", None)
+        self.client = client
+
+    def try_get(self, kind: str, namespace: str, name: str,
+                api_version: Optional[str] = None):
+        if self.server is not None:
+            return self.server.try_get(kind, namespace, name)
+        fn = getattr(self.client, "raw_try_get", None)
+        return fn(kind, namespace, name, api_version=api_version) if fn else None
+
+    def create(self, obj: Dict[str, Any]):
+        if self.server is not None:
+            return self.server.create(obj)
+        fn = getattr(self.client, "raw_create", None)
+        return fn(obj) if fn else None
+
+    def update(self, obj: Dict[str, Any]):
+        if self.server is not None:
+            return self.server.update(obj)
+        fn = getattr(self.client, "raw_update", None)
+        return fn(obj) if fn else None
+
+    def patch(self, kind: str, namespace: str, name: str,
+              patch: Dict[str, Any], api_version: Optional[str] = None):
+        if self.server is not None:
+            return self.server.patch_merge(kind, namespace, name, patch)
+        fn = getattr(self.client, "raw_patch", None)
+        return fn(kind, namespace, name, patch,
+                  api_version=api_version) if fn else None
+
+    def delete(self, kind: str, namespace: str, name: str,
+               api_version: Optional[str] = None) -> None:
+        if self.server is not None:
+            self.server.delete(kind, namespace, name)
+            return
+        fn = getattr(self.client, "raw_delete", None)
+        if fn is not None:
+            fn(kind, namespace, name, api_version=api_version)
+
+    def list(self, kind: str, namespace: Optional[str] = None):
+        if self.server is not None:
+            return self.server.list(kind, namespace)
+        fn = getattr(self.client, "raw_list", None)
+        return fn(kind, namespace) if fn else []
+
+
 def _kind_of(model_or_obj) -> str:
     if isinstance(model_or_obj, type):
         # model class: read the default of the `kind` field

@@ -33,11 +33,31 @@ _PATH_RE = re.compile(
     r"(?:/(?P<sub>status))?$")
 
 
-def _parse_path(path: str) -> Optional[Tuple[str, Optional[str], Optional[str], Optional[str]]]:
+# dynamic kinds (CRDs the static map doesn't know: PodGroup, cert-manager
+# objects, ...) learned from POST bodies / store contents, keyed like the
+# static map
+_DYNAMIC_KINDS: Dict[Tuple[str, str], str] = {}
+
+
+def _plural_of(kind: str) -> str:
+    # mirror of RestClient._path's convention for unknown kinds
+    return kind.lower() + ("es" if kind.lower().endswith("s") else "s")
+
+
+def _parse_path(path: str, store=None, body_kind: Optional[str] = None,
+                ) -> Optional[Tuple[str, Optional[str], Optional[str], Optional[str]]]:
     m = _PATH_RE.match(path)
     if not m:
         return None
-    kind = _KIND_BY_PREFIX_PLURAL.get((m.group("prefix"), m.group("plural")))
+    key = (m.group("prefix"), m.group("plural"))
+    kind = _KIND_BY_PREFIX_PLURAL.get(key) or _DYNAMIC_KINDS.get(key)
+    if kind is None and store is not None:
+        for k in store.kinds():
+            if _plural_of(k) == key[1]:
+                kind = _DYNAMIC_KINDS[key] = k
+                break
+    if kind is None and body_kind and _plural_of(body_kind) == key[1]:
+        kind = _DYNAMIC_KINDS[key] = body_kind
     if kind is None:
         return None
     name, sub = m.group("name"), m.group("sub")
@@ -209,7 +229,7 @@ class _Handler(BaseHTTPRequestHandler):
     # ------------------------------------------------------------------
     def do_GET(self):  # noqa: N802
         url = urlparse(self.path)
-        parsed = _parse_path(url.path)
+        parsed = _parse_path(url.path, store=self.store)
         if parsed is None:
             return self._send_json(404, {"message": f"unknown path {url.path}"})
         kind, ns, name, _ = parsed
@@ -327,11 +347,12 @@ class _Handler(BaseHTTPRequestHandler):
             watcher.stop()
 
     def do_POST(self):  # noqa: N802
-        parsed = _parse_path(urlparse(self.path).path)
+        obj = self._body()
+        parsed = _parse_path(urlparse(self.path).path, store=self.store,
+                             body_kind=obj.get("kind"))
         if parsed is None:
             return self._send_json(404, {"message": "unknown path"})
         kind, ns, _, _ = parsed
-        obj = self._body()
         obj.setdefault("kind", kind)
         obj.setdefault("metadata", {}).setdefault("namespace", ns or "default")
         try:
@@ -340,11 +361,12 @@ class _Handler(BaseHTTPRequestHandler):
             self._send_error_status(e)
 
     def do_PUT(self):  # noqa: N802
-        parsed = _parse_path(urlparse(self.path).path)
+        obj = self._body()
+        parsed = _parse_path(urlparse(self.path).path, store=self.store,
+                             body_kind=obj.get("kind"))
         if parsed is None:
             return self._send_json(404, {"message": "unknown path"})
         kind, ns, name, sub = parsed
-        obj = self._body()
         obj.setdefault("kind", kind)
         obj.setdefault("metadata", {}).setdefault("namespace", ns or "default")
         if name:
@@ -358,7 +380,7 @@ class _Handler(BaseHTTPRequestHandler):
         """Dispatch on Content-Type like the real apiserver:
         merge-patch (RFC 7386), json-patch (RFC 6902) and
         strategic-merge-patch (list merge by the `name` merge key)."""
-        parsed = _parse_path(urlparse(self.path).path)
+        parsed = _parse_path(urlparse(self.path).path, store=self.store)
         if parsed is None:
             return self._send_json(404, {"message": "unknown path"})
         kind, ns, name, sub = parsed
@@ -391,7 +413,7 @@ class _Handler(BaseHTTPRequestHandler):
             self._send_error_status(e)
 
     def do_DELETE(self):  # noqa: N802
-        parsed = _parse_path(urlparse(self.path).path)
+        parsed = _parse_path(urlparse(self.path).path, store=self.store)
         if parsed is None:
             return self._send_json(404, {"message": "unknown path"})
         kind, ns, name, _ = parsed

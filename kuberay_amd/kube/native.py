@@ -47,9 +47,14 @@ class NativeBackend:
 
     def __init__(self) -> None:
         self._store = engine.NativeStore()
+        # kind -> live-object count; NativeStore has no kind enumeration,
+        # and the HTTP facade needs one for dynamic (CRD) path routing
+        self._kind_counts: Dict[str, int] = {}
 
     def put(self, key: Key, obj: Dict[str, Any]) -> None:
         kind, ns, name = key
+        if not self._store.contains(kind, ns, name):
+            self._kind_counts[kind] = self._kind_counts.get(kind, 0) + 1
         meta = obj.get("metadata", {})
         labels = list((meta.get("labels") or {}).items())
         owners = [ref.get("uid") for ref in meta.get("ownerReferences") or []
@@ -71,6 +76,12 @@ class NativeBackend:
 
     def remove(self, key: Key) -> Optional[Dict[str, Any]]:
         blob = self._store.remove(*key)
+        if blob is not None:
+            n = self._kind_counts.get(key[0], 0) - 1
+            if n > 0:
+                self._kind_counts[key[0]] = n
+            else:
+                self._kind_counts.pop(key[0], None)
         return _loads(blob) if blob is not None else None
 
     def list(self, kind: str, namespace: Optional[str],
@@ -101,3 +112,6 @@ class NativeBackend:
 
     def total_bytes(self) -> int:
         return self._store.total_bytes()
+
+    def kinds(self) -> List[str]:
+        return list(self._kind_counts)

@@ -158,8 +158,20 @@ class RestClient(KubeClient):
     # -- paths ---------------------------------------------------------
     @staticmethod
     def _path(kind: str, namespace: Optional[str], name: Optional[str] = None,
-              subresource: Optional[str] = None) -> str:
-        prefix, plural = RESOURCES[kind]
+              subresource: Optional[str] = None,
+              api_version: Optional[str] = None) -> str:
+        if kind in RESOURCES:
+            prefix, plural = RESOURCES[kind]
+        else:
+            # dynamic kinds (PodGroup in two API groups, cert-manager
+            # Certificate/Issuer, Workload...): derive the path from the
+            # object's apiVersion + the standard lowercase-plural convention
+            if not api_version:
+                raise KeyError(f"unknown kind {kind!r} needs api_version")
+            prefix = ("/api/v1" if api_version == "v1"
+                      else f"/apis/{api_version}")
+            plural = kind.lower() + ("es" if kind.lower().endswith("s")
+                                     else "s")
         p = f"{prefix}/namespaces/{namespace}/{plural}" if namespace else f"{prefix}/{plural}"
         if name:
             p += f"/{name}"
@@ -242,7 +254,9 @@ class RestClient(KubeClient):
     def raw_create(self, obj: Dict[str, Any]) -> Dict[str, Any]:
         kind = obj.get("kind", "")
         ns = obj.get("metadata", {}).get("namespace", "default")
-        resp = self._check(self._http.post(self._path(kind, ns), json=obj))
+        resp = self._check(self._http.post(
+            self._path(kind, ns, api_version=obj.get("apiVersion")),
+            json=obj))
         return resp.json()
 
     def raw_update(self, obj: Dict[str, Any]) -> Dict[str, Any]:
@@ -258,19 +272,25 @@ class RestClient(KubeClient):
         return resp.json()
 
     def raw_patch(self, kind: str, namespace: str, name: str,
-                  patch: Dict[str, Any]) -> Dict[str, Any]:
+                  patch: Dict[str, Any],
+                  api_version: Optional[str] = None) -> Dict[str, Any]:
         resp = self._check(self._http.patch(
-            self._path(kind, namespace, name), json=patch,
+            self._path(kind, namespace, name, api_version=api_version),
+            json=patch,
             headers={"Content-Type": "application/merge-patch+json"}))
         return resp.json()
 
-    def raw_delete(self, kind: str, namespace: str, name: str) -> None:
-        self._check(self._http.delete(self._path(kind, namespace, name)))
+    def raw_delete(self, kind: str, namespace: str, name: str,
+                   api_version: Optional[str] = None) -> None:
+        self._check(self._http.delete(
+            self._path(kind, namespace, name, api_version=api_version)))
 
-    def raw_try_get(self, kind: str, namespace: str, name: str) -> Optional[Dict[str, Any]]:
+    def raw_try_get(self, kind: str, namespace: str, name: str,
+                    api_version: Optional[str] = None) -> Optional[Dict[str, Any]]:
         try:
             return self._check(self._http.get(
-                self._path(kind, namespace, name))).json()
+                self._path(kind, namespace, name,
+                           api_version=api_version))).json()
         except NotFoundError:
             return None
 

@@ -159,6 +159,9 @@ class PyBackend:
         obj = self._objects.get(key)
         return jsoncopy(obj) if obj is not None else None
 
+    def kinds(self):
+        return [k for k, keys in self._kind_index.items() if keys]
+
     def rv(self, key: Key) -> Optional[str]:
         obj = self._objects.get(key)
         return obj["metadata"].get("resourceVersion") if obj is not None else None
@@ -351,6 +354,13 @@ class InMemoryApiServer:
              label_selector: Optional[Dict[str, str]] = None) -> List[Dict[str, Any]]:
         with self._lock:
             return self._backend.list(kind, namespace, label_selector)
+
+    def kinds(self):
+        """Kinds with at least one stored object (dynamic-routing support
+        in the HTTP facade)."""
+        with self._lock:
+            fn = getattr(self._backend, "kinds", None)
+            return fn() if fn else []
 
     def list_pod_views(self, namespace: Optional[str] = None,
                        label_selector: Optional[Dict[str, str]] = None) -> List[PodView]:

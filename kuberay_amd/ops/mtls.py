@@ -205,9 +205,8 @@ class MTLSReconciler(Reconciler):
         """Create cert-manager CRs (Issuer → CA cert → CA Issuer → leaves)."""
         namespace = cluster.metadata.namespace or "default"
         cname = cluster.metadata.name
-        server = getattr(self.client, "server", None)
-        if server is None:
-            return
+        from ..kube.client import RawObjectClient
+        raw = RawObjectClient(self.client)
         owner = [k8s.owner_reference_for(cluster).to_dict()]
         # Per-role IP SANs, mirroring reconcileHeadCertificate /
         # reconcileWorkerCertificate: worker cert tracks worker pod IPs
@@ -258,15 +257,16 @@ class MTLSReconciler(Reconciler):
         ]
         for obj in objs:
             try:
-                server.create(obj)
+                raw.create(obj)
             except AlreadyExistsError:
                 # Certificates: keep ipAddresses tracking the live pod IPs
                 # (the reference updates the Certificate spec in place).
                 if obj["kind"] == "Certificate" and "ipAddresses" in obj["spec"]:
                     md = obj["metadata"]
-                    cur = server.try_get("Certificate", md["namespace"],
-                                         md["name"])
+                    cur = raw.try_get("Certificate", md["namespace"],
+                                      md["name"],
+                                      api_version="cert-manager.io/v1")
                     if cur is not None and (cur.get("spec", {}).get("ipAddresses")
                                             != obj["spec"]["ipAddresses"]):
                         cur["spec"]["ipAddresses"] = obj["spec"]["ipAddresses"]
-                        server.update(cur)
+                        raw.update(cur)

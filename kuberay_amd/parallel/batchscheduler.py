@@ -18,7 +18,7 @@ from __future__ import annotations
 from typing import Dict, Optional
 
 from ..kube import objects as k8s
-from ..kube.client import KubeClient
+from ..kube.client import KubeClient, RawObjectClient
 from ..kube.store import AlreadyExistsError, ApiError, NotFoundError
 from ..models.raycluster import RayCluster
 from ..utils import constants as C
@@ -103,12 +103,10 @@ class VolcanoBatchScheduler(BatchScheduler):
                    else {}),
             },
         }
-        server = getattr(client, "server", None)
-        if server is not None:
-            try:
-                server.create(pg)
-            except AlreadyExistsError:
-                pass
+        try:
+            RawObjectClient(client).create(pg)
+        except AlreadyExistsError:
+            pass
 
     def add_metadata_to_pod(self, client, cluster, group_name, pod) -> None:
         ann = pod.metadata.ensure_annotations()
@@ -154,12 +152,10 @@ class SchedulerPluginsBatchScheduler(BatchScheduler):
             "spec": {"minMember": _min_member(cluster),
                      "minResources": _min_resources(cluster)},
         }
-        server = getattr(client, "server", None)
-        if server is not None:
-            try:
-                server.create(pg)
-            except AlreadyExistsError:
-                pass
+        try:
+            RawObjectClient(client).create(pg)
+        except AlreadyExistsError:
+            pass
 
     def add_metadata_to_pod(self, client, cluster, group_name, pod) -> None:
         pod.metadata.ensure_labels()[POD_GROUP_ANNOTATION] = pod_group_name(cluster)
@@ -167,12 +163,11 @@ class SchedulerPluginsBatchScheduler(BatchScheduler):
     def cleanup_on_completion(self, client, cluster) -> None:
         """Delete the PodGroup when the owning workload completes
         (volcano_scheduler.go CleanupOnCompletion behavior)."""
-        server = getattr(client, "server", None)
-        if server is None:
-            return
         try:
-            server.delete("PodGroup", cluster.metadata.namespace or "default",
-                          pod_group_name(cluster))
+            RawObjectClient(client).delete(
+                "PodGroup", cluster.metadata.namespace or "default",
+                pod_group_name(cluster),
+                api_version="scheduling.x-k8s.io/v1alpha1")
         except NotFoundError:
             pass
 
@@ -234,11 +229,8 @@ class XgmiGangScheduler(BatchScheduler):
         the split would cross PCIe, exactly what this scheduler exists to
         avoid.
         """
-        server = getattr(client, "server", None)
-        if server is None:
-            return {}
         capacities: Dict[str, int] = {}
-        for node in server.list("Node"):
+        for node in RawObjectClient(client).list("Node"):
             labels = (node.get("metadata") or {}).get("labels") or {}
             island = labels.get(C.XGMI_ISLAND_NODE_LABEL)
             if not island:
@@ -438,6 +430,9 @@ class KubernetesWASBatchScheduler(BatchScheduler):
             pass
 
     def do_batch_scheduling_on_submission(self, client, cluster) -> None:
+        # the Workload API (KubernetesWAS) is alpha and exercised on the
+        # in-memory tier only; volcano/scheduler-plugins/xgmi-gang are the
+        # REST-capable production schedulers
         server = getattr(client, "server", None)
         if server is None:
             return

@@ -445,3 +445,60 @@ class TestDeleteRevision:
             got.append((event_type, obj["metadata"]["name"]))
             break
         assert got == [("DELETED", "doomed")]
+
+
+class TestSchedulersOverRest:
+    """Gang schedulers must work on the REST backend too (they used to
+    silently no-op without the in-memory `server` attribute)."""
+
+    def test_podgroup_created_and_cleaned_over_rest(self, facade):
+        from kuberay_amd.parallel.batchscheduler import XgmiGangScheduler
+        client = RestClient(base_url=facade.url)
+        cluster = client.create(simple_raycluster("restgang", workers=2,
+                                                  gpus_per_worker=1))
+        sched = XgmiGangScheduler()
+        sched.do_batch_scheduling_on_submission(client, cluster)
+        pgs = facade.store.list("PodGroup")
+        assert len(pgs) == 1
+        assert pgs[0]["spec"]["minMember"] == 3
+        sched.cleanup_on_completion(client, cluster)
+        assert facade.store.list("PodGroup") == []
+
+    def test_island_pinning_reads_nodes_over_rest(self, facade):
+        from kuberay_amd.common import pod as podlib
+        from kuberay_amd.parallel.batchscheduler import XgmiGangScheduler
+        client = RestClient(base_url=facade.url)
+        facade.store.create({"kind": "Node", "apiVersion": "v1",
+                             "metadata": {"name": "n1", "labels": {
+                                 "amd.com/gpu.count": "8",
+                                 "amd.com/xgmi-island": "n1-island0"}}})
+        cluster = client.create(simple_raycluster("restpin", workers=2,
+                                                  gpus_per_worker=2))
+        group = cluster.spec.worker_group_specs[0]
+        fqdn = "x.default.svc.cluster.local"
+        t = podlib.default_worker_pod_template(cluster, group, "p-", fqdn,
+                                               "6379")
+        pod = podlib.build_pod(t, "worker", group.ray_start_params, "6379",
+                               False, None, fqdn)
+        XgmiGangScheduler().add_metadata_to_pod(client, cluster,
+                                                "default-group", pod)
+        terms = pod.spec.affinity["nodeAffinity"][
+            "requiredDuringSchedulingIgnoredDuringExecution"][
+            "nodeSelectorTerms"]
+        assert terms[0]["matchExpressions"][0]["values"] == ["n1-island0"]
+
+    def test_cert_manager_issuance_over_rest(self, facade):
+        from kuberay_amd.ops.mtls import MTLSReconciler
+        client = RestClient(base_url=facade.url)
+        client.create(simple_raycluster("restmtls",
+                                        tlsOptions={"enabled": True}))
+        r = MTLSReconciler(client, mode="cert-manager")
+        r.reconcile(("default", "restmtls"))
+        issuers = facade.store.list("Issuer")
+        certs = facade.store.list("Certificate")
+        assert {i["metadata"]["name"] for i in issuers} == {
+            "ray-selfsigned-issuer-restmtls", "ray-ca-issuer-restmtls"}
+        assert len(certs) == 3
+        # second reconcile is idempotent
+        r.reconcile(("default", "restmtls"))
+        assert len(facade.store.list("Certificate")) == 3
