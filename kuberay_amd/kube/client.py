@@ -159,11 +159,12 @@ class RawObjectClient:
         if fn is not None:
             fn(kind, namespace, name, api_version=api_version)
 
-    def list(self, kind: str, namespace: Optional[str] = None):
+    def list(self, kind: str, namespace: Optional[str] = None,
+             api_version: Optional[str] = None):
         if self.server is not None:
             return self.server.list(kind, namespace)
         fn = getattr(self.client, "raw_list", None)
-        return fn(kind, namespace) if fn else []
+        return fn(kind, namespace, api_version=api_version) if fn else []
 
 
 def _kind_of(model_or_obj) -> str:

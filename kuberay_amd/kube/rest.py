@@ -294,8 +294,10 @@ class RestClient(KubeClient):
         except NotFoundError:
             return None
 
-    def raw_list(self, kind: str, namespace: Optional[str] = None) -> List[Dict[str, Any]]:
-        resp = self._check(self._http.get(self._path(kind, namespace)))
+    def raw_list(self, kind: str, namespace: Optional[str] = None,
+                 api_version: Optional[str] = None) -> List[Dict[str, Any]]:
+        resp = self._check(self._http.get(
+            self._path(kind, namespace, api_version=api_version)))
         items = resp.json().get("items", [])
         for o in items:
             o.setdefault("kind", kind)
