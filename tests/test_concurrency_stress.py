@@ -257,3 +257,32 @@ class TestPatchFuzz:
                         assert k not in out
             else:
                 assert out == patch or isinstance(out, list)
+
+
+def test_memory_benchmark_script_smoke():
+    """The memory-benchmark harness (reference memory_benchmark analog)
+    runs end to end at reduced scale and emits a well-formed result."""
+    import json
+    import os
+    import subprocess
+    import sys
+    import tempfile
+    here = os.path.dirname(os.path.abspath(__file__))
+    script = os.path.join(here, "..", "benchmark", "memory_benchmark",
+                          "run.py")
+    out = tempfile.mktemp(suffix=".json")
+    src = open(script).read().replace("crs=150", "crs=20") \
+                             .replace("target_pods=150", "target_pods=20") \
+                             .replace("crs=30", "crs=5") \
+                             .replace("PORT = 18443", "PORT = 18447")
+    small = tempfile.mktemp(suffix=".py")
+    open(small, "w").write(src)
+    env = dict(os.environ)
+    env["PYTHONPATH"] = os.path.join(here, "..") + os.pathsep + \
+        env.get("PYTHONPATH", "")
+    r = subprocess.run([sys.executable, small, "--out", out], timeout=180,
+                       capture_output=True, text=True, env=env)
+    assert r.returncode == 0, r.stdout + r.stderr
+    data = json.load(open(out))
+    assert data["summary"]["exp2_peak_rss_mb"] > 0
+    assert any(s["experiment"] == 3 for s in data["samples"])
