@@ -327,7 +327,8 @@ class RestClient(KubeClient):
         return items, (body.get("metadata") or {}).get("resourceVersion")
 
     def raw_watch_stream(self, kind: str, resource_version: Optional[str] = None,
-                         allow_bookmarks: bool = False):
+                         allow_bookmarks: bool = False,
+                         api_version: Optional[str] = None):
         """Generator of (event_type, obj) from a K8s watch request.
 
         Raises :class:`GoneError` when the apiserver answers 410 (the
@@ -336,13 +337,13 @@ class RestClient(KubeClient):
         yielded too (type ``"BOOKMARK"``); callers use them to advance
         their rv without real traffic.
         """
-        prefix, plural = RESOURCES[kind]
+        path = self._path(kind, None, api_version=api_version)
         params = {"watch": "true"}
         if resource_version:
             params["resourceVersion"] = resource_version
         if allow_bookmarks:
             params["allowWatchBookmarks"] = "true"
-        with self._http.stream("GET", f"{prefix}/{plural}", params=params,
+        with self._http.stream("GET", path, params=params,
                                timeout=None) as resp:
             if resp.status_code == 410:
                 raise GoneError(f"watch {kind} from rv {resource_version}: "

@@ -502,3 +502,31 @@ class TestSchedulersOverRest:
         # second reconcile is idempotent
         r.reconcile(("default", "restmtls"))
         assert len(facade.store.list("Certificate")) == 3
+
+    def test_dynamic_kind_watch_over_rest(self, facade):
+        """PodGroup (a CRD kind outside the static resource map) is
+        watchable over REST once the facade has learned it."""
+        import threading
+        client = RestClient(base_url=facade.url)
+        client.raw_create({"apiVersion": "scheduling.x-k8s.io/v1alpha1",
+                           "kind": "PodGroup",
+                           "metadata": {"name": "w0", "namespace": "default"},
+                           "spec": {"minMember": 2}})
+        seen = []
+        def consume():
+            for etype, obj in client.raw_watch_stream(
+                    "PodGroup", "0",
+                    api_version="scheduling.x-k8s.io/v1alpha1"):
+                seen.append((etype, obj["metadata"]["name"]))
+                if len(seen) >= 2:
+                    break
+        t = threading.Thread(target=consume, daemon=True)
+        t.start()
+        import time as _t
+        _t.sleep(0.3)
+        client.raw_create({"apiVersion": "scheduling.x-k8s.io/v1alpha1",
+                           "kind": "PodGroup",
+                           "metadata": {"name": "w1", "namespace": "default"},
+                           "spec": {"minMember": 3}})
+        t.join(timeout=10)
+        assert ("ADDED", "w0") in seen and ("ADDED", "w1") in seen
