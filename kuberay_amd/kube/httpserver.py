@@ -163,6 +163,10 @@ def strategic_merge(current: Any, patch: Any, field: str = "") -> Any:
                 out.pop(key, None)
             elif key in out:
                 out[key] = strategic_merge(out[key], val, key)
+            elif isinstance(val, dict):
+                # RFC 7386 MergePatch(target, patch) with target absent:
+                # recurse into {} so nested nulls delete (never materialize)
+                out[key] = strategic_merge({}, val, key)
             else:
                 out[key] = val
         return out
@@ -188,9 +192,12 @@ def strategic_merge(current: Any, patch: Any, field: str = "") -> Any:
                 out[index[key_val]] = strategic_merge(out[index[key_val]],
                                                       elem)
             else:
-                out.append(elem)
+                out.append(strategic_merge({}, elem))
                 index[key_val] = len(out) - 1
         return out
+    if isinstance(patch, dict):
+        # replacing a non-dict with a dict: same null-stripping recursion
+        return strategic_merge({}, patch, field)
     return patch
 
 

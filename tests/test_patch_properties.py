@@ -1,0 +1,101 @@
+"""Property tests for the apiserver patch machinery (RFC 6902 json-patch,
+strategic-merge-patch, RFC 7386 merge-patch) — hypothesis-driven analogs
+of apimachinery's patch fuzzers."""
+import copy
+
+import pytest
+
+pytest.importorskip("hypothesis")
+
+from hypothesis import given, settings, strategies as st
+
+from kuberay_amd.kube.httpserver import apply_json_patch, strategic_merge
+
+KEYS = st.sampled_from(["a", "b", "name", "spec", "items", "replicas"])
+SCALARS = st.one_of(st.integers(-5, 5), st.text(max_size=4), st.booleans(),
+                    st.none())
+
+
+def docs(depth=3):
+    if depth == 0:
+        return SCALARS
+    return st.one_of(
+        SCALARS,
+        st.lists(docs(depth - 1), max_size=3),
+        st.dictionaries(KEYS, docs(depth - 1), max_size=3))
+
+
+DICT_DOCS = st.dictionaries(KEYS, docs(), min_size=0, max_size=4)
+
+
+class TestJsonPatch:
+    @settings(max_examples=80, deadline=None)
+    @given(doc=DICT_DOCS, key=KEYS, value=docs())
+    def test_add_then_remove_roundtrips(self, doc, key, value):
+        before = copy.deepcopy(doc)
+        added = apply_json_patch(doc, [{"op": "add", "path": f"/{key}",
+                                        "value": value}])
+        assert added[key] == value
+        if key in before:
+            # removed slot previously occupied: add acts as replace, remove
+            # deletes — original value is gone (spec semantics), but the
+            # patch input must not be mutated
+            assert doc == before
+        else:
+            removed = apply_json_patch(added, [{"op": "remove",
+                                                "path": f"/{key}"}])
+            assert removed == before
+
+    @settings(max_examples=80, deadline=None)
+    @given(doc=DICT_DOCS)
+    def test_copy_then_test_passes(self, doc):
+        if not doc:
+            return
+        key = sorted(doc)[0]
+        out = apply_json_patch(doc, [
+            {"op": "copy", "from": f"/{key}", "path": "/copied"},
+            {"op": "test", "path": "/copied", "value": doc[key]}])
+        assert out["copied"] == doc[key]
+
+    @settings(max_examples=80, deadline=None)
+    @given(doc=DICT_DOCS)
+    def test_input_never_mutated(self, doc):
+        snapshot = copy.deepcopy(doc)
+        try:
+            apply_json_patch(doc, [{"op": "add", "path": "/x", "value": 1},
+                                   {"op": "remove", "path": "/nope"}])
+        except Exception:
+            pass
+        assert doc == snapshot
+
+
+class TestStrategicMerge:
+    @settings(max_examples=80, deadline=None)
+    @given(doc=DICT_DOCS, patch=DICT_DOCS)
+    def test_idempotent(self, doc, patch):
+        once = strategic_merge(copy.deepcopy(doc), patch)
+        twice = strategic_merge(copy.deepcopy(once), patch)
+        assert once == twice
+
+    @settings(max_examples=80, deadline=None)
+    @given(doc=DICT_DOCS, patch=st.dictionaries(KEYS, SCALARS, max_size=3))
+    def test_scalar_leaves_win(self, doc, patch):
+        out = strategic_merge(copy.deepcopy(doc), patch)
+        for k, v in patch.items():
+            if v is None:
+                assert k not in out  # null deletes, RFC 7386 style
+            else:
+                assert out[k] == v
+
+    def test_merge_key_containers(self):
+        doc = {"containers": [{"name": "a", "image": "x"},
+                              {"name": "b", "image": "y"}]}
+        out = strategic_merge(doc, {"containers": [{"name": "b",
+                                                    "image": "z"}]})
+        assert [c["image"] for c in out["containers"]] == ["x", "z"]
+
+    def test_patch_delete_directive(self):
+        doc = {"containers": [{"name": "a"}, {"name": "b"}]}
+        out = strategic_merge(doc, {"containers": [
+            {"name": "a", "$patch": "delete"}]})
+        assert [c["name"] for c in out["containers"]] == ["b"]
