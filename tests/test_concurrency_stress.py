@@ -255,6 +255,15 @@ class TestPatchFuzz:
                 for k, v in patch.items():
                     if v is None:
                         assert k not in out
+            elif isinstance(patch, dict):
+                # replacement dicts get the RFC 7386 null-stripping
+                # recursion: no null may materialize anywhere in them
+                def no_nulls(node):
+                    if isinstance(node, dict):
+                        return all(v is not None and no_nulls(v)
+                                   for v in node.values())
+                    return True
+                assert no_nulls(out)
             else:
                 assert out == patch or isinstance(out, list)
 
