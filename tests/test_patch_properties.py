@@ -99,3 +99,47 @@ class TestStrategicMerge:
         out = strategic_merge(doc, {"containers": [
             {"name": "a", "$patch": "delete"}]})
         assert [c["name"] for c in out["containers"]] == ["b"]
+
+
+class TestWorkqueueCoalescing:
+    """Property tests for the delayed-entry coalescing fix: no lost keys,
+    and memory bounded by DISTINCT keys regardless of requeue volume."""
+
+    @settings(max_examples=25, deadline=None)
+    @given(ops=st.lists(
+        st.tuples(st.sampled_from(["k1", "k2", "k3"]),
+                  st.floats(min_value=0.0005, max_value=0.004)),
+        min_size=1, max_size=40))
+    def test_every_key_delivered_and_heap_bounded(self, ops):
+        from kuberay_amd.kube.workqueue import RateLimitingQueue
+        q = RateLimitingQueue()
+        keys = {k for k, _ in ops}
+        for k, d in ops:
+            q.add_after(k, d)
+            # the RSS invariant: one live deadline per key, ever
+            assert len(q._delayed_next) <= len(keys)
+        delivered = set()
+        import time as _t
+        deadline = _t.monotonic() + 5
+        while delivered != keys and _t.monotonic() < deadline:
+            item = q.get(timeout=0.5)
+            if item is None:
+                continue
+            delivered.add(item)
+            q.done(item)
+        assert delivered == keys
+        # after full drain no delayed state remains
+        assert len(q._delayed_next) == 0
+
+    def test_requeue_while_processing_redelivers(self):
+        from kuberay_amd.kube.workqueue import RateLimitingQueue
+        q = RateLimitingQueue()
+        q.add("x")
+        item = q.get(timeout=1)
+        assert item == "x"
+        q.add_after("x", 0.001)   # fires while x is processing → dirty
+        import time as _t
+        _t.sleep(0.05)
+        q.get(timeout=0.05)       # drains delayed heap into dirty set
+        q.done("x")
+        assert q.get(timeout=1) == "x"  # dirty redelivery
