@@ -143,3 +143,55 @@ class TestWorkqueueCoalescing:
         q.get(timeout=0.05)       # drains delayed heap into dirty set
         q.done("x")
         assert q.get(timeout=1) == "x"  # dirty redelivery
+
+
+class TestWatchResumeExactness:
+    """Resuming from ANY rv inside the retained window yields exactly the
+    suffix of events after that rv, in order (the two round-2 watch bugs —
+    replay-before-register and delete-rv-reuse — were both violations of
+    this property)."""
+
+    @settings(max_examples=30, deadline=None)
+    @given(ops=st.lists(
+        st.tuples(st.sampled_from(["create", "update", "delete"]),
+                  st.sampled_from(["o1", "o2", "o3", "o4"])),
+        min_size=1, max_size=60))
+    def test_suffix_exactness(self, ops):
+        from kuberay_amd.kube.store import InMemoryApiServer
+        server = InMemoryApiServer()
+        log = []  # (rv_after_op, etype, name)
+        live = set()
+        for op, name in ops:
+            if op == "create" and name not in live:
+                out = server.create({"kind": "ConfigMap", "apiVersion": "v1",
+                                     "metadata": {"name": name}, "data": {}})
+                live.add(name)
+                log.append((int(out["metadata"]["resourceVersion"]),
+                            "ADDED", name))
+            elif op == "update" and name in live:
+                cur = server.get("ConfigMap", "default", name)
+                cur["data"] = {"n": str(len(log))}
+                out = server.update(cur)
+                log.append((int(out["metadata"]["resourceVersion"]),
+                            "MODIFIED", name))
+            elif op == "delete" and name in live:
+                server.delete("ConfigMap", "default", name)
+                live.discard(name)
+                log.append((server.current_rv, "DELETED", name))
+        # resume from every recorded rv plus 0: suffix must match the log
+        checkpoints = [0] + [rv for rv, _, _ in log]
+        for rv in checkpoints:
+            got = server.events_since(rv, kinds={"ConfigMap"})
+            assert got is not None
+            expect = [(etype, name) for (erv, etype, name) in log
+                      if erv > rv]
+            assert [(e, o["metadata"]["name"]) for e, o in got] == expect, rv
+
+    def test_rv_strictly_increases_per_event(self):
+        from kuberay_amd.kube.store import InMemoryApiServer
+        server = InMemoryApiServer()
+        server.create({"kind": "ConfigMap", "apiVersion": "v1",
+                       "metadata": {"name": "a"}})
+        rv1 = server.current_rv
+        server.delete("ConfigMap", "default", "a")
+        assert server.current_rv > rv1  # DELETED must own a fresh rv
