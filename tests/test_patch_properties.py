@@ -195,3 +195,39 @@ class TestWatchResumeExactness:
         rv1 = server.current_rv
         server.delete("ConfigMap", "default", "a")
         assert server.current_rv > rv1  # DELETED must own a fresh rv
+
+
+class TestNamingAndHashProperties:
+    @settings(max_examples=60, deadline=None)
+    @given(name=st.text(alphabet=st.characters(
+        whitelist_categories=("Ll", "Nd"), whitelist_characters="-."),
+        min_size=1, max_size=120))
+    def test_check_name_bounds_and_determinism(self, name):
+        from kuberay_amd.utils.names import check_name, check_name_63
+        out = check_name(name)
+        assert len(out) <= 63 and check_name(name) == out
+        assert len(check_name_63(name)) <= 63
+
+    @settings(max_examples=40, deadline=None)
+    @given(replicas=st.integers(0, 50), to_delete=st.lists(
+        st.sampled_from(["p1", "p2", "p3"]), max_size=3))
+    def test_spec_hash_ignores_scale_churn(self, replicas, to_delete):
+        """Upgrade detection must not fire on autoscaler activity:
+        replicas and workersToDelete never affect the spec hash
+        (reference GenerateHashWithoutReplicasAndWorkersToDelete)."""
+        from kuberay_amd.testing import simple_raycluster
+        from kuberay_amd.utils.hashing import (
+            hash_without_replicas_and_workers_to_delete)
+        base = simple_raycluster("h", workers=1)
+        h0 = hash_without_replicas_and_workers_to_delete(base.spec)
+        c = simple_raycluster("h", workers=1)
+        g = c.spec.worker_group_specs[0]
+        g.replicas = replicas
+        g.max_replicas = max(replicas, g.max_replicas or 0)
+        if to_delete:
+            g.scale_strategy.workers_to_delete = to_delete
+        assert hash_without_replicas_and_workers_to_delete(c.spec) == h0
+        # but a real spec change (image) must change it
+        c.spec.worker_group_specs[0].template.spec.containers[0].image = \
+            "other:tag"
+        assert hash_without_replicas_and_workers_to_delete(c.spec) != h0
