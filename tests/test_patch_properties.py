@@ -231,3 +231,54 @@ class TestNamingAndHashProperties:
         c.spec.worker_group_specs[0].template.spec.containers[0].image = \
             "other:tag"
         assert hash_without_replicas_and_workers_to_delete(c.spec) != h0
+
+
+class TestClientDifferential:
+    """The same verb sequence through InMemoryClient and through
+    RestClient-over-the-facade must converge to the same object tree
+    (modulo resourceVersions, which depend on server-internal counters)."""
+
+    @settings(max_examples=15, deadline=None)
+    @given(ops=st.lists(st.tuples(
+        st.sampled_from(["create", "update", "patch", "delete"]),
+        st.sampled_from(["c1", "c2", "c3"]),
+        st.dictionaries(st.sampled_from(["x", "y", "z"]),
+                        st.text(max_size=3), max_size=2)),
+        min_size=1, max_size=25))
+    def test_configmap_sequences_converge(self, ops):
+        from kuberay_amd.kube.client import InMemoryClient
+        from kuberay_amd.kube.httpserver import KubeApiFacade
+        from kuberay_amd.kube.objects import ConfigMap
+        from kuberay_amd.kube.rest import RestClient
+        from kuberay_amd.kube.store import (AlreadyExistsError, ApiError,
+                                            NotFoundError)
+        mem = InMemoryClient()
+        facade = KubeApiFacade().start()
+        try:
+            rest = RestClient(base_url=facade.url)
+            for client in (mem, rest):
+                for op, name, data in ops:
+                    try:
+                        if op == "create":
+                            client.create(ConfigMap.from_dict(
+                                {"kind": "ConfigMap", "apiVersion": "v1",
+                                 "metadata": {"name": name}, "data": data}))
+                        elif op == "update":
+                            cur = client.get(ConfigMap, "default", name)
+                            cur.data = data
+                            client.update(cur)
+                        elif op == "patch":
+                            client.patch(ConfigMap, "default", name,
+                                         {"data": data})
+                        elif op == "delete":
+                            client.delete(ConfigMap, "default", name)
+                    except (NotFoundError, AlreadyExistsError, ApiError):
+                        pass
+            def snapshot(client):
+                out = {}
+                for cm in client.list(ConfigMap, "default"):
+                    out[cm.metadata.name] = cm.data
+                return out
+            assert snapshot(mem) == snapshot(rest)
+        finally:
+            facade.stop()
