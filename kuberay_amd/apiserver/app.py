@@ -67,6 +67,11 @@ def create_app(client: Optional[KubeClient] = None,
     def list_compute_templates(ns: str):
         return {"computeTemplates": list(_templates(ns).values())}
 
+    @app.get("/apis/v1/compute_templates")
+    def list_all_compute_templates():
+        """grpc-gateway ListAllComputeTemplates route (config.proto:50)."""
+        return {"computeTemplates": list(_templates(None).values())}
+
     @app.get("/apis/v1/namespaces/{ns}/compute_templates/{name}")
     def get_compute_template(ns: str, name: str):
         t = _templates(ns).get(name)
@@ -149,6 +154,12 @@ def create_app(client: Optional[KubeClient] = None,
         return {"clusters": [conv.raycluster_to_api_cluster(rc)
                              for rc in client.list(RayCluster, ns)]}
 
+    @app.get("/apis/v1/clusters")
+    def list_all_clusters():
+        """grpc-gateway ListAllClusters route (cluster.proto:50)."""
+        return {"clusters": [conv.raycluster_to_api_cluster(rc)
+                             for rc in client.list(RayCluster, None)]}
+
     @app.get("/apis/v1/namespaces/{ns}/clusters/{name}")
     def get_cluster(ns: str, name: str):
         rc = client.try_get(RayCluster, ns, name)
@@ -190,6 +201,12 @@ def create_app(client: Optional[KubeClient] = None,
         return {"jobs": [conv.rayjob_to_api_job(j)
                          for j in client.list(RayJob, ns)]}
 
+    @app.get("/apis/v1/jobs")
+    def list_all_jobs():
+        """grpc-gateway ListAllRayJobs route (job.proto:52)."""
+        return {"jobs": [conv.rayjob_to_api_job(j)
+                         for j in client.list(RayJob, None)]}
+
     @app.get("/apis/v1/namespaces/{ns}/jobs/{name}")
     def get_job(ns: str, name: str):
         job = client.try_get(RayJob, ns, name)
@@ -220,6 +237,12 @@ def create_app(client: Optional[KubeClient] = None,
     def list_services(ns: str):
         return {"services": [conv.rayservice_to_api_service(s)
                              for s in client.list(RayService, ns)]}
+
+    @app.get("/apis/v1/services")
+    def list_all_services():
+        """grpc-gateway ListAllRayServices route (service.proto:49)."""
+        return {"services": [conv.rayservice_to_api_service(x)
+                             for x in client.list(RayService, None)]}
 
     @app.get("/apis/v1/namespaces/{ns}/services/{name}")
     def get_service(ns: str, name: str):

@@ -1105,3 +1105,22 @@ class TestImageTemplatesHttp:
             assert got.base_image == "img"
         finally:
             server.stop(0)
+
+
+def test_list_all_routes_cross_namespace(api):
+    """grpc-gateway ListAll* HTTP routes return objects from every
+    namespace (cluster.proto:50 / job.proto:52 / service.proto:49 /
+    config.proto:50)."""
+    t, _, _ = api
+    for ns in ("nsa", "nsb"):
+        r = t.post(f"/apis/v1/namespaces/{ns}/clusters", json={
+            "name": f"c-{ns}", "clusterSpec": {
+                "headGroupSpec": {"computeTemplate": "", "image": "i",
+                                  "rayStartParams": {}},
+                "workerGroupSpec": []}})
+        assert r.status_code == 200, r.text
+    names = {c["name"] for c in t.get("/apis/v1/clusters").json()["clusters"]}
+    assert {"c-nsa", "c-nsb"} <= names
+    assert t.get("/apis/v1/jobs").json()["jobs"] == []
+    assert t.get("/apis/v1/services").json()["services"] == []
+    assert "computeTemplates" in t.get("/apis/v1/compute_templates").json()
