@@ -91,6 +91,10 @@ class MI355XAutoscaler:
             ann = cluster.metadata.annotations or {}
             if ann.get(AMD_AUTOSCALER_ANNOTATION, "").lower() != "true":
                 continue
+            if cluster.spec.suspend:
+                # a suspended cluster has no pods; node telemetry cannot
+                # be attributed to it and scaling it would fight resume
+                continue
             try:
                 decisions.extend(self._evaluate_cluster(cluster, summary))
             except ApiError:

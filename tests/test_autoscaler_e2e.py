@@ -153,3 +153,85 @@ class TestZeroToEight:
             time.sleep(0.12)
         assert cluster_of(cp).status.available_worker_replicas == 0
         assert cluster_of(cp).status.desired_worker_replicas == 0
+
+
+class TestPolicyMechanics:
+    """Pure policy-level tests with a fake clock and telemetry: the
+    stability-window / cooldown / clamp rules that keep rocm-smi-driven
+    scaling from flapping."""
+
+    def _mk(self, util=90.0, hbm=0.2):
+        from kuberay_amd.gpu.autoscaler import (AMD_AUTOSCALER_ANNOTATION,
+                                                AutoscalerPolicy,
+                                                MI355XAutoscaler)
+        from kuberay_amd.kube.client import InMemoryClient
+        from kuberay_amd.testing import simple_raycluster
+        client = InMemoryClient()
+        c = simple_raycluster("as", workers=2, gpus_per_worker=1)
+        c.metadata.annotations = {AMD_AUTOSCALER_ANNOTATION: "true"}
+        c.spec.worker_group_specs[0].min_replicas = 1
+        c.spec.worker_group_specs[0].max_replicas = 3
+        client.create(c)
+        self.now = 1000.0
+        self.summary = {"avg_utilization_pct": util,
+                        "max_vram_used_fraction": hbm}
+        policy = AutoscalerPolicy(up_stable_s=10, idle_timeout_s=60,
+                                  cooldown_s=30)
+        a = MI355XAutoscaler(client, telemetry=lambda: dict(self.summary),
+                             policy=policy, clock=lambda: self.now)
+        return a, client
+
+    def _replicas(self, client):
+        from kuberay_amd.models import RayCluster
+        c = client.get(RayCluster, "default", "as")
+        return c.spec.worker_group_specs[0].replicas
+
+    def test_stability_window_gates_scale_up(self):
+        a, client = self._mk(util=95)
+        assert a.step() == []            # first sighting starts the window
+        self.now += 5
+        assert a.step() == []            # still inside up_stable_s
+        self.now += 6
+        assert any("scale-up" in d for d in a.step())
+        assert self._replicas(client) == 3
+
+    def test_cooldown_limits_action_rate(self):
+        a, client = self._mk(util=95)
+        a.step(); self.now += 11; a.step()
+        assert self._replicas(client) == 3
+        # high again immediately: cooldown (30 s) must hold even after
+        # another full stability window
+        self.now += 12
+        assert a.step() == []
+        assert self._replicas(client) == 3
+
+    def test_max_replicas_clamps(self):
+        a, client = self._mk(util=95)
+        for _ in range(6):
+            a.step(); self.now += 50
+        assert self._replicas(client) == 3  # maxReplicas
+
+    def test_min_replicas_clamps_scale_down(self):
+        a, client = self._mk(util=1, hbm=0.05)
+        for _ in range(8):
+            a.step(); self.now += 100
+        assert self._replicas(client) == 1  # minReplicas floor
+
+    def test_flapping_utilization_never_acts(self):
+        a, client = self._mk()
+        for i in range(20):
+            self.summary["avg_utilization_pct"] = 95 if i % 2 == 0 else 40
+            assert a.step() == []
+            self.now += 6   # window resets before up_stable_s elapses
+        assert self._replicas(client) == 2
+
+    def test_suspended_cluster_skipped(self):
+        from kuberay_amd.models import RayCluster
+        a, client = self._mk(util=95)
+        def suspend(c):
+            c.spec.suspend = True
+        client.update_with_retry(RayCluster, "default", "as", suspend)
+        for _ in range(4):
+            assert a.step() == []
+            self.now += 50
+        assert self._replicas(client) == 2
