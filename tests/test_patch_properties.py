@@ -282,3 +282,42 @@ class TestClientDifferential:
             assert snapshot(mem) == snapshot(rest)
         finally:
             facade.stop()
+
+
+class TestOwnerGCCascade:
+    """Random ownership forests: deleting a root garbage-collects exactly
+    its transitive dependents — nothing orphaned, nothing over-deleted
+    (kube garbage-collector background-cascade semantics)."""
+
+    @settings(max_examples=25, deadline=None)
+    @given(parents=st.lists(st.integers(0, 9), min_size=10, max_size=10),
+           root=st.integers(0, 9))
+    def test_cascade_exactness(self, parents, root):
+        from kuberay_amd.kube.store import InMemoryApiServer
+        server = InMemoryApiServer()
+        uids = {}
+        # node i is owned by parents[i] when parents[i] < i (forest, no
+        # cycles); otherwise it is a root
+        for i in range(10):
+            refs = []
+            p = parents[i]
+            if p < i:
+                refs = [{"apiVersion": "v1", "kind": "ConfigMap",
+                         "name": f"n{p}", "uid": uids[p]}]
+            out = server.create({
+                "kind": "ConfigMap", "apiVersion": "v1",
+                "metadata": {"name": f"n{i}",
+                             "ownerReferences": refs or None}})
+            uids[i] = out["metadata"]["uid"]
+        reach = {root}
+        changed = True
+        while changed:
+            changed = False
+            for i in range(10):
+                if parents[i] < i and parents[i] in reach and i not in reach:
+                    reach.add(i)
+                    changed = True
+        server.delete("ConfigMap", "default", f"n{root}")
+        left = {o["metadata"]["name"]
+                for o in server.list("ConfigMap", "default")}
+        assert left == {f"n{i}" for i in range(10) if i not in reach}
