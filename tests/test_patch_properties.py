@@ -321,3 +321,13 @@ class TestOwnerGCCascade:
         left = {o["metadata"]["name"]
                 for o in server.list("ConfigMap", "default")}
         assert left == {f"n{i}" for i in range(10) if i not in reach}
+
+
+class TestQuantityRoundTrip:
+    @settings(max_examples=80, deadline=None)
+    @given(n=st.integers(0, 10**12),
+           suffix=st.sampled_from(["", "m", "k", "M", "G", "Ki", "Mi", "Gi"]))
+    def test_format_parse_roundtrip(self, n, suffix):
+        from kuberay_amd.utils.quantity import format_quantity, parse_quantity
+        v = parse_quantity(f"{n}{suffix}")
+        assert parse_quantity(format_quantity(v)) == v
