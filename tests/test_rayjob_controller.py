@@ -439,3 +439,41 @@ class TestCronSuspendAndCatchup:
         now["t"] += dt.timedelta(minutes=2)
         rec.reconcile(("default", "cj"))
         assert client.server.count("RayJob") == 1
+
+
+class TestSuspendResumeStorm:
+    def test_random_suspend_toggles_always_recover(self, control_plane):
+        """Adversarial lifecycle fuzz: random suspend/resume flips while
+        the job progresses must always converge — job lands in a valid
+        state, no reconciler exception, no leaked cluster after the final
+        suspend."""
+        import random
+        rng = random.Random(7)
+        control_plane.dashboard.get_job_info_mock = lambda job_id: {
+            "submission_id": job_id, "status": "RUNNING"}
+        control_plane.client.create(make_rayjob(
+            shutdownAfterJobFinishes=True))
+        assert wait_deployment_status(control_plane, "job1", "Running")
+        from kuberay_amd.models import RayJob
+        for _ in range(6):
+            want = rng.random() < 0.5
+            control_plane.client.update_with_retry(
+                RayJob, "default", "job1",
+                lambda j, w=want: setattr(j.spec, "suspend", w))
+            time.sleep(rng.uniform(0.05, 0.3))
+        # final state: suspend and verify full teardown
+        control_plane.client.update_with_retry(
+            RayJob, "default", "job1",
+            lambda j: setattr(j.spec, "suspend", True))
+        assert wait_deployment_status(control_plane, "job1", "Suspended",
+                                      timeout=30)
+        assert control_plane.wait_for(
+            lambda: control_plane.server.count("RayCluster") == 0,
+            timeout=30)
+        # and resume one last time: it must come back to life
+        control_plane.client.update_with_retry(
+            RayJob, "default", "job1",
+            lambda j: setattr(j.spec, "suspend", False))
+        assert control_plane.wait_for(
+            lambda: job_of(control_plane).status.job_deployment_status
+            in ("Initializing", "Running", "Complete"), timeout=30)
