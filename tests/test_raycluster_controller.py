@@ -530,3 +530,39 @@ class TestIngressReconcile:
         cp.client.create(simple_raycluster("noing", workers=0))
         assert cp.wait_cluster_state("default", "noing", "ready")
         assert cp.server.count("Ingress") == 0
+
+
+class TestReplicaRandomWalk:
+    def test_replica_random_walk_converges_exactly(self, control_plane):
+        """Random replica walk (0..5) with interleaved waits: after the
+        final target the cluster must converge to EXACTLY that many ready
+        workers — no stragglers from superseded targets, no over-delete."""
+        import random
+        rng = random.Random(13)
+        from kuberay_amd.models import RayCluster
+        control_plane.client.create(simple_raycluster("walk", workers=1))
+        assert control_plane.wait_cluster_state("default", "walk", "ready")
+        target = 1
+        for _ in range(8):
+            target = rng.randint(0, 5)
+            control_plane.client.update_with_retry(
+                RayCluster, "default", "walk",
+                lambda c, t=target: (
+                    setattr(c.spec.worker_group_specs[0], "replicas", t),
+                    setattr(c.spec.worker_group_specs[0], "max_replicas",
+                            max(t, c.spec.worker_group_specs[0]
+                                .max_replicas or 0))))
+            time.sleep(rng.uniform(0.02, 0.25))
+        assert control_plane.wait_for(
+            lambda: control_plane.client.get(
+                RayCluster, "default", "walk"
+            ).status.ready_worker_replicas == target, timeout=40), target
+        # pod-level exactness, not just status
+        def worker_pods():
+            return [v for v in control_plane.client.list_pod_views("default")
+                    if v.labels.get("ray.io/node-type") == "worker"
+                    and v.labels.get("ray.io/cluster") == "walk"
+                    and not v.deletion_timestamp]
+        assert control_plane.wait_for(
+            lambda: len(worker_pods()) == target, timeout=30), \
+            (target, len(worker_pods()))
