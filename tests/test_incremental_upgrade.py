@@ -212,3 +212,58 @@ class TestGatewayReadinessGate:
             subresource="status")
         assert up.step_traffic(svc, active, pending) is False  # 50%
         assert svc.status.pending_service_status.traffic_routed_percent == 50
+
+
+class TestSteppingInvariants:
+    def _ready_gateway(self, client, svc):
+        ns = "default"
+        client.server.patch_merge(
+            "Gateway", ns, f"{svc.metadata.name}-gateway",
+            {"status": {"conditions": [
+                {"type": "Accepted", "status": "True"},
+                {"type": "Programmed", "status": "True"}]}},
+            subresource="status")
+        client.server.patch_merge(
+            "HTTPRoute", ns, f"{svc.metadata.name}-route",
+            {"status": {"parents": [{
+                "parentRef": {"name": f"{svc.metadata.name}-gateway"},
+                "conditions": [
+                    {"type": "Accepted", "status": "True"},
+                    {"type": "ResolvedRefs", "status": "True"}]}]}},
+            subresource="status")
+
+    @pytest.mark.parametrize("step", [1, 7, 25, 40, 100])
+    def test_weight_walk_invariants(self, step):
+        """For ANY stepSizePercent: weights are monotone, bounded by the
+        step, active+pending always sum to 100, capacity leads traffic,
+        promotion fires exactly at 100, and the HTTPRoute carries the
+        same split (rayservice_controller.go:1644-1843 invariants)."""
+        from kuberay_amd.kube.client import InMemoryClient
+        from kuberay_amd.kube.events import StoreRecorder
+        from kuberay_amd.ops.incremental import IncrementalUpgrader
+        client = InMemoryClient()
+        svc = client.create(make_service(interval=0, step=step))
+        active = client.create(simple_raycluster("act", workers=1))
+        pending = client.create(simple_raycluster("pen", workers=1))
+        up = IncrementalUpgrader(client, StoreRecorder(client.server))
+        up.ensure_gateway_infra(svc, active, pending)
+        self._ready_gateway(client, svc)
+        prev = 0
+        for i in range(0, 120):
+            done = up.step_traffic(svc, active, pending)
+            ps = svc.status.pending_service_status
+            As = svc.status.active_service_status
+            w = ps.traffic_routed_percent or 0
+            assert prev < w <= min(100, prev + step)
+            assert (As.traffic_routed_percent or 0) + w == 100
+            assert (ps.target_capacity or 0) >= w
+            route = client.server.get(
+                "HTTPRoute", "default", f"{svc.metadata.name}-route")
+            weights = {r["name"]: r["weight"] for r in
+                       route["spec"]["rules"][0]["backendRefs"]}
+            assert sum(weights.values()) == 100
+            assert done is (w >= 100)
+            if done:
+                break
+            prev = w
+        assert prev < 100 and w == 100
