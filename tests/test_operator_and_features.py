@@ -739,3 +739,79 @@ class TestGateDefaultParity:
     def test_defaults_match_reference(self):
         for gate, want in self.REFERENCE_DEFAULTS.items():
             assert features.enabled(gate) is want, gate
+
+
+class TestNetworkPolicyModes:
+    """networkpolicy_controller.go behavior matrix: deny modes, always-on
+    intra-cluster + DNS allowances, user rule append, per-group overrides,
+    stale policy GC on mode change."""
+
+    def _cluster(self, mode, **np_extra):
+        c = simple_raycluster("npx", workers=1)
+        c.spec.network_policy = {"mode": mode, **np_extra}
+        from kuberay_amd.models import RayCluster
+        return RayCluster.from_dict(c.to_dict())
+
+    def _reconcile(self, client, cluster):
+        from kuberay_amd.ops.networkpolicy import NetworkPolicyReconciler
+        NetworkPolicyReconciler(client).reconcile(
+            ("default", cluster.metadata.name))
+
+    def test_deny_all_ingress_only_sets_ingress_type(self):
+        from kuberay_amd.ops.networkpolicy import build_head_network_policy
+        p = build_head_network_policy(self._cluster("DenyAllIngress"))
+        assert p.spec["policyTypes"] == ["Ingress"]
+        assert "egress" not in p.spec
+        # intra-cluster traffic always allowed
+        sel = p.spec["ingress"][0]["from"][0]["podSelector"]["matchLabels"]
+        assert sel["ray.io/cluster"] == "npx"
+
+    def test_deny_all_egress_keeps_dns_open(self):
+        from kuberay_amd.ops.networkpolicy import build_head_network_policy
+        p = build_head_network_policy(self._cluster("DenyAllEgress"))
+        assert p.spec["policyTypes"] == ["Egress"]
+        ports = [pt["port"] for r in p.spec["egress"]
+                 for pt in r.get("ports", [])]
+        assert ports.count(53) == 2  # UDP + TCP DNS
+
+    def test_user_rules_appended_after_builtins(self):
+        from kuberay_amd.ops.networkpolicy import build_head_network_policy
+        p = build_head_network_policy(self._cluster(
+            "DenyAll", head={"ingressRules": [
+                {"from": [{"ipBlock": {"cidr": "10.0.0.0/8"}}]}]}))
+        assert p.spec["ingress"][-1]["from"][0]["ipBlock"][
+            "cidr"] == "10.0.0.0/8"
+
+    def test_per_group_override_layers_on_general_policy(self):
+        from kuberay_amd.ops.networkpolicy import (
+            build_worker_network_policies)
+        ps = build_worker_network_policies(self._cluster(
+            "DenyAll", workerGroups=[{"groupName": "default-group",
+                                      "egressRules": [
+                {"to": [{"ipBlock": {"cidr": "192.168.0.0/16"}}]}]}]))
+        names = [p.metadata.name for p in ps]
+        assert names == ["npx-workers", "npx-workers-default-group"]
+        grp = ps[1].spec["podSelector"]["matchLabels"]
+        assert grp["ray.io/group"] == "default-group"
+
+    def test_stale_policy_gc_on_mode_change(self):
+        from kuberay_amd.kube import objects as k8s
+        from kuberay_amd.kube.client import InMemoryClient
+        client = InMemoryClient()
+        c = self._cluster("DenyAll", workerGroups=[
+            {"groupName": "default-group"}])
+        client.create(c)
+        self._reconcile(client, c)
+        assert len(client.list(k8s.NetworkPolicy, "default")) == 3
+        # drop the per-group override: its policy must be GC'd
+        c2 = self._cluster("DenyAll")
+        c2.metadata.resource_version = None
+        from kuberay_amd.models import RayCluster
+        client.update_with_retry(
+            RayCluster, "default", "npx",
+            lambda cur: setattr(cur.spec, "network_policy",
+                                c2.spec.network_policy))
+        self._reconcile(client, client.get(RayCluster, "default", "npx"))
+        left = {p.metadata.name
+                for p in client.list(k8s.NetworkPolicy, "default")}
+        assert left == {"npx-head", "npx-workers"}
