@@ -35,6 +35,13 @@ from kuberay_amd.testing import simple_raycluster  # noqa: E402
 PORT = 18443
 
 
+def free_port() -> int:
+    import socket
+    with socket.socket() as sk:
+        sk.bind(("127.0.0.1", 0))
+        return sk.getsockname()[1]
+
+
 def start_operator():
     env = dict(os.environ, PYTHONUNBUFFERED="1")
     proc = subprocess.Popen(
@@ -125,7 +132,11 @@ def main() -> int:
     ap = argparse.ArgumentParser()
     ap.add_argument("--out", default=os.path.join(
         os.path.dirname(__file__), "results-memory.json"))
+    ap.add_argument("--port", type=int, default=0,
+                    help="facade port (0 = pick a free one)")
     args = ap.parse_args()
+    global PORT
+    PORT = args.port or free_port()
     proc = start_operator()
     out = []
     try:

@@ -282,8 +282,7 @@ def test_memory_benchmark_script_smoke():
     out = tempfile.mktemp(suffix=".json")
     src = open(script).read().replace("crs=150", "crs=20") \
                              .replace("target_pods=150", "target_pods=20") \
-                             .replace("crs=30", "crs=5") \
-                             .replace("PORT = 18443", "PORT = 18447")
+                             .replace("crs=30", "crs=5")
     small = tempfile.mktemp(suffix=".py")
     open(small, "w").write(src)
     env = dict(os.environ)
