@@ -530,3 +530,18 @@ class TestSchedulersOverRest:
                            "spec": {"minMember": 3}})
         t.join(timeout=10)
         assert ("ADDED", "w0") in seen and ("ADDED", "w1") in seen
+
+
+def test_rest_op_latency_has_no_delayed_ack_stall(facade):
+    """100 sequential REST gets must beat 3 s by a wide margin. With the
+    Nagle/delayed-ACK interaction (no TCP_NODELAY) each op stalls ~40 ms
+    and this takes >4 s; post-fix it is ~0.1 s. Generous bound so loaded
+    CI never flakes, tight enough to catch the 40 ms/op failure mode."""
+    import time as _t
+    client = RestClient(base_url=facade.url)
+    facade.store.create({"kind": "ConfigMap", "apiVersion": "v1",
+                         "metadata": {"name": "lat"}, "data": {}})
+    t0 = _t.perf_counter()
+    for _ in range(100):
+        client.raw_try_get("ConfigMap", "default", "lat")
+    assert _t.perf_counter() - t0 < 3.0
