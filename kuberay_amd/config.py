@@ -21,6 +21,7 @@ class Configuration(K8sModel):
     probe_addr: str = ":8082"
     enable_leader_election: bool = True
     leader_election_namespace: str = ""
+    leader_lease_seconds: float = 15.0
     # horizontal sharding: run N operator processes against one cluster,
     # each owning CRs whose crc32(ns/name) %% shards == shard_index; every
     # shard elects its own leader Lease (kuberay-amd-operator-shard-I)
@@ -74,6 +75,7 @@ def load_config(argv: Optional[List[str]] = None) -> Configuration:
     parser.add_argument("--log-file")
     parser.add_argument("--state-file")
     parser.add_argument("--api-port", type=int, default=None)
+    parser.add_argument("--leader-lease-seconds", type=float, default=None)
     parser.add_argument("--no-metrics", action="store_true", default=None)
     args = parser.parse_args(argv)
 
@@ -95,6 +97,7 @@ def load_config(argv: Optional[List[str]] = None) -> Configuration:
         ("backend", "backend"), ("kubeconfig", "kubeconfig"),
         ("log_file", "log_file"), ("state_file", "state_file"),
         ("api_port", "api_port"),
+        ("leader_lease_seconds", "leader_lease_seconds"),
     ]:
         val = getattr(args, flag, None)
         if val is not None:

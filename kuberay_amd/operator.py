@@ -222,6 +222,7 @@ def main(argv=None) -> int:
             lease_name=lease,
             namespace=cfg.leader_election_namespace or "ray-system",
             identity=f"{socket.gethostname()}-{os.getpid()}",
+            lease_duration_s=cfg.leader_lease_seconds,
             on_started_leading=started.set)
         elector.start()
         logger.info("waiting for leader election (identity=%s)",
