@@ -268,7 +268,8 @@ class RestClient(KubeClient):
         md = obj.get("metadata", {})
         ns = md.get("namespace", "default")
         resp = self._check(self._http.put(
-            self._path(kind, ns, md.get("name")), json=obj))
+            self._path(kind, ns, md.get("name"),
+                       api_version=obj.get("apiVersion")), json=obj))
         return resp.json()
 
     def raw_patch(self, kind: str, namespace: str, name: str,

@@ -412,7 +412,7 @@ class KubernetesWASBatchScheduler(BatchScheduler):
         return (node.get("gang") or {}).get("minCount") \
             if isinstance(node, dict) else None
 
-    def _delete_pod_group(self, server, pg: dict) -> None:
+    def _delete_pod_group(self, raw, pg: dict) -> None:
         """Strip the kubelet-style protection finalizer, then delete."""
         meta = pg.setdefault("metadata", {})
         fins = meta.get("finalizers") or []
@@ -420,29 +420,27 @@ class KubernetesWASBatchScheduler(BatchScheduler):
             meta["finalizers"] = [f for f in fins
                                   if f != self.PROTECTION_FINALIZER]
             try:
-                server.update(pg)
+                raw.update(pg)
             except ApiError:
                 pass
         ns = meta.get("namespace") or "default"
         try:
-            server.delete("PodGroup", ns, meta["name"])
+            raw.delete("PodGroup", ns, meta["name"],
+                       api_version=self.GROUP_VERSION)
         except ApiError:
             pass
 
     def do_batch_scheduling_on_submission(self, client, cluster) -> None:
-        # the Workload API (KubernetesWAS) is alpha and exercised on the
-        # in-memory tier only; volcano/scheduler-plugins/xgmi-gang are the
-        # REST-capable production schedulers
-        server = getattr(client, "server", None)
-        if server is None:
-            return
+        server = RawObjectClient(client)
         if self._skip_reason(cluster) is not None:
             self.cleanup_on_completion(client, cluster)
             return
         ns = cluster.metadata.namespace or "default"
         workload, pod_group = self._build(cluster)
 
-        existing = server.try_get("Workload", ns, workload["metadata"]["name"])
+        existing = server.try_get("Workload", ns,
+                                  workload["metadata"]["name"],
+                                  api_version=self.GROUP_VERSION)
         if existing is None:
             server.create(workload)
         elif not self._owned(existing, cluster):
@@ -454,18 +452,22 @@ class KubernetesWASBatchScheduler(BatchScheduler):
                                     "schedulingPolicy")
               != _min_member(cluster)):
             # stale/terminating Workload: drop dependent PodGroup first
-            pg = server.try_get("PodGroup", ns, pod_group["metadata"]["name"])
+            pg = server.try_get("PodGroup", ns,
+                                pod_group["metadata"]["name"],
+                                api_version=self.GROUP_VERSION)
             if self._owned(pg, cluster):
                 self._delete_pod_group(server, pg)
             try:
-                server.delete("Workload", ns, workload["metadata"]["name"])
+                server.delete("Workload", ns, workload["metadata"]["name"],
+                              api_version=self.GROUP_VERSION)
             except ApiError:
                 pass
             raise SchedulingRetry(
                 f"replaced stale Workload {ns}/{workload['metadata']['name']}"
                 "; retrying after deletion completes")
 
-        pg = server.try_get("PodGroup", ns, pod_group["metadata"]["name"])
+        pg = server.try_get("PodGroup", ns, pod_group["metadata"]["name"],
+                            api_version=self.GROUP_VERSION)
         if pg is None:
             server.create(pod_group)
         elif not self._owned(pg, cluster):
@@ -489,17 +491,18 @@ class KubernetesWASBatchScheduler(BatchScheduler):
         pod.spec.schedulingGroup = {"podGroupName": self._pod_group_name(cluster)}
 
     def cleanup_on_completion(self, client, cluster) -> None:
-        server = getattr(client, "server", None)
-        if server is None:
-            return
+        server = RawObjectClient(client)
         ns = cluster.metadata.namespace or "default"
-        pg = server.try_get("PodGroup", ns, self._pod_group_name(cluster))
+        pg = server.try_get("PodGroup", ns, self._pod_group_name(cluster),
+                            api_version=self.GROUP_VERSION)
         if self._owned(pg, cluster):
             self._delete_pod_group(server, pg)
-        wl = server.try_get("Workload", ns, self._workload_name(cluster))
+        wl = server.try_get("Workload", ns, self._workload_name(cluster),
+                            api_version=self.GROUP_VERSION)
         if self._owned(wl, cluster):
             try:
-                server.delete("Workload", ns, self._workload_name(cluster))
+                server.delete("Workload", ns, self._workload_name(cluster),
+                              api_version=self.GROUP_VERSION)
             except ApiError:
                 pass
 

@@ -545,3 +545,24 @@ def test_rest_op_latency_has_no_delayed_ack_stall(facade):
     for _ in range(100):
         client.raw_try_get("ConfigMap", "default", "lat")
     assert _t.perf_counter() - t0 < 3.0
+
+
+def test_was_scheduler_over_rest(facade):
+    """The Workload-API (kubernetes-was) scheduler manages Workload +
+    PodGroup over the REST backend, including the protection-finalizer
+    strip on cleanup."""
+    from kuberay_amd.parallel.batchscheduler import KubernetesWASBatchScheduler
+    from kuberay_amd.utils import constants as C
+    client = RestClient(base_url=facade.url)
+    c = simple_raycluster("wasrest", workers=2, gpus_per_worker=1)
+    c.metadata.labels = {**(c.metadata.labels or {}),
+                         C.RAY_GANG_SCHEDULING_ENABLED: "true"}
+    cluster = client.create(c)
+    sched = KubernetesWASBatchScheduler()
+    sched.do_batch_scheduling_on_submission(client, cluster)
+    wls = facade.store.list("Workload")
+    pgs = facade.store.list("PodGroup")
+    assert len(wls) == 1 and len(pgs) == 1
+    sched.cleanup_on_completion(client, cluster)
+    assert facade.store.list("Workload") == []
+    assert facade.store.list("PodGroup") == []
