@@ -166,15 +166,15 @@ def create_app(client: Optional[KubeClient] = None,
         if rc is None:
             raise HTTPException(404, f"cluster {name} not found")
         out = conv.raycluster_to_api_cluster(rc)
-        # operator events for this cluster (reference returns cluster events)
-        server = getattr(client, "server", None)
-        if server is not None:
-            out["events"] = [
-                {"reason": e.get("reason"), "type": e.get("type"),
-                 "message": e.get("message"), "count": e.get("count", 1),
-                 "lastTimestamp": e.get("lastTimestamp")}
-                for e in server.list("Event", ns)
-                if (e.get("involvedObject") or {}).get("name") == name]
+        # operator events for this cluster (reference returns cluster
+        # events); typed list works on both backends
+        events = [e.to_dict() for e in client.list(k8s.Event, ns)]
+        out["events"] = [
+            {"reason": e.get("reason"), "type": e.get("type"),
+             "message": e.get("message"), "count": e.get("count", 1),
+             "lastTimestamp": e.get("lastTimestamp")}
+            for e in events
+            if (e.get("involvedObject") or {}).get("name") == name]
         return out
 
     @app.delete("/apis/v1/namespaces/{ns}/clusters/{name}")
