@@ -60,6 +60,9 @@ RESOURCES: Dict[str, Tuple[str, str]] = {
     "RayCronJob": ("/apis/ray.io/v1", "raycronjobs"),
 }
 
+# cluster-scoped kinds: never a /namespaces/{ns}/ segment in the path
+CLUSTER_SCOPED = {"Node"}
+
 SERVICE_ACCOUNT_DIR = "/var/run/secrets/kubernetes.io/serviceaccount"
 
 
@@ -172,6 +175,8 @@ class RestClient(KubeClient):
                       else f"/apis/{api_version}")
             plural = kind.lower() + ("es" if kind.lower().endswith("s")
                                      else "s")
+        if kind in CLUSTER_SCOPED:
+            namespace = None
         p = f"{prefix}/namespaces/{namespace}/{plural}" if namespace else f"{prefix}/{plural}"
         if name:
             p += f"/{name}"

@@ -108,9 +108,8 @@ class VirtualKubeletPodPool:
 
     # -- node registration (runVirtualKubelet node.NewNodeController) ---
     def register_node(self) -> None:
-        server = getattr(self.client, "server", None)
-        if server is None:
-            return
+        from ..kube.client import RawObjectClient
+        server = RawObjectClient(self.client)
         capacity = {"pods": str(sum(size for _, size in
                                     self.manager._pools.values()))}
         node = {
@@ -138,9 +137,8 @@ class VirtualKubeletPodPool:
             server.update(existing)
 
     def heartbeat(self) -> None:
-        server = getattr(self.client, "server", None)
-        if server is None:
-            return
+        from ..kube.client import RawObjectClient
+        server = RawObjectClient(self.client)
         node = server.try_get("Node", "default", self.node_name)
         if node is None:
             self.register_node()

@@ -566,3 +566,17 @@ def test_was_scheduler_over_rest(facade):
     sched.cleanup_on_completion(client, cluster)
     assert facade.store.list("Workload") == []
     assert facade.store.list("PodGroup") == []
+
+
+def test_podpool_vk_node_registers_over_rest(facade):
+    from kuberay_amd.ops.podpool import PodPoolManager, VirtualKubeletPodPool
+    client = RestClient(base_url=facade.url)
+    mgr = PodPoolManager(client)
+    vk = VirtualKubeletPodPool(client, mgr, node_name="vk-rest")
+    vk.register_node()
+    node = facade.store.try_get("Node", "default", "vk-rest")
+    assert node is not None
+    assert node["metadata"]["labels"]["type"] == "virtual-kubelet"
+    vk.heartbeat()
+    assert facade.store.try_get("Node", "default", "vk-rest")[
+        "status"]["conditions"][0]["status"] == "True"
