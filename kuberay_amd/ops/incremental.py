@@ -84,41 +84,9 @@ def is_http_route_ready(gw: Optional[Dict[str, Any]],
     return False
 
 
-class _RawObjects:
-    """Dict-object verbs over either backend: the in-memory server directly,
-    or a RestClient's raw_* methods (Gateway/HTTPRoute have no typed model)."""
-
-    def __init__(self, client):
-        self.server = getattr(client, "server", None)
-        self.client = client
-
-    def try_get(self, kind, namespace, name):
-        if self.server is not None:
-            return self.server.try_get(kind, namespace, name)
-        fn = getattr(self.client, "raw_try_get", None)
-        return fn(kind, namespace, name) if fn else None
-
-    def create(self, obj):
-        if self.server is not None:
-            return self.server.create(obj)
-        fn = getattr(self.client, "raw_create", None)
-        if fn is not None:
-            return fn(obj)
-        return None
-
-    def patch(self, kind, namespace, name, patch):
-        if self.server is not None:
-            return self.server.patch_merge(kind, namespace, name, patch)
-        fn = getattr(self.client, "raw_patch", None)
-        return fn(kind, namespace, name, patch) if fn else None
-
-    def delete(self, kind, namespace, name):
-        if self.server is not None:
-            self.server.delete(kind, namespace, name)
-            return
-        fn = getattr(self.client, "raw_delete", None)
-        if fn is not None:
-            fn(kind, namespace, name)
+# Gateway/HTTPRoute have no typed model; the shared raw-object seam
+# covers both backends (kube/client.py)
+from ..kube.client import RawObjectClient as _RawObjects  # noqa: E402
 
 
 class IncrementalUpgrader:
