@@ -580,3 +580,52 @@ def test_podpool_vk_node_registers_over_rest(facade):
     vk.heartbeat()
     assert facade.store.try_get("Node", "default", "vk-rest")[
         "status"]["conditions"][0]["status"] == "True"
+
+
+def test_incremental_upgrade_gateway_over_rest(facade):
+    """Gateway/HTTPRoute management and weight stepping run over the REST
+    backend (production topology for the Gateway-API migration)."""
+    import yaml as _yaml
+    from kuberay_amd.kube.events import NullRecorder
+    from kuberay_amd.models import RayService
+    from kuberay_amd.ops.incremental import IncrementalUpgrader
+    client = RestClient(base_url=facade.url)
+    cluster_spec = simple_raycluster("x", workers=1).spec.to_dict()
+    cluster_spec["enableInTreeAutoscaling"] = True
+    svc = client.create(RayService.from_dict({
+        "apiVersion": "ray.io/v1", "kind": "RayService",
+        "metadata": {"name": "irsvc", "namespace": "default"},
+        "spec": {"serveConfigV2": _yaml.safe_dump(
+                     {"applications": [{"name": "a", "import_path": "m:x"}]}),
+                 "rayClusterConfig": cluster_spec,
+                 "upgradeStrategy": {
+                     "type": "NewClusterWithIncrementalUpgrade",
+                     "clusterUpgradeOptions": {
+                         "gatewayClassName": "istio",
+                         "stepSizePercent": 50, "intervalSeconds": 0,
+                         "maxSurgePercent": 100}}}}))
+    active = client.create(simple_raycluster("ir-act", workers=1))
+    pending = client.create(simple_raycluster("ir-pen", workers=1))
+    up = IncrementalUpgrader(client, NullRecorder())
+    up.ensure_gateway_infra(svc, active, pending)
+    assert facade.store.try_get("Gateway", "default",
+                                "irsvc-gateway") is not None
+    # no gateway controller status yet -> step holds
+    assert up.step_traffic(svc, active, pending) is False
+    facade.store.patch_merge("Gateway", "default", "irsvc-gateway",
+        {"status": {"conditions": [
+            {"type": "Accepted", "status": "True"},
+            {"type": "Programmed", "status": "True"}]}},
+        subresource="status")
+    facade.store.patch_merge("HTTPRoute", "default", "irsvc-route",
+        {"status": {"parents": [{
+            "parentRef": {"name": "irsvc-gateway"},
+            "conditions": [{"type": "Accepted", "status": "True"},
+                           {"type": "ResolvedRefs", "status": "True"}]}]}},
+        subresource="status")
+    assert up.step_traffic(svc, active, pending) is False   # 50
+    assert up.step_traffic(svc, active, pending) is True    # 100
+    route = facade.store.get("HTTPRoute", "default", "irsvc-route")
+    weights = {r["name"]: r["weight"]
+               for r in route["spec"]["rules"][0]["backendRefs"]}
+    assert any(w == 100 for w in weights.values())
