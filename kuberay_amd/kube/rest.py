@@ -396,6 +396,22 @@ class _PodViewCache:
         with self._lock:
             return (namespace, name) in self._views
 
+    def replace(self, objs: List[Dict[str, Any]]) -> List[Tuple[str, str]]:
+        """Reflector Replace(): swap in the fresh list and report which
+        previously-cached pods vanished during the watch gap (the caller
+        delivers synthetic DELETED events for them)."""
+        fresh = {}
+        for obj in objs:
+            if obj.get("kind") != "Pod":
+                continue
+            meta = obj.get("metadata", {})
+            fresh[(meta.get("namespace", "default"),
+                   meta.get("name", ""))] = compute_pod_view(obj)
+        with self._lock:
+            gone = [k for k in self._views if k not in fresh]
+            self._views = fresh
+        return gone
+
 
 class RestApiServerAdapter:
     """Manager-facing surface over a real cluster: watch streams + list
@@ -451,8 +467,16 @@ class RestApiServerAdapter:
         rv: Optional[str] = None
         while not self._stopped.is_set():
             try:
-                # list first (seed + resourceVersion to watch from)
+                # list first (seed + resourceVersion to watch from);
+                # Replace() semantics — anything deleted during a watch
+                # gap gets a synthetic DELETED so caches never go stale
                 items, rv = self._client.raw_list_with_rv(kind)
+                if kind == "Pod":
+                    gone = self._view_cache.replace(items)
+                    for ns, name in gone:
+                        watcher.push("DELETED", {
+                            "kind": "Pod", "apiVersion": "v1",
+                            "metadata": {"namespace": ns, "name": name}})
                 for obj in items:
                     self._view_cache.apply("ADDED", obj)
                     watcher.push("ADDED", obj)

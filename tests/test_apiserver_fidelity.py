@@ -629,3 +629,48 @@ def test_incremental_upgrade_gateway_over_rest(facade):
     weights = {r["name"]: r["weight"]
                for r in route["spec"]["rules"][0]["backendRefs"]}
     assert any(w == 100 for w in weights.values())
+
+
+def test_reflector_relist_purges_deleted_pods(facade):
+    """Objects deleted while a watch is DOWN must leave the informer
+    cache on re-list (client-go Reflector Replace semantics): ghost pods
+    would make reconcilers refuse to recreate. The watch is genuinely
+    severed here — the stream raises GoneError while the pod is deleted
+    behind its back, so only the Replace() path can purge it."""
+    from kuberay_amd.kube.rest import RestApiServerAdapter
+    from kuberay_amd.kube.store import GoneError as _Gone
+    client = RestClient(base_url=facade.url)
+    adapter = RestApiServerAdapter(rest_client=client)
+    facade.store.create({"kind": "Pod", "apiVersion": "v1",
+                         "metadata": {"name": "ghost", "labels": {}},
+                         "spec": {}, "status": {"phase": "Running"}})
+    import threading as _th
+    severed = _th.Event()
+    deleted = _th.Event()
+    real_stream = client.raw_watch_stream
+
+    def breaking_stream(kind, rv=None, allow_bookmarks=False, **kw):
+        if not severed.is_set():
+            severed.set()
+            deleted.wait(timeout=10)   # pod is deleted while "down"
+            raise _Gone("watch severed by test")
+        return real_stream(kind, rv, allow_bookmarks=allow_bookmarks, **kw)
+
+    client.raw_watch_stream = breaking_stream
+    w = adapter.watch({"Pod"})
+    assert severed.wait(timeout=10)
+    # first list seeded the cache before the stream call
+    assert adapter._client.pod_cache_contains("default", "ghost")
+    facade.store.delete("Pod", "default", "ghost")
+    deleted.set()
+    events = []
+    deadline = time.time() + 10
+    while time.time() < deadline:
+        ev = w.next(timeout=0.2)
+        if ev and ev[0] == "DELETED" and \
+                ev[1]["metadata"]["name"] == "ghost":
+            events.append(ev)
+            break
+    assert events, "no synthetic DELETED delivered on re-list"
+    assert not adapter._client.pod_cache_contains("default", "ghost")
+    adapter.stop()
