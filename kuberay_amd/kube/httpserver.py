@@ -14,6 +14,8 @@ import binascii
 import json
 import re
 import threading
+import time
+import uuid
 from http.server import BaseHTTPRequestHandler, ThreadingHTTPServer
 from typing import Any, Dict, List, Optional, Tuple
 from urllib.parse import parse_qs, urlparse
@@ -260,24 +262,56 @@ class _Handler(BaseHTTPRequestHandler):
             selector = dict(kv.split("=", 1)
                             for kv in params["labelSelector"][0].split(","))
         items = self.store.list(kind, ns, selector)
-        # chunked lists: ?limit=N&continue=token (apiserver pagination)
+        # chunked lists: ?limit=N&continue=token. Real-apiserver
+        # semantics: the continue token pins a consistent SNAPSHOT (etcd
+        # revision) — later pages never skip or duplicate items mutated
+        # between requests, and an expired token answers 410 Expired.
         list_meta = {"resourceVersion": str(self.store.current_rv)}
         limit = int(params.get("limit", ["0"])[0] or 0)
         if limit > 0:
-            offset = 0
+            snaps = getattr(self.server, "_page_snaps", None)
+            if snaps is None:
+                snaps = self.server._page_snaps = {}  # type: ignore[attr-defined]
             token = params.get("continue", [None])[0]
             if token:
                 try:
-                    offset = int(base64.b64decode(token).decode())
+                    snap_id, offset = base64.b64decode(
+                        token).decode().split(":", 1)
+                    offset = int(offset)
                 except (ValueError, binascii.Error):
                     return self._send_json(400, {
                         "kind": "Status", "status": "Failure",
                         "reason": "BadRequest",
                         "message": "invalid continue token", "code": 400})
+                snap = snaps.get(snap_id)
+                if snap is None or time.time() - snap[2] > 300:
+                    return self._send_json(410, {
+                        "kind": "Status", "status": "Failure",
+                        "reason": "Expired",
+                        "message": "The provided continue parameter is too "
+                                   "old to display a consistent list result.",
+                        "code": 410})
+                items, rv, _ = snap
+            else:
+                snap_id = uuid.uuid4().hex[:12]
+                rv = str(self.store.current_rv)
+                offset = 0
+                if len(items) > limit:
+                    # bound the snapshot table: drop expired, then oldest
+                    now = time.time()
+                    for k in [k for k, v in snaps.items()
+                              if now - v[2] > 300]:
+                        snaps.pop(k, None)
+                    while len(snaps) >= 64:
+                        snaps.pop(next(iter(snaps)), None)
+                    snaps[snap_id] = (items, rv, now)
+            list_meta["resourceVersion"] = rv
             page = items[offset:offset + limit]
             if offset + limit < len(items):
                 list_meta["continue"] = base64.b64encode(
-                    str(offset + limit).encode()).decode()
+                    f"{snap_id}:{offset + limit}".encode()).decode()
+            else:
+                snaps.pop(snap_id, None)  # fully consumed
             items = page
         self._send_json(200, {"kind": f"{kind}List", "apiVersion": "v1",
                               "metadata": list_meta, "items": items})

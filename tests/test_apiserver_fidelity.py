@@ -674,3 +674,42 @@ def test_reflector_relist_purges_deleted_pods(facade):
     assert events, "no synthetic DELETED delivered on re-list"
     assert not adapter._client.pod_cache_contains("default", "ghost")
     adapter.stop()
+
+
+class TestPaginationSnapshots:
+    def test_pages_are_churn_consistent(self, facade):
+        """Continue tokens pin a snapshot: mutations between page fetches
+        never cause skips or duplicates (etcd-revision semantics)."""
+        import httpx
+        for i in range(30):
+            facade.store.create(_cm(f"pg{i:02d}"))
+        base = facade.url + "/api/v1/namespaces/default/configmaps"
+        r1 = httpx.get(base, params={"limit": 10}).json()
+        assert len(r1["items"]) == 10
+        token = r1["metadata"]["continue"]
+        # churn between pages: delete one already-served and one
+        # yet-to-be-served item, create new ones
+        facade.store.delete("ConfigMap", "default", "pg03")
+        facade.store.delete("ConfigMap", "default", "pg25")
+        facade.store.create(_cm("zz-new"))
+        r2 = httpx.get(base, params={"limit": 10, "continue": token}).json()
+        token = r2["metadata"]["continue"]
+        r3 = httpx.get(base, params={"limit": 10, "continue": token}).json()
+        names = [o["metadata"]["name"] for r in (r1, r2, r3)
+                 for o in r["items"]]
+        assert names == [f"pg{i:02d}" for i in range(30)]
+        assert "continue" not in r3["metadata"]
+        # all three pages report the snapshot's rv
+        assert (r1["metadata"]["resourceVersion"]
+                == r3["metadata"]["resourceVersion"])
+
+    def test_expired_token_answers_410(self, facade):
+        import base64 as b64
+        import httpx
+        for i in range(5):
+            facade.store.create(_cm(f"ex{i}"))
+        base = facade.url + "/api/v1/namespaces/default/configmaps"
+        bogus = b64.b64encode(b"deadbeef0000:2").decode()
+        r = httpx.get(base, params={"limit": 2, "continue": bogus})
+        assert r.status_code == 410
+        assert r.json()["reason"] == "Expired"
