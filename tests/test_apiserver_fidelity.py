@@ -713,3 +713,25 @@ class TestPaginationSnapshots:
         r = httpx.get(base, params={"limit": 2, "continue": bogus})
         assert r.status_code == 410
         assert r.json()["reason"] == "Expired"
+
+
+def test_arbitrary_crd_kind_full_round_trip(facade):
+    """Any CRD kind works over REST with zero registration: paths derive
+    from apiVersion, the facade learns the kind from the POST body."""
+    from kuberay_amd.kube.client import RawObjectClient
+    client = RestClient(base_url=facade.url)
+    raw = RawObjectClient(client)
+    av = "argoproj.io/v1alpha1"
+    raw.create({"apiVersion": av, "kind": "Workflow",
+                "metadata": {"name": "wf1", "namespace": "default"},
+                "spec": {"entrypoint": "main"}})
+    got = raw.try_get("Workflow", "default", "wf1", api_version=av)
+    assert got["spec"]["entrypoint"] == "main"
+    raw.patch("Workflow", "default", "wf1", {"spec": {"parallelism": 3}},
+              api_version=av)
+    assert raw.try_get("Workflow", "default", "wf1",
+                       api_version=av)["spec"]["parallelism"] == 3
+    assert len(raw.list("Workflow", "default", api_version=av)) == 1
+    raw.delete("Workflow", "default", "wf1", api_version=av)
+    assert raw.try_get("Workflow", "default", "wf1",
+                       api_version=av) is None
