@@ -418,8 +418,14 @@ class InMemoryApiServer:
 
             def merge(dst, src):
                 for k, v in src.items():
-                    if isinstance(v, dict) and isinstance(dst.get(k), dict):
-                        merge(dst[k], v)
+                    if isinstance(v, dict):
+                        # RFC 7386: patch objects merge into the existing
+                        # value or {} — nested nulls delete, never
+                        # materialize (keeps merge patch idempotent)
+                        node = dst.get(k)
+                        if not isinstance(node, dict):
+                            node = dst[k] = {}
+                        merge(node, v)
                     elif v is None:
                         dst.pop(k, None)
                     else:

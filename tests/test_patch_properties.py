@@ -331,3 +331,25 @@ class TestQuantityRoundTrip:
         from kuberay_amd.utils.quantity import format_quantity, parse_quantity
         v = parse_quantity(f"{n}{suffix}")
         assert parse_quantity(format_quantity(v)) == v
+
+
+class TestStoreMergePatch:
+    @settings(max_examples=60, deadline=None)
+    @given(patch=DICT_DOCS)
+    def test_store_merge_patch_idempotent_and_null_free(self, patch):
+        from kuberay_amd.kube.store import InMemoryApiServer
+        server = InMemoryApiServer()
+        server.create({"kind": "ConfigMap", "apiVersion": "v1",
+                       "metadata": {"name": "m"}, "data": {}})
+        once = server.patch_merge("ConfigMap", "default", "m",
+                                  {"data": dict(patch)})
+        twice = server.patch_merge("ConfigMap", "default", "m",
+                                   {"data": dict(patch)})
+        assert once["data"] == twice["data"]
+
+        def no_nulls(node):
+            if isinstance(node, dict):
+                return all(v is not None and no_nulls(v)
+                           for v in node.values())
+            return True
+        assert no_nulls(once["data"])
