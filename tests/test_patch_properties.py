@@ -353,3 +353,30 @@ class TestStoreMergePatch:
                            for v in node.values())
             return True
         assert no_nulls(once["data"])
+
+
+class TestUnknownFieldPreservation:
+    def test_exotic_and_future_fields_survive_round_trip(self):
+        """The operator must never wipe fields it doesn't model when it
+        writes a CR back (kubectl-applied topologySpreadConstraints,
+        lifecycle hooks, future API fields)."""
+        from kuberay_amd.models import RayCluster
+        from kuberay_amd.testing import simple_raycluster
+        d = simple_raycluster("u", workers=1).to_dict()
+        ps = d["spec"]["headGroupSpec"]["template"]["spec"]
+        ps["topologySpreadConstraints"] = [
+            {"maxSkew": 1, "topologyKey": "zone",
+             "whenUnsatisfiable": "DoNotSchedule"}]
+        ps["containers"][0]["lifecycle"] = {
+            "preStop": {"exec": {"command": ["/bin/sleep", "5"]}}}
+        d["spec"]["workerGroupSpecs"][0]["template"]["metadata"] = {
+            "annotations": {"custom.io/x": "1"}}
+        d["spec"]["someFutureField"] = {"a": 1}
+        rt = RayCluster.from_dict(d).to_dict()
+        hs = rt["spec"]["headGroupSpec"]["template"]["spec"]
+        assert hs["topologySpreadConstraints"][0]["topologyKey"] == "zone"
+        assert hs["containers"][0]["lifecycle"]["preStop"]["exec"][
+            "command"] == ["/bin/sleep", "5"]
+        assert rt["spec"]["workerGroupSpecs"][0]["template"]["metadata"][
+            "annotations"] == {"custom.io/x": "1"}
+        assert rt["spec"]["someFutureField"] == {"a": 1}
