@@ -237,7 +237,12 @@ class _Handler(BaseHTTPRequestHandler):
 
     def _body(self):
         length = int(self.headers.get("Content-Length", 0))
-        return json.loads(self.rfile.read(length)) if length else {}
+        if not length:
+            return {}
+        try:
+            return json.loads(self.rfile.read(length))
+        except ValueError as e:
+            raise ApiError(400, f"invalid JSON body: {e}")
 
     # ------------------------------------------------------------------
     def do_GET(self):  # noqa: N802
@@ -392,7 +397,10 @@ class _Handler(BaseHTTPRequestHandler):
             watcher.stop()
 
     def do_POST(self):  # noqa: N802
-        obj = self._body()
+        try:
+            obj = self._body()
+        except ApiError as e:
+            return self._send_error_status(e)
         parsed = _parse_path(urlparse(self.path).path, store=self.store,
                              body_kind=obj.get("kind"))
         if parsed is None:
@@ -406,7 +414,10 @@ class _Handler(BaseHTTPRequestHandler):
             self._send_error_status(e)
 
     def do_PUT(self):  # noqa: N802
-        obj = self._body()
+        try:
+            obj = self._body()
+        except ApiError as e:
+            return self._send_error_status(e)
         parsed = _parse_path(urlparse(self.path).path, store=self.store,
                              body_kind=obj.get("kind"))
         if parsed is None:
@@ -430,7 +441,10 @@ class _Handler(BaseHTTPRequestHandler):
             return self._send_json(404, {"message": "unknown path"})
         kind, ns, name, sub = parsed
         content_type = (self.headers.get("Content-Type") or "").split(";")[0]
-        body = self._body()
+        try:
+            body = self._body()
+        except ApiError as e:
+            return self._send_error_status(e)
         try:
             if content_type == "application/json-patch+json":
                 current = self.store.get(kind, ns or "default", name)

@@ -735,3 +735,59 @@ def test_arbitrary_crd_kind_full_round_trip(facade):
     raw.delete("Workflow", "default", "wf1", api_version=av)
     assert raw.try_get("Workflow", "default", "wf1",
                        api_version=av) is None
+
+
+class TestFacadeRobustness:
+    """Malformed requests get structured Status errors, never 500s."""
+
+    def test_malformed_json_body(self, facade):
+        import httpx
+        r = httpx.post(facade.url + "/api/v1/namespaces/default/configmaps",
+                       content=b"{not json", headers={
+                           "Content-Type": "application/json"})
+        assert r.status_code in (400, 422), r.status_code
+
+    def test_unknown_paths_404(self, facade):
+        import httpx
+        for path in ("/", "/api", "/api/v2/zzz", "/apis/x",
+                     "/api/v1/namespaces/default/configmaps/a/b/c"):
+            r = httpx.get(facade.url + path)
+            assert r.status_code == 404, path
+
+    def test_delete_nonexistent_is_status_404(self, facade):
+        import httpx
+        r = httpx.delete(facade.url +
+                         "/api/v1/namespaces/default/configmaps/nope")
+        assert r.status_code == 404
+        assert r.json()["kind"] == "Status"
+
+    def test_put_with_stale_rv_conflicts(self, facade):
+        import httpx
+        facade.store.create(_cm("rvx"))
+        cur = facade.store.get("ConfigMap", "default", "rvx")
+        facade.store.update({**cur, "data": {"v": "2"}})
+        stale = dict(cur)
+        stale["data"] = {"v": "stale"}
+        r = httpx.put(facade.url +
+                      "/api/v1/namespaces/default/configmaps/rvx",
+                      json=stale)
+        assert r.status_code == 409
+
+    def test_bad_json_patch_is_422(self, facade):
+        import httpx
+        facade.store.create(_cm("jp"))
+        r = httpx.patch(facade.url +
+                        "/api/v1/namespaces/default/configmaps/jp",
+                        json=[{"op": "frobnicate", "path": "/x"}],
+                        headers={"Content-Type":
+                                 "application/json-patch+json"})
+        assert r.status_code == 422
+
+    def test_watch_unknown_rv_format_starts_from_now(self, facade):
+        import httpx
+        # non-numeric rv: facade treats it as 0/now rather than crashing
+        with httpx.stream(
+                "GET", facade.url + "/api/v1/namespaces/default/configmaps",
+                params={"watch": "true", "resourceVersion": "abc"},
+                timeout=2) as resp:
+            assert resp.status_code in (200, 400, 410)
