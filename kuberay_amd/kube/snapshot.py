@@ -52,6 +52,10 @@ def load_snapshot(server: InMemoryApiServer, path: str) -> int:
         meta = json.loads(first).get("__meta__", {}) if first.strip() else {}
         with server._lock:
             server._rv = int(meta.get("resourceVersion", 0))
+            # events before the restore point are unreplayable: watchers
+            # resuming below this must 410 + re-list
+            server._history_base = server._rv
+            server._event_history.clear()
             for line in f:
                 line = line.strip()
                 if not line:

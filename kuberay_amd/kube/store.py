@@ -261,6 +261,10 @@ class InMemoryApiServer:
         self._rv = 0
         self._watchers: List["Watcher"] = []
         self._event_history: deque = deque(maxlen=self.EVENT_HISTORY_LIMIT)
+        # lowest rv from which a watch can replay with proven completeness;
+        # raised on snapshot restore (pre-restart events are gone — clients
+        # below it must get 410 and re-list, never a silent gap)
+        self._history_base: int = 0
 
     @property
     def backend_name(self) -> str:
@@ -304,6 +308,8 @@ class InMemoryApiServer:
         import json as _json
         with self._lock:
             hist = self._event_history
+            if rv < self._history_base:
+                return None
             if len(hist) == hist.maxlen and hist[0][0] > rv + 1:
                 return None
             return [(etype, _json.loads(payload))

@@ -380,3 +380,27 @@ class TestUnknownFieldPreservation:
         assert rt["spec"]["workerGroupSpecs"][0]["template"]["metadata"][
             "annotations"] == {"custom.io/x": "1"}
         assert rt["spec"]["someFutureField"] == {"a": 1}
+
+
+class TestSnapshotWatchSemantics:
+    def test_resume_below_restore_point_gets_410(self, tmp_path):
+        """A watcher holding a pre-restart rv cannot be silently given a
+        partial replay after snapshot restore — it must see Gone and
+        re-list (events between its rv and the restore point are lost)."""
+        from kuberay_amd.kube.snapshot import load_snapshot, save_snapshot
+        from kuberay_amd.kube.store import InMemoryApiServer
+        s1 = InMemoryApiServer()
+        for i in range(5):
+            s1.create({"kind": "ConfigMap", "apiVersion": "v1",
+                       "metadata": {"name": f"c{i}"}})
+        path = str(tmp_path / "state.jsonl")
+        save_snapshot(s1, path)
+        s2 = InMemoryApiServer()
+        load_snapshot(s2, path)
+        assert s2.events_since(1) is None          # pre-restart rv: 410
+        assert s2.events_since(s2.current_rv) == []  # at restore point: ok
+        out = s2.create({"kind": "ConfigMap", "apiVersion": "v1",
+                         "metadata": {"name": "after"}})
+        got = s2.events_since(s2.current_rv - 1)
+        assert [o["metadata"]["name"] for _, o in got] == ["after"]
+        assert int(out["metadata"]["resourceVersion"]) > 5  # rv monotone
