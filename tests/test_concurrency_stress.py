@@ -268,9 +268,12 @@ class TestPatchFuzz:
                 assert out == patch or isinstance(out, list)
 
 
+@pytest.mark.timeout(420)
 def test_memory_benchmark_script_smoke():
     """The memory-benchmark harness (reference memory_benchmark analog)
-    runs end to end at reduced scale and emits a well-formed result."""
+    runs end to end at reduced scale and emits a well-formed result.
+    Timeouts are generous: under a fully loaded parallel test run the
+    operator subprocess can take tens of seconds just to start."""
     import json
     import os
     import subprocess
@@ -282,13 +285,15 @@ def test_memory_benchmark_script_smoke():
     out = tempfile.mktemp(suffix=".json")
     src = open(script).read().replace("crs=150", "crs=20") \
                              .replace("target_pods=150", "target_pods=20") \
-                             .replace("crs=30", "crs=5")
+                             .replace("crs=30", "crs=5") \
+                             .replace("timeout=60", "timeout=150") \
+                             .replace("range(100)", "range(400)")
     small = tempfile.mktemp(suffix=".py")
     open(small, "w").write(src)
     env = dict(os.environ)
     env["PYTHONPATH"] = os.path.join(here, "..") + os.pathsep + \
         env.get("PYTHONPATH", "")
-    r = subprocess.run([sys.executable, small, "--out", out], timeout=180,
+    r = subprocess.run([sys.executable, small, "--out", out], timeout=360,
                        capture_output=True, text=True, env=env)
     assert r.returncode == 0, r.stdout + r.stderr
     data = json.load(open(out))
