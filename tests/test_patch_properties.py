@@ -404,3 +404,40 @@ class TestSnapshotWatchSemantics:
         got = s2.events_since(s2.current_rv - 1)
         assert [o["metadata"]["name"] for _, o in got] == ["after"]
         assert int(out["metadata"]["resourceVersion"]) > 5  # rv monotone
+
+
+class TestSelectorDifferential:
+    @settings(max_examples=10, deadline=None)
+    @given(labelings=st.lists(
+        st.dictionaries(st.sampled_from(["app", "tier", "env"]),
+                        st.sampled_from(["a", "b", "c"]), max_size=3),
+        min_size=4, max_size=8),
+        sel=st.dictionaries(st.sampled_from(["app", "tier"]),
+                            st.sampled_from(["a", "b"]), min_size=1,
+                            max_size=2))
+    def test_label_selector_lists_match(self, labelings, sel):
+        """Label-selector list results agree between the in-memory client
+        and REST-over-facade for random labelings and selectors."""
+        from kuberay_amd.kube.client import InMemoryClient
+        from kuberay_amd.kube.httpserver import KubeApiFacade
+        from kuberay_amd.kube.objects import ConfigMap
+        from kuberay_amd.kube.rest import RestClient
+        mem = InMemoryClient()
+        facade = KubeApiFacade().start()
+        try:
+            rest = RestClient(base_url=facade.url)
+            for client in (mem, rest):
+                for i, labels in enumerate(labelings):
+                    client.create(ConfigMap.from_dict(
+                        {"kind": "ConfigMap", "apiVersion": "v1",
+                         "metadata": {"name": f"s{i}", "labels": labels}}))
+            a = {c.metadata.name
+                 for c in mem.list(ConfigMap, "default", sel)}
+            b = {c.metadata.name
+                 for c in rest.list(ConfigMap, "default", sel)}
+            assert a == b
+            expect = {f"s{i}" for i, lb in enumerate(labelings)
+                      if all(lb.get(k) == v for k, v in sel.items())}
+            assert a == expect
+        finally:
+            facade.stop()
