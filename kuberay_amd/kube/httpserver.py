@@ -304,11 +304,16 @@ class _Handler(BaseHTTPRequestHandler):
                 if len(items) > limit:
                     # bound the snapshot table: drop expired, then oldest
                     now = time.time()
-                    for k in [k for k, v in snaps.items()
-                              if now - v[2] > 300]:
-                        snaps.pop(k, None)
+                    # concurrent handler threads mutate snaps: iterate a
+                    # copy, pop defensively
+                    for k, v in list(snaps.items()):
+                        if now - v[2] > 300:
+                            snaps.pop(k, None)
                     while len(snaps) >= 64:
-                        snaps.pop(next(iter(snaps)), None)
+                        try:
+                            snaps.pop(next(iter(snaps)), None)
+                        except (StopIteration, RuntimeError):
+                            break
                     snaps[snap_id] = (items, rv, now)
             list_meta["resourceVersion"] = rv
             page = items[offset:offset + limit]
