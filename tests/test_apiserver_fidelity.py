@@ -791,3 +791,38 @@ class TestFacadeRobustness:
                 params={"watch": "true", "resourceVersion": "abc"},
                 timeout=2) as resp:
             assert resp.status_code in (200, 400, 410)
+
+
+def test_snapshot_table_bounded_and_watch_ordered(facade):
+    """Pagination snapshots are bounded (old tokens expire as 410 once
+    evicted) and watch events always arrive in resourceVersion order."""
+    import httpx
+    for i in range(80):
+        facade.store.create(_cm(f"bnd{i:03d}"))
+    base = facade.url + "/api/v1/namespaces/default/configmaps"
+    # open 70 paginated lists without consuming them: table caps at 64
+    tokens = []
+    for _ in range(70):
+        r = httpx.get(base, params={"limit": 10}).json()
+        tokens.append(r["metadata"]["continue"])
+    snaps = getattr(facade._httpd, "_page_snaps", {})
+    assert len(snaps) <= 64
+    r = httpx.get(base, params={"limit": 10, "continue": tokens[0]})
+    assert r.status_code == 410  # evicted token expires, never mis-pages
+    # watch ordering: rv strictly increases over a live stream
+    client = RestClient(base_url=facade.url)
+    rv0 = str(facade.store.current_rv)
+    seen = []
+    import threading
+    def consume():
+        for _, obj in client.raw_watch_stream("ConfigMap", rv0):
+            seen.append(int(obj["metadata"]["resourceVersion"]))
+            if len(seen) >= 10:
+                break
+    t = threading.Thread(target=consume, daemon=True)
+    t.start()
+    time.sleep(0.2)
+    for i in range(10):
+        facade.store.create(_cm(f"ord{i}"))
+    t.join(timeout=10)
+    assert seen == sorted(seen) and len(seen) == 10
