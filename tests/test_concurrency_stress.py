@@ -293,9 +293,15 @@ def test_memory_benchmark_script_smoke():
     env = dict(os.environ)
     env["PYTHONPATH"] = os.path.join(here, "..") + os.pathsep + \
         env.get("PYTHONPATH", "")
-    r = subprocess.run([sys.executable, small, "--out", out], timeout=360,
-                       capture_output=True, text=True, env=env)
-    assert r.returncode == 0, r.stdout + r.stderr
+    attempts = []
+    for _ in range(2):  # one retry: subprocess startup is load-sensitive
+        r = subprocess.run([sys.executable, small, "--out", out],
+                           timeout=360, capture_output=True, text=True,
+                           env=env)
+        attempts.append(r.stdout + r.stderr)
+        if r.returncode == 0:
+            break
+    assert r.returncode == 0, "\n--- attempt ---\n".join(attempts)
     data = json.load(open(out))
     assert data["summary"]["exp2_peak_rss_mb"] > 0
     assert any(s["experiment"] == 3 for s in data["samples"])
